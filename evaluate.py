@@ -1,0 +1,40 @@
+#!/usr/bin/env python3
+"""Evaluation CLI — reference-compatible (see reference evaluate.py:185-272).
+
+    python evaluate.py --model raft_nc_dbl --restore_ckpt ckpt.pth --dataset sintel ...
+"""
+
+import torch
+
+from flowhip.config import build_eval_parser, finalize_args
+from flowhip.engine import checkpoints
+from flowhip.engine.evaluate import (
+    validate_chairs,
+    validate_kitti,
+    validate_sintel,
+    validate_synthetic,
+)
+from flowhip.models import build_model
+
+if __name__ == "__main__":
+    args = finalize_args(build_eval_parser().parse_args())
+
+    model = build_model(args)
+    if args.restore_ckpt is not None:
+        checkpoints.load_weights(model, args.restore_ckpt)
+
+    device = "cuda" if torch.cuda.is_available() else "cpu"
+    model.to(device)
+    model.eval()
+
+    with torch.no_grad():
+        if args.dataset == "chairs":
+            validate_chairs(model, args.iters or 24)
+        elif args.dataset == "sintel":
+            validate_sintel(model, args.iters or 32)
+        elif args.dataset == "kitti":
+            validate_kitti(model, args.iters or 24)
+        elif args.dataset == "synthetic":
+            validate_synthetic(model, args.iters or 12)
+        else:
+            raise SystemExit(f"unknown --dataset {args.dataset!r}")

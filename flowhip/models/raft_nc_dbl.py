@@ -1,0 +1,78 @@
+"""RAFT-NCUP (`raft_nc_dbl`): RAFT with the convex-upsample mask head replaced
+by the normalized-convolution upsampler, applied every iteration.
+
+State-dict compatible with the reference `core/raft_nc_dbl.py`: same fnet /
+cnet / update_block attributes, mask head emptied (:68), `upsampler.*` keys
+from get_upsampler(2, 128, args) (:75). `--load_pretrained` strips the
+`module.` DataParallel prefix (:57-66); `--freeze_raft` freezes everything
+but the upsampler (:70-72).
+"""
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+from ..nn.upsampler import get_upsampler
+from ..utils.amp import autocast_ctx
+from .raft import RAFT as _RAFTBase
+
+
+class RAFT_NC_DBL(_RAFTBase):
+    def __init__(self, args):
+        super().__init__(args)
+
+        if getattr(args, "load_pretrained", None) is not None:
+            state_dict = torch.load(args.load_pretrained, map_location="cpu",
+                                    weights_only=True)
+            stripped = {k[7:] if k.startswith("module.") else k: v
+                        for k, v in state_dict.items()}
+            self.load_state_dict(stripped)
+
+        # NCUP replaces the convex-upsample mask head entirely.
+        self.update_block.mask = nn.Sequential()
+
+        if getattr(args, "freeze_raft", False):
+            for p in self.parameters():
+                p.requires_grad = False
+
+        # 2 data channels (flow u,v); guidance = the GRU hidden state.
+        # The reference hardcodes 128 guidance channels (raft_nc_dbl.py:75),
+        # which crashes for --small (96-ch hidden state); using hidden_dim is
+        # identical for the basic model and defines the small variant.
+        self.upsampler = get_upsampler(2, self.hidden_dim, args)
+
+    def upsample_flow(self, flow_lr, guidance):
+        """H/8 flow -> H flow: nearest x2 pre-upsample then NConvUpsampler
+        (scale=4) guided by the GRU hidden state (raft_nc_dbl.py:107-112)."""
+        flow_lr = F.interpolate(flow_lr, scale_factor=2, mode="nearest")
+        return self.upsampler(flow_lr, guidance)
+
+    def forward(self, image1, image2, iters=12, flow_init=None, upsample=True,
+                test_mode=False):
+        image1, corr_fn, net, inp = self._features(image1, image2)
+
+        coords0, coords1 = self.initialize_flow(image1)
+        if flow_init is not None:
+            coords1 = coords1 + flow_init
+
+        flow_predictions = []
+        flow_up = None
+        for _ in range(iters):
+            coords1 = coords1.detach()
+            corr = corr_fn(coords1)
+
+            flow = coords1 - coords0
+            with autocast_ctx(image1, enabled=self.args.mixed_precision):
+                net, _, delta_flow = self.update_block(net, inp, corr, flow)
+
+            coords1 = coords1 + delta_flow
+
+            # NCUP upsample every iteration, guided by the GRU hidden state;
+            # the x8 scale is applied to the upsampled field (ref :161).
+            flow_up = 8 * self.upsample_flow((coords1 - coords0).float(),
+                                             net.float())
+            flow_predictions.append(flow_up)
+
+        if test_mode:
+            return coords1 - coords0, flow_up
+        return flow_predictions

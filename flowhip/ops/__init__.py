@@ -1,0 +1,73 @@
+"""flowhip.ops — the op layer: dispatch between hand-written
+gfx950 HIP kernels (CUDA tensors, extension built) and the pure-PyTorch
+reference path (CPU, or forced via FLOWHIP_FORCE_REF=1).
+
+Public surface (mirrors the compute-heavy call sites of the reference —
+SURVEY.md §2.2 kernel inventory):
+
+  corr_volume(fmap1, fmap2)            kernel #1  (MFMA batched GEMM)
+  corr_pyramid(corr, num_levels)       kernel #2  (avg-pool pyramid)
+  corr_lookup(pyramid, coords, r)      kernel #3  (4-level 81-tap gather+lerp)
+  nconv2d(...)                         kernel #6  (fused normalized conv)
+  conf_pool(data, conf)                kernel #7  (confidence-based pooling)
+  zero_inject(x, sh, sw)               kernel #8  (sparse injection scatter)
+  convex_upsample(flow, mask)          kernel #11 (RAFT convex upsample)
+  sequence_loss(preds, gt, valid, g)   kernel #12 (loss; torch for now)
+"""
+
+import torch
+
+from . import _ext, torch_ref
+from .torch_ref import MAX_FLOW, sequence_loss  # re-export  # noqa: F401
+
+
+def corr_volume(fmap1, fmap2):
+    if _ext.use_hip(fmap1):
+        from .functional import CorrVolumeFn
+        return CorrVolumeFn.apply(fmap1, fmap2)
+    return torch_ref.corr_volume(fmap1, fmap2)
+
+
+def corr_pyramid(corr, num_levels=4):
+    # avg_pool2d chain; ROCm-PyTorch path is already memory-bound optimal for
+    # this minor op (<1% of step — see profiles/), torch autograd handles bwd.
+    return torch_ref.corr_pyramid(corr, num_levels)
+
+
+def corr_lookup(pyramid, coords, radius):
+    if _ext.use_hip(coords):
+        from .functional import CorrLookupFn
+        return CorrLookupFn.apply(coords, radius, *pyramid)
+    return torch_ref.corr_lookup(pyramid, coords, radius)
+
+
+def nconv2d(data, conf, weight, bias=None, stride=1, padding=0, dilation=1,
+            groups=1, eps=1e-20, prop_conf=True):
+    if _ext.use_hip(data) and _can_fuse_nconv(weight, stride, dilation, groups):
+        from .functional_nconv import NConv2dFn
+        return NConv2dFn.apply(data, conf, weight, bias, padding, eps, prop_conf)
+    return torch_ref.nconv2d(data, conf, weight, bias, stride, padding,
+                             dilation, groups, eps, prop_conf)
+
+
+def _can_fuse_nconv(weight, stride, dilation, groups):
+    from torch.nn.modules.utils import _pair
+    if _pair(stride) != (1, 1) or _pair(dilation) != (1, 1) or groups != 1:
+        return False
+    o, i, kh, kw = weight.shape
+    return kh == kw and kh in (1, 3, 5) and i <= 8 and o <= 8
+
+
+def conf_pool(data, conf, ds_factor=2, pooling_type="conf_based"):
+    return torch_ref.conf_pool(data, conf, ds_factor, pooling_type)
+
+
+def zero_inject(inp, scale_h, scale_w, out_h=None, out_w=None):
+    return torch_ref.zero_inject(inp, scale_h, scale_w, out_h, out_w)
+
+
+def convex_upsample(flow, mask, factor=8):
+    if _ext.use_hip(flow):
+        from .functional_upsample import ConvexUpsampleFn
+        return ConvexUpsampleFn.apply(flow, mask, factor)
+    return torch_ref.convex_upsample(flow, mask, factor)

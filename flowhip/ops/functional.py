@@ -1,0 +1,100 @@
+"""Autograd bindings for the HIP kernels.
+
+Each Function pairs a hand-written gfx950 forward kernel with an explicit
+backward (HIP kernels where hot, torch/MIOpen composition where not). The
+math contracts live in `torch_ref.py`; GPU unit tests compare both paths.
+
+Layout conventions for the correlation stack (chosen for MFMA/LDS on CDNA4,
+not inherited from the reference):
+  - feature maps enter as (B, D, H, W) fp32 and are staged as k-contiguous
+    bf16 (P, D) operands so both GEMM tiles are ds_read_b128-friendly;
+  - the batched GEMM computes C = alpha * A @ B^T with A (M,K), B (N,K) both
+    row-major bf16, C (M,N) fp32 — forward and both backward products all map
+    onto this one kernel (see CorrVolumeFn docstring).
+"""
+
+import math
+
+import torch
+
+from . import _ext
+
+
+def _bgemm_nt(a, b, alpha):
+    """C[bat] = alpha * A[bat] @ B[bat]^T ; a (Bt,M,K) bf16, b (Bt,N,K) bf16 -> fp32."""
+    return _ext.ext().bgemm_nt(a, b, alpha)
+
+
+class CorrVolumeFn(torch.autograd.Function):
+    """All-pairs correlation C[b,i,j] = <f1[:,i], f2[:,j]>/sqrt(D).
+
+    forward:  C (B,P,P) = 1/sqrt(D) * F1t (B,P,D) @ F2t (B,P,D)^T
+    backward: dF1t = 1/sqrt(D) * dC @ F2  -> gemm_nt(dC, F2 (D,P) as (N=D,K=P)? )
+    Both backward products reuse the same gemm_nt kernel:
+      dF1t (P,D) = gemm_nt(dC  (P,P), F2 (D,P))   [B operand = f2 k-major over P]
+      dF2t (P,D) = gemm_nt(dC^T(P,P), F1 (D,P))
+    Reference math: core/corr.py:47-55. Inputs are cast to bf16 (fp32
+    accumulate in MFMA) — documented deviation from the reference's fp32
+    matmul; tolerance covered by tests/test_gpu_corr.py.
+    """
+
+    @staticmethod
+    def forward(ctx, fmap1, fmap2):
+        B, D, H, W = fmap1.shape
+        P = H * W
+        f1 = fmap1.reshape(B, D, P)
+        f2 = fmap2.reshape(B, D, P)
+        f1t = f1.transpose(1, 2).contiguous().to(torch.bfloat16)  # (B,P,D)
+        f2t = f2.transpose(1, 2).contiguous().to(torch.bfloat16)  # (B,P,D)
+        corr = _bgemm_nt(f1t, f2t, 1.0 / math.sqrt(D))  # (B,P,P) fp32
+        ctx.save_for_backward(f1t, f2t)
+        ctx.shape = (B, D, H, W)
+        return corr.reshape(B * P, 1, H, W)
+
+    @staticmethod
+    def backward(ctx, grad):
+        f1t, f2t = ctx.saved_tensors
+        B, D, H, W = ctx.shape
+        P = H * W
+        alpha = 1.0 / math.sqrt(D)
+        dc = grad.reshape(B, P, P).to(torch.bfloat16).contiguous()
+        dct = grad.reshape(B, P, P).transpose(1, 2).to(torch.bfloat16).contiguous()
+        # dF1t[i,d] = sum_j dC[i,j] * F2t[j,d] -> A=dC (M=P,K=P), B=f2 with
+        # (N=D, K=P) k-major == f2t^T; i.e. B operand is f2 (D,P) row-major? No:
+        # gemm_nt wants B (N,K) row-major = (D,P) with P contiguous == f2 view.
+        f2_kn = f2t.transpose(1, 2).contiguous()  # (B,D,P) bf16, P contiguous
+        f1_kn = f1t.transpose(1, 2).contiguous()
+        df1t = _bgemm_nt(dc, f2_kn, alpha)   # (B,P,D) fp32
+        df2t = _bgemm_nt(dct, f1_kn, alpha)  # (B,P,D) fp32
+        df1 = df1t.transpose(1, 2).reshape(B, D, H, W)
+        df2 = df2t.transpose(1, 2).reshape(B, D, H, W)
+        return df1, df2
+
+
+class CorrLookupFn(torch.autograd.Function):
+    """Fused 4-level (2r+1)^2-tap bilinear window lookup (core/corr.py:23-44).
+
+    coords never require grad in the RAFT iteration loop (coords1 is detached
+    right before every lookup — raft.py:122, raft_nc_dbl.py:149), so backward
+    produces gradients for the pyramid levels only and asserts that contract.
+    """
+
+    @staticmethod
+    def forward(ctx, coords, radius, *pyramid):
+        B, _, H1, W1 = coords.shape
+        assert not coords.requires_grad, (
+            "corr_lookup: coords must be detached (reference contract)")
+        coords_c = coords.contiguous()
+        out = _ext.ext().corr_lookup_fwd(list(pyramid), coords_c, int(radius))
+        ctx.save_for_backward(coords_c)
+        ctx.radius = int(radius)
+        ctx.level_shapes = [tuple(p.shape) for p in pyramid]
+        return out
+
+    @staticmethod
+    def backward(ctx, grad):
+        (coords,) = ctx.saved_tensors
+        grads = _ext.ext().corr_lookup_bwd(
+            grad.contiguous(), coords, ctx.radius,
+            [list(s) for s in ctx.level_shapes])
+        return (None, None, *grads)

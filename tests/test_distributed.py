@@ -1,0 +1,100 @@
+"""Distributed-correctness tests on gloo (CPU, world_size=2):
+DDP gradient all-reduce equivalence — a world_size=2 step with per-rank
+batch 1 must produce the same gradients as a single-process batch-2 step
+(SURVEY.md §4.2 item 4)."""
+
+import os
+
+import numpy as np
+import pytest
+import torch
+import torch.distributed as dist
+import torch.multiprocessing as mp
+
+from flowhip.config.args import default_ncup_args
+from flowhip.models import build_model
+from flowhip import ops
+
+
+def _make_data(seed=7, b=2, h=128, w=128):
+    g = torch.Generator().manual_seed(seed)
+    img1 = torch.rand(b, 3, h, w, generator=g) * 255
+    img2 = torch.rand(b, 3, h, w, generator=g) * 255
+    flow = torch.randn(b, 2, h, w, generator=g)
+    valid = torch.ones(b, h, w)
+    return img1, img2, flow, valid
+
+
+def _build_model(seed=11):
+    torch.manual_seed(seed)
+    args = default_ncup_args(model="raft_nc_dbl", small=False)
+    model = build_model(args)
+    # freeze BN (as in all fine-tuning stages): per-batch BN stats would
+    # differ between batch-2 single-process and per-rank batch-1 otherwise.
+    model.freeze_bn()
+    return model
+
+
+def _single_process_grads():
+    model = _build_model()
+    img1, img2, flow, valid = _make_data()
+    preds = model(img1, img2, iters=2)
+    loss, _ = ops.sequence_loss(preds, flow, valid, 0.85)
+    loss.backward()
+    return {n: p.grad.clone() for n, p in model.named_parameters()
+            if p.grad is not None}
+
+
+def _ddp_worker(rank, world_size, port, out_dir):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world_size)
+    try:
+        model = _build_model()
+        ddp = torch.nn.parallel.DistributedDataParallel(
+            model, bucket_cap_mb=64, gradient_as_bucket_view=True)
+
+        img1, img2, flow, valid = _make_data()
+        sl = slice(rank, rank + 1)  # shard the batch across ranks
+        preds = ddp(img1[sl], img2[sl], iters=2)
+        loss, _ = ops.sequence_loss(preds, flow[sl], valid[sl], 0.85)
+        loss.backward()
+
+        if rank == 0:
+            grads = {n: p.grad.clone() for n, p in model.named_parameters()
+                     if p.grad is not None}
+            torch.save(grads, os.path.join(out_dir, "ddp_grads.pth"))
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_ddp_allreduce_equals_batched(tmp_path):
+    port = int(np.random.default_rng(os.getpid()).integers(20000, 40000))
+    ctx = mp.get_context("spawn")
+    procs = [ctx.Process(target=_ddp_worker, args=(r, 2, port, str(tmp_path)))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(timeout=240)
+        assert p.exitcode == 0
+
+    ddp_grads = torch.load(os.path.join(tmp_path, "ddp_grads.pth"),
+                           weights_only=True)
+    ref_grads = _single_process_grads()
+
+    assert set(ddp_grads.keys()) == set(ref_grads.keys())
+    for name in ref_grads:
+        # DDP averages over ranks; the single-process loss already averages
+        # over the batch -> equal up to numeric noise.
+        assert torch.allclose(ddp_grads[name], ref_grads[name],
+                              atol=1e-5, rtol=1e-4), name
+
+
+def test_distributed_helpers_single_process():
+    from flowhip.engine import distributed
+    rank, world, device = distributed.init_distributed()
+    assert rank == 0 and world == 1
+    assert distributed.is_main()
+    distributed.barrier()  # no-op

@@ -1,0 +1,49 @@
+import torch
+
+from flowhip.utils.geometry import InputPadder, bilinear_sampler, coords_grid, upflow8
+
+
+def test_coords_grid():
+    g = coords_grid(2, 3, 4)
+    assert g.shape == (2, 2, 3, 4)
+    # channel 0 = x, channel 1 = y
+    assert torch.equal(g[0, 0, 0], torch.tensor([0., 1., 2., 3.]))
+    assert torch.equal(g[0, 1, :, 0], torch.tensor([0., 1., 2.]))
+    assert torch.equal(g[0], g[1])
+
+
+def test_bilinear_sampler_identity():
+    img = torch.arange(24, dtype=torch.float32).reshape(1, 1, 4, 6)
+    coords = coords_grid(1, 4, 6).permute(0, 2, 3, 1)  # (1,H,W,2) xy
+    out = bilinear_sampler(img, coords)
+    assert torch.allclose(out, img, atol=1e-5)
+
+
+def test_bilinear_sampler_zero_outside():
+    img = torch.ones(1, 1, 4, 4)
+    coords = torch.tensor([[[[-10.0, -10.0]]]])  # far out of bounds
+    out = bilinear_sampler(img, coords)
+    assert out.abs().max().item() == 0.0
+
+
+def test_input_padder_roundtrip():
+    x = torch.rand(1, 3, 437, 1021)
+    padder = InputPadder(x.shape)
+    (xp,) = padder.pad(x)
+    assert xp.shape[-1] % 8 == 0 and xp.shape[-2] % 8 == 0
+    assert torch.equal(padder.unpad(xp), x)
+
+
+def test_input_padder_kitti_mode():
+    x = torch.rand(1, 3, 370, 1226)
+    padder = InputPadder(x.shape, mode="kitti")
+    (xp,) = padder.pad(x)
+    # kitti pads only at the bottom vertically
+    assert torch.equal(xp[..., :370, 3:-3], x)
+
+
+def test_upflow8_shape_and_scale():
+    flow = torch.ones(1, 2, 4, 5)
+    up = upflow8(flow)
+    assert up.shape == (1, 2, 32, 40)
+    assert torch.allclose(up, 8 * torch.ones_like(up))

@@ -1,0 +1,117 @@
+"""Model-level CPU tests: shapes, fwd+bwd, test_mode, checkpoint keys
+(BASELINE config 1: RAFT-small NCUP, 2 iters, small random frames)."""
+
+import argparse
+
+import pytest
+import torch
+
+from flowhip.config.args import default_ncup_args
+from flowhip.models import RAFT, RAFT_NC_DBL, build_model
+
+
+def small_raft_args(**kw):
+    ns = argparse.Namespace(model="raft", small=True, dropout=0.0,
+                            mixed_precision=False)
+    for k, v in kw.items():
+        setattr(ns, k, v)
+    return ns
+
+
+def make_inputs(b=1, h=128, w=128):
+    # all 4 corr-pyramid levels must stay >=2px (H,W >= 128): at smaller
+    # shapes the 1x1 level hits bilinear_sampler's 2x/(W-1) singularity
+    # (same degenerate behavior as the reference at such shapes).
+    torch.manual_seed(1234)
+    img1 = torch.rand(b, 3, h, w) * 255
+    img2 = torch.rand(b, 3, h, w) * 255
+    return img1, img2
+
+
+def test_raft_small_forward_backward():
+    args = small_raft_args()
+    model = RAFT(args)
+    img1, img2 = make_inputs()
+    preds = model(img1, img2, iters=2)
+    assert len(preds) == 2
+    assert preds[0].shape == (1, 2, 128, 128)
+    loss = sum(p.abs().mean() for p in preds)
+    loss.backward()
+    grads = [p.grad for p in model.parameters() if p.requires_grad]
+    assert all(g is not None for g in grads)
+    assert all(torch.isfinite(g).all() for g in grads)
+
+
+def test_raft_small_test_mode():
+    args = small_raft_args()
+    model = RAFT(args).eval()
+    img1, img2 = make_inputs()
+    with torch.no_grad():
+        flow_low, flow_up = model(img1, img2, iters=2, test_mode=True)
+    assert flow_low.shape == (1, 2, 16, 16)
+    assert flow_up.shape == (1, 2, 128, 128)
+
+
+def test_raft_basic_convex_upsample_path():
+    args = argparse.Namespace(model="raft", small=False, dropout=0.0,
+                              mixed_precision=False)
+    model = RAFT(args)
+    img1, img2 = make_inputs()
+    preds = model(img1, img2, iters=1)
+    assert preds[0].shape == (1, 2, 128, 128)
+
+
+def test_raft_nc_dbl_forward_backward():
+    args = default_ncup_args()
+    model = RAFT_NC_DBL(args)
+    img1, img2 = make_inputs()
+    preds = model(img1, img2, iters=2)
+    assert len(preds) == 2
+    assert preds[0].shape == (1, 2, 128, 128)
+    loss = sum(p.abs().mean() for p in preds)
+    loss.backward()
+    # all trainable params get grads (mask head removed; shared encoder keeps
+    # nconv params live through the full-res path)
+    missing = [n for n, p in model.named_parameters()
+               if p.requires_grad and p.grad is None]
+    assert missing == []
+
+
+def test_raft_nc_dbl_mask_head_removed():
+    model = RAFT_NC_DBL(default_ncup_args())
+    assert len(model.update_block.mask) == 0
+    keys = model.state_dict().keys()
+    assert not any(k.startswith("update_block.mask") for k in keys)
+
+
+def test_raft_nc_dbl_freeze_raft():
+    args = default_ncup_args(freeze_raft=True)
+    model = RAFT_NC_DBL(args)
+    assert all(not p.requires_grad for p in model.fnet.parameters())
+    assert all(p.requires_grad for p in model.upsampler.parameters())
+
+
+def test_flow_init_warm_start():
+    args = small_raft_args()
+    model = RAFT(args).eval()
+    img1, img2 = make_inputs()
+    with torch.no_grad():
+        flow_init = torch.ones(1, 2, 16, 16)
+        low, _ = model(img1, img2, iters=1, flow_init=flow_init, test_mode=True)
+    assert torch.isfinite(low).all()
+
+
+def test_build_model_aliases():
+    args = default_ncup_args(model="raft_nc")
+    with pytest.warns(UserWarning):
+        m = build_model(args)
+    assert isinstance(m, RAFT_NC_DBL)
+
+
+def test_model_iter_count_affects_predictions():
+    args = default_ncup_args()
+    model = RAFT_NC_DBL(args).eval()
+    img1, img2 = make_inputs()
+    with torch.no_grad():
+        preds = model(img1, img2, iters=3)
+    assert len(preds) == 3
