@@ -35,7 +35,8 @@ def corr_pyramid(corr, num_levels=4):
 
 
 def corr_lookup(pyramid, coords, radius):
-    if _ext.use_hip(coords):
+    # HIP kernel instantiates the RAFT radii (3 small / 4 basic)
+    if radius in (3, 4) and _ext.use_hip(coords):
         from .functional import CorrLookupFn
         return CorrLookupFn.apply(coords, radius, *pyramid)
     return torch_ref.corr_lookup(pyramid, coords, radius)

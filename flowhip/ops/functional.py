@@ -20,9 +20,20 @@ import torch
 from . import _ext
 
 
+def _pad_k(t):
+    """Zero-pad the trailing (K) dim to a multiple of 64 (kernel tile depth).
+    Zero columns contribute nothing to the dot products."""
+    k = t.shape[-1]
+    pad = (-k) % 64
+    if pad == 0:
+        return t
+    return torch.nn.functional.pad(t, (0, pad))
+
+
 def _bgemm_nt(a, b, alpha):
     """C[bat] = alpha * A[bat] @ B[bat]^T ; a (Bt,M,K) bf16, b (Bt,N,K) bf16 -> fp32."""
-    return _ext.ext().bgemm_nt(a, b, alpha)
+    return _ext.ext().bgemm_nt(_pad_k(a).contiguous(), _pad_k(b).contiguous(),
+                               alpha)
 
 
 class CorrVolumeFn(torch.autograd.Function):

@@ -1,0 +1,184 @@
+"""GPU kernel tests (MI355X): HIP kernels vs the pure-torch fp32 oracle
+(SURVEY.md §4.2 item 2). Run via gpurun: pytest tests -m gpu."""
+
+import math
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+
+def _dev():
+    return torch.device("cuda:0")
+
+
+@pytest.fixture(scope="module")
+def ext():
+    import flowhip._C as C
+    return C
+
+
+class TestBgemmNT:
+    def test_square_multiple_of_tile(self, ext):
+        torch.manual_seed(0)
+        a = torch.randn(2, 256, 256, device=_dev()).to(torch.bfloat16)
+        b = torch.randn(2, 256, 256, device=_dev()).to(torch.bfloat16)
+        c = ext.bgemm_nt(a, b, 0.5)
+        ref = 0.5 * torch.matmul(a.float(), b.float().transpose(1, 2))
+        assert c.shape == ref.shape
+        torch.testing.assert_close(c, ref, atol=2e-3, rtol=2e-3)
+
+    def test_ragged_edges(self, ext):
+        # M, N not multiples of 128 (FlyingChairs P=2852 case)
+        torch.manual_seed(1)
+        a = torch.randn(1, 300, 128, device=_dev()).to(torch.bfloat16)
+        b = torch.randn(1, 177, 128, device=_dev()).to(torch.bfloat16)
+        c = ext.bgemm_nt(a, b, 1.0)
+        ref = torch.matmul(a.float(), b.float().transpose(1, 2))
+        torch.testing.assert_close(c, ref, atol=2e-3, rtol=2e-3)
+
+    def test_asymmetric_catches_transpose(self, ext):
+        # asymmetric operands catch a row/col swap (guide §3 A=I-check)
+        a = torch.zeros(1, 128, 64, device=_dev())
+        a[0, 3, :] = 1.0
+        b = torch.zeros(1, 128, 64, device=_dev())
+        b[0, 7, 0] = 2.0
+        c = ext.bgemm_nt(a.to(torch.bfloat16), b.to(torch.bfloat16), 1.0)
+        assert c[0, 3, 7].item() == pytest.approx(2.0, abs=1e-2)
+        assert c[0, 7, 3].item() == 0.0
+
+
+class TestCorrVolume:
+    def test_forward_matches_ref(self):
+        from flowhip.ops import torch_ref
+        from flowhip.ops.functional import CorrVolumeFn
+        torch.manual_seed(2)
+        B, D, H, W = 2, 256, 14, 32
+        f1 = torch.randn(B, D, H, W, device=_dev())
+        f2 = torch.randn(B, D, H, W, device=_dev())
+        out = CorrVolumeFn.apply(f1, f2)
+        ref = torch_ref.corr_volume(f1, f2)
+        assert out.shape == ref.shape
+        # bf16 inputs, fp32 accumulate: |err| ~ sqrt(D)*eps_bf16*|f|^2
+        torch.testing.assert_close(out, ref, atol=5e-2, rtol=5e-2)
+
+    def test_backward_matches_ref(self):
+        from flowhip.ops import torch_ref
+        from flowhip.ops.functional import CorrVolumeFn
+        torch.manual_seed(3)
+        B, D, H, W = 1, 128, 10, 23  # P=230: exercises K padding in backward
+        f1 = torch.randn(B, D, H, W, device=_dev(), requires_grad=True)
+        f2 = torch.randn(B, D, H, W, device=_dev(), requires_grad=True)
+
+        out = CorrVolumeFn.apply(f1, f2)
+        g = torch.randn_like(out)
+        d1, d2 = torch.autograd.grad(out, (f1, f2), g)
+
+        f1r = f1.detach().clone().requires_grad_(True)
+        f2r = f2.detach().clone().requires_grad_(True)
+        ref = torch_ref.corr_volume(f1r, f2r)
+        r1, r2 = torch.autograd.grad(ref, (f1r, f2r), g)
+
+        torch.testing.assert_close(d1, r1, atol=0.5, rtol=5e-2)
+        torch.testing.assert_close(d2, r2, atol=0.5, rtol=5e-2)
+
+
+class TestCorrLookup:
+    @pytest.mark.parametrize("radius", [3, 4])
+    def test_forward_matches_ref(self, radius):
+        from flowhip import ops
+        from flowhip.ops import torch_ref
+        from flowhip.ops.functional import CorrLookupFn
+        torch.manual_seed(4)
+        B, H, W = 1, 16, 24
+        P = H * W
+        l0 = torch.randn(B * P, 1, H, W, device=_dev())
+        pyramid = [p.detach() for p in torch_ref.corr_pyramid(l0, 4)]
+        coords = (torch.rand(B, 2, H, W, device=_dev()) *
+                  torch.tensor([W, H], device=_dev()).view(1, 2, 1, 1))
+
+        out = CorrLookupFn.apply(coords, radius, *pyramid)
+        ref = torch_ref.corr_lookup(pyramid, coords, radius)
+        assert out.shape == ref.shape
+        torch.testing.assert_close(out, ref, atol=1e-4, rtol=1e-4)
+
+    def test_backward_matches_ref(self):
+        from flowhip.ops import torch_ref
+        from flowhip.ops.functional import CorrLookupFn
+        torch.manual_seed(5)
+        B, H, W, radius = 1, 12, 16, 4
+        P = H * W
+        l0 = torch.randn(B * P, 1, H, W, device=_dev(), requires_grad=True)
+        pyramid = torch_ref.corr_pyramid(l0, 4)
+        coords = (torch.rand(B, 2, H, W, device=_dev()) *
+                  torch.tensor([W, H], device=_dev()).view(1, 2, 1, 1))
+
+        out = CorrLookupFn.apply(coords, radius, *pyramid)
+        g = torch.randn_like(out)
+        (dl0,) = torch.autograd.grad(out, l0, g)
+
+        l0r = l0.detach().clone().requires_grad_(True)
+        pyr_r = torch_ref.corr_pyramid(l0r, 4)
+        ref = torch_ref.corr_lookup(pyr_r, coords, radius)
+        (dl0r,) = torch.autograd.grad(ref, l0r, g)
+
+        torch.testing.assert_close(dl0, dl0r, atol=1e-3, rtol=1e-3)
+
+    def test_coords_must_be_detached(self):
+        from flowhip.ops.functional import CorrLookupFn
+        coords = torch.rand(1, 2, 8, 8, device=_dev(), requires_grad=True)
+        l0 = torch.randn(64, 1, 8, 8, device=_dev())
+        with pytest.raises(AssertionError):
+            CorrLookupFn.apply(coords, 4, l0)
+
+
+class TestModelGPU:
+    def test_raft_nc_dbl_train_step(self):
+        from flowhip import ops
+        from flowhip.config.args import default_ncup_args
+        from flowhip.models import build_model
+
+        torch.manual_seed(1234)
+        args = default_ncup_args(model="raft_nc_dbl", mixed_precision=True)
+        model = build_model(args).to(_dev())
+        model.train()
+        model.freeze_bn()
+
+        h, w = 128, 256
+        img1 = torch.rand(2, 3, h, w, device=_dev()) * 255
+        img2 = torch.rand(2, 3, h, w, device=_dev()) * 255
+        flow_gt = torch.randn(2, 2, h, w, device=_dev())
+        valid = torch.ones(2, h, w, device=_dev())
+
+        preds = model(img1, img2, iters=3)
+        loss, metrics = ops.sequence_loss(preds, flow_gt, valid, 0.85)
+        loss.backward()
+        torch.cuda.synchronize()
+        assert torch.isfinite(loss).item()
+        for n, p in model.named_parameters():
+            if p.requires_grad:
+                assert p.grad is not None and torch.isfinite(p.grad).all(), n
+
+    def test_gpu_matches_cpu_forward(self):
+        """Same weights + inputs: GPU (HIP kernels, fp32 autocast off) vs CPU
+        reference path agree within bf16-corr tolerance."""
+        from flowhip.config.args import default_ncup_args
+        from flowhip.models import build_model
+
+        torch.manual_seed(1234)
+        args = default_ncup_args(model="raft_nc_dbl", mixed_precision=False)
+        model = build_model(args)
+        model.eval()
+
+        img1 = torch.rand(1, 3, 128, 128) * 255
+        img2 = torch.rand(1, 3, 128, 128) * 255
+
+        with torch.no_grad():
+            low_cpu, up_cpu = model(img1, img2, iters=2, test_mode=True)
+            model_gpu = model.to(_dev())
+            low_gpu, up_gpu = model_gpu(img1.to(_dev()), img2.to(_dev()),
+                                        iters=2, test_mode=True)
+
+        torch.testing.assert_close(low_gpu.cpu(), low_cpu, atol=0.05, rtol=0.05)
+        torch.testing.assert_close(up_gpu.cpu(), up_cpu, atol=0.5, rtol=0.1)
