@@ -1,0 +1,81 @@
+"""Autograd binding for the fused normalized-convolution kernel.
+
+Forward: one fused HIP kernel (csrc/nconv.hip) producing (nconv, cout).
+Backward: torch/MIOpen composition — the op's backward is conv-shaped and
+runs a small fraction of step time; denom is reconstructed from the saved
+cout (denom = cout * sum(w)) so no extra forward tensor is stored.
+
+Math (d denotes upstream grads):
+  nomin = (out − bias) · (denom + eps)
+  dnomin = dout / (denom + eps)
+  ddenom = −dout · nomin / (denom+eps)² + dcout / s
+  ddata  = conf · convT(dnomin, w)
+  dconf  = data · convT(dnomin, w) + convT(ddenom, w)
+  dw     = ∇w[conv(data·conf) vs dnomin] + ∇w[conv(conf) vs ddenom]
+           − Σ(cout·dcout)/s   (through s = Σw per out-channel)
+  dbias  = Σ dout
+Reference math: nconv_modules.py:164-199.
+"""
+
+import torch
+import torch.nn.functional as F
+
+from . import _ext
+
+
+class NConv2dFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, data, conf, weight, bias, padding, eps, prop_conf):
+        data = data.contiguous()
+        conf = conf.contiguous()
+        weight = weight.contiguous()
+        b = bias.contiguous() if bias is not None else None
+        out, cout = _ext.ext().nconv_fwd(data, conf, weight, b)
+        if b is not None:
+            ctx.save_for_backward(data, conf, weight, out, cout, b)
+        else:
+            ctx.save_for_backward(data, conf, weight, out, cout)
+        ctx.has_bias = bias is not None
+        ctx.padding = padding
+        ctx.eps = eps
+        return out, cout
+
+    @staticmethod
+    def backward(ctx, gout, gcout):
+        data, conf, weight, out, cout, *maybe_bias = ctx.saved_tensors
+        pad = ctx.padding if isinstance(ctx.padding, int) else ctx.padding[0]
+        eps = ctx.eps
+
+        s = weight.sum(dim=(1, 2, 3))                      # (Co)
+        denom = cout * s.view(1, -1, 1, 1)
+        de = denom + eps
+
+        gout = gout.contiguous()
+        gcout = gcout.contiguous() if gcout is not None else None
+
+        # nomin/de = out (− bias when present)
+        ratio = out
+        if ctx.has_bias:
+            ratio = out - maybe_bias[0].view(1, -1, 1, 1)
+
+        dnomin = gout / de
+        ddenom = -gout * ratio / de
+        if gcout is not None:
+            ddenom = ddenom + gcout / s.view(1, -1, 1, 1)
+
+        g_dc = F.conv_transpose2d(dnomin, weight, padding=pad)
+        ddata = conf * g_dc
+        dconf = data * g_dc + F.conv_transpose2d(ddenom, weight, padding=pad)
+
+        dweight = None
+        if ctx.needs_input_grad[2]:
+            dweight = torch.nn.grad.conv2d_weight(
+                (data * conf), weight.shape, dnomin, stride=1, padding=pad)
+            dweight = dweight + torch.nn.grad.conv2d_weight(
+                conf, weight.shape, ddenom, stride=1, padding=pad)
+            if gcout is not None:
+                ds = -(cout * gcout).sum(dim=(0, 2, 3)) / s
+                dweight = dweight + ds.view(-1, 1, 1, 1)
+
+        dbias = gout.sum(dim=(0, 2, 3)) if ctx.has_bias else None
+        return ddata, dconf, dweight, dbias, None, None, None
