@@ -278,3 +278,33 @@ class TestNConv:
             del os.environ["FLOWHIP_FORCE_REF"]
         torch.testing.assert_close(out, rout, atol=1e-4, rtol=1e-4)
         torch.testing.assert_close(cout, rcout, atol=1e-4, rtol=1e-4)
+
+
+class TestHipGraph:
+    def test_graphed_inference_matches_eager(self):
+        from flowhip.config.args import default_ncup_args
+        from flowhip.engine.graph import GraphedInference
+        from flowhip.models import build_model
+
+        torch.manual_seed(1234)
+        args = default_ncup_args(model="raft_nc_dbl", mixed_precision=True)
+        model = build_model(args).to(_dev()).eval()
+
+        shape = (1, 3, 128, 256)
+        img1 = torch.rand(shape, device=_dev()) * 255
+        img2 = torch.rand(shape, device=_dev()) * 255
+
+        with torch.no_grad():
+            low_e, up_e = model(img1, img2, iters=4, test_mode=True)
+        g = GraphedInference(model, shape, iters=4)
+        low_g, up_g = g(img1, img2)
+
+        torch.testing.assert_close(low_g, low_e, atol=1e-3, rtol=1e-3)
+        torch.testing.assert_close(up_g, up_e, atol=1e-2, rtol=1e-2)
+
+        # replay with different inputs must track eager
+        img3 = torch.rand(shape, device=_dev()) * 255
+        with torch.no_grad():
+            low_e2, up_e2 = model(img3, img2, iters=4, test_mode=True)
+        low_g2, up_g2 = g(img3, img2)
+        torch.testing.assert_close(up_g2, up_e2, atol=1e-2, rtol=1e-2)
