@@ -53,6 +53,8 @@ def module_classes_to_dict(module, include_classes="*", exclude_classes=()):
 def _type_from_default(value):
     if isinstance(value, bool):
         return str2bool
+    if isinstance(value, list):
+        return str2intlist
     return type(value)
 
 
@@ -113,16 +115,23 @@ def add_arguments_for_module(parser, module, name, default_class,
 
 def add_ncup_module_flags(parser, argv=None):
     """The three reflective families of the reference CLI
-    (train.py:299-342, evaluate.py:198-241)."""
+    (train.py:299-342, evaluate.py:198-241).
+
+    Unlike the reference (which crashes on a bare `--model raft_nc_dbl` run
+    because no flags exist until a class is chosen), the families default to
+    the shipped NCUP configuration — the canonical scripts
+    pass every flag explicitly, so their behavior is unchanged."""
     from ..nn import interp_weights_est as interp_weights_est_mod
     from ..nn import nconv as nconv_mod
     from ..nn import upsampler as upsampler_mod
 
     add_arguments_for_module(
-        parser, upsampler_mod, name="final_upsampling", default_class=None,
+        parser, upsampler_mod, name="final_upsampling",
+        default_class="NConvUpsampler",
         exclude_classes=["_*"],
         exclude_params=["self", "args", "interpolation_net", "weights_est_net",
                         "size"],
+        param_defaults={"scale": 4},
         forced_default_types={"scale": int,
                               "use_data_for_guidance": str2bool,
                               "channels_to_batch": str2bool,
@@ -131,9 +140,10 @@ def add_ncup_module_flags(parser, argv=None):
         argv=argv)
 
     add_arguments_for_module(
-        parser, nconv_mod, name="interp_net", default_class=None,
+        parser, nconv_mod, name="interp_net", default_class="NConvUNet",
         exclude_classes=["_*"],
         exclude_params=["self", "args"],
+        param_defaults={"num_downsampling": 1, "use_double_conv": False},
         forced_default_types={"encoder_filter_sz": int,
                               "decoder_filter_sz": int,
                               "out_filter_sz": int,
@@ -143,8 +153,10 @@ def add_ncup_module_flags(parser, argv=None):
 
     add_arguments_for_module(
         parser, interp_weights_est_mod, name="weights_est_net",
-        default_class=None, exclude_classes=["_*"],
+        default_class="Simple", exclude_classes=["_*"],
         exclude_params=["self", "args", "out_ch", "final_act"],
+        param_defaults={"num_ch": [64, 32], "filter_sz": [3, 3, 1],
+                        "dilation": [1, 1, 1]},
         unknown_default_types={"num_ch": str2intlist,
                                "filter_sz": str2intlist},
         forced_default_types={"dilation": str2intlist},

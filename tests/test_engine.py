@@ -1,0 +1,64 @@
+"""Engine tests (CPU): full train() smoke on the synthetic stage, synthetic
+validation, logger file outputs."""
+
+import json
+import os
+
+import pytest
+import torch
+
+from flowhip.config import build_train_parser, finalize_args
+from flowhip.config.args import default_ncup_args
+from flowhip.engine.evaluate import validate_synthetic
+from flowhip.models import build_model
+
+
+@pytest.mark.timeout(600)
+def test_train_synthetic_two_steps(tmp_path, monkeypatch):
+    """train() runs end-to-end for 2 steps on CPU (BASELINE config 1
+    plumbing: small model, 2 iters, 128x128 synthetic pairs) and writes the
+    reference checkpoint layout."""
+    monkeypatch.chdir(tmp_path)
+
+    argv = ["--name", "smoke", "--model", "raft_nc_dbl", "--stage", "synthetic",
+            "--small", "--num_steps", "2", "--batch_size", "1",
+            "--image_size", "128", "128", "--iters", "2", "--lr", "1e-4",
+            "--num_workers", "0"]
+    parser = build_train_parser(argv=argv)
+    args = finalize_args(parser.parse_args(argv))
+
+    from flowhip.engine.train import train
+    path = train(args)
+    assert path == "checkpoints/smoke/final_model.pth"
+    assert os.path.exists(path)
+
+    sd = torch.load(path, weights_only=True)
+    assert all(k.startswith("module.") for k in sd)
+
+    log = open("checkpoints/smoke/log.txt").read()
+    assert "Parameter Count" in log
+
+
+def test_validate_synthetic_runs():
+    args = default_ncup_args(small=True)
+    model = build_model(args)
+    res = validate_synthetic(model, iters=2, n_samples=1, image_size=(128, 128))
+    assert "synthetic" in res and res["synthetic"] > 0
+
+
+def test_logger_files(tmp_path):
+    import argparse
+    from flowhip.engine.logger import Logger
+
+    args = argparse.Namespace(name="logtest")
+    logger = Logger(None, args, sum_freq=2, run_dir=str(tmp_path / "run"))
+    logger.scheduler = type("S", (), {"get_last_lr": lambda self: [1e-4]})()
+    for i in range(4):
+        logger.push({"loss": float(i)}, n_imgs=2)
+    logger.write_dict({"synthetic": 1.5})
+    logger.close()
+
+    jsonl = [json.loads(l) for l in open(tmp_path / "run" / "metrics.jsonl")]
+    assert any("loss" in r for r in jsonl)
+    assert any("validation" in r for r in jsonl)
+    assert (tmp_path / "run" / "log.txt").exists()
