@@ -44,6 +44,9 @@ def parse_args():
     p.add_argument("--profile-steps", type=int, default=0,
                    help="if >0, run only this many timed steps per rank "
                    "without JSON (for rocprofv3 kernel capture)")
+    p.add_argument("--torch-profile", default=None,
+                   help="write torch.profiler kernel table for steady-state "
+                   "steps to this path (after warmup; excludes MIOpen tuning)")
     return p.parse_args()
 
 
@@ -106,6 +109,25 @@ def main():
 
     for _ in range(cli.warmup):
         step()
+
+    if cli.torch_profile:
+        from torch.profiler import ProfilerActivity, profile
+        if on_gpu:
+            torch.cuda.synchronize()
+        with profile(activities=[ProfilerActivity.CPU, ProfilerActivity.CUDA],
+                     record_shapes=False) as prof:
+            for _ in range(max(cli.steps, 2)):
+                step()
+            if on_gpu:
+                torch.cuda.synchronize()
+        table = prof.key_averages().table(
+            sort_by="self_cuda_time_total" if on_gpu else "self_cpu_time_total",
+            row_limit=60)
+        with open(cli.torch_profile, "w") as f:
+            f.write(table)
+        if rank == 0:
+            print(table[:4000])
+        return
 
     if cli.profile_steps > 0:
         if on_gpu:
