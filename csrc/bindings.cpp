@@ -31,6 +31,15 @@ void flowhip_nconv_fwd_launch(const float* data, const float* conf,
                               const float* weight, const float* bias,
                               float* out, float* cout, int N, int Ci, int Co,
                               int H, int W, int K, hipStream_t stream);
+void flowhip_nconv_bwd_data_launch(const float* dnomin, const float* ddenom,
+                                   const float* data, const float* conf,
+                                   const float* weight, float* ddata,
+                                   float* dconf, int N, int Ci, int Co, int H,
+                                   int W, int K, hipStream_t stream);
+void flowhip_nconv_wrw_launch(const float* dnomin, const float* ddenom,
+                              const float* data, const float* conf,
+                              float* dweight, int N, int Ci, int Co, int H,
+                              int W, int K, hipStream_t stream);
 
 namespace {
 
@@ -180,6 +189,35 @@ std::vector<torch::Tensor> nconv_fwd(torch::Tensor data, torch::Tensor conf,
   return {out, cout};
 }
 
+std::vector<torch::Tensor> nconv_bwd(torch::Tensor dnomin,
+                                     torch::Tensor ddenom, torch::Tensor data,
+                                     torch::Tensor conf,
+                                     torch::Tensor weight) {
+  for (auto* t : {&dnomin, &ddenom, &data, &conf, &weight}) {
+    TORCH_CHECK(t->is_cuda() && t->is_contiguous() &&
+                t->dtype() == torch::kFloat32);
+  }
+  const int N = data.size(0), Ci = data.size(1), H = data.size(2),
+            W = data.size(3);
+  const int Co = weight.size(0), K = weight.size(2);
+
+  auto ddata = torch::empty_like(data);
+  auto dconf = torch::empty_like(conf);
+  auto dweight = torch::zeros_like(weight);
+  const c10::cuda::CUDAGuard guard(data.device());
+  hipStream_t stream = at::cuda::getCurrentCUDAStream().stream();
+  flowhip_nconv_bwd_data_launch(
+      dnomin.data_ptr<float>(), ddenom.data_ptr<float>(),
+      data.data_ptr<float>(), conf.data_ptr<float>(),
+      weight.data_ptr<float>(), ddata.data_ptr<float>(),
+      dconf.data_ptr<float>(), N, Ci, Co, H, W, K, stream);
+  flowhip_nconv_wrw_launch(
+      dnomin.data_ptr<float>(), ddenom.data_ptr<float>(),
+      data.data_ptr<float>(), conf.data_ptr<float>(),
+      dweight.data_ptr<float>(), N, Ci, Co, H, W, K, stream);
+  return {ddata, dconf, dweight};
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -194,4 +232,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("convex_up_bwd", &convex_up_bwd, "backward of convex_up_fwd");
   m.def("nconv_fwd", &nconv_fwd,
         "fused normalized convolution forward (out, cout)");
+  m.def("nconv_bwd", &nconv_bwd,
+        "fused normalized convolution backward (ddata, dconf, dweight)");
 }

@@ -1,9 +1,12 @@
 """Autograd binding for the fused normalized-convolution kernel.
 
 Forward: one fused HIP kernel (csrc/nconv.hip) producing (nconv, cout).
-Backward: torch/MIOpen composition — the op's backward is conv-shaped and
-runs a small fraction of step time; denom is reconstructed from the saved
-cout (denom = cout * sum(w)) so no extra forward tensor is stored.
+Backward: two fused HIP kernels (bwd-data + weight-grad); only the tiny
+elementwise dnomin/ddenom pre-computation stays in torch. denom is
+reconstructed from the saved cout (denom = cout * sum(w)) so no extra
+forward tensor is stored. (The original torch/MIOpen composition fell into
+naive_conv wrw fallbacks at ~240 ms/call for these tiny-channel full-res
+shapes — see profiles/.)
 
 Math (d denotes upstream grads):
   nomin = (out − bias) · (denom + eps)
@@ -18,7 +21,6 @@ Reference math: nconv_modules.py:164-199.
 """
 
 import torch
-import torch.nn.functional as F
 
 from . import _ext
 
@@ -58,24 +60,20 @@ class NConv2dFn(torch.autograd.Function):
         if ctx.has_bias:
             ratio = out - maybe_bias[0].view(1, -1, 1, 1)
 
-        dnomin = gout / de
+        dnomin = (gout / de).contiguous()
         ddenom = -gout * ratio / de
         if gcout is not None:
             ddenom = ddenom + gcout / s.view(1, -1, 1, 1)
+        ddenom = ddenom.contiguous()
 
-        g_dc = F.conv_transpose2d(dnomin, weight, padding=pad)
-        ddata = conf * g_dc
-        dconf = data * g_dc + F.conv_transpose2d(ddenom, weight, padding=pad)
-
-        dweight = None
-        if ctx.needs_input_grad[2]:
-            dweight = torch.nn.grad.conv2d_weight(
-                (data * conf), weight.shape, dnomin, stride=1, padding=pad)
-            dweight = dweight + torch.nn.grad.conv2d_weight(
-                conf, weight.shape, ddenom, stride=1, padding=pad)
-            if gcout is not None:
-                ds = -(cout * gcout).sum(dim=(0, 2, 3)) / s
-                dweight = dweight + ds.view(-1, 1, 1, 1)
+        ddata, dconf, dweight = _ext.ext().nconv_bwd(dnomin, ddenom, data,
+                                                     conf, weight)
+        if not ctx.needs_input_grad[2]:
+            dweight = None
+        elif gcout is not None:
+            # cout = denom/s depends on s = sum(w) per out channel
+            ds = -(cout * gcout).sum(dim=(0, 2, 3)) / s
+            dweight = dweight + ds.view(-1, 1, 1, 1)
 
         dbias = gout.sum(dim=(0, 2, 3)) if ctx.has_bias else None
         return ddata, dconf, dweight, dbias, None, None, None
