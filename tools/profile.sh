@@ -1,0 +1,40 @@
+#!/bin/bash
+# rocprofv3 capture recipes for the MI355X box (run via gpurun).
+#
+# IMPORTANT pool rule: never combine --pmc (or -i counter files) with
+# -s/--sys-trace, -r/--runtime-trace or hip/hsa/memory-copy/scratch-memory/
+# marker trace domains in ONE invocation — collect counters in their own run.
+#
+# Usage:  bash tools/profile.sh trace   [out_dir]   # kernel trace + stats
+#         bash tools/profile.sh pmc     [out_dir]   # MFMA/LDS/HBM counters
+#         bash tools/profile.sh infer   [out_dir]   # inference trace
+set -e
+MODE=${1:-trace}
+OUT=${2:-gpurun_out/prof_$MODE}
+cd /tmp && export TMPDIR=/tmp
+REPO=${GRAFT_REPO_ROOT:-/root/repo}
+mkdir -p "$REPO/$OUT"
+
+case "$MODE" in
+  trace)
+    rocprofv3 --kernel-trace --stats --output-format csv -d /tmp/prof -o step \
+      -- python "$REPO/bench.py" --steps 4 --warmup 8 --profile-steps 4
+    cp /tmp/prof/*stats*.csv "$REPO/$OUT/" 2>/dev/null || true
+    ;;
+  pmc)
+    # counters only (no trace domains!); SQ has 8 slots, TCC 4
+    rocprofv3 --pmc SQ_WAVE_CYCLES SQ_WAIT_ANY SQ_WAIT_INST_ANY \
+      SQ_LDS_BANK_CONFLICT SQ_VALU_MFMA_BUSY_CYCLES \
+      --output-format csv -d /tmp/pmc -o step \
+      -- python "$REPO/bench.py" --steps 2 --warmup 6 --profile-steps 2
+    cp /tmp/pmc/*.csv "$REPO/$OUT/" 2>/dev/null || true
+    ;;
+  infer)
+    rocprofv3 --kernel-trace --stats --output-format csv -d /tmp/prof -o infer \
+      -- python "$REPO/tools/bench_infer.py"
+    cp /tmp/prof/*stats*.csv "$REPO/$OUT/" 2>/dev/null || true
+    ;;
+  *)
+    echo "unknown mode $MODE"; exit 1;;
+esac
+ls -la "$REPO/$OUT/"
