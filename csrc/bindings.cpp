@@ -40,6 +40,21 @@ void flowhip_nconv_wrw_launch(const float* dnomin, const float* ddenom,
                               const float* data, const float* conf,
                               float* dweight, int N, int Ci, int Co, int H,
                               int W, int K, hipStream_t stream);
+int flowhip_nconv_tiled_nblocks(int N, int H, int W);
+bool flowhip_nconv_fwd_tiled_launch(const float* data, const float* conf,
+                                    const float* weight, const float* bias,
+                                    float* out, float* cout, int N, int Ci,
+                                    int Co, int H, int W, int K,
+                                    hipStream_t stream);
+bool flowhip_nconv_bwd_data_tiled_launch(
+    const float* dnomin, const float* ddenom, const float* data,
+    const float* conf, const float* weight, float* ddata, float* dconf, int N,
+    int Ci, int Co, int H, int W, int K, hipStream_t stream);
+bool flowhip_nconv_wrw_tiled_launch(const float* dnomin, const float* ddenom,
+                                    const float* data, const float* conf,
+                                    float* partials, float* dweight, int N,
+                                    int Ci, int Co, int H, int W, int K,
+                                    hipStream_t stream);
 
 namespace {
 
@@ -182,10 +197,15 @@ std::vector<torch::Tensor> nconv_fwd(torch::Tensor data, torch::Tensor conf,
   }
   const c10::cuda::CUDAGuard guard(data.device());
   hipStream_t stream = at::cuda::getCurrentCUDAStream().stream();
-  flowhip_nconv_fwd_launch(data.data_ptr<float>(), conf.data_ptr<float>(),
-                           weight.data_ptr<float>(), bptr,
-                           out.data_ptr<float>(), cout.data_ptr<float>(), N,
-                           Ci, Co, H, W, K, stream);
+  if (!flowhip_nconv_fwd_tiled_launch(
+          data.data_ptr<float>(), conf.data_ptr<float>(),
+          weight.data_ptr<float>(), bptr, out.data_ptr<float>(),
+          cout.data_ptr<float>(), N, Ci, Co, H, W, K, stream)) {
+    flowhip_nconv_fwd_launch(data.data_ptr<float>(), conf.data_ptr<float>(),
+                             weight.data_ptr<float>(), bptr,
+                             out.data_ptr<float>(), cout.data_ptr<float>(), N,
+                             Ci, Co, H, W, K, stream);
+  }
   return {out, cout};
 }
 
@@ -206,15 +226,34 @@ std::vector<torch::Tensor> nconv_bwd(torch::Tensor dnomin,
   auto dweight = torch::zeros_like(weight);
   const c10::cuda::CUDAGuard guard(data.device());
   hipStream_t stream = at::cuda::getCurrentCUDAStream().stream();
-  flowhip_nconv_bwd_data_launch(
-      dnomin.data_ptr<float>(), ddenom.data_ptr<float>(),
-      data.data_ptr<float>(), conf.data_ptr<float>(),
-      weight.data_ptr<float>(), ddata.data_ptr<float>(),
-      dconf.data_ptr<float>(), N, Ci, Co, H, W, K, stream);
-  flowhip_nconv_wrw_launch(
-      dnomin.data_ptr<float>(), ddenom.data_ptr<float>(),
-      data.data_ptr<float>(), conf.data_ptr<float>(),
-      dweight.data_ptr<float>(), N, Ci, Co, H, W, K, stream);
+  if (!flowhip_nconv_bwd_data_tiled_launch(
+          dnomin.data_ptr<float>(), ddenom.data_ptr<float>(),
+          data.data_ptr<float>(), conf.data_ptr<float>(),
+          weight.data_ptr<float>(), ddata.data_ptr<float>(),
+          dconf.data_ptr<float>(), N, Ci, Co, H, W, K, stream)) {
+    flowhip_nconv_bwd_data_launch(
+        dnomin.data_ptr<float>(), ddenom.data_ptr<float>(),
+        data.data_ptr<float>(), conf.data_ptr<float>(),
+        weight.data_ptr<float>(), ddata.data_ptr<float>(),
+        dconf.data_ptr<float>(), N, Ci, Co, H, W, K, stream);
+  }
+  bool wrw_done = false;
+  if (K == 3 || K == 5) {
+    const int nblocks = flowhip_nconv_tiled_nblocks(N, H, W);
+    auto partials = torch::empty({nblocks, (long)Co * Ci * K * K},
+                                 data.options());
+    wrw_done = flowhip_nconv_wrw_tiled_launch(
+        dnomin.data_ptr<float>(), ddenom.data_ptr<float>(),
+        data.data_ptr<float>(), conf.data_ptr<float>(),
+        partials.data_ptr<float>(), dweight.data_ptr<float>(), N, Ci, Co, H,
+        W, K, stream);
+  }
+  if (!wrw_done) {
+    flowhip_nconv_wrw_launch(
+        dnomin.data_ptr<float>(), ddenom.data_ptr<float>(),
+        data.data_ptr<float>(), conf.data_ptr<float>(),
+        dweight.data_ptr<float>(), N, Ci, Co, H, W, K, stream);
+  }
   return {ddata, dconf, dweight};
 }
 

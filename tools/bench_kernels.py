@@ -92,6 +92,29 @@ def main():
     shapes = [list(p.shape) for p in pyramid]
     t = timeit(lambda: C.corr_lookup_bwd(g, coords, 4, shapes))
     print(json.dumps({"probe": "corr_lookup_bwd", "ms": t * 1e3}))
+    del corr, pyramid, g
+
+    # --- nconv family (kernels #6) at the NCUP bench shapes:
+    # channels_to_batch full-res grids, N = 2*batch, H,W = 8*h8, 8*w8 ---
+    Hf, Wf = 8 * H, 8 * W
+    Nf = 2 * B
+    for (ci, co, k) in [(1, 2, 5), (2, 2, 5), (4, 2, 3), (2, 1, 1)]:
+        data = torch.randn(Nf, ci, Hf, Wf, device=dev)
+        conf = torch.rand(Nf, ci, Hf, Wf, device=dev)
+        wt = torch.rand(co, ci, k, k, device=dev) + 0.05
+        out, cout = C.nconv_fwd(data, conf, wt, None)
+        ref_o, ref_c = torch_ref.nconv2d(data, conf, wt, padding=k // 2,
+                                         prop_conf=True)
+        err_o = (out - ref_o).abs().max().item()
+        err_c = (cout - ref_c).abs().max().item()
+        t = timeit(lambda: C.nconv_fwd(data, conf, wt, None))
+        gn = torch.randn_like(out)
+        gd = torch.randn_like(out)
+        tb = timeit(lambda: C.nconv_bwd(gn, gd, data, conf, wt))
+        print(json.dumps({"probe": f"nconv_k{k}_ci{ci}_co{co}",
+                          "shape": [Nf, ci, Hf, Wf], "fwd_ms": t * 1e3,
+                          "bwd_ms": tb * 1e3, "fwd_err": err_o,
+                          "cout_err": err_c}))
 
 
 if __name__ == "__main__":
