@@ -14,11 +14,11 @@ void flowhip_bgemm_nt_launch(const void* A, const void* B, void* C,
                              hipStream_t stream);
 void flowhip_corr_lookup_fwd_launch(const float* level, const float* coords,
                                     float* out, int BP, int P, int Hl, int Wl,
-                                    int l, int L, int radius,
+                                    int l, int L, int radius, int cl,
                                     hipStream_t stream);
 void flowhip_corr_lookup_bwd_launch(const float* gout, const float* coords,
                                     float* glevel, int BP, int P, int Hl,
-                                    int Wl, int l, int L, int radius,
+                                    int Wl, int l, int L, int radius, int cl,
                                     hipStream_t stream);
 void flowhip_convex_up_fwd_launch(const float* flow, const float* mask,
                                   float* out, int N, int H, int W, int factor,
@@ -79,7 +79,8 @@ torch::Tensor bgemm_nt(torch::Tensor a, torch::Tensor b, double alpha) {
 }
 
 torch::Tensor corr_lookup_fwd(std::vector<torch::Tensor> pyramid,
-                              torch::Tensor coords, int64_t radius) {
+                              torch::Tensor coords, int64_t radius,
+                              bool channels_last) {
   TORCH_CHECK(!pyramid.empty());
   TORCH_CHECK(coords.is_cuda() && coords.dtype() == torch::kFloat32 &&
               coords.is_contiguous());
@@ -88,8 +89,12 @@ torch::Tensor corr_lookup_fwd(std::vector<torch::Tensor> pyramid,
   const int L = (int)pyramid.size();
   const int K = 2 * (int)radius + 1;
 
-  auto out = torch::empty({B, (long)L * K * K, H, W},
-                          coords.options().dtype(torch::kFloat32));
+  auto out = channels_last
+                 ? torch::empty({B, (long)L * K * K, H, W},
+                                coords.options().dtype(torch::kFloat32)
+                                    .memory_format(torch::MemoryFormat::ChannelsLast))
+                 : torch::empty({B, (long)L * K * K, H, W},
+                                coords.options().dtype(torch::kFloat32));
   const c10::cuda::CUDAGuard guard(coords.device());
   hipStream_t stream = at::cuda::getCurrentCUDAStream().stream();
 
@@ -102,7 +107,7 @@ torch::Tensor corr_lookup_fwd(std::vector<torch::Tensor> pyramid,
     const int Hl = lvl.size(-2), Wl = lvl.size(-1);
     flowhip_corr_lookup_fwd_launch(
         lvl.data_ptr<float>(), coords.data_ptr<float>(), out.data_ptr<float>(),
-        B * P, P, Hl, Wl, l, L, (int)radius, stream);
+        B * P, P, Hl, Wl, l, L, (int)radius, channels_last ? 1 : 0, stream);
   }
   return out;
 }
@@ -111,9 +116,12 @@ std::vector<torch::Tensor> corr_lookup_bwd(torch::Tensor gout,
                                            torch::Tensor coords,
                                            int64_t radius,
                                            std::vector<std::vector<int64_t>>
-                                               level_shapes) {
-  TORCH_CHECK(gout.is_cuda() && gout.is_contiguous() &&
-              gout.dtype() == torch::kFloat32);
+                                               level_shapes,
+                                           bool channels_last) {
+  TORCH_CHECK(gout.is_cuda() && gout.dtype() == torch::kFloat32);
+  TORCH_CHECK(channels_last
+                  ? gout.is_contiguous(torch::MemoryFormat::ChannelsLast)
+                  : gout.is_contiguous());
   const int B = coords.size(0), H = coords.size(2), W = coords.size(3);
   const int P = H * W;
   const int L = (int)level_shapes.size();
@@ -121,15 +129,28 @@ std::vector<torch::Tensor> corr_lookup_bwd(torch::Tensor gout,
   const c10::cuda::CUDAGuard guard(coords.device());
   hipStream_t stream = at::cuda::getCurrentCUDAStream().stream();
 
+  // one flat zero-fill for all levels (4x fewer fill kernels), sliced into
+  // per-level views
+  int64_t total = 0;
+  std::vector<int64_t> sizes(L);
+  for (int l = 0; l < L; ++l) {
+    int64_t n = 1;
+    for (auto d : level_shapes[l]) n *= d;
+    sizes[l] = n;
+    total += n;
+  }
+  auto flat = torch::zeros({total}, gout.options().dtype(torch::kFloat32));
+
   std::vector<torch::Tensor> grads;
   grads.reserve(L);
+  int64_t off = 0;
   for (int l = 0; l < L; ++l) {
-    auto g = torch::zeros(level_shapes[l],
-                          gout.options().dtype(torch::kFloat32));
+    auto g = flat.narrow(0, off, sizes[l]).view(level_shapes[l]);
+    off += sizes[l];
     const int Hl = g.size(-2), Wl = g.size(-1);
     flowhip_corr_lookup_bwd_launch(
         gout.data_ptr<float>(), coords.data_ptr<float>(), g.data_ptr<float>(),
-        B * P, P, Hl, Wl, l, L, (int)radius, stream);
+        B * P, P, Hl, Wl, l, L, (int)radius, channels_last ? 1 : 0, stream);
     grads.push_back(g);
   }
   return grads;
@@ -238,7 +259,7 @@ std::vector<torch::Tensor> nconv_bwd(torch::Tensor dnomin,
         dconf.data_ptr<float>(), N, Ci, Co, H, W, K, stream);
   }
   bool wrw_done = false;
-  if (K == 3 || K == 5) {
+  if (K == 1 || K == 3 || K == 5) {
     const int nblocks = flowhip_nconv_tiled_nblocks(N, H, W);
     auto partials = torch::empty({nblocks, (long)Co * Ci * K * K},
                                  data.options());

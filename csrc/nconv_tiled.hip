@@ -60,18 +60,18 @@ __device__ inline void nct_stage_tile(const float* __restrict__ data,
 //           cout = conv(c, w) / sum_w   (confidence propagation)
 // Grid: (ntx*nty*N); each thread computes NCT_TH/4 pixels for ALL Co.
 // ---------------------------------------------------------------------------
-template <int K, int CI>
+template <int K, int CI, int CO>
 __global__ __launch_bounds__(NCT_THREADS) void nconv_fwd_tiled_kernel(
     const float* __restrict__ data, const float* __restrict__ conf,
     const float* __restrict__ weight, const float* __restrict__ bias,
     float* __restrict__ out, float* __restrict__ cout,
-    int N, int Co, int H, int W, int ntx, int nty) {
+    int N, int H, int W, int ntx, int nty) {
   constexpr int LW = NCT_TW + K - 1;
   constexpr int LH = NCT_TH + K - 1;
   __shared__ float lds_c[CI * LH * LW];
   __shared__ float lds_dc[CI * LH * LW];
-  __shared__ float wsh[8 * CI * K * K];
-  __shared__ float winv[8];
+  __shared__ float wsh[CO * CI * K * K];
+  __shared__ float winv[CO];
 
   int t = blockIdx.x;
   const int tx = t % ntx; t /= ntx;
@@ -79,10 +79,10 @@ __global__ __launch_bounds__(NCT_THREADS) void nconv_fwd_tiled_kernel(
   const int n = t;
   const int x0 = tx * NCT_TW, y0 = ty * NCT_TH;
 
-  const int nw = Co * CI * K * K;
+  constexpr int nw = CO * CI * K * K;
   for (int i = threadIdx.x; i < nw; i += NCT_THREADS) wsh[i] = weight[i];
   __syncthreads();
-  if (threadIdx.x < Co) {
+  if (threadIdx.x < CO) {
     float s = 0.f;
     for (int i = 0; i < CI * K * K; ++i) s += wsh[threadIdx.x * CI * K * K + i];
     winv[threadIdx.x] = 1.0f / s;
@@ -95,25 +95,28 @@ __global__ __launch_bounds__(NCT_THREADS) void nconv_fwd_tiled_kernel(
   const int ly0 = threadIdx.x >> 6;         // 0..3
   const int x = x0 + lx;
 
-#pragma unroll
+#pragma unroll 1
   for (int j = 0; j < NCT_TH / 4; ++j) {
     const int lyy = ly0 + 4 * j;
     const int y = y0 + lyy;
     if (x >= W || y >= H) continue;
-    float denom[8], nomin[8];
+    float denom[CO], nomin[CO];
 #pragma unroll
-    for (int co = 0; co < 8; ++co) { denom[co] = 0.f; nomin[co] = 0.f; }
+    for (int co = 0; co < CO; ++co) { denom[co] = 0.f; nomin[co] = 0.f; }
 #pragma unroll
     for (int ci = 0; ci < CI; ++ci) {
       const float* lc = lds_c + ci * LH * LW;
       const float* ldc = lds_dc + ci * LH * LW;
-#pragma unroll
+      // bounded unroll (K*CO*2 fma + K*2 LDS loads in flight per row):
+      // full K*K unrolling costs 256 VGPRs -> 1 wave/SIMD
+#pragma unroll 1
       for (int ky = 0; ky < K; ++ky) {
 #pragma unroll
         for (int kx = 0; kx < K; ++kx) {
           const float c = lc[(lyy + ky) * LW + lx + kx];
           const float dc = ldc[(lyy + ky) * LW + lx + kx];
-          for (int co = 0; co < Co; ++co) {
+#pragma unroll
+          for (int co = 0; co < CO; ++co) {
             const float w = wsh[((co * CI + ci) * K + ky) * K + kx];
             denom[co] += w * c;
             nomin[co] += w * dc;
@@ -121,10 +124,11 @@ __global__ __launch_bounds__(NCT_THREADS) void nconv_fwd_tiled_kernel(
         }
       }
     }
-    for (int co = 0; co < Co; ++co) {
+#pragma unroll
+    for (int co = 0; co < CO; ++co) {
       float v = nomin[co] / (denom[co] + 1e-20f);
       if (bias != nullptr) v += bias[co];
-      const long o = ((long)n * Co + co) * plane + (long)y * W + x;
+      const long o = ((long)n * CO + co) * plane + (long)y * W + x;
       out[o] = v;
       cout[o] = denom[co] * winv[co];
     }
@@ -330,17 +334,12 @@ __global__ __launch_bounds__(NCT_THREADS) void nconv_wrw_reduce_kernel(
 // falls back to the v1 kernels).
 // ---------------------------------------------------------------------------
 
-int flowhip_nconv_tiled_nblocks(int N, int H, int W) {
-  return fh_cdiv(W, NCT_TW) * fh_cdiv(H, NCT_TH) * N;
-}
-
-#define NCT_DISPATCH_CI(KK, CIV, kernel, ...)                                  \
-  switch (CIV) {                                                               \
-    case 1: hipLaunchKernelGGL((kernel<KK, 1>), grid, block, 0, stream, __VA_ARGS__); return true; \
-    case 2: hipLaunchKernelGGL((kernel<KK, 2>), grid, block, 0, stream, __VA_ARGS__); return true; \
-    case 4: hipLaunchKernelGGL((kernel<KK, 4>), grid, block, 0, stream, __VA_ARGS__); return true; \
-    case 8: hipLaunchKernelGGL((kernel<KK, 8>), grid, block, 0, stream, __VA_ARGS__); return true; \
-    default: return false;                                                     \
+#define NCT_CASE_FWD(KK, CIV, COV)                                             \
+  if (K == KK && Ci == CIV && Co == COV) {                                     \
+    hipLaunchKernelGGL((nconv_fwd_tiled_kernel<KK, CIV, COV>), grid, block, 0, \
+                       stream, data, conf, weight, bias, out, cout, N, H, W,   \
+                       ntx, nty);                                              \
+    return true;                                                               \
   }
 
 bool flowhip_nconv_fwd_tiled_launch(const float* data, const float* conf,
@@ -348,18 +347,26 @@ bool flowhip_nconv_fwd_tiled_launch(const float* data, const float* conf,
                                     float* out, float* cout, int N, int Ci,
                                     int Co, int H, int W, int K,
                                     hipStream_t stream) {
-  if (Co > 8) return false;
   const int ntx = fh_cdiv(W, NCT_TW), nty = fh_cdiv(H, NCT_TH);
   dim3 grid(ntx * nty * N), block(NCT_THREADS);
-  if (K == 3) {
-    NCT_DISPATCH_CI(3, Ci, nconv_fwd_tiled_kernel, data, conf, weight, bias,
-                    out, cout, N, Co, H, W, ntx, nty)
-  } else if (K == 5) {
-    NCT_DISPATCH_CI(5, Ci, nconv_fwd_tiled_kernel, data, conf, weight, bias,
-                    out, cout, N, Co, H, W, ntx, nty)
-  }
+  NCT_CASE_FWD(5, 1, 1) NCT_CASE_FWD(5, 1, 2) NCT_CASE_FWD(5, 1, 4)
+  NCT_CASE_FWD(5, 2, 1) NCT_CASE_FWD(5, 2, 2) NCT_CASE_FWD(5, 2, 4)
+  NCT_CASE_FWD(5, 4, 2) NCT_CASE_FWD(5, 4, 4)
+  NCT_CASE_FWD(3, 1, 1) NCT_CASE_FWD(3, 1, 2) NCT_CASE_FWD(3, 1, 4)
+  NCT_CASE_FWD(3, 2, 1) NCT_CASE_FWD(3, 2, 2) NCT_CASE_FWD(3, 2, 4)
+  NCT_CASE_FWD(3, 4, 2) NCT_CASE_FWD(3, 4, 4) NCT_CASE_FWD(3, 8, 2)
+  NCT_CASE_FWD(1, 1, 1) NCT_CASE_FWD(1, 1, 2) NCT_CASE_FWD(1, 2, 1)
+  NCT_CASE_FWD(1, 2, 2) NCT_CASE_FWD(1, 4, 2)
   return false;
 }
+
+#define NCT_CASE_BWD(KK, COV)                                                  \
+  if (K == KK && Co == COV) {                                                  \
+    hipLaunchKernelGGL((nconv_bwd_data_tiled_kernel<KK, COV>), grid, block, 0, \
+                       stream, dnomin, ddenom, data, conf, weight, ddata,      \
+                       dconf, N, Ci, H, W, ntx, nty);                          \
+    return true;                                                               \
+  }
 
 bool flowhip_nconv_bwd_data_tiled_launch(
     const float* dnomin, const float* ddenom, const float* data,
@@ -367,14 +374,22 @@ bool flowhip_nconv_bwd_data_tiled_launch(
     int Ci, int Co, int H, int W, int K, hipStream_t stream) {
   const int ntx = fh_cdiv(W, NCT_TW), nty = fh_cdiv(H, NCT_TH);
   dim3 grid(ntx * nty * N), block(NCT_THREADS);
-  if (K == 3) {
-    NCT_DISPATCH_CI(3, Co, nconv_bwd_data_tiled_kernel, dnomin, ddenom, data,
-                    conf, weight, ddata, dconf, N, Ci, H, W, ntx, nty)
-  } else if (K == 5) {
-    NCT_DISPATCH_CI(5, Co, nconv_bwd_data_tiled_kernel, dnomin, ddenom, data,
-                    conf, weight, ddata, dconf, N, Ci, H, W, ntx, nty)
-  }
+  NCT_CASE_BWD(5, 1) NCT_CASE_BWD(5, 2) NCT_CASE_BWD(5, 4)
+  NCT_CASE_BWD(3, 1) NCT_CASE_BWD(3, 2) NCT_CASE_BWD(3, 4) NCT_CASE_BWD(3, 8)
+  NCT_CASE_BWD(1, 1) NCT_CASE_BWD(1, 2) NCT_CASE_BWD(1, 4)
   return false;
+}
+
+#define NCT_CASE_WRW(KK, CIV)                                                  \
+  if (K == KK && Ci == CIV) {                                                  \
+    hipLaunchKernelGGL((nconv_wrw_tiled_kernel<KK, CIV>), grid, block, 0,      \
+                       stream, dnomin, ddenom, data, conf, partials, N, Co, H, \
+                       W, ntx, nty);                                           \
+    launched = true;                                                           \
+  }
+
+int flowhip_nconv_tiled_nblocks(int N, int H, int W) {
+  return fh_cdiv(W, NCT_TW) * fh_cdiv(H, NCT_TH) * N;
 }
 
 bool flowhip_nconv_wrw_tiled_launch(const float* dnomin, const float* ddenom,
@@ -387,20 +402,9 @@ bool flowhip_nconv_wrw_tiled_launch(const float* dnomin, const float* ddenom,
   const int nw = Co * Ci * K * K;
   dim3 grid(nblocks), block(NCT_THREADS);
   bool launched = false;
-  if (K == 3) {
-    switch (Ci) {
-      case 1: hipLaunchKernelGGL((nconv_wrw_tiled_kernel<3, 1>), grid, block, 0, stream, dnomin, ddenom, data, conf, partials, N, Co, H, W, ntx, nty); launched = true; break;
-      case 2: hipLaunchKernelGGL((nconv_wrw_tiled_kernel<3, 2>), grid, block, 0, stream, dnomin, ddenom, data, conf, partials, N, Co, H, W, ntx, nty); launched = true; break;
-      case 4: hipLaunchKernelGGL((nconv_wrw_tiled_kernel<3, 4>), grid, block, 0, stream, dnomin, ddenom, data, conf, partials, N, Co, H, W, ntx, nty); launched = true; break;
-      case 8: hipLaunchKernelGGL((nconv_wrw_tiled_kernel<3, 8>), grid, block, 0, stream, dnomin, ddenom, data, conf, partials, N, Co, H, W, ntx, nty); launched = true; break;
-    }
-  } else if (K == 5) {
-    switch (Ci) {
-      case 1: hipLaunchKernelGGL((nconv_wrw_tiled_kernel<5, 1>), grid, block, 0, stream, dnomin, ddenom, data, conf, partials, N, Co, H, W, ntx, nty); launched = true; break;
-      case 2: hipLaunchKernelGGL((nconv_wrw_tiled_kernel<5, 2>), grid, block, 0, stream, dnomin, ddenom, data, conf, partials, N, Co, H, W, ntx, nty); launched = true; break;
-      case 4: hipLaunchKernelGGL((nconv_wrw_tiled_kernel<5, 4>), grid, block, 0, stream, dnomin, ddenom, data, conf, partials, N, Co, H, W, ntx, nty); launched = true; break;
-    }
-  }
+  NCT_CASE_WRW(5, 1) NCT_CASE_WRW(5, 2) NCT_CASE_WRW(5, 4)
+  NCT_CASE_WRW(3, 1) NCT_CASE_WRW(3, 2) NCT_CASE_WRW(3, 4) NCT_CASE_WRW(3, 8)
+  NCT_CASE_WRW(1, 1) NCT_CASE_WRW(1, 2) NCT_CASE_WRW(1, 4) NCT_CASE_WRW(1, 8)
   if (!launched) return false;
   hipLaunchKernelGGL(nconv_wrw_reduce_kernel, dim3(nw), dim3(NCT_THREADS), 0,
                      stream, partials, dweight, nblocks, nw);

@@ -11,6 +11,7 @@ import warnings
 
 from .raft import RAFT
 from .raft_nc_dbl import RAFT_NC_DBL
+from ..utils.layout import apply_channels_last
 
 MODEL_NAMES = ("raft", "raft_nc", "raft_nc_sep", "raft_nc_dbl")
 
@@ -18,15 +19,19 @@ MODEL_NAMES = ("raft", "raft_nc", "raft_nc_sep", "raft_nc_dbl")
 def build_model(args):
     name = getattr(args, "model", "raft")
     if name == "raft":
-        return RAFT(args)
-    if name == "raft_nc_dbl":
-        return RAFT_NC_DBL(args)
-    if name in ("raft_nc", "raft_nc_sep"):
+        model = RAFT(args)
+    elif name == "raft_nc_dbl":
+        model = RAFT_NC_DBL(args)
+    elif name in ("raft_nc", "raft_nc_sep"):
         warnings.warn(
             f"model {name!r} is missing from the reference snapshot; "
             "resolving to raft_nc_dbl (see SURVEY.md §2.9).")
-        return RAFT_NC_DBL(args)
-    raise NotImplementedError(f"Model {name!r} not found!")
+        model = RAFT_NC_DBL(args)
+    else:
+        raise NotImplementedError(f"Model {name!r} not found!")
+    # GPU path runs NHWC (see utils/layout.py); layout has no effect on
+    # state-dict values or CPU semantics.
+    return apply_channels_last(model)
 
 
 __all__ = ["RAFT", "RAFT_NC_DBL", "build_model", "MODEL_NAMES"]

@@ -12,7 +12,7 @@ import torch.nn as nn
 from ..nn.corr import CorrBlock
 from ..nn.extractor import BasicEncoder, SmallEncoder
 from ..nn.update import BasicUpdateBlock, SmallUpdateBlock
-from ..utils.amp import autocast_ctx
+from ..utils.amp import autocast_ctx, autocast_off_ctx
 from ..utils.geometry import coords_grid, upflow8
 from .. import ops
 
@@ -63,23 +63,23 @@ class RAFT(nn.Module):
 
     def _features(self, image1, image2):
         """Normalize images and run feature + context networks."""
+        from ..utils.layout import to_model_layout
         image1 = 2 * (image1 / 255.0) - 1.0
         image2 = 2 * (image2 / 255.0) - 1.0
-        image1 = image1.contiguous()
-        image2 = image2.contiguous()
+        image1 = to_model_layout(image1)
+        image2 = to_model_layout(image2)
 
-        with autocast_ctx(image1, enabled=self.args.mixed_precision):
-            fmap1, fmap2 = self.fnet([image1, image2])
+        fmap1, fmap2 = self.fnet([image1, image2])
 
-        fmap1 = fmap1.float()
-        fmap2 = fmap2.float()
-        corr_fn = CorrBlock(fmap1, fmap2, radius=self.args.corr_radius)
+        with autocast_off_ctx(image1):
+            fmap1 = fmap1.float()
+            fmap2 = fmap2.float()
+            corr_fn = CorrBlock(fmap1, fmap2, radius=self.args.corr_radius)
 
-        with autocast_ctx(image1, enabled=self.args.mixed_precision):
-            cnet = self.cnet(image1)
-            net, inp = torch.split(cnet, [self.hidden_dim, self.context_dim], dim=1)
-            net = torch.tanh(net)
-            inp = torch.relu(inp)
+        cnet = self.cnet(image1)
+        net, inp = torch.split(cnet, [self.hidden_dim, self.context_dim], dim=1)
+        net = torch.tanh(net)
+        inp = torch.relu(inp)
         return image1, corr_fn, net, inp
 
     def forward(self, image1, image2, iters=12, flow_init=None, upsample=True,
@@ -89,30 +89,33 @@ class RAFT(nn.Module):
         `upsample` is accepted for API compatibility and ignored, as in the
         reference (raft.py:87 — SURVEY.md §2.9 quirk 4).
         """
-        image1, corr_fn, net, inp = self._features(image1, image2)
+        with autocast_ctx(image1, enabled=self.args.mixed_precision):
+            image1, corr_fn, net, inp = self._features(image1, image2)
 
-        coords0, coords1 = self.initialize_flow(image1)
-        if flow_init is not None:
-            coords1 = coords1 + flow_init
+            coords0, coords1 = self.initialize_flow(image1)
+            if flow_init is not None:
+                coords1 = coords1 + flow_init
 
-        flow_predictions = []
-        flow_up = None
-        for _ in range(iters):
-            coords1 = coords1.detach()
-            corr = corr_fn(coords1)
+            flow_predictions = []
+            flow_up = None
+            for _ in range(iters):
+                coords1 = coords1.detach()
+                with autocast_off_ctx(image1):
+                    corr = corr_fn(coords1)
 
-            flow = coords1 - coords0
-            with autocast_ctx(image1, enabled=self.args.mixed_precision):
-                net, up_mask, delta_flow = self.update_block(net, inp, corr, flow)
+                flow = coords1 - coords0
+                net, up_mask, delta_flow = self.update_block(net, inp, corr,
+                                                             flow)
 
-            coords1 = coords1 + delta_flow
+                coords1 = coords1 + delta_flow.float()
 
-            if up_mask is None:
-                flow_up = upflow8(coords1 - coords0)
-            else:
-                flow_up = self.upsample_flow((coords1 - coords0).float(),
-                                             up_mask.float())
-            flow_predictions.append(flow_up)
+                with autocast_off_ctx(image1):
+                    if up_mask is None:
+                        flow_up = upflow8(coords1 - coords0)
+                    else:
+                        flow_up = self.upsample_flow(
+                            (coords1 - coords0).float(), up_mask.float())
+                flow_predictions.append(flow_up)
 
         if test_mode:
             return coords1 - coords0, flow_up

@@ -13,7 +13,7 @@ import torch.nn as nn
 import torch.nn.functional as F
 
 from ..nn.upsampler import get_upsampler
-from ..utils.amp import autocast_ctx
+from ..utils.amp import autocast_ctx, autocast_off_ctx
 from .raft import RAFT as _RAFTBase
 
 
@@ -49,29 +49,32 @@ class RAFT_NC_DBL(_RAFTBase):
 
     def forward(self, image1, image2, iters=12, flow_init=None, upsample=True,
                 test_mode=False):
-        image1, corr_fn, net, inp = self._features(image1, image2)
+        with autocast_ctx(image1, enabled=self.args.mixed_precision):
+            image1, corr_fn, net, inp = self._features(image1, image2)
 
-        coords0, coords1 = self.initialize_flow(image1)
-        if flow_init is not None:
-            coords1 = coords1 + flow_init
+            coords0, coords1 = self.initialize_flow(image1)
+            if flow_init is not None:
+                coords1 = coords1 + flow_init
 
-        flow_predictions = []
-        flow_up = None
-        for _ in range(iters):
-            coords1 = coords1.detach()
-            corr = corr_fn(coords1)
+            flow_predictions = []
+            flow_up = None
+            for _ in range(iters):
+                coords1 = coords1.detach()
+                with autocast_off_ctx(image1):
+                    corr = corr_fn(coords1)
 
-            flow = coords1 - coords0
-            with autocast_ctx(image1, enabled=self.args.mixed_precision):
+                flow = coords1 - coords0
                 net, _, delta_flow = self.update_block(net, inp, corr, flow)
 
-            coords1 = coords1 + delta_flow
+                coords1 = coords1 + delta_flow.float()
 
-            # NCUP upsample every iteration, guided by the GRU hidden state;
-            # the x8 scale is applied to the upsampled field (ref :161).
-            flow_up = 8 * self.upsample_flow((coords1 - coords0).float(),
-                                             net.float())
-            flow_predictions.append(flow_up)
+                # NCUP upsample every iteration, guided by the GRU hidden
+                # state; x8 applied to the upsampled field (ref :161).
+                # fp32 island: the upsampler runs fp32 as in the reference.
+                with autocast_off_ctx(image1):
+                    flow_up = 8 * self.upsample_flow(
+                        (coords1 - coords0).float(), net.float())
+                flow_predictions.append(flow_up)
 
         if test_mode:
             return coords1 - coords0, flow_up

@@ -53,13 +53,21 @@ class CorrVolumeFn(torch.autograd.Function):
     def forward(ctx, fmap1, fmap2):
         B, D, H, W = fmap1.shape
         P = H * W
-        f1 = fmap1.reshape(B, D, P)
-        f2 = fmap2.reshape(B, D, P)
-        f1t = f1.transpose(1, 2).contiguous().to(torch.bfloat16)  # (B,P,D)
-        f2t = f2.transpose(1, 2).contiguous().to(torch.bfloat16)  # (B,P,D)
+
+        def as_pd(f):
+            # channels_last (B,H,W,D) memory IS the (B,P,D) operand — no
+            # transpose kernel, just the bf16 cast
+            if f.is_contiguous(memory_format=torch.channels_last):
+                return f.permute(0, 2, 3, 1).reshape(B, P, D).to(torch.bfloat16)
+            return f.reshape(B, D, P).transpose(1, 2).contiguous().to(
+                torch.bfloat16)
+
+        f1t = as_pd(fmap1)  # (B,P,D)
+        f2t = as_pd(fmap2)
         corr = _bgemm_nt(f1t, f2t, 1.0 / math.sqrt(D))  # (B,P,P) fp32
         ctx.save_for_backward(f1t, f2t)
         ctx.shape = (B, D, H, W)
+        ctx.cl = fmap1.is_contiguous(memory_format=torch.channels_last)
         return corr.reshape(B * P, 1, H, W)
 
     @staticmethod
@@ -77,8 +85,13 @@ class CorrVolumeFn(torch.autograd.Function):
         f1_kn = f1t.transpose(1, 2).contiguous()
         df1t = _bgemm_nt(dc, f2_kn, alpha)   # (B,P,D) fp32
         df2t = _bgemm_nt(dct, f1_kn, alpha)  # (B,P,D) fp32
-        df1 = df1t.transpose(1, 2).reshape(B, D, H, W)
-        df2 = df2t.transpose(1, 2).reshape(B, D, H, W)
+        if ctx.cl:
+            # (B,P,D) memory == channels_last (B,D,H,W): zero-copy view
+            df1 = df1t.reshape(B, H, W, D).permute(0, 3, 1, 2)
+            df2 = df2t.reshape(B, H, W, D).permute(0, 3, 1, 2)
+        else:
+            df1 = df1t.transpose(1, 2).reshape(B, D, H, W)
+            df2 = df2t.transpose(1, 2).reshape(B, D, H, W)
         return df1, df2
 
 
@@ -92,20 +105,28 @@ class CorrLookupFn(torch.autograd.Function):
 
     @staticmethod
     def forward(ctx, coords, radius, *pyramid):
+        from ..utils.layout import channels_last_enabled
         B, _, H1, W1 = coords.shape
         assert not coords.requires_grad, (
             "corr_lookup: coords must be detached (reference contract)")
         coords_c = coords.contiguous()
-        out = _ext.ext().corr_lookup_fwd(list(pyramid), coords_c, int(radius))
+        cl = channels_last_enabled()
+        out = _ext.ext().corr_lookup_fwd(list(pyramid), coords_c, int(radius),
+                                         cl)
         ctx.save_for_backward(coords_c)
         ctx.radius = int(radius)
+        ctx.cl = cl
         ctx.level_shapes = [tuple(p.shape) for p in pyramid]
         return out
 
     @staticmethod
     def backward(ctx, grad):
         (coords,) = ctx.saved_tensors
+        if ctx.cl:
+            grad = grad.contiguous(memory_format=torch.channels_last)
+        else:
+            grad = grad.contiguous()
         grads = _ext.ext().corr_lookup_bwd(
-            grad.contiguous(), coords, ctx.radius,
-            [list(s) for s in ctx.level_shapes])
+            grad, coords, ctx.radius,
+            [list(s) for s in ctx.level_shapes], ctx.cl)
         return (None, None, *grads)

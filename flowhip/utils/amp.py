@@ -20,6 +20,20 @@ def autocast_ctx(ref_tensor, enabled=True, dtype=torch.bfloat16):
     return torch.autocast(device_type=device_type, dtype=dtype, enabled=True)
 
 
+def autocast_off_ctx(ref_tensor):
+    """fp32 island inside an enclosing autocast region.
+
+    The model forward runs under ONE autocast region (keeping autocast's
+    bf16 weight-cast cache alive across the 12-32 refinement iterations —
+    per-iteration regions re-cast every conv weight every iteration, ~430
+    extra cast kernels/step in profiles/r01). The correlation volume and
+    the NCUP upsampler stay fp32 through this context, mirroring the
+    reference's fp32 sections (raft.py:103-104, raft_nc_dbl.py:161).
+    """
+    device_type = "cuda" if ref_tensor.is_cuda else "cpu"
+    return torch.autocast(device_type=device_type, enabled=False)
+
+
 class NoOpGradScaler:
     """API-compatible stand-in for torch.cuda.amp.GradScaler under bf16
     (bf16 has fp32's exponent range — no scaling required)."""

@@ -1,0 +1,44 @@
+"""Memory-format policy for the GPU path.
+
+MIOpen's bf16 conv kernels are NHWC-native: running the model NCHW costs a
+batched_transpose pair around every conv call (~1,470 transpose launches per
+training step measured in profiles/r01). The GPU path therefore runs the
+whole network channels_last:
+
+- conv modules + activations carry torch.channels_last strides;
+- the corr GEMM (ops.functional.CorrVolumeFn) reads (B,H,W,D)-contiguous
+  feature maps directly as its (B,P,D) row-major operands — the explicit
+  transpose of the NCHW path disappears;
+- the fused corr-lookup kernel writes its (B, L*K*K, H, W) output with
+  channels_last strides (per-pixel-contiguous taps, fully coalesced stores)
+  so the motion-encoder convs consume it natively.
+
+CPU keeps NCHW (the torch reference path; layout has no semantic effect).
+"""
+
+import torch
+
+_CHANNELS_LAST = True
+
+
+def set_channels_last(enabled):
+    global _CHANNELS_LAST
+    _CHANNELS_LAST = bool(enabled)
+
+
+def channels_last_enabled():
+    return _CHANNELS_LAST
+
+
+def apply_channels_last(model):
+    """Convert a CUDA-resident model's conv weights to channels_last."""
+    if _CHANNELS_LAST:
+        model.to(memory_format=torch.channels_last)
+    return model
+
+
+def to_model_layout(x):
+    """Bring a CUDA input batch into the model's memory format."""
+    if _CHANNELS_LAST and x.is_cuda:
+        return x.contiguous(memory_format=torch.channels_last)
+    return x.contiguous()

@@ -79,7 +79,7 @@ def main():
     pyramid = [p.contiguous() for p in torch_ref.corr_pyramid(corr, 4)]
     coords = (torch.rand(B, 2, H, W, device=dev) *
               torch.tensor([W, H], device=dev).view(1, 2, 1, 1)).contiguous()
-    t = timeit(lambda: C.corr_lookup_fwd(pyramid, coords, 4))
+    t = timeit(lambda: C.corr_lookup_fwd(pyramid, coords, 4, False))
     out_bytes = B * 4 * 81 * P * 4
     print(json.dumps({"probe": "corr_lookup_fwd", "ms": t * 1e3,
                       "out_gbps": out_bytes / t / 1e9}))
@@ -90,7 +90,7 @@ def main():
 
     g = torch.randn(B, 4 * 81, H, W, device=dev)
     shapes = [list(p.shape) for p in pyramid]
-    t = timeit(lambda: C.corr_lookup_bwd(g, coords, 4, shapes))
+    t = timeit(lambda: C.corr_lookup_bwd(g, coords, 4, shapes, False))
     print(json.dumps({"probe": "corr_lookup_bwd", "ms": t * 1e3}))
     del corr, pyramid, g
 

@@ -21,9 +21,9 @@ pyr = [p.detach().contiguous() for p in torch_ref.corr_pyramid(l0, 4)]
 coords = (torch.rand(B, 2, H, W, device=dev) *
           torch.tensor([W, H], device=dev).view(1, 2, 1, 1)).contiguous()
 
-out = C.corr_lookup_fwd(pyr, coords, R)
+out = C.corr_lookup_fwd(pyr, coords, R, False)
 g = torch.randn_like(out)
-grads = C.corr_lookup_bwd(g, coords, R, [list(p.shape) for p in pyr])
+grads = C.corr_lookup_bwd(g, coords, R, [list(p.shape) for p in pyr], False)
 
 lhs = (out * g).sum().item()
 rhs = sum((pyr[l] * grads[l]).sum().item() for l in range(4))
