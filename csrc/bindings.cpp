@@ -55,6 +55,29 @@ bool flowhip_nconv_wrw_tiled_launch(const float* dnomin, const float* ddenom,
                                     float* partials, float* dweight, int N,
                                     int Ci, int Co, int H, int W, int K,
                                     hipStream_t stream);
+void flowhip_seq_loss_fwd_launch(const float* const* preds, int n,
+                                 const float* gt, const float* valid,
+                                 float* partials, float* out, long npix,
+                                 long plane, float max_flow, float gamma,
+                                 long numel_full, hipStream_t stream);
+void flowhip_seq_loss_bwd_launch(const float* const* preds, float* const* dst,
+                                 int n, const float* gt, const float* valid,
+                                 const float* gloss, long npix, long plane,
+                                 float max_flow, float gamma, long numel_full,
+                                 hipStream_t stream);
+void flowhip_nconv_bwd_prep_launch(const float* gout, const float* gcout,
+                                   const float* out, const float* cout,
+                                   const float* wsum, const float* bias,
+                                   float* dnomin, float* ddenom, long total,
+                                   long plane, int Co, float eps,
+                                   hipStream_t stream);
+bool flowhip_corr_pyramid_fwd_launch(const float* corr, float* l1, float* l2,
+                                     float* l3, int BP, int H0, int W0,
+                                     int nlev, hipStream_t stream);
+void flowhip_corr_pyramid_bwd_launch(const float* g0, const float* g1,
+                                     const float* g2, const float* g3,
+                                     float* dcorr, long total, int H0, int W0,
+                                     hipStream_t stream);
 
 namespace {
 
@@ -154,6 +177,61 @@ std::vector<torch::Tensor> corr_lookup_bwd(torch::Tensor gout,
     grads.push_back(g);
   }
   return grads;
+}
+
+std::vector<torch::Tensor> corr_pyramid_fwd(torch::Tensor corr,
+                                            int64_t num_levels) {
+  TORCH_CHECK(corr.is_cuda() && corr.dtype() == torch::kFloat32 &&
+              corr.is_contiguous());
+  TORCH_CHECK(corr.dim() == 4 && corr.size(1) == 1);
+  const long BP = corr.size(0);
+  const int H0 = corr.size(2), W0 = corr.size(3);
+  int H = H0, W = W0;
+  std::vector<torch::Tensor> levels;
+  for (int l = 1; l < num_levels; ++l) {
+    H /= 2; W /= 2;
+    levels.push_back(torch::empty({BP, 1, H, W}, corr.options()));
+  }
+  const c10::cuda::CUDAGuard guard(corr.device());
+  hipStream_t stream = at::cuda::getCurrentCUDAStream().stream();
+  float* p1 = levels.size() > 0 ? levels[0].data_ptr<float>() : nullptr;
+  float* p2 = levels.size() > 1 ? levels[1].data_ptr<float>() : nullptr;
+  float* p3 = levels.size() > 2 ? levels[2].data_ptr<float>() : nullptr;
+  bool ok = flowhip_corr_pyramid_fwd_launch(
+      corr.data_ptr<float>(), p1, p2, p3, (int)BP, H0, W0, (int)num_levels,
+      stream);
+  TORCH_CHECK(ok, "corr_pyramid_fwd: unsupported shape (fall back to torch)");
+  return levels;
+}
+
+torch::Tensor corr_pyramid_bwd(std::vector<c10::optional<torch::Tensor>> grads,
+                               std::vector<int64_t> corr_shape) {
+  TORCH_CHECK(grads.size() >= 1 && grads.size() <= 4);
+  const long BP = corr_shape[0];
+  const int H0 = corr_shape[2], W0 = corr_shape[3];
+  const float* g[4] = {nullptr, nullptr, nullptr, nullptr};
+  torch::TensorOptions opts;
+  torch::Device dev(torch::kCUDA);
+  bool have = false;
+  for (size_t l = 0; l < grads.size(); ++l) {
+    if (grads[l].has_value()) {
+      auto& t = grads[l].value();
+      TORCH_CHECK(t.is_cuda() && t.is_contiguous() &&
+                  t.dtype() == torch::kFloat32);
+      g[l] = t.data_ptr<float>();
+      opts = t.options();
+      dev = t.device();
+      have = true;
+    }
+  }
+  TORCH_CHECK(have, "corr_pyramid_bwd: all grads missing");
+  auto dcorr = torch::empty(corr_shape, opts);
+  const c10::cuda::CUDAGuard guard(dev);
+  hipStream_t stream = at::cuda::getCurrentCUDAStream().stream();
+  flowhip_corr_pyramid_bwd_launch(g[0], g[1], g[2], g[3],
+                                  dcorr.data_ptr<float>(),
+                                  (long)BP * H0 * W0, H0, W0, stream);
+  return dcorr;
 }
 
 torch::Tensor convex_up_fwd(torch::Tensor flow, torch::Tensor mask,
@@ -278,6 +356,86 @@ std::vector<torch::Tensor> nconv_bwd(torch::Tensor dnomin,
   return {ddata, dconf, dweight};
 }
 
+std::vector<torch::Tensor> nconv_bwd_prep(
+    torch::Tensor gout, c10::optional<torch::Tensor> gcout, torch::Tensor out,
+    torch::Tensor cout, torch::Tensor wsum, c10::optional<torch::Tensor> bias,
+    double eps) {
+  TORCH_CHECK(gout.is_cuda() && gout.is_contiguous() &&
+              gout.dtype() == torch::kFloat32);
+  const int Co = gout.size(1);
+  const long plane = (long)gout.size(2) * gout.size(3);
+  const long total = gout.numel();
+  auto dnomin = torch::empty_like(gout);
+  auto ddenom = torch::empty_like(gout);
+  const c10::cuda::CUDAGuard guard(gout.device());
+  hipStream_t stream = at::cuda::getCurrentCUDAStream().stream();
+  flowhip_nconv_bwd_prep_launch(
+      gout.data_ptr<float>(),
+      gcout.has_value() ? gcout->data_ptr<float>() : nullptr,
+      out.data_ptr<float>(), cout.data_ptr<float>(), wsum.data_ptr<float>(),
+      bias.has_value() ? bias->data_ptr<float>() : nullptr,
+      dnomin.data_ptr<float>(), ddenom.data_ptr<float>(), total, plane, Co,
+      (float)eps, stream);
+  return {dnomin, ddenom};
+}
+
+torch::Tensor seq_loss_fwd(std::vector<torch::Tensor> preds,
+                           torch::Tensor gt, torch::Tensor valid,
+                           double gamma, double max_flow) {
+  const int n = (int)preds.size();
+  TORCH_CHECK(n >= 1 && n <= 32, "seq_loss: 1..32 predictions");
+  TORCH_CHECK(gt.is_cuda() && gt.is_contiguous() &&
+              gt.dtype() == torch::kFloat32);
+  TORCH_CHECK(valid.is_cuda() && valid.is_contiguous());
+  const long B = gt.size(0), H = gt.size(2), W = gt.size(3);
+  const long plane = H * W, npix = B * plane;
+  std::vector<const float*> ptrs(n);
+  for (int i = 0; i < n; ++i) {
+    TORCH_CHECK(preds[i].is_cuda() && preds[i].is_contiguous() &&
+                preds[i].dtype() == torch::kFloat32 &&
+                preds[i].sizes() == gt.sizes());
+    ptrs[i] = preds[i].data_ptr<float>();
+  }
+  auto partials = torch::empty({1024, (long)n + 5},
+                               gt.options().dtype(torch::kFloat32));
+  auto out = torch::empty({5}, gt.options().dtype(torch::kFloat32));
+  const c10::cuda::CUDAGuard guard(gt.device());
+  hipStream_t stream = at::cuda::getCurrentCUDAStream().stream();
+  flowhip_seq_loss_fwd_launch(ptrs.data(), n, gt.data_ptr<float>(),
+                              valid.data_ptr<float>(),
+                              partials.data_ptr<float>(),
+                              out.data_ptr<float>(), npix, plane,
+                              (float)max_flow, (float)gamma, gt.numel(),
+                              stream);
+  return out;
+}
+
+std::vector<torch::Tensor> seq_loss_bwd(std::vector<torch::Tensor> preds,
+                                        torch::Tensor gt, torch::Tensor valid,
+                                        torch::Tensor gloss, double gamma,
+                                        double max_flow) {
+  const int n = (int)preds.size();
+  const long B = gt.size(0), H = gt.size(2), W = gt.size(3);
+  const long plane = H * W, npix = B * plane;
+  std::vector<const float*> ptrs(n);
+  std::vector<float*> gptrs(n);
+  std::vector<torch::Tensor> grads;
+  grads.reserve(n);
+  for (int i = 0; i < n; ++i) {
+    ptrs[i] = preds[i].data_ptr<float>();
+    grads.push_back(torch::empty_like(preds[i]));
+    gptrs[i] = grads[i].data_ptr<float>();
+  }
+  const c10::cuda::CUDAGuard guard(gt.device());
+  hipStream_t stream = at::cuda::getCurrentCUDAStream().stream();
+  flowhip_seq_loss_bwd_launch(ptrs.data(), gptrs.data(), n,
+                              gt.data_ptr<float>(), valid.data_ptr<float>(),
+                              gloss.data_ptr<float>(), npix, plane,
+                              (float)max_flow, (float)gamma, gt.numel(),
+                              stream);
+  return grads;
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -288,10 +446,20 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "fused multi-level correlation window lookup");
   m.def("corr_lookup_bwd", &corr_lookup_bwd,
         "backward of corr_lookup_fwd (pyramid grads)");
+  m.def("corr_pyramid_fwd", &corr_pyramid_fwd,
+        "fused avg-pool pyramid build (levels 1..n-1)");
+  m.def("corr_pyramid_bwd", &corr_pyramid_bwd,
+        "fused pyramid backward combine -> dcorr");
   m.def("convex_up_fwd", &convex_up_fwd, "fused convex-combination upsample");
   m.def("convex_up_bwd", &convex_up_bwd, "backward of convex_up_fwd");
   m.def("nconv_fwd", &nconv_fwd,
         "fused normalized convolution forward (out, cout)");
+  m.def("seq_loss_fwd", &seq_loss_fwd,
+        "fused sequence loss forward -> [loss, epe, 1px, 3px, 5px]");
+  m.def("seq_loss_bwd", &seq_loss_bwd,
+        "fused sequence loss backward -> per-prediction grads");
+  m.def("nconv_bwd_prep", &nconv_bwd_prep,
+        "fused elementwise preamble of nconv backward (dnomin, ddenom)");
   m.def("nconv_bwd", &nconv_bwd,
         "fused normalized convolution backward (ddata, dconf, dweight)");
 }

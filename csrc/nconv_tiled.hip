@@ -410,3 +410,45 @@ bool flowhip_nconv_wrw_tiled_launch(const float* dnomin, const float* ddenom,
                      stream, partials, dweight, nblocks, nw);
   return true;
 }
+
+// ---------------------------------------------------------------------------
+// Backward elementwise preamble, fused (was ~7 torch elementwise kernels per
+// nconv backward — gout/de, -gout*ratio/de, +gcout/s, contiguous copies):
+//   de     = cout * s[co] + eps          (denom reconstructed from cout)
+//   dnomin = gout / de
+//   ddenom = -gout * (out - bias[co]) / de  [+ gcout / s[co]]
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(NCT_THREADS) void nconv_bwd_prep_kernel(
+    const float* __restrict__ gout, const float* __restrict__ gcout,
+    const float* __restrict__ out, const float* __restrict__ cout,
+    const float* __restrict__ wsum,   // (Co) sum of weights per out channel
+    const float* __restrict__ bias,   // (Co) or nullptr
+    float* __restrict__ dnomin, float* __restrict__ ddenom,
+    long total, long plane, int Co, float eps) {
+  for (long idx = (long)blockIdx.x * NCT_THREADS + threadIdx.x; idx < total;
+       idx += (long)gridDim.x * NCT_THREADS) {
+    const int co = (int)((idx / plane) % Co);
+    const float s = wsum[co];
+    const float go = gout[idx];
+    const float de = cout[idx] * s + eps;
+    const float inv_de = 1.0f / de;
+    const float ratio = bias ? out[idx] - bias[co] : out[idx];
+    float dd = -go * ratio * inv_de;
+    if (gcout != nullptr) dd += gcout[idx] / s;
+    dnomin[idx] = go * inv_de;
+    ddenom[idx] = dd;
+  }
+}
+
+void flowhip_nconv_bwd_prep_launch(const float* gout, const float* gcout,
+                                   const float* out, const float* cout,
+                                   const float* wsum, const float* bias,
+                                   float* dnomin, float* ddenom, long total,
+                                   long plane, int Co, float eps,
+                                   hipStream_t stream) {
+  long blocks = (total + NCT_THREADS - 1) / NCT_THREADS;
+  if (blocks > 32768) blocks = 32768;
+  hipLaunchKernelGGL(nconv_bwd_prep_kernel, dim3((int)blocks),
+                     dim3(NCT_THREADS), 0, stream, gout, gcout, out, cout,
+                     wsum, bias, dnomin, ddenom, total, plane, Co, eps);
+}

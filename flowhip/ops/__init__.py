@@ -18,7 +18,19 @@ SURVEY.md §2.2 kernel inventory):
 import torch
 
 from . import _ext, torch_ref
-from .torch_ref import MAX_FLOW, sequence_loss  # re-export  # noqa: F401
+from .torch_ref import MAX_FLOW  # re-export  # noqa: F401
+
+
+def sequence_loss(flow_preds, flow_gt, valid, gamma=0.8, max_flow=MAX_FLOW):
+    """gamma-weighted L1 over the prediction sequence + final-pred metrics
+    (kernel #12; reference train.py:46-71)."""
+    if (_ext.use_hip(flow_gt) and 1 <= len(flow_preds) <= 32
+            and all(p.dtype == torch.float32 for p in flow_preds)):
+        from .functional_loss import sequence_loss_fused
+        return sequence_loss_fused(flow_preds, flow_gt, valid, gamma,
+                                   max_flow)
+    return torch_ref.sequence_loss(flow_preds, flow_gt, valid, gamma,
+                                   max_flow)
 
 
 def corr_volume(fmap1, fmap2):
@@ -29,8 +41,12 @@ def corr_volume(fmap1, fmap2):
 
 
 def corr_pyramid(corr, num_levels=4):
-    # avg_pool2d chain; ROCm-PyTorch path is already memory-bound optimal for
-    # this minor op (<1% of step — see profiles/), torch autograd handles bwd.
+    # fused single-pass build + single-kernel backward combine (LDS cap:
+    # maps over 56x128 fall back to the torch avg_pool chain)
+    if (_ext.use_hip(corr) and 2 <= num_levels <= 4
+            and corr.shape[-2] * corr.shape[-1] <= 56 * 128):
+        from .functional import CorrPyramidFn
+        return list(CorrPyramidFn.apply(corr, num_levels))
     return torch_ref.corr_pyramid(corr, num_levels)
 
 

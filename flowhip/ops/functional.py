@@ -130,3 +130,26 @@ class CorrLookupFn(torch.autograd.Function):
             grad, coords, ctx.radius,
             [list(s) for s in ctx.level_shapes], ctx.cl)
         return (None, None, *grads)
+
+
+class CorrPyramidFn(torch.autograd.Function):
+    """Fused avg-pool pyramid (reference corr.py:19-21).
+
+    forward: one LDS-staged kernel emits levels 1..n-1 (level 0 = input,
+    passed through). backward: one gather kernel combines the per-level
+    grads, dcorr[y,x] = g0 + g1[y/2,x/2]/4 + g2[..]/16 + g3[..]/64 —
+    replacing the 3-deep avg_pool2d-backward chain.
+    """
+
+    @staticmethod
+    def forward(ctx, corr, num_levels):
+        corr = corr.contiguous()
+        levels = _ext.ext().corr_pyramid_fwd(corr, int(num_levels))
+        ctx.corr_shape = list(corr.shape)
+        return (corr, *levels)
+
+    @staticmethod
+    def backward(ctx, *grads):
+        gs = [g.contiguous() if g is not None else None for g in grads]
+        dcorr = _ext.ext().corr_pyramid_bwd(gs, ctx.corr_shape)
+        return dcorr, None

@@ -48,23 +48,16 @@ class NConv2dFn(torch.autograd.Function):
         pad = ctx.padding if isinstance(ctx.padding, int) else ctx.padding[0]
         eps = ctx.eps
 
-        s = weight.sum(dim=(1, 2, 3))                      # (Co)
-        denom = cout * s.view(1, -1, 1, 1)
-        de = denom + eps
+        s = weight.sum(dim=(1, 2, 3)).contiguous()         # (Co)
 
         gout = gout.contiguous()
         gcout = gcout.contiguous() if gcout is not None else None
 
-        # nomin/de = out (− bias when present)
-        ratio = out
-        if ctx.has_bias:
-            ratio = out - maybe_bias[0].view(1, -1, 1, 1)
-
-        dnomin = (gout / de).contiguous()
-        ddenom = -gout * ratio / de
-        if gcout is not None:
-            ddenom = ddenom + gcout / s.view(1, -1, 1, 1)
-        ddenom = ddenom.contiguous()
+        # fused elementwise preamble (one kernel): denom reconstructed from
+        # cout; dnomin = gout/de, ddenom = -gout*(out-bias)/de + gcout/s
+        dnomin, ddenom = _ext.ext().nconv_bwd_prep(
+            gout, gcout, out, cout, s,
+            maybe_bias[0] if ctx.has_bias else None, eps)
 
         ddata, dconf, dweight = _ext.ext().nconv_bwd(dnomin, ddenom, data,
                                                      conf, weight)

@@ -312,3 +312,60 @@ class TestHipGraph:
             low_e2, up_e2 = model(img3, img2, iters=4, test_mode=True)
         low_g2, up_g2 = g(img3, img2)
         torch.testing.assert_close(up_g2, up_e2, atol=0.4, rtol=0.1)
+
+
+class TestCorrPyramidFused:
+    def test_forward_matches_avgpool_chain(self):
+        from flowhip import ops
+        from flowhip.ops import torch_ref
+        torch.manual_seed(5)
+        for (h, w) in [(56, 128), (23, 31), (36, 120)]:
+            corr = torch.randn(64, 1, h, w, device=_dev())
+            got = ops.corr_pyramid(corr, 4)
+            ref = torch_ref.corr_pyramid(corr.cpu(), 4)
+            assert len(got) == len(ref)
+            for g, r in zip(got, ref):
+                torch.testing.assert_close(g.cpu(), r, atol=1e-6, rtol=1e-6)
+
+    def test_backward_matches_autograd(self):
+        from flowhip import ops
+        from flowhip.ops import torch_ref
+        torch.manual_seed(6)
+        corr = torch.randn(16, 1, 24, 40, device=_dev(), requires_grad=True)
+        corr_cpu = corr.detach().cpu().requires_grad_(True)
+        levels = ops.corr_pyramid(corr, 4)
+        ref_levels = torch_ref.corr_pyramid(corr_cpu, 4)
+        gs = [torch.randn_like(l) for l in levels]
+        torch.autograd.backward(levels, gs)
+        torch.autograd.backward(ref_levels, [g.cpu() for g in gs])
+        torch.testing.assert_close(corr.grad.cpu(), corr_cpu.grad,
+                                   atol=1e-5, rtol=1e-5)
+
+
+class TestSequenceLossFused:
+    def test_matches_torch_ref(self):
+        from flowhip import ops
+        from flowhip.ops import torch_ref
+        torch.manual_seed(7)
+        B, H, W, n = 2, 64, 96, 5
+        preds = [torch.randn(B, 2, H, W, device=_dev(), requires_grad=True)
+                 for _ in range(n)]
+        gt = torch.randn(B, 2, H, W, device=_dev()) * 30
+        gt[0, :, :8, :8] = 500.0  # exercise the max-flow exclusion
+        valid = (torch.rand(B, H, W, device=_dev()) > 0.2).float()
+
+        loss, metrics = ops.sequence_loss(preds, gt, valid, gamma=0.85)
+        preds_cpu = [p.detach().cpu().requires_grad_(True) for p in preds]
+        ref_loss, ref_metrics = torch_ref.sequence_loss(
+            preds_cpu, gt.cpu(), valid.cpu(), gamma=0.85)
+
+        assert loss.item() == pytest.approx(ref_loss.item(), rel=1e-4)
+        for k in ("epe", "1px", "3px", "5px"):
+            assert metrics[k] == pytest.approx(ref_metrics[k], rel=1e-4,
+                                               abs=1e-6)
+
+        loss.backward()
+        ref_loss.backward()
+        for p, pc in zip(preds, preds_cpu):
+            torch.testing.assert_close(p.grad.cpu(), pc.grad,
+                                       atol=1e-6, rtol=1e-5)
