@@ -55,6 +55,20 @@ bool flowhip_nconv_wrw_tiled_launch(const float* dnomin, const float* ddenom,
                                     float* partials, float* dweight, int N,
                                     int Ci, int Co, int H, int W, int K,
                                     hipStream_t stream);
+void flowhip_gru_gate1_fwd_launch(const void* zr, const void* h, void* z,
+                                  void* rh, long total, int C, long P,
+                                  int is_bf16, int cl, hipStream_t stream);
+void flowhip_gru_gate1_bwd_launch(const void* dz, const void* drh,
+                                  const void* zr, const void* h, void* dzr,
+                                  void* dh, long total, int C, long P,
+                                  int is_bf16, int cl, hipStream_t stream);
+void flowhip_gru_gate2_fwd_launch(const void* qp, const void* z,
+                                  const void* h, void* hnew, long total,
+                                  int is_bf16, hipStream_t stream);
+void flowhip_gru_gate2_bwd_launch(const void* dhnew, const void* qp,
+                                  const void* z, const void* h, void* dqp,
+                                  void* dz, void* dh, long total, int is_bf16,
+                                  hipStream_t stream);
 void flowhip_seq_loss_fwd_launch(const float* const* preds, int n,
                                  const float* gt, const float* valid,
                                  float* partials, float* out, long npix,
@@ -436,6 +450,89 @@ std::vector<torch::Tensor> seq_loss_bwd(std::vector<torch::Tensor> preds,
   return grads;
 }
 
+namespace gg {
+
+bool is_cl(const torch::Tensor& t) {
+  return t.is_contiguous(torch::MemoryFormat::ChannelsLast);
+}
+
+torch::Tensor make_like(const torch::Tensor& t, bool cl) {
+  return cl ? torch::empty(t.sizes(), t.options(),
+                           torch::MemoryFormat::ChannelsLast)
+            : torch::empty(t.sizes(), t.options());
+}
+
+}  // namespace gg
+
+std::vector<torch::Tensor> gru_gate1_fwd(torch::Tensor zr, torch::Tensor h) {
+  TORCH_CHECK(zr.is_cuda() && h.is_cuda());
+  TORCH_CHECK(zr.scalar_type() == h.scalar_type());
+  const bool bf16 = zr.scalar_type() == torch::kBFloat16;
+  TORCH_CHECK(bf16 || zr.scalar_type() == torch::kFloat32);
+  const bool cl = gg::is_cl(zr);
+  TORCH_CHECK(cl ? gg::is_cl(h) : h.is_contiguous(),
+              "gru_gate1: layouts must match");
+  const int C = h.size(1);
+  const long P = (long)h.size(2) * h.size(3);
+  TORCH_CHECK(zr.size(1) == 2 * C && zr.size(0) == h.size(0));
+  auto z = gg::make_like(h, cl);
+  auto rh = gg::make_like(h, cl);
+  const c10::cuda::CUDAGuard guard(h.device());
+  hipStream_t stream = at::cuda::getCurrentCUDAStream().stream();
+  flowhip_gru_gate1_fwd_launch(zr.data_ptr(), h.data_ptr(), z.data_ptr(),
+                               rh.data_ptr(), h.numel(), C, P, bf16 ? 1 : 0,
+                               cl ? 1 : 0, stream);
+  return {z, rh};
+}
+
+std::vector<torch::Tensor> gru_gate1_bwd(c10::optional<torch::Tensor> dz,
+                                         torch::Tensor drh, torch::Tensor zr,
+                                         torch::Tensor h) {
+  const bool bf16 = zr.scalar_type() == torch::kBFloat16;
+  const bool cl = gg::is_cl(zr);
+  const int C = h.size(1);
+  const long P = (long)h.size(2) * h.size(3);
+  auto dzr = gg::make_like(zr, cl);
+  auto dh = gg::make_like(h, cl);
+  const c10::cuda::CUDAGuard guard(h.device());
+  hipStream_t stream = at::cuda::getCurrentCUDAStream().stream();
+  flowhip_gru_gate1_bwd_launch(
+      dz.has_value() ? dz->data_ptr() : nullptr, drh.data_ptr(),
+      zr.data_ptr(), h.data_ptr(), dzr.data_ptr(), dh.data_ptr(), h.numel(),
+      C, P, bf16 ? 1 : 0, cl ? 1 : 0, stream);
+  return {dzr, dh};
+}
+
+torch::Tensor gru_gate2_fwd(torch::Tensor qp, torch::Tensor z,
+                            torch::Tensor h) {
+  const bool bf16 = qp.scalar_type() == torch::kBFloat16;
+  const bool cl = gg::is_cl(qp);
+  auto hnew = gg::make_like(h, cl);
+  const c10::cuda::CUDAGuard guard(h.device());
+  hipStream_t stream = at::cuda::getCurrentCUDAStream().stream();
+  flowhip_gru_gate2_fwd_launch(qp.data_ptr(), z.data_ptr(), h.data_ptr(),
+                               hnew.data_ptr(), h.numel(), bf16 ? 1 : 0,
+                               stream);
+  return hnew;
+}
+
+std::vector<torch::Tensor> gru_gate2_bwd(torch::Tensor dhnew,
+                                         torch::Tensor qp, torch::Tensor z,
+                                         torch::Tensor h) {
+  const bool bf16 = qp.scalar_type() == torch::kBFloat16;
+  const bool cl = gg::is_cl(qp);
+  auto dqp = gg::make_like(qp, cl);
+  auto dz = gg::make_like(z, cl);
+  auto dh = gg::make_like(h, cl);
+  const c10::cuda::CUDAGuard guard(h.device());
+  hipStream_t stream = at::cuda::getCurrentCUDAStream().stream();
+  flowhip_gru_gate2_bwd_launch(dhnew.data_ptr(), qp.data_ptr(), z.data_ptr(),
+                               h.data_ptr(), dqp.data_ptr(), dz.data_ptr(),
+                               dh.data_ptr(), h.numel(), bf16 ? 1 : 0,
+                               stream);
+  return {dqp, dz, dh};
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -454,6 +551,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("convex_up_bwd", &convex_up_bwd, "backward of convex_up_fwd");
   m.def("nconv_fwd", &nconv_fwd,
         "fused normalized convolution forward (out, cout)");
+  m.def("gru_gate1_fwd", &gru_gate1_fwd, "fused GRU z/r gates + r*h");
+  m.def("gru_gate1_bwd", &gru_gate1_bwd, "backward of gru_gate1");
+  m.def("gru_gate2_fwd", &gru_gate2_fwd, "fused GRU tanh + lerp update");
+  m.def("gru_gate2_bwd", &gru_gate2_bwd, "backward of gru_gate2");
   m.def("seq_loss_fwd", &seq_loss_fwd,
         "fused sequence loss forward -> [loss, epe, 1px, 3px, 5px]");
   m.def("seq_loss_bwd", &seq_loss_bwd,

@@ -43,9 +43,14 @@ class RAFT_NC_DBL(_RAFTBase):
 
     def upsample_flow(self, flow_lr, guidance):
         """H/8 flow -> H flow: nearest x2 pre-upsample then NConvUpsampler
-        (scale=4) guided by the GRU hidden state (raft_nc_dbl.py:107-112)."""
-        flow_lr = F.interpolate(flow_lr, scale_factor=2, mode="nearest")
-        return self.upsampler(flow_lr, guidance)
+        (scale=4) guided by the GRU hidden state (raft_nc_dbl.py:107-112).
+
+        The upsampler subtree runs NCHW (see utils/layout.py) — one explicit
+        layout conversion of the guidance here instead of mixed-layout
+        cat/copy churn inside."""
+        flow_lr = F.interpolate(flow_lr.contiguous(), scale_factor=2,
+                                mode="nearest")
+        return self.upsampler(flow_lr, guidance.contiguous())
 
     def forward(self, image1, image2, iters=12, flow_init=None, upsample=True,
                 test_mode=False):

@@ -40,6 +40,11 @@ class ConvGRU(nn.Module):
         hx = torch.cat([h, x], dim=1)
         zr = F.conv2d(hx, torch.cat([self.convz.weight, self.convr.weight]),
                       torch.cat([self.convz.bias, self.convr.bias]), padding=1)
+        if h.is_cuda:
+            from ..ops.functional_gru import GruGate1Fn, GruGate2Fn
+            z, rh = GruGate1Fn.apply(zr, h)
+            qp = self.convq(torch.cat([rh, x], dim=1))
+            return GruGate2Fn.apply(qp, z, h)
         z, r = torch.sigmoid(zr).chunk(2, dim=1)
         q = torch.tanh(self.convq(torch.cat([r * h, x], dim=1)))
         return (1 - z) * h + z * q
@@ -63,6 +68,13 @@ class SepConvGRU(nn.Module):
         hx = torch.cat([h, x], dim=1)
         zr = F.conv2d(hx, torch.cat([convz.weight, convr.weight]),
                       torch.cat([convz.bias, convr.bias]), padding=padding)
+        if h.is_cuda:
+            # fused gate kernels (ops/functional_gru): one kernel for
+            # sigmoid/chunk/r*h, one for tanh + lerp, fused backwards
+            from ..ops.functional_gru import GruGate1Fn, GruGate2Fn
+            z, rh = GruGate1Fn.apply(zr, h)
+            qp = convq(torch.cat([rh, x], dim=1))
+            return GruGate2Fn.apply(qp, z, h)
         z, r = torch.sigmoid(zr).chunk(2, dim=1)
         q = torch.tanh(convq(torch.cat([r * h, x], dim=1)))
         return (1 - z) * h + z * q

@@ -31,9 +31,18 @@ def channels_last_enabled():
 
 
 def apply_channels_last(model):
-    """Convert a CUDA-resident model's conv weights to channels_last."""
+    """Convert conv weights to channels_last — except the NCUP upsampler.
+
+    The upsampler subtree (nconv kernels + the tiny fp32 weights-est CNN)
+    runs NCHW: the fused nconv kernels are NCHW-native, and MIOpen's fp32
+    NCHW Winograd (Sp3AsmConv) beats its NHWC fp32 igemm ~3x at the
+    weights-est shapes (profiles/r01 trace3 vs trace1). Keeping the subtree
+    NCHW also removes the mixed-layout cat/copy storm at the boundary."""
     if _CHANNELS_LAST:
         model.to(memory_format=torch.channels_last)
+        ups = getattr(model, "upsampler", None)
+        if ups is not None:
+            ups.to(memory_format=torch.contiguous_format)
     return model
 
 

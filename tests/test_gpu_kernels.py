@@ -369,3 +369,42 @@ class TestSequenceLossFused:
         for p, pc in zip(preds, preds_cpu):
             torch.testing.assert_close(p.grad.cpu(), pc.grad,
                                        atol=1e-6, rtol=1e-5)
+
+
+class TestGruGatesFused:
+    @pytest.mark.parametrize("dtype,cl", [
+        (torch.float32, False), (torch.float32, True),
+        (torch.bfloat16, True)])
+    def test_matches_torch_math(self, dtype, cl):
+        from flowhip.ops.functional_gru import GruGate1Fn, GruGate2Fn
+        torch.manual_seed(11)
+        B, C, H, W = 2, 32, 14, 18
+        mf = torch.channels_last if cl else torch.contiguous_format
+
+        def mk(c, grad=True):
+            t = torch.randn(B, c, H, W, device=_dev()).to(dtype) \
+                .contiguous(memory_format=mf)
+            return t.requires_grad_(grad)
+
+        zr, h, qp = mk(2 * C), mk(C), mk(C)
+        z, rh = GruGate1Fn.apply(zr, h)
+        hnew = GruGate2Fn.apply(qp, z, h)
+
+        zr2 = zr.detach().float().requires_grad_(True)
+        h2 = h.detach().float().requires_grad_(True)
+        qp2 = qp.detach().float().requires_grad_(True)
+        z2, r2 = torch.sigmoid(zr2).chunk(2, dim=1)
+        hnew2 = (1 - z2) * h2 + z2 * torch.tanh(qp2)
+
+        tol = dict(atol=1e-5, rtol=1e-5) if dtype == torch.float32 else \
+            dict(atol=2e-2, rtol=2e-2)
+        torch.testing.assert_close(hnew.float(), hnew2, **tol)
+        torch.testing.assert_close(rh.float(), (r2 * h2), **tol)
+
+        g = torch.randn_like(hnew2)
+        # route rh's grad too (as convq would): use rh.sum()*0 + hnew path
+        (hnew.float() * g).sum().backward()
+        (hnew2 * g).sum().backward()
+        torch.testing.assert_close(zr.grad.float(), zr2.grad, **tol)
+        torch.testing.assert_close(qp.grad.float(), qp2.grad, **tol)
+        torch.testing.assert_close(h.grad.float(), h2.grad, **tol)
