@@ -115,16 +115,19 @@ def main():
         if on_gpu:
             torch.cuda.synchronize()
         with profile(activities=[ProfilerActivity.CPU, ProfilerActivity.CUDA],
-                     record_shapes=False) as prof:
+                     record_shapes=True) as prof:
             for _ in range(max(cli.steps, 2)):
                 step()
             if on_gpu:
                 torch.cuda.synchronize()
-        table = prof.key_averages().table(
-            sort_by="self_cuda_time_total" if on_gpu else "self_cpu_time_total",
-            row_limit=60)
+        sort = "self_cuda_time_total" if on_gpu else "self_cpu_time_total"
+        table = prof.key_averages().table(sort_by=sort, row_limit=60)
+        shapes = prof.key_averages(group_by_input_shape=True).table(
+            sort_by=sort, row_limit=80)
         with open(cli.torch_profile, "w") as f:
             f.write(table)
+            f.write("\n\n==== grouped by input shape ====\n\n")
+            f.write(shapes)
         if rank == 0:
             print(table[:4000])
         return

@@ -55,6 +55,13 @@ bool flowhip_nconv_wrw_tiled_launch(const float* dnomin, const float* ddenom,
                                     float* partials, float* dweight, int N,
                                     int Ci, int Co, int H, int W, int K,
                                     hipStream_t stream);
+void flowhip_zero_inject_fwd_launch(const float* inp, float* out, long total,
+                                    int ih, int iw, int oh, int ow, int sH,
+                                    int sW, hipStream_t stream);
+void flowhip_zero_inject_bwd_launch(const float* gout, float* dinp,
+                                    long total, int ih, int iw, int oh,
+                                    int ow, int sH, int sW,
+                                    hipStream_t stream);
 void flowhip_gru_gate1_fwd_launch(const void* zr, const void* h, void* z,
                                   void* rh, long total, int C, long P,
                                   int is_bf16, int cl, hipStream_t stream);
@@ -462,6 +469,35 @@ torch::Tensor make_like(const torch::Tensor& t, bool cl) {
             : torch::empty(t.sizes(), t.options());
 }
 
+torch::Tensor zero_inject_fwd(torch::Tensor inp, int64_t sH, int64_t sW,
+                              int64_t oh, int64_t ow) {
+  TORCH_CHECK(inp.is_cuda() && inp.dtype() == torch::kFloat32);
+  auto x = inp.contiguous();
+  const long N = x.size(0), C = x.size(1);
+  const int ih = x.size(2), iw = x.size(3);
+  auto out = torch::empty({N, C, oh, ow}, x.options());
+  const c10::cuda::CUDAGuard guard(x.device());
+  hipStream_t stream = at::cuda::getCurrentCUDAStream().stream();
+  flowhip_zero_inject_fwd_launch(x.data_ptr<float>(), out.data_ptr<float>(),
+                                 out.numel(), ih, iw, (int)oh, (int)ow,
+                                 (int)sH, (int)sW, stream);
+  return out;
+}
+
+torch::Tensor zero_inject_bwd(torch::Tensor gout, int64_t sH, int64_t sW,
+                              int64_t ih, int64_t iw) {
+  auto g = gout.contiguous();
+  const long N = g.size(0), C = g.size(1);
+  const int oh = g.size(2), ow = g.size(3);
+  auto dinp = torch::empty({N, C, ih, iw}, g.options());
+  const c10::cuda::CUDAGuard guard(g.device());
+  hipStream_t stream = at::cuda::getCurrentCUDAStream().stream();
+  flowhip_zero_inject_bwd_launch(g.data_ptr<float>(), dinp.data_ptr<float>(),
+                                 dinp.numel(), (int)ih, (int)iw, oh, ow,
+                                 (int)sH, (int)sW, stream);
+  return dinp;
+}
+
 }  // namespace gg
 
 std::vector<torch::Tensor> gru_gate1_fwd(torch::Tensor zr, torch::Tensor h) {
@@ -533,6 +569,35 @@ std::vector<torch::Tensor> gru_gate2_bwd(torch::Tensor dhnew,
   return {dqp, dz, dh};
 }
 
+torch::Tensor zero_inject_fwd(torch::Tensor inp, int64_t sH, int64_t sW,
+                              int64_t oh, int64_t ow) {
+  TORCH_CHECK(inp.is_cuda() && inp.dtype() == torch::kFloat32);
+  auto x = inp.contiguous();
+  const long N = x.size(0), C = x.size(1);
+  const int ih = x.size(2), iw = x.size(3);
+  auto out = torch::empty({N, C, oh, ow}, x.options());
+  const c10::cuda::CUDAGuard guard(x.device());
+  hipStream_t stream = at::cuda::getCurrentCUDAStream().stream();
+  flowhip_zero_inject_fwd_launch(x.data_ptr<float>(), out.data_ptr<float>(),
+                                 out.numel(), ih, iw, (int)oh, (int)ow,
+                                 (int)sH, (int)sW, stream);
+  return out;
+}
+
+torch::Tensor zero_inject_bwd(torch::Tensor gout, int64_t sH, int64_t sW,
+                              int64_t ih, int64_t iw) {
+  auto g = gout.contiguous();
+  const long N = g.size(0), C = g.size(1);
+  const int oh = g.size(2), ow = g.size(3);
+  auto dinp = torch::empty({N, C, ih, iw}, g.options());
+  const c10::cuda::CUDAGuard guard(g.device());
+  hipStream_t stream = at::cuda::getCurrentCUDAStream().stream();
+  flowhip_zero_inject_bwd_launch(g.data_ptr<float>(), dinp.data_ptr<float>(),
+                                 dinp.numel(), (int)ih, (int)iw, oh, ow,
+                                 (int)sH, (int)sW, stream);
+  return dinp;
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -551,6 +616,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("convex_up_bwd", &convex_up_bwd, "backward of convex_up_fwd");
   m.def("nconv_fwd", &nconv_fwd,
         "fused normalized convolution forward (out, cout)");
+  m.def("zero_inject_fwd", &zero_inject_fwd, "sparse injection scatter");
+  m.def("zero_inject_bwd", &zero_inject_bwd, "backward gather of inject");
   m.def("gru_gate1_fwd", &gru_gate1_fwd, "fused GRU z/r gates + r*h");
   m.def("gru_gate1_bwd", &gru_gate1_bwd, "backward of gru_gate1");
   m.def("gru_gate2_fwd", &gru_gate2_fwd, "fused GRU tanh + lerp update");

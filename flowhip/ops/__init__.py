@@ -80,6 +80,11 @@ def conf_pool(data, conf, ds_factor=2, pooling_type="conf_based"):
 
 
 def zero_inject(inp, scale_h, scale_w, out_h=None, out_w=None):
+    if _ext.use_hip(inp) and inp.dtype == torch.float32:
+        from .functional_upsample import ZeroInjectFn
+        oh = out_h if out_h is not None else inp.shape[2] * scale_h
+        ow = out_w if out_w is not None else inp.shape[3] * scale_w
+        return ZeroInjectFn.apply(inp, scale_h, scale_w, oh, ow)
     return torch_ref.zero_inject(inp, scale_h, scale_w, out_h, out_w)
 
 

@@ -22,3 +22,19 @@ class ConvexUpsampleFn(torch.autograd.Function):
         gflow, gmask = _ext.ext().convex_up_bwd(gout.contiguous(), flow, mask,
                                                 ctx.factor)
         return gflow, gmask, None
+
+
+class ZeroInjectFn(torch.autograd.Function):
+    """Sparse zero-injection (kernel #8; ref upsampler.py:179-210): one
+    coalesced write pass forward, one strided gather backward."""
+
+    @staticmethod
+    def forward(ctx, inp, sH, sW, oh, ow):
+        ctx.meta = (int(sH), int(sW), inp.shape[2], inp.shape[3])
+        return _ext.ext().zero_inject_fwd(inp, int(sH), int(sW), int(oh),
+                                          int(ow))
+
+    @staticmethod
+    def backward(ctx, gout):
+        sH, sW, ih, iw = ctx.meta
+        return _ext.ext().zero_inject_bwd(gout, sH, sW, ih, iw), None, None, None, None
