@@ -55,6 +55,14 @@ bool flowhip_nconv_wrw_tiled_launch(const float* dnomin, const float* ddenom,
                                     float* partials, float* dweight, int N,
                                     int Ci, int Co, int H, int W, int K,
                                     hipStream_t stream);
+void flowhip_instnorm_cl_fwd_launch(const void* x, void* y, float* mean,
+                                    float* rstd, int N, int C, long P,
+                                    float eps, int is_bf16,
+                                    hipStream_t stream);
+void flowhip_instnorm_cl_bwd_launch(const void* x, const void* dy,
+                                    const float* mean, const float* rstd,
+                                    void* dx, int N, int C, long P,
+                                    int is_bf16, hipStream_t stream);
 void flowhip_zero_inject_fwd_launch(const float* inp, float* out, long total,
                                     int ih, int iw, int oh, int ow, int sH,
                                     int sW, hipStream_t stream);
@@ -498,6 +506,44 @@ torch::Tensor zero_inject_bwd(torch::Tensor gout, int64_t sH, int64_t sW,
   return dinp;
 }
 
+std::vector<torch::Tensor> instnorm_cl_fwd(torch::Tensor x, double eps) {
+  TORCH_CHECK(x.is_cuda() && x.dim() == 4);
+  TORCH_CHECK(x.is_contiguous(torch::MemoryFormat::ChannelsLast),
+              "instnorm_cl: channels_last input required");
+  const bool bf16 = x.scalar_type() == torch::kBFloat16;
+  TORCH_CHECK(bf16 || x.scalar_type() == torch::kFloat32);
+  const int N = x.size(0), C = x.size(1);
+  const long P = (long)x.size(2) * x.size(3);
+  auto y = torch::empty(x.sizes(), x.options(),
+                        torch::MemoryFormat::ChannelsLast);
+  auto mean = torch::empty({(long)N, (long)C},
+                           x.options().dtype(torch::kFloat32));
+  auto rstd = torch::empty_like(mean);
+  const c10::cuda::CUDAGuard guard(x.device());
+  hipStream_t stream = at::cuda::getCurrentCUDAStream().stream();
+  flowhip_instnorm_cl_fwd_launch(x.data_ptr(), y.data_ptr(),
+                                 mean.data_ptr<float>(),
+                                 rstd.data_ptr<float>(), N, C, P, (float)eps,
+                                 bf16 ? 1 : 0, stream);
+  return {y, mean, rstd};
+}
+
+torch::Tensor instnorm_cl_bwd(torch::Tensor x, torch::Tensor dy,
+                              torch::Tensor mean, torch::Tensor rstd) {
+  const bool bf16 = x.scalar_type() == torch::kBFloat16;
+  const int N = x.size(0), C = x.size(1);
+  const long P = (long)x.size(2) * x.size(3);
+  auto dx = torch::empty(x.sizes(), x.options(),
+                         torch::MemoryFormat::ChannelsLast);
+  const c10::cuda::CUDAGuard guard(x.device());
+  hipStream_t stream = at::cuda::getCurrentCUDAStream().stream();
+  flowhip_instnorm_cl_bwd_launch(x.data_ptr(), dy.data_ptr(),
+                                 mean.data_ptr<float>(),
+                                 rstd.data_ptr<float>(), dx.data_ptr(), N, C,
+                                 P, bf16 ? 1 : 0, stream);
+  return dx;
+}
+
 }  // namespace gg
 
 std::vector<torch::Tensor> gru_gate1_fwd(torch::Tensor zr, torch::Tensor h) {
@@ -598,6 +644,44 @@ torch::Tensor zero_inject_bwd(torch::Tensor gout, int64_t sH, int64_t sW,
   return dinp;
 }
 
+std::vector<torch::Tensor> instnorm_cl_fwd(torch::Tensor x, double eps) {
+  TORCH_CHECK(x.is_cuda() && x.dim() == 4);
+  TORCH_CHECK(x.is_contiguous(torch::MemoryFormat::ChannelsLast),
+              "instnorm_cl: channels_last input required");
+  const bool bf16 = x.scalar_type() == torch::kBFloat16;
+  TORCH_CHECK(bf16 || x.scalar_type() == torch::kFloat32);
+  const int N = x.size(0), C = x.size(1);
+  const long P = (long)x.size(2) * x.size(3);
+  auto y = torch::empty(x.sizes(), x.options(),
+                        torch::MemoryFormat::ChannelsLast);
+  auto mean = torch::empty({(long)N, (long)C},
+                           x.options().dtype(torch::kFloat32));
+  auto rstd = torch::empty_like(mean);
+  const c10::cuda::CUDAGuard guard(x.device());
+  hipStream_t stream = at::cuda::getCurrentCUDAStream().stream();
+  flowhip_instnorm_cl_fwd_launch(x.data_ptr(), y.data_ptr(),
+                                 mean.data_ptr<float>(),
+                                 rstd.data_ptr<float>(), N, C, P, (float)eps,
+                                 bf16 ? 1 : 0, stream);
+  return {y, mean, rstd};
+}
+
+torch::Tensor instnorm_cl_bwd(torch::Tensor x, torch::Tensor dy,
+                              torch::Tensor mean, torch::Tensor rstd) {
+  const bool bf16 = x.scalar_type() == torch::kBFloat16;
+  const int N = x.size(0), C = x.size(1);
+  const long P = (long)x.size(2) * x.size(3);
+  auto dx = torch::empty(x.sizes(), x.options(),
+                         torch::MemoryFormat::ChannelsLast);
+  const c10::cuda::CUDAGuard guard(x.device());
+  hipStream_t stream = at::cuda::getCurrentCUDAStream().stream();
+  flowhip_instnorm_cl_bwd_launch(x.data_ptr(), dy.data_ptr(),
+                                 mean.data_ptr<float>(),
+                                 rstd.data_ptr<float>(), dx.data_ptr(), N, C,
+                                 P, bf16 ? 1 : 0, stream);
+  return dx;
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -616,6 +700,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("convex_up_bwd", &convex_up_bwd, "backward of convex_up_fwd");
   m.def("nconv_fwd", &nconv_fwd,
         "fused normalized convolution forward (out, cout)");
+  m.def("instnorm_cl_fwd", &instnorm_cl_fwd,
+        "channels-last InstanceNorm2d forward (y, mean, rstd)");
+  m.def("instnorm_cl_bwd", &instnorm_cl_bwd,
+        "channels-last InstanceNorm2d backward");
   m.def("zero_inject_fwd", &zero_inject_fwd, "sparse injection scatter");
   m.def("zero_inject_bwd", &zero_inject_bwd, "backward gather of inject");
   m.def("gru_gate1_fwd", &gru_gate1_fwd, "fused GRU z/r gates + r*h");

@@ -25,7 +25,8 @@ def _norm(norm_fn, planes, groups_planes=None):
     if norm_fn == "batch":
         return nn.BatchNorm2d(planes)
     if norm_fn == "instance":
-        return nn.InstanceNorm2d(planes)
+        from .norm import InstanceNorm2d
+        return InstanceNorm2d(planes)
     if norm_fn == "none":
         return nn.Sequential()
     raise ValueError(f"unknown norm_fn {norm_fn!r}")

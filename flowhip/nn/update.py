@@ -97,6 +97,8 @@ class SmallMotionEncoder(nn.Module):
         self.conv = nn.Conv2d(128, 80, 3, padding=1)
 
     def forward(self, flow, corr):
+        if corr.is_contiguous(memory_format=torch.channels_last):
+            flow = flow.contiguous(memory_format=torch.channels_last)
         cor = F.relu(self.convc1(corr))
         flo = F.relu(self.convf2(F.relu(self.convf1(flow))))
         out = F.relu(self.conv(torch.cat([cor, flo], dim=1)))
@@ -116,6 +118,11 @@ class BasicMotionEncoder(nn.Module):
         self.conv = nn.Conv2d(64 + 192, 128 - 2, 3, padding=1)
 
     def forward(self, flow, corr):
+        if corr.is_contiguous(memory_format=torch.channels_last):
+            # keep the whole motion-feature chain in one layout: a mixed
+            # cat([cl, nchw]) falls back to NCHW and costs an uncoalesced
+            # layout copy on every GRU conv input (profiles/, tprof6)
+            flow = flow.contiguous(memory_format=torch.channels_last)
         cor = F.relu(self.convc2(F.relu(self.convc1(corr))))
         flo = F.relu(self.convf2(F.relu(self.convf1(flow))))
         out = F.relu(self.conv(torch.cat([cor, flo], dim=1)))

@@ -408,3 +408,44 @@ class TestGruGatesFused:
         torch.testing.assert_close(zr.grad.float(), zr2.grad, **tol)
         torch.testing.assert_close(qp.grad.float(), qp2.grad, **tol)
         torch.testing.assert_close(h.grad.float(), h2.grad, **tol)
+
+
+class TestInstanceNormCL:
+    @pytest.mark.parametrize("dtype,C", [(torch.float32, 64),
+                                         (torch.float32, 96),
+                                         (torch.bfloat16, 128)])
+    def test_matches_torch(self, dtype, C):
+        from flowhip.nn.norm import InstanceNorm2d
+        torch.manual_seed(13)
+        x = torch.randn(3, C, 20, 34, device=_dev()).to(dtype) \
+            .contiguous(memory_format=torch.channels_last).requires_grad_(True)
+        m = InstanceNorm2d(C).to(_dev())
+        y = m(x)
+
+        x2 = x.detach().float().requires_grad_(True)
+        y2 = torch.nn.functional.instance_norm(x2, eps=m.eps)
+
+        tol = dict(atol=1e-5, rtol=1e-5) if dtype == torch.float32 else \
+            dict(atol=3e-2, rtol=3e-2)
+        torch.testing.assert_close(y.float(), y2, **tol)
+
+        g = torch.randn_like(y2)
+        (y.float() * g).sum().backward()
+        (y2 * g).sum().backward()
+        torch.testing.assert_close(x.grad.float(), x2.grad, **tol)
+
+
+class TestZeroInjectHIP:
+    def test_matches_torch_ref(self):
+        from flowhip import ops
+        from flowhip.ops import torch_ref
+        torch.manual_seed(14)
+        x = torch.randn(4, 2, 112, 256, device=_dev(), requires_grad=True)
+        out = ops.zero_inject(x, 4, 4)
+        x2 = x.detach().cpu().requires_grad_(True)
+        ref = torch_ref.zero_inject(x2, 4, 4)
+        torch.testing.assert_close(out.cpu(), ref)
+        g = torch.randn_like(ref)
+        (out * g.to(_dev())).sum().backward()
+        (ref * g).sum().backward()
+        torch.testing.assert_close(x.grad.cpu(), x2.grad)
