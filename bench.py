@@ -23,6 +23,13 @@ import time
 import numpy as np
 import torch
 
+import os as _os
+
+if torch.cuda.is_available() and _os.environ.get("FLOWHIP_MIOPEN_FIND", "1") != "0":
+    # MIOpen find mode: benchmark=True lets MIOpen search its perf-db for
+    # the fastest conv solver per (static) shape during warmup
+    torch.backends.cudnn.benchmark = True
+
 from flowhip.config.args import default_ncup_args
 from flowhip.engine import distributed
 from flowhip.engine.train import fetch_optimizer

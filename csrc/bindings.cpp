@@ -55,14 +55,16 @@ bool flowhip_nconv_wrw_tiled_launch(const float* dnomin, const float* ddenom,
                                     float* partials, float* dweight, int N,
                                     int Ci, int Co, int H, int W, int K,
                                     hipStream_t stream);
+int flowhip_instnorm_partial_rows(int N, int C, long P);
 void flowhip_instnorm_cl_fwd_launch(const void* x, void* y, float* mean,
-                                    float* rstd, int N, int C, long P,
-                                    float eps, int is_bf16,
+                                    float* rstd, float* partials, int N,
+                                    int C, long P, float eps, int is_bf16,
                                     hipStream_t stream);
 void flowhip_instnorm_cl_bwd_launch(const void* x, const void* dy,
                                     const float* mean, const float* rstd,
-                                    void* dx, int N, int C, long P,
-                                    int is_bf16, hipStream_t stream);
+                                    float* gmean, float* gxmean,
+                                    float* partials, void* dx, int N, int C,
+                                    long P, int is_bf16, hipStream_t stream);
 void flowhip_zero_inject_fwd_launch(const float* inp, float* out, long total,
                                     int ih, int iw, int oh, int ow, int sH,
                                     int sW, hipStream_t stream);
@@ -521,10 +523,14 @@ std::vector<torch::Tensor> instnorm_cl_fwd(torch::Tensor x, double eps) {
   auto rstd = torch::empty_like(mean);
   const c10::cuda::CUDAGuard guard(x.device());
   hipStream_t stream = at::cuda::getCurrentCUDAStream().stream();
+  auto partials = torch::empty({(long)flowhip_instnorm_partial_rows(N, C, P),
+                                64},
+                               x.options().dtype(torch::kFloat32));
   flowhip_instnorm_cl_fwd_launch(x.data_ptr(), y.data_ptr(),
                                  mean.data_ptr<float>(),
-                                 rstd.data_ptr<float>(), N, C, P, (float)eps,
-                                 bf16 ? 1 : 0, stream);
+                                 rstd.data_ptr<float>(),
+                                 partials.data_ptr<float>(), N, C, P,
+                                 (float)eps, bf16 ? 1 : 0, stream);
   return {y, mean, rstd};
 }
 
@@ -535,12 +541,20 @@ torch::Tensor instnorm_cl_bwd(torch::Tensor x, torch::Tensor dy,
   const long P = (long)x.size(2) * x.size(3);
   auto dx = torch::empty(x.sizes(), x.options(),
                          torch::MemoryFormat::ChannelsLast);
+  auto gmean = torch::empty_like(mean);
+  auto gxmean = torch::empty_like(mean);
+  auto partials = torch::empty({(long)flowhip_instnorm_partial_rows(N, C, P),
+                                64},
+                               x.options().dtype(torch::kFloat32));
   const c10::cuda::CUDAGuard guard(x.device());
   hipStream_t stream = at::cuda::getCurrentCUDAStream().stream();
   flowhip_instnorm_cl_bwd_launch(x.data_ptr(), dy.data_ptr(),
                                  mean.data_ptr<float>(),
-                                 rstd.data_ptr<float>(), dx.data_ptr(), N, C,
-                                 P, bf16 ? 1 : 0, stream);
+                                 rstd.data_ptr<float>(),
+                                 gmean.data_ptr<float>(),
+                                 gxmean.data_ptr<float>(),
+                                 partials.data_ptr<float>(), dx.data_ptr(),
+                                 N, C, P, bf16 ? 1 : 0, stream);
   return dx;
 }
 
@@ -659,10 +673,14 @@ std::vector<torch::Tensor> instnorm_cl_fwd(torch::Tensor x, double eps) {
   auto rstd = torch::empty_like(mean);
   const c10::cuda::CUDAGuard guard(x.device());
   hipStream_t stream = at::cuda::getCurrentCUDAStream().stream();
+  auto partials = torch::empty({(long)flowhip_instnorm_partial_rows(N, C, P),
+                                64},
+                               x.options().dtype(torch::kFloat32));
   flowhip_instnorm_cl_fwd_launch(x.data_ptr(), y.data_ptr(),
                                  mean.data_ptr<float>(),
-                                 rstd.data_ptr<float>(), N, C, P, (float)eps,
-                                 bf16 ? 1 : 0, stream);
+                                 rstd.data_ptr<float>(),
+                                 partials.data_ptr<float>(), N, C, P,
+                                 (float)eps, bf16 ? 1 : 0, stream);
   return {y, mean, rstd};
 }
 
@@ -673,12 +691,20 @@ torch::Tensor instnorm_cl_bwd(torch::Tensor x, torch::Tensor dy,
   const long P = (long)x.size(2) * x.size(3);
   auto dx = torch::empty(x.sizes(), x.options(),
                          torch::MemoryFormat::ChannelsLast);
+  auto gmean = torch::empty_like(mean);
+  auto gxmean = torch::empty_like(mean);
+  auto partials = torch::empty({(long)flowhip_instnorm_partial_rows(N, C, P),
+                                64},
+                               x.options().dtype(torch::kFloat32));
   const c10::cuda::CUDAGuard guard(x.device());
   hipStream_t stream = at::cuda::getCurrentCUDAStream().stream();
   flowhip_instnorm_cl_bwd_launch(x.data_ptr(), dy.data_ptr(),
                                  mean.data_ptr<float>(),
-                                 rstd.data_ptr<float>(), dx.data_ptr(), N, C,
-                                 P, bf16 ? 1 : 0, stream);
+                                 rstd.data_ptr<float>(),
+                                 gmean.data_ptr<float>(),
+                                 gxmean.data_ptr<float>(),
+                                 partials.data_ptr<float>(), dx.data_ptr(),
+                                 N, C, P, bf16 ? 1 : 0, stream);
   return dx;
 }
 

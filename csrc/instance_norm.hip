@@ -5,155 +5,220 @@
 // PyTorch lowers InstanceNorm to batch_norm on a (1, N*C, H, W) view; on a
 // channels-last tensor that view is non-contiguous, so every call pays an
 // uncoalesced NCHW round-trip copy (~97 µs at (6,64,224,512), ~37 of them
-// per training step — tprof6 in profiles/). This kernel reduces the (N,P,C)
-// layout directly: lanes = consecutive channels (coalesced), waves split P.
+// per training step — tprof6 in profiles/). These kernels reduce the
+// (N,P,C) layout directly: lanes = consecutive channels (coalesced).
+//
+// Parallel structure (v2 — v1 used one workgroup per (n, c-block), i.e.
+// 6-12 workgroups on a 256-CU chip, a 10x step regression): three stages,
+//   1. partial: grid (N * CB * PCHUNKS), each block reduces a P-chunk ->
+//      partials (n, cb, chunk, {s1,s2}, 64)
+//   2. finalize: one small kernel -> mean/rstd (N, C)
+//   3. apply: grid-stride elementwise normalize / backward-apply
+// All passes are coalesced in the channels-last layout.
 //
 //   fwd: y = (x - mu_{n,c}) * rstd_{n,c},  rstd = 1/sqrt(var + eps)
 //   bwd: dx = rstd * (dy - mean_p(dy) - (x-mu)*rstd^2 * mean_p(dy*(x-mu)))
-//
-// One workgroup per (n, 64-channel block); two passes over P per kernel
-// (reduce, then apply) — 3 HBM passes total, the memory-bound optimum for
-// an unfused norm.
 
 #include "common.h"
 
 #define IN_THREADS 256
-#define IN_CB 64  // channels per workgroup (= lanes per wave)
+#define IN_CB 64      // channels per block column
+#define IN_PCHUNK 4096  // pixels per partial-reduce block
 
-template <typename T>
-__global__ __launch_bounds__(IN_THREADS) void instnorm_cl_fwd_kernel(
-    const T* __restrict__ x, T* __restrict__ y, float* __restrict__ mean,
-    float* __restrict__ rstd, int N, int C, long P, float eps) {
-  const int n = blockIdx.x / ((C + IN_CB - 1) / IN_CB);
-  const int cb = blockIdx.x % ((C + IN_CB - 1) / IN_CB);
+// mode 0: s1 = x, s2 = x*x          (fwd statistics)
+// mode 1: s1 = dy, s2 = dy*(x-mu)   (bwd reductions; a = dy)
+template <typename T, int MODE>
+__global__ __launch_bounds__(IN_THREADS) void instnorm_partial_kernel(
+    const T* __restrict__ x, const T* __restrict__ a,
+    const float* __restrict__ mean, float* __restrict__ partials,
+    int N, int C, long P, int nchunk) {
+  const int ncb = (C + IN_CB - 1) / IN_CB;
+  int b = blockIdx.x;
+  const int chunk = b % nchunk; b /= nchunk;
+  const int cb = b % ncb; b /= ncb;
+  const int n = b;
   const int c = cb * IN_CB + (threadIdx.x & 63);
   const int wave = threadIdx.x >> 6;
   const bool cv = c < C;
 
-  const T* xb = x + (long)n * P * C;
+  const long p0 = (long)chunk * IN_PCHUNK;
+  const long p1 = min(p0 + IN_PCHUNK, P);
+  const float mu =
+      (MODE == 1 && cv) ? mean[(long)n * C + c] : 0.f;
+
   float s1 = 0.f, s2 = 0.f;
   if (cv) {
-    for (long p = wave; p < P; p += 4) {
-      const float v = (float)xb[p * C + c];
-      s1 += v;
-      s2 += v * v;
+    const T* xb = x + (long)n * P * C;
+    const T* ab = (MODE == 1) ? a + (long)n * P * C : nullptr;
+    for (long p = p0 + wave; p < p1; p += 4) {
+      if (MODE == 0) {
+        const float v = (float)xb[p * C + c];
+        s1 += v;
+        s2 += v * v;
+      } else {
+        const float g = (float)ab[p * C + c];
+        s1 += g;
+        s2 += g * ((float)xb[p * C + c] - mu);
+      }
     }
   }
-  // cross-wave reduce: 4 partials per channel
   __shared__ float red1[4][IN_CB];
   __shared__ float red2[4][IN_CB];
   red1[wave][threadIdx.x & 63] = s1;
   red2[wave][threadIdx.x & 63] = s2;
   __syncthreads();
-  __shared__ float smu[IN_CB], srs[IN_CB];
   if (threadIdx.x < IN_CB) {
-    const float t1 = red1[0][threadIdx.x] + red1[1][threadIdx.x] +
-                     red1[2][threadIdx.x] + red1[3][threadIdx.x];
-    const float t2 = red2[0][threadIdx.x] + red2[1][threadIdx.x] +
-                     red2[2][threadIdx.x] + red2[3][threadIdx.x];
-    const float mu = t1 / (float)P;
-    const float var = fmaxf(t2 / (float)P - mu * mu, 0.f);
-    const float rs = rsqrtf(var + eps);
-    smu[threadIdx.x] = mu;
-    srs[threadIdx.x] = rs;
-    const int cc = cb * IN_CB + (int)threadIdx.x;
-    if (cc < C) {
-      mean[(long)n * C + cc] = mu;
-      rstd[(long)n * C + cc] = rs;
-    }
+    const long row = (((long)n * ncb + cb) * nchunk + chunk) * 2;
+    partials[(row + 0) * IN_CB + threadIdx.x] =
+        red1[0][threadIdx.x] + red1[1][threadIdx.x] + red1[2][threadIdx.x] +
+        red1[3][threadIdx.x];
+    partials[(row + 1) * IN_CB + threadIdx.x] =
+        red2[0][threadIdx.x] + red2[1][threadIdx.x] + red2[2][threadIdx.x] +
+        red2[3][threadIdx.x];
   }
-  __syncthreads();
+}
 
-  if (cv) {
-    const float mu = smu[threadIdx.x & 63];
-    const float rs = srs[threadIdx.x & 63];
-    T* yb = y + (long)n * P * C;
-    for (long p = wave; p < P; p += 4)
-      yb[p * C + c] = (T)(((float)xb[p * C + c] - mu) * rs);
+// finalize fwd: mean/rstd from partials. grid (N*ncb), 64 threads.
+__global__ __launch_bounds__(64) void instnorm_finalize_fwd_kernel(
+    const float* __restrict__ partials, float* __restrict__ mean,
+    float* __restrict__ rstd, int N, int C, long P, int nchunk, float eps) {
+  const int ncb = (C + IN_CB - 1) / IN_CB;
+  const int cb = blockIdx.x % ncb;
+  const int n = blockIdx.x / ncb;
+  const int c = cb * IN_CB + threadIdx.x;
+  float s1 = 0.f, s2 = 0.f;
+  for (int ch = 0; ch < nchunk; ++ch) {
+    const long row = (((long)n * ncb + cb) * nchunk + ch) * 2;
+    s1 += partials[(row + 0) * IN_CB + threadIdx.x];
+    s2 += partials[(row + 1) * IN_CB + threadIdx.x];
+  }
+  if (c < C) {
+    const float mu = s1 / (float)P;
+    const float var = fmaxf(s2 / (float)P - mu * mu, 0.f);
+    mean[(long)n * C + c] = mu;
+    rstd[(long)n * C + c] = rsqrtf(var + eps);
+  }
+}
+
+// finalize bwd: means of (dy, dy*(x-mu)). grid (N*ncb), 64 threads.
+__global__ __launch_bounds__(64) void instnorm_finalize_bwd_kernel(
+    const float* __restrict__ partials, float* __restrict__ gmean,
+    float* __restrict__ gxmean, int N, int C, long P, int nchunk) {
+  const int ncb = (C + IN_CB - 1) / IN_CB;
+  const int cb = blockIdx.x % ncb;
+  const int n = blockIdx.x / ncb;
+  const int c = cb * IN_CB + threadIdx.x;
+  float s1 = 0.f, s2 = 0.f;
+  for (int ch = 0; ch < nchunk; ++ch) {
+    const long row = (((long)n * ncb + cb) * nchunk + ch) * 2;
+    s1 += partials[(row + 0) * IN_CB + threadIdx.x];
+    s2 += partials[(row + 1) * IN_CB + threadIdx.x];
+  }
+  if (c < C) {
+    gmean[(long)n * C + c] = s1 / (float)P;
+    gxmean[(long)n * C + c] = s2 / (float)P;
   }
 }
 
 template <typename T>
-__global__ __launch_bounds__(IN_THREADS) void instnorm_cl_bwd_kernel(
-    const T* __restrict__ x, const T* __restrict__ dy,
+__global__ __launch_bounds__(IN_THREADS) void instnorm_apply_fwd_kernel(
+    const T* __restrict__ x, T* __restrict__ y,
     const float* __restrict__ mean, const float* __restrict__ rstd,
-    T* __restrict__ dx, int N, int C, long P) {
-  const int n = blockIdx.x / ((C + IN_CB - 1) / IN_CB);
-  const int cb = blockIdx.x % ((C + IN_CB - 1) / IN_CB);
-  const int c = cb * IN_CB + (threadIdx.x & 63);
-  const int wave = threadIdx.x >> 6;
-  const bool cv = c < C;
-
-  const float mu = cv ? mean[(long)n * C + c] : 0.f;
-  const float rs = cv ? rstd[(long)n * C + c] : 0.f;
-
-  const T* xb = x + (long)n * P * C;
-  const T* gb = dy + (long)n * P * C;
-  float s1 = 0.f, s2 = 0.f;
-  if (cv) {
-    for (long p = wave; p < P; p += 4) {
-      const float g = (float)gb[p * C + c];
-      const float xc = (float)xb[p * C + c] - mu;
-      s1 += g;
-      s2 += g * xc;
-    }
+    long total, int C, long P) {
+  for (long idx = (long)blockIdx.x * IN_THREADS + threadIdx.x; idx < total;
+       idx += (long)gridDim.x * IN_THREADS) {
+    const long n = idx / (P * C);
+    const int c = (int)(idx % C);
+    const long s = n * C + c;
+    y[idx] = (T)(((float)x[idx] - mean[s]) * rstd[s]);
   }
-  __shared__ float red1[4][IN_CB];
-  __shared__ float red2[4][IN_CB];
-  red1[wave][threadIdx.x & 63] = s1;
-  red2[wave][threadIdx.x & 63] = s2;
-  __syncthreads();
-  __shared__ float sm1[IN_CB], sm2[IN_CB];
-  if (threadIdx.x < IN_CB) {
-    sm1[threadIdx.x] = (red1[0][threadIdx.x] + red1[1][threadIdx.x] +
-                        red1[2][threadIdx.x] + red1[3][threadIdx.x]) /
-                       (float)P;
-    sm2[threadIdx.x] = (red2[0][threadIdx.x] + red2[1][threadIdx.x] +
-                        red2[2][threadIdx.x] + red2[3][threadIdx.x]) /
-                       (float)P;
-  }
-  __syncthreads();
+}
 
-  if (cv) {
-    const float gmean = sm1[threadIdx.x & 63];
-    const float gxmean = sm2[threadIdx.x & 63];
-    T* db = dx + (long)n * P * C;
-    for (long p = wave; p < P; p += 4) {
-      const float g = (float)gb[p * C + c];
-      const float xc = (float)xb[p * C + c] - mu;
-      db[p * C + c] = (T)(rs * (g - gmean - xc * rs * rs * gxmean));
-    }
+template <typename T>
+__global__ __launch_bounds__(IN_THREADS) void instnorm_apply_bwd_kernel(
+    const T* __restrict__ x, const T* __restrict__ dy, T* __restrict__ dx,
+    const float* __restrict__ mean, const float* __restrict__ rstd,
+    const float* __restrict__ gmean, const float* __restrict__ gxmean,
+    long total, int C, long P) {
+  for (long idx = (long)blockIdx.x * IN_THREADS + threadIdx.x; idx < total;
+       idx += (long)gridDim.x * IN_THREADS) {
+    const long n = idx / (P * C);
+    const int c = (int)(idx % C);
+    const long s = n * C + c;
+    const float rs = rstd[s];
+    const float xc = (float)x[idx] - mean[s];
+    dx[idx] = (T)(rs * ((float)dy[idx] - gmean[s] - xc * rs * rs * gxmean[s]));
   }
+}
+
+static inline int in_nchunk(long P) { return (int)((P + IN_PCHUNK - 1) / IN_PCHUNK); }
+
+int flowhip_instnorm_partial_rows(int N, int C, long P) {
+  return N * ((C + IN_CB - 1) / IN_CB) * in_nchunk(P) * 2;
+}
+
+template <typename T>
+static void instnorm_fwd_t(const T* x, T* y, float* mean, float* rstd,
+                           float* partials, int N, int C, long P, float eps,
+                           hipStream_t stream) {
+  const int ncb = (C + IN_CB - 1) / IN_CB;
+  const int nchunk = in_nchunk(P);
+  hipLaunchKernelGGL((instnorm_partial_kernel<T, 0>),
+                     dim3(N * ncb * nchunk), dim3(IN_THREADS), 0, stream, x,
+                     (const T*)nullptr, (const float*)nullptr, partials, N, C,
+                     P, nchunk);
+  hipLaunchKernelGGL(instnorm_finalize_fwd_kernel, dim3(N * ncb), dim3(64), 0,
+                     stream, partials, mean, rstd, N, C, P, nchunk, eps);
+  const long total = (long)N * P * C;
+  long blocks = (total + IN_THREADS - 1) / IN_THREADS;
+  if (blocks > 16384) blocks = 16384;
+  hipLaunchKernelGGL((instnorm_apply_fwd_kernel<T>), dim3((int)blocks),
+                     dim3(IN_THREADS), 0, stream, x, y, mean, rstd, total, C,
+                     P);
+}
+
+template <typename T>
+static void instnorm_bwd_t(const T* x, const T* dy, const float* mean,
+                           const float* rstd, float* gmean, float* gxmean,
+                           float* partials, T* dx, int N, int C, long P,
+                           hipStream_t stream) {
+  const int ncb = (C + IN_CB - 1) / IN_CB;
+  const int nchunk = in_nchunk(P);
+  hipLaunchKernelGGL((instnorm_partial_kernel<T, 1>),
+                     dim3(N * ncb * nchunk), dim3(IN_THREADS), 0, stream, x,
+                     dy, mean, partials, N, C, P, nchunk);
+  hipLaunchKernelGGL(instnorm_finalize_bwd_kernel, dim3(N * ncb), dim3(64), 0,
+                     stream, partials, gmean, gxmean, N, C, P, nchunk);
+  const long total = (long)N * P * C;
+  long blocks = (total + IN_THREADS - 1) / IN_THREADS;
+  if (blocks > 16384) blocks = 16384;
+  hipLaunchKernelGGL((instnorm_apply_bwd_kernel<T>), dim3((int)blocks),
+                     dim3(IN_THREADS), 0, stream, x, dy, dx, mean, rstd,
+                     gmean, gxmean, total, C, P);
 }
 
 void flowhip_instnorm_cl_fwd_launch(const void* x, void* y, float* mean,
-                                    float* rstd, int N, int C, long P,
-                                    float eps, int is_bf16,
+                                    float* rstd, float* partials, int N,
+                                    int C, long P, float eps, int is_bf16,
                                     hipStream_t stream) {
-  const int grid = N * ((C + IN_CB - 1) / IN_CB);
   if (is_bf16)
-    hipLaunchKernelGGL((instnorm_cl_fwd_kernel<__hip_bfloat16>), dim3(grid),
-                       dim3(IN_THREADS), 0, stream,
-                       (const __hip_bfloat16*)x, (__hip_bfloat16*)y, mean,
-                       rstd, N, C, P, eps);
+    instnorm_fwd_t((const __hip_bfloat16*)x, (__hip_bfloat16*)y, mean, rstd,
+                   partials, N, C, P, eps, stream);
   else
-    hipLaunchKernelGGL((instnorm_cl_fwd_kernel<float>), dim3(grid),
-                       dim3(IN_THREADS), 0, stream, (const float*)x,
-                       (float*)y, mean, rstd, N, C, P, eps);
+    instnorm_fwd_t((const float*)x, (float*)y, mean, rstd, partials, N, C, P,
+                   eps, stream);
 }
 
 void flowhip_instnorm_cl_bwd_launch(const void* x, const void* dy,
                                     const float* mean, const float* rstd,
-                                    void* dx, int N, int C, long P,
-                                    int is_bf16, hipStream_t stream) {
-  const int grid = N * ((C + IN_CB - 1) / IN_CB);
+                                    float* gmean, float* gxmean,
+                                    float* partials, void* dx, int N, int C,
+                                    long P, int is_bf16, hipStream_t stream) {
   if (is_bf16)
-    hipLaunchKernelGGL((instnorm_cl_bwd_kernel<__hip_bfloat16>), dim3(grid),
-                       dim3(IN_THREADS), 0, stream,
-                       (const __hip_bfloat16*)x, (const __hip_bfloat16*)dy,
-                       mean, rstd, (__hip_bfloat16*)dx, N, C, P);
+    instnorm_bwd_t((const __hip_bfloat16*)x, (const __hip_bfloat16*)dy, mean,
+                   rstd, gmean, gxmean, partials, (__hip_bfloat16*)dx, N, C,
+                   P, stream);
   else
-    hipLaunchKernelGGL((instnorm_cl_bwd_kernel<float>), dim3(grid),
-                       dim3(IN_THREADS), 0, stream, (const float*)x,
-                       (const float*)dy, mean, rstd, (float*)dx, N, C, P);
+    instnorm_bwd_t((const float*)x, (const float*)dy, mean, rstd, gmean,
+                   gxmean, partials, (float*)dx, N, C, P, stream);
 }

@@ -20,6 +20,13 @@ import os
 
 import numpy as np
 import torch
+
+import os as _os
+
+if torch.cuda.is_available() and _os.environ.get("FLOWHIP_MIOPEN_FIND", "1") != "0":
+    # MIOpen find mode: benchmark=True lets MIOpen search its perf-db for
+    # the fastest conv solver per (static) shape during warmup
+    torch.backends.cudnn.benchmark = True
 import torch.optim as optim
 
 from .. import ops
