@@ -56,6 +56,18 @@ bool flowhip_nconv_wrw_tiled_launch(const float* dnomin, const float* ddenom,
                                     int Ci, int Co, int H, int W, int K,
                                     hipStream_t stream);
 int flowhip_instnorm_partial_rows(int N, int C, long P);
+void flowhip_conv_gemm_fwd_launch(const void* x, const void* wpk,
+                                  const float* bias, void* out,
+                                  const void* zpage, long Mtot, int HH,
+                                  int WW, int ld_x, int Cin, int Cout,
+                                  int cpad, int KH, int KW, int padH,
+                                  int padW, int act, hipStream_t stream);
+void flowhip_conv_gemm_wrw_launch(const void* dy, const void* x,
+                                  float* partials, float* dw,
+                                  const void* zpage, long Mtot, int HH,
+                                  int WW, int ld_x, int Cin, int Cout,
+                                  int cpad, int KH, int KW, int padH,
+                                  int padW, int nchunk, hipStream_t stream);
 void flowhip_instnorm_cl_fwd_launch(const void* x, void* y, float* mean,
                                     float* rstd, float* partials, int N,
                                     int C, long P, float eps, int is_bf16,
@@ -558,6 +570,83 @@ torch::Tensor instnorm_cl_bwd(torch::Tensor x, torch::Tensor dy,
   return dx;
 }
 
+static const void* cg_zero_page() {
+  static void* p = nullptr;
+  if (p == nullptr) {
+    hipMalloc(&p, 256);
+    hipMemset(p, 0, 256);
+  }
+  return p;
+}
+
+// x: (N, C, H, W) with channels-last-like strides (stride(1)==1); a
+// channel-narrowed view is allowed (ld_x = stride at dim 3 >= C).
+static void cg_check_x(const torch::Tensor& x, int& ld) {
+  TORCH_CHECK(x.is_cuda() && x.dim() == 4 &&
+              x.scalar_type() == torch::kBFloat16);
+  TORCH_CHECK(x.stride(1) == 1, "conv_gemm: channels-last layout required");
+  ld = (int)x.stride(3);
+  TORCH_CHECK(ld >= x.size(1) && x.stride(2) == (long)ld * x.size(3) &&
+              x.stride(0) == (long)ld * x.size(3) * x.size(2));
+}
+
+torch::Tensor conv_gemm_fwd(torch::Tensor x, torch::Tensor wpk,
+                            c10::optional<torch::Tensor> bias, int64_t Cout,
+                            int64_t KH, int64_t KW, int64_t act) {
+  int ld_x;
+  cg_check_x(x, ld_x);
+  TORCH_CHECK(wpk.is_cuda() && wpk.is_contiguous() &&
+              wpk.scalar_type() == torch::kBFloat16 && wpk.dim() == 3);
+  const int N = x.size(0), Cin = x.size(1), H = x.size(2), W = x.size(3);
+  const int cpad = wpk.size(2);
+  TORCH_CHECK(wpk.size(0) == KH * KW && wpk.size(1) == Cout);
+  const long Mtot = (long)N * H * W;
+  auto out = torch::empty({(long)N, Cout, (long)H, (long)W},
+                          x.options(), torch::MemoryFormat::ChannelsLast);
+  const float* bptr = nullptr;
+  if (bias.has_value()) {
+    TORCH_CHECK(bias->is_contiguous() &&
+                bias->scalar_type() == torch::kFloat32 &&
+                bias->numel() == Cout);
+    bptr = bias->data_ptr<float>();
+  }
+  const c10::cuda::CUDAGuard guard(x.device());
+  hipStream_t stream = at::cuda::getCurrentCUDAStream().stream();
+  flowhip_conv_gemm_fwd_launch(x.data_ptr(), wpk.data_ptr(), bptr,
+                               out.data_ptr(), cg_zero_page(), Mtot, H, W,
+                               ld_x, Cin, (int)Cout, cpad, (int)KH, (int)KW,
+                               (int)KH / 2, (int)KW / 2, (int)act, stream);
+  return out;
+}
+
+torch::Tensor conv_gemm_wrw(torch::Tensor dy, torch::Tensor x, int64_t KH,
+                            int64_t KW) {
+  int ld_x;
+  cg_check_x(x, ld_x);
+  TORCH_CHECK(dy.is_cuda() && dy.scalar_type() == torch::kBFloat16 &&
+              dy.stride(1) == 1 && dy.stride(3) == dy.size(1),
+              "conv_gemm_wrw: dy must be channels-last contiguous");
+  const int N = x.size(0), Cin = x.size(1), H = x.size(2), W = x.size(3);
+  const int Cout = dy.size(1);
+  const long Mtot = (long)N * H * W;
+  const int cpad = (int)((Cin + 63) / 64) * 64;
+  const int tiles_o = (Cout + 63) / 64;
+  const int nchunk = 8;
+  auto partials = torch::empty(
+      {(long)nchunk * KH * KW * tiles_o * 64 * cpad},
+      x.options().dtype(torch::kFloat32));
+  auto dw = torch::empty({(long)Cout, (long)Cin, KH, KW},
+                         x.options().dtype(torch::kFloat32));
+  const c10::cuda::CUDAGuard guard(x.device());
+  hipStream_t stream = at::cuda::getCurrentCUDAStream().stream();
+  flowhip_conv_gemm_wrw_launch(dy.data_ptr(), x.data_ptr(),
+                               partials.data_ptr<float>(),
+                               dw.data_ptr<float>(), cg_zero_page(), Mtot, H,
+                               W, ld_x, Cin, Cout, cpad, (int)KH, (int)KW,
+                               (int)KH / 2, (int)KW / 2, nchunk, stream);
+  return dw;
+}
+
 }  // namespace gg
 
 std::vector<torch::Tensor> gru_gate1_fwd(torch::Tensor zr, torch::Tensor h) {
@@ -708,6 +797,83 @@ torch::Tensor instnorm_cl_bwd(torch::Tensor x, torch::Tensor dy,
   return dx;
 }
 
+static const void* cg_zero_page() {
+  static void* p = nullptr;
+  if (p == nullptr) {
+    hipMalloc(&p, 256);
+    hipMemset(p, 0, 256);
+  }
+  return p;
+}
+
+// x: (N, C, H, W) with channels-last-like strides (stride(1)==1); a
+// channel-narrowed view is allowed (ld_x = stride at dim 3 >= C).
+static void cg_check_x(const torch::Tensor& x, int& ld) {
+  TORCH_CHECK(x.is_cuda() && x.dim() == 4 &&
+              x.scalar_type() == torch::kBFloat16);
+  TORCH_CHECK(x.stride(1) == 1, "conv_gemm: channels-last layout required");
+  ld = (int)x.stride(3);
+  TORCH_CHECK(ld >= x.size(1) && x.stride(2) == (long)ld * x.size(3) &&
+              x.stride(0) == (long)ld * x.size(3) * x.size(2));
+}
+
+torch::Tensor conv_gemm_fwd(torch::Tensor x, torch::Tensor wpk,
+                            c10::optional<torch::Tensor> bias, int64_t Cout,
+                            int64_t KH, int64_t KW, int64_t act) {
+  int ld_x;
+  cg_check_x(x, ld_x);
+  TORCH_CHECK(wpk.is_cuda() && wpk.is_contiguous() &&
+              wpk.scalar_type() == torch::kBFloat16 && wpk.dim() == 3);
+  const int N = x.size(0), Cin = x.size(1), H = x.size(2), W = x.size(3);
+  const int cpad = wpk.size(2);
+  TORCH_CHECK(wpk.size(0) == KH * KW && wpk.size(1) == Cout);
+  const long Mtot = (long)N * H * W;
+  auto out = torch::empty({(long)N, Cout, (long)H, (long)W},
+                          x.options(), torch::MemoryFormat::ChannelsLast);
+  const float* bptr = nullptr;
+  if (bias.has_value()) {
+    TORCH_CHECK(bias->is_contiguous() &&
+                bias->scalar_type() == torch::kFloat32 &&
+                bias->numel() == Cout);
+    bptr = bias->data_ptr<float>();
+  }
+  const c10::cuda::CUDAGuard guard(x.device());
+  hipStream_t stream = at::cuda::getCurrentCUDAStream().stream();
+  flowhip_conv_gemm_fwd_launch(x.data_ptr(), wpk.data_ptr(), bptr,
+                               out.data_ptr(), cg_zero_page(), Mtot, H, W,
+                               ld_x, Cin, (int)Cout, cpad, (int)KH, (int)KW,
+                               (int)KH / 2, (int)KW / 2, (int)act, stream);
+  return out;
+}
+
+torch::Tensor conv_gemm_wrw(torch::Tensor dy, torch::Tensor x, int64_t KH,
+                            int64_t KW) {
+  int ld_x;
+  cg_check_x(x, ld_x);
+  TORCH_CHECK(dy.is_cuda() && dy.scalar_type() == torch::kBFloat16 &&
+              dy.stride(1) == 1 && dy.stride(3) == dy.size(1),
+              "conv_gemm_wrw: dy must be channels-last contiguous");
+  const int N = x.size(0), Cin = x.size(1), H = x.size(2), W = x.size(3);
+  const int Cout = dy.size(1);
+  const long Mtot = (long)N * H * W;
+  const int cpad = (int)((Cin + 63) / 64) * 64;
+  const int tiles_o = (Cout + 63) / 64;
+  const int nchunk = 8;
+  auto partials = torch::empty(
+      {(long)nchunk * KH * KW * tiles_o * 64 * cpad},
+      x.options().dtype(torch::kFloat32));
+  auto dw = torch::empty({(long)Cout, (long)Cin, KH, KW},
+                         x.options().dtype(torch::kFloat32));
+  const c10::cuda::CUDAGuard guard(x.device());
+  hipStream_t stream = at::cuda::getCurrentCUDAStream().stream();
+  flowhip_conv_gemm_wrw_launch(dy.data_ptr(), x.data_ptr(),
+                               partials.data_ptr<float>(),
+                               dw.data_ptr<float>(), cg_zero_page(), Mtot, H,
+                               W, ld_x, Cin, Cout, cpad, (int)KH, (int)KW,
+                               (int)KH / 2, (int)KW / 2, nchunk, stream);
+  return dw;
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -726,6 +892,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("convex_up_bwd", &convex_up_bwd, "backward of convex_up_fwd");
   m.def("nconv_fwd", &nconv_fwd,
         "fused normalized convolution forward (out, cout)");
+  m.def("conv_gemm_fwd", &conv_gemm_fwd,
+        "implicit-GEMM NHWC bf16 conv forward (also bwd-data with flipped "
+        "packed weights)");
+  m.def("conv_gemm_wrw", &conv_gemm_wrw,
+        "implicit-GEMM conv weight gradient (split-M + reduce)");
   m.def("instnorm_cl_fwd", &instnorm_cl_fwd,
         "channels-last InstanceNorm2d forward (y, mean, rstd)");
   m.def("instnorm_cl_bwd", &instnorm_cl_bwd,

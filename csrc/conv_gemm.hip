@@ -1,0 +1,453 @@
+// Implicit-GEMM NHWC bf16 convolution family (kernel #5 of SURVEY.md §2.2)
+// for the iterative update block (motion encoder, SepConvGRU, flow head —
+// reference update.py:6-146). Stride 1, dilation 1, "same" padding, odd or
+// 1xK/Kx1 kernels, Cin % 8 == 0 (callers pad the one 324-channel case).
+//
+// Why not MIOpen here: at the RAFT working set (M = B*H/8*W/8 = 21504 at
+// batch 3) MIOpen's 128x128/256x128 igemm tiles launch ~170-340
+// workgroups on a 256-CU chip — underfilled — and every call drags
+// SubTensorOp cast/Set kernels with it (profiles/). This kernel uses a
+// 64x64 tile (>=1300 workgroups at Cout 256), stages the im2col gather
+// through LDS with 16-byte global_load_lds (lane-linear image, XOR bank
+// swizzle on the SOURCE address — guide §5 rule 21), double-buffered, and
+// fuses bias + ReLU into the epilogue.
+//
+//   fwd:       out[m, o] = act( sum_{ky,kx,c} x[m+off, c] * w[o, ky,kx, c] + b[o] )
+//   bwd-data:  SAME kernel with w' = flip(w).T packed as [kyx][ci][co]
+//              (stride-1 same-pad transposed conv == conv with flipped w)
+//   wrw:       dW[o, kyx, c] = sum_m dy[m, o] * x[m+off, c]
+//              (split-M partials + small reduce; MFMA over the m axis with
+//              LDS-transposed fragment reads)
+//
+// Weight packing (host side, cached per weight version):
+//   wpk[kyx][o][c_pad]  (c_pad = roundup(Cin, 64), zero-filled)
+// Out-of-bounds taps and the c >= Cin tail read a 16-B zero page through
+// the per-lane glds source address.
+
+#include "common.h"
+
+#define CG_BM 64
+#define CG_BN 64
+#define CG_BK 64
+#define CG_THREADS 256
+
+#define CG_SLOTS 8                       // 16-B pieces per row of a 64-k tile
+#define CG_PIECES (CG_BM * CG_SLOTS)     // 512 per tile
+#define CG_PPT (CG_PIECES / CG_THREADS)  // 2 per thread
+
+__device__ __forceinline__ unsigned cg_swz(unsigned row, unsigned slot) {
+  return slot ^ ((row >> 1) & 7u);
+}
+
+// Stage the A (im2col) tile: rows = 64 consecutive output pixels, k = one
+// 64-channel slab of one (ky,kx) section. OOB rows read the zero page.
+__device__ __forceinline__ void cg_stage_a(
+    const __bf16* __restrict__ x, const __bf16* __restrict__ zpage,
+    char* lds_buf, int wave, int lane, long m0, long Mtot, int HH, int WW,
+    int ld_x, int Cin, int dy, int dx, int c0) {
+#pragma unroll
+  for (int j = 0; j < CG_PPT; ++j) {
+    const int piece0 = wave * 64 + CG_THREADS * j;
+    const int piece = piece0 + lane;
+    const int row = piece >> 3;          // 0..63
+    const int slot = piece & 7;
+    const int sslot = cg_swz(row, slot);
+    long m = m0 + row;
+    if (m >= Mtot) m = Mtot - 1;
+    const int xx = (int)(m % WW);
+    const int yy = (int)((m / WW) % HH);
+    const long n = m / ((long)WW * HH);
+    const int sy = yy + dy, sx = xx + dx;
+    const int c = c0 + sslot * 8;
+    const __bf16* src;
+    if (sy >= 0 && sy < HH && sx >= 0 && sx < WW && c < Cin)
+      src = x + (((long)n * HH + sy) * WW + sx) * ld_x + c;
+    else
+      src = zpage;
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) void*)src,
+        (__attribute__((address_space(3))) void*)(lds_buf + piece0 * 16),
+        16, 0, 0);
+  }
+}
+
+// Stage the B (weight) tile: rows = 64 output channels (clamped), k = the
+// same 64-channel slab. wsec = wpk section base [kyx][.][.].
+__device__ __forceinline__ void cg_stage_b(const __bf16* __restrict__ wsec,
+                                           char* lds_buf, int wave, int lane,
+                                           int n0, int Cout, int cpad,
+                                           int c0) {
+#pragma unroll
+  for (int j = 0; j < CG_PPT; ++j) {
+    const int piece0 = wave * 64 + CG_THREADS * j;
+    const int piece = piece0 + lane;
+    const int row = piece >> 3;
+    const int slot = piece & 7;
+    const int sslot = cg_swz(row, slot);
+    int o = n0 + row;
+    if (o >= Cout) o = Cout - 1;
+    const __bf16* src = wsec + (long)o * cpad + c0 + sslot * 8;
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) void*)src,
+        (__attribute__((address_space(3))) void*)(lds_buf + piece0 * 16),
+        16, 0, 0);
+  }
+}
+
+// ACT: 0 = none, 1 = ReLU
+template <int ACT>
+__global__ __launch_bounds__(CG_THREADS, 2) void conv_gemm_fwd_kernel(
+    const __bf16* __restrict__ x,     // (Mtot, ld_x) NHWC rows
+    const __bf16* __restrict__ wpk,   // (KYX, Cout, cpad)
+    const float* __restrict__ bias,   // (Cout) or nullptr
+    __bf16* __restrict__ out,         // (Mtot, Cout)
+    const __bf16* __restrict__ zpage,
+    long Mtot, int HH, int WW, int ld_x, int Cin, int Cout, int cpad,
+    int KH, int KW, int padH, int padW, int tiles_m) {
+  __shared__ __attribute__((aligned(16))) char lds[2 * 2 * CG_BM * CG_BK * 2];
+
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6;
+  const int lane = tid & 63;
+
+  const int tm = blockIdx.x % tiles_m;
+  const int tn = blockIdx.x / tiles_m;
+  const long m0 = (long)tm * CG_BM;
+  const int n0 = tn * CG_BN;
+
+  const int wr = (wave >> 1) * 32;  // M offset of this wave's quadrant
+  const int wc = (wave & 1) * 32;   // N offset
+
+  f32x4 acc[2][2];
+#pragma unroll
+  for (int i = 0; i < 2; ++i)
+#pragma unroll
+    for (int j = 0; j < 2; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  const int cslabs = cpad / CG_BK;
+  const int nsub = KH * KW * cslabs;
+  const unsigned TS = CG_BM * CG_BK * 2;  // 8 KB
+
+  // subtile s -> (kyx = s / cslabs, c0 = (s % cslabs) * 64)
+  {
+    const int kyx = 0, cs = 0;
+    const int dy = -padH, dx = -padW;
+    cg_stage_a(x, zpage, lds, wave, lane, m0, Mtot, HH, WW, ld_x, Cin, dy, dx,
+               cs * CG_BK);
+    cg_stage_b(wpk, lds + TS, wave, lane, n0, Cout, cpad, cs * CG_BK);
+    (void)kyx;
+  }
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __syncthreads();
+
+  int cur = 0;
+  for (int s = 0; s < nsub; ++s) {
+    if (s + 1 < nsub) {
+      const int kyx = (s + 1) / cslabs;
+      const int cs = (s + 1) - kyx * cslabs;
+      const int ky = kyx / KW, kx = kyx - ky * KW;
+      cg_stage_a(x, zpage, lds + (cur ^ 1) * 2 * TS, wave, lane, m0, Mtot, HH,
+                 WW, ld_x, Cin, ky - padH, kx - padW, cs * CG_BK);
+      cg_stage_b(wpk + (long)kyx * Cout * cpad, lds + (cur ^ 1) * 2 * TS + TS,
+                 wave, lane, n0, Cout, cpad, cs * CG_BK);
+    }
+
+    const char* abuf = lds + cur * 2 * TS;
+    const char* bbuf = abuf + TS;
+    const int frow = lane & 15;
+    const int fk = lane >> 4;
+
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+      bf16x8 af[2], bf[2];
+#pragma unroll
+      for (int i = 0; i < 2; ++i) {
+        const unsigned row = wr + i * 16 + frow;
+        const unsigned slot = kk * 4 + fk;
+        af[i] = *(const bf16x8*)(abuf + row * (CG_BK * 2) +
+                                 cg_swz(row, slot) * 16);
+      }
+#pragma unroll
+      for (int j = 0; j < 2; ++j) {
+        const unsigned row = wc + j * 16 + frow;
+        const unsigned slot = kk * 4 + fk;
+        bf[j] = *(const bf16x8*)(bbuf + row * (CG_BK * 2) +
+                                 cg_swz(row, slot) * 16);
+      }
+#pragma unroll
+      for (int i = 0; i < 2; ++i)
+#pragma unroll
+        for (int j = 0; j < 2; ++j)
+          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af[i], bf[j],
+                                                              acc[i][j], 0,
+                                                              0, 0);
+    }
+
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __syncthreads();
+    cur ^= 1;
+  }
+
+  // epilogue: bias + activation, bf16 store (m, Cout), edge-masked
+  const int fcol = lane & 15;
+  const int frow0 = (lane >> 4) * 4;
+#pragma unroll
+  for (int i = 0; i < 2; ++i) {
+#pragma unroll
+    for (int j = 0; j < 2; ++j) {
+      const int gco = n0 + wc + j * 16 + fcol;
+      if (gco >= Cout) continue;
+      const float b = bias ? bias[gco] : 0.f;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const long gm = m0 + wr + i * 16 + frow0 + r;
+        if (gm >= Mtot) continue;
+        float v = acc[i][j][r] + b;
+        if (ACT == 1) v = fmaxf(v, 0.f);
+        out[gm * Cout + gco] = (__bf16)v;
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Weight gradient: dWp[o, kyx, c] partial over an M chunk.
+// grid: (tiles_o * tiles_c, KYX, CHUNKS). LDS tiles stay m-major (glds
+// lane-linear); fragments are read transposed element-wise (u16 loads) —
+// correctness-first; the m-contraction MFMA dominates.
+// partials layout: (CHUNKS, KYX, tiles_o*64, cpad) fp32.
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(CG_THREADS, 2) void conv_gemm_wrw_kernel(
+    const __bf16* __restrict__ dy,  // (Mtot, Cout)
+    const __bf16* __restrict__ x,   // (Mtot, ld_x)
+    float* __restrict__ partials,
+    const __bf16* __restrict__ zpage,
+    long Mtot, int HH, int WW, int ld_x, int Cin, int Cout, int cpad,
+    int KH, int KW, int padH, int padW, int tiles_o, int tiles_c,
+    int nchunk) {
+  __shared__ __attribute__((aligned(16))) char lds[2 * 2 * CG_BM * CG_BK * 2];
+
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6;
+  const int lane = tid & 63;
+
+  const int to = blockIdx.x % tiles_o;
+  const int tc = blockIdx.x / tiles_o;
+  const int kyx = blockIdx.y;
+  const int chunk = blockIdx.z;
+  const int ky = kyx / KW, kx = kyx - ky * KW;
+  const int dyo = ky - padH, dxo = kx - padW;
+
+  const int o0 = to * 64;
+  const int c0 = tc * 64;
+
+  const long mtiles = (Mtot + CG_BM - 1) / CG_BM;
+  const long t0 = (mtiles * chunk) / nchunk;
+  const long t1 = (mtiles * (chunk + 1)) / nchunk;
+
+  const int wr = (wave >> 1) * 32;  // o offset
+  const int wc = (wave & 1) * 32;   // c offset
+
+  f32x4 acc[2][2];
+#pragma unroll
+  for (int i = 0; i < 2; ++i)
+#pragma unroll
+    for (int j = 0; j < 2; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  const unsigned TS = CG_BM * CG_BK * 2;
+  auto stage_dy = [&](long mt, char* buf) {
+#pragma unroll
+    for (int j = 0; j < CG_PPT; ++j) {
+      const int piece0 = wave * 64 + CG_THREADS * j;
+      const int piece = piece0 + lane;
+      const int row = piece >> 3;
+      const int slot = piece & 7;
+      const int sslot = cg_swz(row, slot);
+      long m = mt * CG_BM + row;
+      const __bf16* src;
+      const int o = o0 + sslot * 8;
+      if (m < Mtot && o < Cout)
+        src = dy + m * Cout + o;
+      else
+        src = zpage;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)src,
+          (__attribute__((address_space(3))) void*)(buf + piece0 * 16),
+          16, 0, 0);
+    }
+  };
+  auto stage_x = [&](long mt, char* buf) {
+#pragma unroll
+    for (int j = 0; j < CG_PPT; ++j) {
+      const int piece0 = wave * 64 + CG_THREADS * j;
+      const int piece = piece0 + lane;
+      const int row = piece >> 3;
+      const int slot = piece & 7;
+      const int sslot = cg_swz(row, slot);
+      const long m = mt * CG_BM + row;
+      const __bf16* src = zpage;
+      if (m < Mtot) {
+        const int xx = (int)(m % WW);
+        const int yy = (int)((m / WW) % HH);
+        const long n = m / ((long)WW * HH);
+        const int sy = yy + dyo, sx = xx + dxo;
+        const int c = c0 + sslot * 8;
+        if (sy >= 0 && sy < HH && sx >= 0 && sx < WW && c < Cin)
+          src = x + (((long)n * HH + sy) * WW + sx) * ld_x + c;
+      }
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)src,
+          (__attribute__((address_space(3))) void*)(buf + piece0 * 16),
+          16, 0, 0);
+    }
+  };
+
+  if (t0 < t1) {
+    stage_dy(t0, lds);
+    stage_x(t0, lds + TS);
+  }
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __syncthreads();
+
+  int cur = 0;
+  for (long t = t0; t < t1; ++t) {
+    if (t + 1 < t1) {
+      stage_dy(t + 1, lds + (cur ^ 1) * 2 * TS);
+      stage_x(t + 1, lds + (cur ^ 1) * 2 * TS + TS);
+    }
+
+    // transposed fragment reads from the m-major images:
+    //   A-frag (o rows, k=m): a[i] lane reads 8 m-consecutive u16 at
+    //   column (wr + i*16 + frow); image element (m, col) at m*128B + col*2
+    const char* dbuf = lds + cur * 2 * TS;
+    const char* xbuf = dbuf + TS;
+    const int frow = lane & 15;
+    const int fk = lane >> 4;  // which 8-m chunk
+
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+      bf16x8 af[2], bf[2];
+      const int mbase = kk * 32 + fk * 8;
+#pragma unroll
+      for (int i = 0; i < 2; ++i) {
+        const int col = wr + i * 16 + frow;  // o column
+        bf16x8 v;
+#pragma unroll
+        for (int e = 0; e < 8; ++e) {
+          const int m = mbase + e;
+          const int slot = col >> 3;
+          v[e] = *(const __bf16*)(dbuf + m * (CG_BK * 2) +
+                                  cg_swz(m, slot) * 16 + (col & 7) * 2);
+        }
+        af[i] = v;
+      }
+#pragma unroll
+      for (int j = 0; j < 2; ++j) {
+        const int col = wc + j * 16 + frow;  // c column
+        bf16x8 v;
+#pragma unroll
+        for (int e = 0; e < 8; ++e) {
+          const int m = mbase + e;
+          const int slot = col >> 3;
+          v[e] = *(const __bf16*)(xbuf + m * (CG_BK * 2) +
+                                  cg_swz(m, slot) * 16 + (col & 7) * 2);
+        }
+        bf[j] = v;
+      }
+#pragma unroll
+      for (int i = 0; i < 2; ++i)
+#pragma unroll
+        for (int j = 0; j < 2; ++j)
+          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af[i], bf[j],
+                                                              acc[i][j], 0,
+                                                              0, 0);
+    }
+
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __syncthreads();
+    cur ^= 1;
+  }
+
+  // store partial tile: partials[chunk][kyx][o0+...][c0+...]
+  const int KYX = KH * KW;
+  const int orows = tiles_o * 64;
+  float* pbase = partials +
+                 (((long)chunk * KYX + kyx) * orows) * cpad;
+  const int fcol = lane & 15;
+  const int frow0 = (lane >> 4) * 4;
+#pragma unroll
+  for (int i = 0; i < 2; ++i) {
+#pragma unroll
+    for (int j = 0; j < 2; ++j) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int o = o0 + wr + i * 16 + frow0 + r;
+        const int c = c0 + wc + j * 16 + fcol;
+        pbase[(long)o * cpad + c] = acc[i][j][r];
+      }
+    }
+  }
+}
+
+// reduce partials over chunks and scatter into dW (Cout, Cin, KH, KW) fp32
+__global__ __launch_bounds__(CG_THREADS) void conv_gemm_wrw_reduce_kernel(
+    const float* __restrict__ partials, float* __restrict__ dw,
+    int nchunk, int KYX, int orows, int cpad, int Cout, int Cin) {
+  const long total = (long)Cout * Cin * KYX;
+  for (long idx = (long)blockIdx.x * CG_THREADS + threadIdx.x; idx < total;
+       idx += (long)gridDim.x * CG_THREADS) {
+    long t = idx;
+    const int kyx = (int)(t % KYX); t /= KYX;
+    const int c = (int)(t % Cin); t /= Cin;
+    const int o = (int)t;
+    float s = 0.f;
+    for (int ch = 0; ch < nchunk; ++ch)
+      s += partials[((((long)ch * KYX + kyx) * orows) + o) * cpad + c];
+    // dw layout (O, I, KH, KW): kyx = ky*KW + kx
+    dw[((long)o * Cin + c) * KYX + kyx] = s;
+  }
+}
+
+// ---------------------------------------------------------------------------
+
+void flowhip_conv_gemm_fwd_launch(const void* x, const void* wpk,
+                                  const float* bias, void* out,
+                                  const void* zpage, long Mtot, int HH,
+                                  int WW, int ld_x, int Cin, int Cout,
+                                  int cpad, int KH, int KW, int padH,
+                                  int padW, int act, hipStream_t stream) {
+  const int tiles_m = (int)((Mtot + CG_BM - 1) / CG_BM);
+  const int tiles_n = fh_cdiv(Cout, CG_BN);
+  dim3 grid(tiles_m * tiles_n), block(CG_THREADS);
+  if (act == 1)
+    hipLaunchKernelGGL((conv_gemm_fwd_kernel<1>), grid, block, 0, stream,
+                       (const __bf16*)x, (const __bf16*)wpk, bias,
+                       (__bf16*)out, (const __bf16*)zpage, Mtot, HH, WW,
+                       ld_x, Cin, Cout, cpad, KH, KW, padH, padW, tiles_m);
+  else
+    hipLaunchKernelGGL((conv_gemm_fwd_kernel<0>), grid, block, 0, stream,
+                       (const __bf16*)x, (const __bf16*)wpk, bias,
+                       (__bf16*)out, (const __bf16*)zpage, Mtot, HH, WW,
+                       ld_x, Cin, Cout, cpad, KH, KW, padH, padW, tiles_m);
+}
+
+void flowhip_conv_gemm_wrw_launch(const void* dy, const void* x,
+                                  float* partials, float* dw,
+                                  const void* zpage, long Mtot, int HH,
+                                  int WW, int ld_x, int Cin, int Cout,
+                                  int cpad, int KH, int KW, int padH,
+                                  int padW, int nchunk, hipStream_t stream) {
+  const int tiles_o = fh_cdiv(Cout, 64);
+  const int tiles_c = fh_cdiv(cpad, 64);
+  dim3 grid(tiles_o * tiles_c, KH * KW, nchunk), block(CG_THREADS);
+  hipLaunchKernelGGL(conv_gemm_wrw_kernel, grid, block, 0, stream,
+                     (const __bf16*)dy, (const __bf16*)x, partials,
+                     (const __bf16*)zpage, Mtot, HH, WW, ld_x, Cin, Cout,
+                     cpad, KH, KW, padH, padW, tiles_o, tiles_c, nchunk);
+  const long total = (long)Cout * Cin * KH * KW;
+  long rblocks = (total + CG_THREADS - 1) / CG_THREADS;
+  if (rblocks > 4096) rblocks = 4096;
+  hipLaunchKernelGGL(conv_gemm_wrw_reduce_kernel, dim3((int)rblocks), block,
+                     0, stream, partials, dw, nchunk, KH * KW, tiles_o * 64,
+                     cpad, Cout, Cin);
+}

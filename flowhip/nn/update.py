@@ -15,12 +15,29 @@ import torch
 import torch.nn as nn
 import torch.nn.functional as F
 
+from ..ops.functional_conv import fused_conv2d
+
+
+class FusedConv2d(nn.Conv2d):
+    """nn.Conv2d whose GPU bf16/NHWC stride-1 path runs the hand-written
+    implicit-GEMM MFMA kernel (csrc/conv_gemm.hip, kernel #5). Parameter
+    names/shapes are untouched — state-dict identical to nn.Conv2d."""
+
+    def __init__(self, *a, **k):
+        super().__init__(*a, **k)
+        self._cg_cache = {}
+
+    def forward(self, x):
+        return fused_conv2d(x, self.weight, self.bias, self.stride,
+                            self.padding, self.dilation, self.groups,
+                            self._cg_cache)
+
 
 class FlowHead(nn.Module):
     def __init__(self, input_dim=128, hidden_dim=256):
         super().__init__()
-        self.conv1 = nn.Conv2d(input_dim, hidden_dim, 3, padding=1)
-        self.conv2 = nn.Conv2d(hidden_dim, 2, 3, padding=1)
+        self.conv1 = FusedConv2d(input_dim, hidden_dim, 3, padding=1)
+        self.conv2 = FusedConv2d(hidden_dim, 2, 3, padding=1)
         self.relu = nn.ReLU(inplace=True)
 
     def forward(self, x):
@@ -32,14 +49,19 @@ class ConvGRU(nn.Module):
 
     def __init__(self, hidden_dim=128, input_dim=192 + 128):
         super().__init__()
-        self.convz = nn.Conv2d(hidden_dim + input_dim, hidden_dim, 3, padding=1)
-        self.convr = nn.Conv2d(hidden_dim + input_dim, hidden_dim, 3, padding=1)
-        self.convq = nn.Conv2d(hidden_dim + input_dim, hidden_dim, 3, padding=1)
+        self.convz = FusedConv2d(hidden_dim + input_dim, hidden_dim, 3, padding=1)
+        self.convr = FusedConv2d(hidden_dim + input_dim, hidden_dim, 3, padding=1)
+        self.convq = FusedConv2d(hidden_dim + input_dim, hidden_dim, 3, padding=1)
 
     def forward(self, h, x):
         hx = torch.cat([h, x], dim=1)
-        zr = F.conv2d(hx, torch.cat([self.convz.weight, self.convr.weight]),
-                      torch.cat([self.convz.bias, self.convr.bias]), padding=1)
+        if not hasattr(self, "_zr_cache"):
+            self._zr_cache = {}
+        zr = fused_conv2d(
+            hx, torch.cat([self.convz.weight, self.convr.weight]),
+            torch.cat([self.convz.bias, self.convr.bias]), 1, 1, 1, 1,
+            self._zr_cache,
+            key=(self.convz.weight._version, self.convr.weight._version))
         if h.is_cuda:
             from ..ops.functional_gru import GruGate1Fn, GruGate2Fn
             z, rh = GruGate1Fn.apply(zr, h)
@@ -55,19 +77,22 @@ class SepConvGRU(nn.Module):
 
     def __init__(self, hidden_dim=128, input_dim=192 + 128):
         super().__init__()
-        self.convz1 = nn.Conv2d(hidden_dim + input_dim, hidden_dim, (1, 5), padding=(0, 2))
-        self.convr1 = nn.Conv2d(hidden_dim + input_dim, hidden_dim, (1, 5), padding=(0, 2))
-        self.convq1 = nn.Conv2d(hidden_dim + input_dim, hidden_dim, (1, 5), padding=(0, 2))
+        self.convz1 = FusedConv2d(hidden_dim + input_dim, hidden_dim, (1, 5), padding=(0, 2))
+        self.convr1 = FusedConv2d(hidden_dim + input_dim, hidden_dim, (1, 5), padding=(0, 2))
+        self.convq1 = FusedConv2d(hidden_dim + input_dim, hidden_dim, (1, 5), padding=(0, 2))
 
-        self.convz2 = nn.Conv2d(hidden_dim + input_dim, hidden_dim, (5, 1), padding=(2, 0))
-        self.convr2 = nn.Conv2d(hidden_dim + input_dim, hidden_dim, (5, 1), padding=(2, 0))
-        self.convq2 = nn.Conv2d(hidden_dim + input_dim, hidden_dim, (5, 1), padding=(2, 0))
+        self.convz2 = FusedConv2d(hidden_dim + input_dim, hidden_dim, (5, 1), padding=(2, 0))
+        self.convr2 = FusedConv2d(hidden_dim + input_dim, hidden_dim, (5, 1), padding=(2, 0))
+        self.convq2 = FusedConv2d(hidden_dim + input_dim, hidden_dim, (5, 1), padding=(2, 0))
 
-    @staticmethod
-    def _pass(h, x, convz, convr, convq, padding):
+    def _pass(self, h, x, convz, convr, convq, padding, zr_cache):
         hx = torch.cat([h, x], dim=1)
-        zr = F.conv2d(hx, torch.cat([convz.weight, convr.weight]),
-                      torch.cat([convz.bias, convr.bias]), padding=padding)
+        zr_w = torch.cat([convz.weight, convr.weight])
+        zr_b = torch.cat([convz.bias, convr.bias])
+        # the packed-weight cache is keyed by the source params' versions
+        # (the cat tensor itself is fresh every call)
+        zr = fused_conv2d(hx, zr_w, zr_b, 1, padding, 1, 1, zr_cache,
+                          key=(convz.weight._version, convr.weight._version))
         if h.is_cuda:
             # fused gate kernels (ops/functional_gru): one kernel for
             # sigmoid/chunk/r*h, one for tanh + lerp, fused backwards
@@ -80,8 +105,12 @@ class SepConvGRU(nn.Module):
         return (1 - z) * h + z * q
 
     def forward(self, h, x):
-        h = self._pass(h, x, self.convz1, self.convr1, self.convq1, (0, 2))  # horizontal
-        h = self._pass(h, x, self.convz2, self.convr2, self.convq2, (2, 0))  # vertical
+        if not hasattr(self, "_zr1_cache"):
+            self._zr1_cache, self._zr2_cache = {}, {}
+        h = self._pass(h, x, self.convz1, self.convr1, self.convq1, (0, 2),
+                       self._zr1_cache)  # horizontal
+        h = self._pass(h, x, self.convz2, self.convr2, self.convq2, (2, 0),
+                       self._zr2_cache)  # vertical
         return h
 
 
@@ -91,10 +120,10 @@ class SmallMotionEncoder(nn.Module):
     def __init__(self, args):
         super().__init__()
         cor_planes = args.corr_levels * (2 * args.corr_radius + 1) ** 2
-        self.convc1 = nn.Conv2d(cor_planes, 96, 1, padding=0)
-        self.convf1 = nn.Conv2d(2, 64, 7, padding=3)
-        self.convf2 = nn.Conv2d(64, 32, 3, padding=1)
-        self.conv = nn.Conv2d(128, 80, 3, padding=1)
+        self.convc1 = FusedConv2d(cor_planes, 96, 1, padding=0)
+        self.convf1 = FusedConv2d(2, 64, 7, padding=3)
+        self.convf2 = FusedConv2d(64, 32, 3, padding=1)
+        self.conv = FusedConv2d(128, 80, 3, padding=1)
 
     def forward(self, flow, corr):
         if corr.is_contiguous(memory_format=torch.channels_last):
@@ -111,11 +140,11 @@ class BasicMotionEncoder(nn.Module):
     def __init__(self, args):
         super().__init__()
         cor_planes = args.corr_levels * (2 * args.corr_radius + 1) ** 2
-        self.convc1 = nn.Conv2d(cor_planes, 256, 1, padding=0)
-        self.convc2 = nn.Conv2d(256, 192, 3, padding=1)
-        self.convf1 = nn.Conv2d(2, 128, 7, padding=3)
-        self.convf2 = nn.Conv2d(128, 64, 3, padding=1)
-        self.conv = nn.Conv2d(64 + 192, 128 - 2, 3, padding=1)
+        self.convc1 = FusedConv2d(cor_planes, 256, 1, padding=0)
+        self.convc2 = FusedConv2d(256, 192, 3, padding=1)
+        self.convf1 = FusedConv2d(2, 128, 7, padding=3)
+        self.convf2 = FusedConv2d(128, 64, 3, padding=1)
+        self.conv = FusedConv2d(64 + 192, 128 - 2, 3, padding=1)
 
     def forward(self, flow, corr):
         if corr.is_contiguous(memory_format=torch.channels_last):

@@ -1,0 +1,101 @@
+"""Implicit-GEMM conv (kernel #5) autograd binding for the update block.
+
+Routes the stride-1 same-padding NHWC bf16 convolutions of the motion
+encoder / SepConvGRU / flow head (reference update.py:6-146) onto the
+hand-written MFMA kernel in csrc/conv_gemm.hip. The packed bf16 weights
+([kyx][o][c_pad] for forward, flipped/transposed for backward-data) are
+cached per weight version so they are rebuilt only after optimizer steps.
+
+  forward:   out = conv(x, w) + b          (one kernel)
+  bwd-data:  dx  = conv(dy, flipT(w))      (same kernel, swapped packing)
+  bwd-wrw:   dW  = split-M MFMA partials + reduce (fp32)
+  bwd-bias:  dy.sum((0,2,3))               (torch reduce)
+"""
+
+import torch
+import torch.nn.functional as F
+
+from . import _ext
+
+
+def _pad64(n):
+    return (n + 63) // 64 * 64
+
+
+def _pack_fwd(weight):
+    """(O, I, KH, KW) fp32 -> (KYX, O, pad64(I)) bf16 contiguous."""
+    O, I, KH, KW = weight.shape
+    w = weight.detach().to(torch.bfloat16).permute(2, 3, 0, 1).reshape(
+        KH * KW, O, I)
+    return F.pad(w, (0, _pad64(I) - I)).contiguous()
+
+
+def _pack_bwd(weight):
+    """flip + transpose: (O, I, KH, KW) -> (KYX, I, pad64(O)) bf16."""
+    O, I, KH, KW = weight.shape
+    w = weight.detach().to(torch.bfloat16).flip(2, 3).permute(2, 3, 1, 0)
+    w = w.reshape(KH * KW, I, O)
+    return F.pad(w, (0, _pad64(O) - O)).contiguous()
+
+
+def _packs(weight, cache, key=None):
+    if key is None:
+        key = (weight._version, weight.data_ptr())
+    if cache.get("k") != key:
+        cache["k"] = key
+        cache["fwd"] = _pack_fwd(weight)
+        cache["bwd"] = _pack_bwd(weight)
+    return cache["fwd"], cache["bwd"]
+
+
+class ConvGemmFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, weight, bias, wpk_fwd, wpk_bwd):
+        O, I, KH, KW = weight.shape
+        b = bias.detach().float().contiguous() if bias is not None else None
+        out = _ext.ext().conv_gemm_fwd(x, wpk_fwd, b, O, KH, KW, 0)
+        ctx.save_for_backward(x, wpk_bwd)
+        ctx.meta = (O, I, KH, KW, bias is not None)
+        return out
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, wpk_bwd = ctx.saved_tensors
+        O, I, KH, KW, has_bias = ctx.meta
+        dy = dy.contiguous(memory_format=torch.channels_last)
+        if dy.dtype != torch.bfloat16:
+            dy = dy.to(torch.bfloat16)
+        dx = _ext.ext().conv_gemm_fwd(dy, wpk_bwd, None, I, KH, KW, 0)
+        dw = _ext.ext().conv_gemm_wrw(dy, x, KH, KW)
+        dbias = dy.float().sum(dim=(0, 2, 3)) if has_bias else None
+        return dx, dw, dbias, None, None
+
+
+def can_fuse_conv(x, weight, stride, padding, dilation, groups):
+    if not (x.is_cuda and x.dim() == 4 and x.dtype == torch.bfloat16):
+        return False
+    if x.stride(1) != 1:  # channels-last-like only
+        return False
+    if _ext.ext() is None:
+        return False
+    from torch.nn.modules.utils import _pair
+    if (_pair(stride) != (1, 1) or _pair(dilation) != (1, 1) or groups != 1):
+        return False
+    O, I, KH, KW = weight.shape
+    if _pair(padding) != (KH // 2, KW // 2):
+        return False
+    if I != x.shape[1] or O > 512 or KH * KW > 49:
+        return False
+    # channel-narrowed views are fine; anything else must be cl-contiguous
+    if x.stride(3) < x.size(1):
+        return False
+    return True
+
+
+def fused_conv2d(x, weight, bias, stride, padding, dilation, groups, cache,
+                 key=None):
+    """F.conv2d drop-in that routes supported shapes to the MFMA kernel."""
+    if can_fuse_conv(x, weight, stride, padding, dilation, groups):
+        wf, wb = _packs(weight, cache, key)
+        return ConvGemmFn.apply(x, weight, bias, wf, wb)
+    return F.conv2d(x, weight, bias, stride, padding, dilation, groups)

@@ -449,3 +449,65 @@ class TestZeroInjectHIP:
         (out * g.to(_dev())).sum().backward()
         (ref * g).sum().backward()
         torch.testing.assert_close(x.grad.cpu(), x2.grad)
+
+
+class TestConvGemm:
+    SHAPES = [
+        # (Cin, Cout, KH, KW) — the update-block inventory (update.py:6-146)
+        (324, 256, 1, 1),   # convc1 (Cin % 8 != 0: zero-weight tail path)
+        (256, 192, 3, 3),   # convc2
+        (2, 128, 7, 7),     # convf1 (tiny Cin)
+        (128, 64, 3, 3),    # convf2
+        (256, 126, 3, 3),   # conv (ragged Cout)
+        (384, 256, 1, 5),   # packed zr horizontal
+        (384, 128, 5, 1),   # q vertical
+        (256, 2, 3, 3),     # flow head out (Cout=2)
+    ]
+
+    @pytest.mark.parametrize("ci,co,kh,kw", SHAPES)
+    def test_fwd_bwd_matches_miopen(self, ci, co, kh, kw):
+        from flowhip.ops.functional_conv import fused_conv2d, can_fuse_conv
+        torch.manual_seed(17)
+        B, H, W = 2, 24, 40
+        x = (torch.randn(B, ci, H, W, device=_dev()) / 8).to(torch.bfloat16) \
+            .contiguous(memory_format=torch.channels_last).requires_grad_(True)
+        w = (torch.randn(co, ci, kh, kw, device=_dev()) /
+             (ci * kh * kw) ** 0.5).requires_grad_(True)
+        b = torch.randn(co, device=_dev()).requires_grad_(True)
+        pad = (kh // 2, kw // 2)
+        assert can_fuse_conv(x, w, 1, pad, 1, 1)
+
+        cache = {}
+        out = fused_conv2d(x, w, b, 1, pad, 1, 1, cache)
+        ref = torch.nn.functional.conv2d(
+            x.detach().float(), w.detach().float(), b.detach().float(),
+            padding=pad)
+        tol = dict(atol=5e-2, rtol=5e-2)  # bf16 inputs, fp32 accumulate
+        torch.testing.assert_close(out.float(), ref, **tol)
+
+        g = torch.randn_like(ref).to(torch.bfloat16) \
+            .contiguous(memory_format=torch.channels_last)
+        out.backward(g)
+        x2 = x.detach().float().requires_grad_(True)
+        w2 = w.detach().float().requires_grad_(True)
+        b2 = b.detach().float().requires_grad_(True)
+        torch.nn.functional.conv2d(x2, w2, b2, padding=pad).backward(
+            g.float())
+        torch.testing.assert_close(x.grad.float(), x2.grad, **tol)
+        torch.testing.assert_close(w.grad.float(), w2.grad,
+                                   atol=1e-1, rtol=5e-2)
+        torch.testing.assert_close(b.grad.float(), b2.grad,
+                                   atol=1e-1, rtol=5e-2)
+
+    def test_narrowed_input_view(self):
+        # a channel-narrowed channels-last view (the 324-of-328 corr case)
+        from flowhip.ops.functional_conv import fused_conv2d
+        torch.manual_seed(18)
+        B, H, W = 2, 16, 24
+        full = (torch.randn(B, 328, H, W, device=_dev()) / 8) \
+            .to(torch.bfloat16).contiguous(memory_format=torch.channels_last)
+        x = full[:, :324]
+        w = (torch.randn(64, 324, 1, 1, device=_dev()) / 18)
+        out = fused_conv2d(x, w, None, 1, 0, 1, 1, {})
+        ref = torch.nn.functional.conv2d(x.float(), w.float())
+        torch.testing.assert_close(out.float(), ref, atol=5e-2, rtol=5e-2)
