@@ -56,6 +56,8 @@ bool flowhip_nconv_wrw_tiled_launch(const float* dnomin, const float* ddenom,
                                     int Ci, int Co, int H, int W, int K,
                                     hipStream_t stream);
 int flowhip_instnorm_partial_rows(int N, int C, long P);
+void flowhip_transpose_cast_launch(const float* in, void* out, int B, int M,
+                                   int N, hipStream_t stream);
 void flowhip_conv_gemm_fwd_launch(const void* x, const void* wpk,
                                   const float* bias, void* out,
                                   const void* zpage, long Mtot, int HH,
@@ -647,6 +649,19 @@ torch::Tensor conv_gemm_wrw(torch::Tensor dy, torch::Tensor x, int64_t KH,
   return dw;
 }
 
+torch::Tensor transpose_cast_bf16(torch::Tensor in) {
+  TORCH_CHECK(in.is_cuda() && in.dim() == 3 && in.is_contiguous() &&
+              in.dtype() == torch::kFloat32);
+  const int B = in.size(0), M = in.size(1), N = in.size(2);
+  auto out = torch::empty({(long)B, (long)N, (long)M},
+                          in.options().dtype(torch::kBFloat16));
+  const c10::cuda::CUDAGuard guard(in.device());
+  hipStream_t stream = at::cuda::getCurrentCUDAStream().stream();
+  flowhip_transpose_cast_launch(in.data_ptr<float>(), out.data_ptr(), B, M,
+                                N, stream);
+  return out;
+}
+
 }  // namespace gg
 
 std::vector<torch::Tensor> gru_gate1_fwd(torch::Tensor zr, torch::Tensor h) {
@@ -874,6 +889,19 @@ torch::Tensor conv_gemm_wrw(torch::Tensor dy, torch::Tensor x, int64_t KH,
   return dw;
 }
 
+torch::Tensor transpose_cast_bf16(torch::Tensor in) {
+  TORCH_CHECK(in.is_cuda() && in.dim() == 3 && in.is_contiguous() &&
+              in.dtype() == torch::kFloat32);
+  const int B = in.size(0), M = in.size(1), N = in.size(2);
+  auto out = torch::empty({(long)B, (long)N, (long)M},
+                          in.options().dtype(torch::kBFloat16));
+  const c10::cuda::CUDAGuard guard(in.device());
+  hipStream_t stream = at::cuda::getCurrentCUDAStream().stream();
+  flowhip_transpose_cast_launch(in.data_ptr<float>(), out.data_ptr(), B, M,
+                                N, stream);
+  return out;
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -892,6 +920,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("convex_up_bwd", &convex_up_bwd, "backward of convex_up_fwd");
   m.def("nconv_fwd", &nconv_fwd,
         "fused normalized convolution forward (out, cout)");
+  m.def("transpose_cast_bf16", &transpose_cast_bf16,
+        "(B,M,N) fp32 -> (B,N,M) bf16 tiled transpose");
   m.def("conv_gemm_fwd", &conv_gemm_fwd,
         "implicit-GEMM NHWC bf16 conv forward (also bwd-data with flipped "
         "packed weights)");

@@ -76,8 +76,11 @@ class CorrVolumeFn(torch.autograd.Function):
         B, D, H, W = ctx.shape
         P = H * W
         alpha = 1.0 / math.sqrt(D)
-        dc = grad.reshape(B, P, P).to(torch.bfloat16).contiguous()
-        dct = grad.reshape(B, P, P).transpose(1, 2).to(torch.bfloat16).contiguous()
+        g3 = grad.reshape(B, P, P).contiguous()
+        dc = g3.to(torch.bfloat16)
+        # tiled transpose+cast kernel (eager transpose().to(bf16) is an
+        # uncoalesced ~300us elementwise op at P=7168)
+        dct = _ext.ext().transpose_cast_bf16(g3)
         # dF1t[i,d] = sum_j dC[i,j] * F2t[j,d] -> A=dC (M=P,K=P), B=f2 with
         # (N=D, K=P) k-major == f2t^T; i.e. B operand is f2 (D,P) row-major? No:
         # gemm_nt wants B (N,K) row-major = (D,P) with P contiguous == f2 view.

@@ -511,3 +511,14 @@ class TestConvGemm:
         out = fused_conv2d(x, w, None, 1, 0, 1, 1, {})
         ref = torch.nn.functional.conv2d(x.float(), w.float())
         torch.testing.assert_close(out.float(), ref, atol=5e-2, rtol=5e-2)
+
+
+class TestTransposeCast:
+    def test_matches_eager(self):
+        import flowhip._C as C
+        torch.manual_seed(19)
+        for (B, M, N) in [(2, 128, 128), (1, 300, 177), (3, 64, 96)]:
+            x = torch.randn(B, M, N, device=_dev())
+            got = C.transpose_cast_bf16(x)
+            ref = x.transpose(1, 2).to(torch.bfloat16).contiguous()
+            torch.testing.assert_close(got, ref)
