@@ -58,6 +58,15 @@ bool flowhip_nconv_wrw_tiled_launch(const float* dnomin, const float* ddenom,
 int flowhip_instnorm_partial_rows(int N, int C, long P);
 void flowhip_transpose_cast_launch(const float* in, void* out, int B, int M,
                                    int N, hipStream_t stream);
+void flowhip_conf_pool_fwd_launch(const float* data, const float* conf,
+                                  float* data_ds, float* conf_ds,
+                                  unsigned char* code, long total, int H,
+                                  int W, int OH, int OW, hipStream_t stream);
+void flowhip_conf_pool_bwd_launch(const float* gdata_ds,
+                                  const float* gconf_ds,
+                                  const unsigned char* code, float* gdata,
+                                  float* gconf, long total_in, int H, int W,
+                                  int OH, int OW, hipStream_t stream);
 void flowhip_conv_gemm_fwd_launch(const void* x, const void* wpk,
                                   const float* bias, void* out,
                                   const void* zpage, long Mtot, int HH,
@@ -662,6 +671,49 @@ torch::Tensor transpose_cast_bf16(torch::Tensor in) {
   return out;
 }
 
+std::vector<torch::Tensor> conf_pool_fwd(torch::Tensor data,
+                                         torch::Tensor conf) {
+  TORCH_CHECK(data.is_cuda() && data.is_contiguous() &&
+              data.dtype() == torch::kFloat32);
+  TORCH_CHECK(conf.is_cuda() && conf.is_contiguous() &&
+              conf.sizes() == data.sizes());
+  const long N = data.size(0), C = data.size(1);
+  const int H = data.size(2), W = data.size(3);
+  const int OH = H / 2, OW = W / 2;
+  auto dds = torch::empty({N, C, (long)OH, (long)OW}, data.options());
+  auto cds = torch::empty_like(dds);
+  auto code = torch::empty({N, C, (long)OH, (long)OW},
+                           data.options().dtype(torch::kUInt8));
+  const c10::cuda::CUDAGuard guard(data.device());
+  hipStream_t stream = at::cuda::getCurrentCUDAStream().stream();
+  flowhip_conf_pool_fwd_launch(data.data_ptr<float>(),
+                               conf.data_ptr<float>(), dds.data_ptr<float>(),
+                               cds.data_ptr<float>(),
+                               code.data_ptr<unsigned char>(), dds.numel(),
+                               H, W, OH, OW, stream);
+  return {dds, cds, code};
+}
+
+std::vector<torch::Tensor> conf_pool_bwd(c10::optional<torch::Tensor> gdds,
+                                         c10::optional<torch::Tensor> gcds,
+                                         torch::Tensor code,
+                                         std::vector<int64_t> in_shape) {
+  TORCH_CHECK(gdds.has_value() || gcds.has_value());
+  auto& any = gdds.has_value() ? gdds.value() : gcds.value();
+  const int H = in_shape[2], W = in_shape[3];
+  const int OH = code.size(2), OW = code.size(3);
+  auto gdata = torch::empty(in_shape, any.options());
+  auto gconf = torch::empty(in_shape, any.options());
+  const c10::cuda::CUDAGuard guard(any.device());
+  hipStream_t stream = at::cuda::getCurrentCUDAStream().stream();
+  flowhip_conf_pool_bwd_launch(
+      gdds.has_value() ? gdds->data_ptr<float>() : nullptr,
+      gcds.has_value() ? gcds->data_ptr<float>() : nullptr,
+      code.data_ptr<unsigned char>(), gdata.data_ptr<float>(),
+      gconf.data_ptr<float>(), gdata.numel(), H, W, OH, OW, stream);
+  return {gdata, gconf};
+}
+
 }  // namespace gg
 
 std::vector<torch::Tensor> gru_gate1_fwd(torch::Tensor zr, torch::Tensor h) {
@@ -902,6 +954,49 @@ torch::Tensor transpose_cast_bf16(torch::Tensor in) {
   return out;
 }
 
+std::vector<torch::Tensor> conf_pool_fwd(torch::Tensor data,
+                                         torch::Tensor conf) {
+  TORCH_CHECK(data.is_cuda() && data.is_contiguous() &&
+              data.dtype() == torch::kFloat32);
+  TORCH_CHECK(conf.is_cuda() && conf.is_contiguous() &&
+              conf.sizes() == data.sizes());
+  const long N = data.size(0), C = data.size(1);
+  const int H = data.size(2), W = data.size(3);
+  const int OH = H / 2, OW = W / 2;
+  auto dds = torch::empty({N, C, (long)OH, (long)OW}, data.options());
+  auto cds = torch::empty_like(dds);
+  auto code = torch::empty({N, C, (long)OH, (long)OW},
+                           data.options().dtype(torch::kUInt8));
+  const c10::cuda::CUDAGuard guard(data.device());
+  hipStream_t stream = at::cuda::getCurrentCUDAStream().stream();
+  flowhip_conf_pool_fwd_launch(data.data_ptr<float>(),
+                               conf.data_ptr<float>(), dds.data_ptr<float>(),
+                               cds.data_ptr<float>(),
+                               code.data_ptr<unsigned char>(), dds.numel(),
+                               H, W, OH, OW, stream);
+  return {dds, cds, code};
+}
+
+std::vector<torch::Tensor> conf_pool_bwd(c10::optional<torch::Tensor> gdds,
+                                         c10::optional<torch::Tensor> gcds,
+                                         torch::Tensor code,
+                                         std::vector<int64_t> in_shape) {
+  TORCH_CHECK(gdds.has_value() || gcds.has_value());
+  auto& any = gdds.has_value() ? gdds.value() : gcds.value();
+  const int H = in_shape[2], W = in_shape[3];
+  const int OH = code.size(2), OW = code.size(3);
+  auto gdata = torch::empty(in_shape, any.options());
+  auto gconf = torch::empty(in_shape, any.options());
+  const c10::cuda::CUDAGuard guard(any.device());
+  hipStream_t stream = at::cuda::getCurrentCUDAStream().stream();
+  flowhip_conf_pool_bwd_launch(
+      gdds.has_value() ? gdds->data_ptr<float>() : nullptr,
+      gcds.has_value() ? gcds->data_ptr<float>() : nullptr,
+      code.data_ptr<unsigned char>(), gdata.data_ptr<float>(),
+      gconf.data_ptr<float>(), gdata.numel(), H, W, OH, OW, stream);
+  return {gdata, gconf};
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -920,6 +1015,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("convex_up_bwd", &convex_up_bwd, "backward of convex_up_fwd");
   m.def("nconv_fwd", &nconv_fwd,
         "fused normalized convolution forward (out, cout)");
+  m.def("conf_pool_fwd", &conf_pool_fwd,
+        "confidence-based 2x pooling forward (data_ds, conf_ds, argmax code)");
+  m.def("conf_pool_bwd", &conf_pool_bwd, "backward of conf_pool");
   m.def("transpose_cast_bf16", &transpose_cast_bf16,
         "(B,M,N) fp32 -> (B,N,M) bf16 tiled transpose");
   m.def("conv_gemm_fwd", &conv_gemm_fwd,

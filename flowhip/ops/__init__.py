@@ -76,6 +76,11 @@ def _can_fuse_nconv(weight, stride, dilation, groups):
 
 
 def conf_pool(data, conf, ds_factor=2, pooling_type="conf_based"):
+    if (_ext.use_hip(data) and ds_factor == 2
+            and pooling_type == "conf_based"
+            and data.dtype == torch.float32):
+        from .functional_nconv import ConfPoolFn
+        return ConfPoolFn.apply(data, conf)
     return torch_ref.conf_pool(data, conf, ds_factor, pooling_type)
 
 

@@ -70,3 +70,26 @@ class NConv2dFn(torch.autograd.Function):
 
         dbias = gout.sum(dim=(0, 2, 3)) if ctx.has_bias else None
         return ddata, dconf, dweight, dbias, None, None, None
+
+
+
+class ConfPoolFn(torch.autograd.Function):
+    """Confidence-based 2x pooling (kernel #7; nconv_modules.py:94-104):
+    one fused kernel each way, full-write backward (no scatter)."""
+
+    @staticmethod
+    def forward(ctx, data, conf):
+        dds, cds, code = _ext.ext().conf_pool_fwd(data.contiguous(),
+                                                  conf.contiguous())
+        ctx.save_for_backward(code)
+        ctx.in_shape = list(data.shape)
+        return dds, cds
+
+    @staticmethod
+    def backward(ctx, gdds, gcds):
+        (code,) = ctx.saved_tensors
+        gdds = gdds.contiguous() if gdds is not None else None
+        gcds = gcds.contiguous() if gcds is not None else None
+        gdata, gconf = _ext.ext().conf_pool_bwd(gdds, gcds, code,
+                                                ctx.in_shape)
+        return gdata, gconf

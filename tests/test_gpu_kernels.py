@@ -522,3 +522,23 @@ class TestTransposeCast:
             got = C.transpose_cast_bf16(x)
             ref = x.transpose(1, 2).to(torch.bfloat16).contiguous()
             torch.testing.assert_close(got, ref)
+
+
+class TestConfPoolFused:
+    def test_matches_torch_ref(self):
+        from flowhip import ops
+        from flowhip.ops import torch_ref
+        torch.manual_seed(21)
+        data = torch.randn(3, 2, 30, 44, device=_dev(), requires_grad=True)
+        conf = torch.rand(3, 2, 30, 44, device=_dev(), requires_grad=True)
+        dds, cds = ops.conf_pool(data, conf)
+        d2 = data.detach().cpu().requires_grad_(True)
+        c2 = conf.detach().cpu().requires_grad_(True)
+        rdds, rcds = torch_ref.conf_pool(d2, c2)
+        torch.testing.assert_close(dds.cpu(), rdds)
+        torch.testing.assert_close(cds.cpu(), rcds)
+        g1, g2 = torch.randn_like(rdds), torch.randn_like(rcds)
+        (dds * g1.to(_dev()) + cds * g2.to(_dev())).sum().backward()
+        (rdds * g1 + rcds * g2).sum().backward()
+        torch.testing.assert_close(data.grad.cpu(), d2.grad)
+        torch.testing.assert_close(conf.grad.cpu(), c2.grad)
