@@ -22,12 +22,16 @@ case "$MODE" in
     cp /tmp/prof/*stats*.csv "$REPO/$OUT/" 2>/dev/null || true
     ;;
   pmc)
-    # counters only (no trace domains!); SQ has 8 slots, TCC 4
+    # counters only (no trace domains!); SQ has 8 slots, TCC 4.
+    # The per-dispatch CSV is ~100 MB — summarize on the box and ship only
+    # the per-kernel aggregate (gpurun merge cap is 64 MiB).
     rocprofv3 --pmc SQ_WAVE_CYCLES SQ_WAIT_ANY SQ_WAIT_INST_ANY \
       SQ_LDS_BANK_CONFLICT SQ_VALU_MFMA_BUSY_CYCLES \
       --output-format csv -d /tmp/pmc -o step \
       -- python "$REPO/bench.py" --steps 2 --warmup 6 --profile-steps 2
-    cp /tmp/pmc/*.csv "$REPO/$OUT/" 2>/dev/null || true
+    python "$REPO/tools/pmc_summarize.py" /tmp/pmc/*counter_collection*.csv \
+      > "$REPO/$OUT/pmc_summary.csv"
+    cp /tmp/pmc/*agent_info*.csv "$REPO/$OUT/" 2>/dev/null || true
     ;;
   infer)
     rocprofv3 --kernel-trace --stats --output-format csv -d /tmp/prof -o infer \
