@@ -17,6 +17,10 @@ mkdir -p "$REPO/$OUT"
 
 case "$MODE" in
   trace)
+    # FLOWHIP_MIOPEN_FIND=0: the trace covers the whole process including
+    # warmup; benchmark-mode MIOpen find would dominate it with solver
+    # trials (throughput is identical either way — bench10 A/B)
+    FLOWHIP_MIOPEN_FIND=0 \
     rocprofv3 --kernel-trace --stats --output-format csv -d /tmp/prof -o step \
       -- python "$REPO/bench.py" --steps 4 --warmup 8 --profile-steps 4
     cp /tmp/prof/*stats*.csv "$REPO/$OUT/" 2>/dev/null || true
@@ -25,6 +29,7 @@ case "$MODE" in
     # counters only (no trace domains!); SQ has 8 slots, TCC 4.
     # The per-dispatch CSV is ~100 MB — summarize on the box and ship only
     # the per-kernel aggregate (gpurun merge cap is 64 MiB).
+    FLOWHIP_MIOPEN_FIND=0 \
     rocprofv3 --pmc SQ_WAVE_CYCLES SQ_WAIT_ANY SQ_WAIT_INST_ANY \
       SQ_LDS_BANK_CONFLICT SQ_VALU_MFMA_BUSY_CYCLES \
       --output-format csv -d /tmp/pmc -o step \
