@@ -67,18 +67,21 @@ void flowhip_conf_pool_bwd_launch(const float* gdata_ds,
                                   const unsigned char* code, float* gdata,
                                   float* gconf, long total_in, int H, int W,
                                   int OH, int OW, hipStream_t stream);
-void flowhip_conv_gemm_fwd_launch(const void* x, const void* wpk,
-                                  const float* bias, void* out,
-                                  const void* zpage, long Mtot, int HH,
-                                  int WW, int ld_x, int Cin, int Cout,
+void flowhip_conv_gemm_fwd_launch(const void* x, const void* x2,
+                                  const void* wpk, const float* bias,
+                                  void* out, void* out2, const void* zpage,
+                                  long Mtot, int HH, int WW, int ld_x,
+                                  int ld_x2, int C1, int Cin, int Cout,
                                   int cpad, int KH, int KW, int padH,
-                                  int padW, int act, hipStream_t stream);
+                                  int padW, int osplit, int act,
+                                  hipStream_t stream);
 void flowhip_conv_gemm_wrw_launch(const void* dy, const void* x,
-                                  float* partials, float* dw,
+                                  const void* x2, float* partials, float* dw,
                                   const void* zpage, long Mtot, int HH,
-                                  int WW, int ld_x, int Cin, int Cout,
-                                  int cpad, int KH, int KW, int padH,
-                                  int padW, int nchunk, hipStream_t stream);
+                                  int WW, int ld_x, int ld_x2, int C1,
+                                  int Cin, int Cout, int cpad, int KH,
+                                  int KW, int padH, int padW, int nchunk,
+                                  hipStream_t stream);
 void flowhip_instnorm_cl_fwd_launch(const void* x, void* y, float* mean,
                                     float* rstd, float* partials, int N,
                                     int C, long P, float eps, int is_bf16,
@@ -502,218 +505,6 @@ torch::Tensor make_like(const torch::Tensor& t, bool cl) {
             : torch::empty(t.sizes(), t.options());
 }
 
-torch::Tensor zero_inject_fwd(torch::Tensor inp, int64_t sH, int64_t sW,
-                              int64_t oh, int64_t ow) {
-  TORCH_CHECK(inp.is_cuda() && inp.dtype() == torch::kFloat32);
-  auto x = inp.contiguous();
-  const long N = x.size(0), C = x.size(1);
-  const int ih = x.size(2), iw = x.size(3);
-  auto out = torch::empty({N, C, oh, ow}, x.options());
-  const c10::cuda::CUDAGuard guard(x.device());
-  hipStream_t stream = at::cuda::getCurrentCUDAStream().stream();
-  flowhip_zero_inject_fwd_launch(x.data_ptr<float>(), out.data_ptr<float>(),
-                                 out.numel(), ih, iw, (int)oh, (int)ow,
-                                 (int)sH, (int)sW, stream);
-  return out;
-}
-
-torch::Tensor zero_inject_bwd(torch::Tensor gout, int64_t sH, int64_t sW,
-                              int64_t ih, int64_t iw) {
-  auto g = gout.contiguous();
-  const long N = g.size(0), C = g.size(1);
-  const int oh = g.size(2), ow = g.size(3);
-  auto dinp = torch::empty({N, C, ih, iw}, g.options());
-  const c10::cuda::CUDAGuard guard(g.device());
-  hipStream_t stream = at::cuda::getCurrentCUDAStream().stream();
-  flowhip_zero_inject_bwd_launch(g.data_ptr<float>(), dinp.data_ptr<float>(),
-                                 dinp.numel(), (int)ih, (int)iw, oh, ow,
-                                 (int)sH, (int)sW, stream);
-  return dinp;
-}
-
-std::vector<torch::Tensor> instnorm_cl_fwd(torch::Tensor x, double eps) {
-  TORCH_CHECK(x.is_cuda() && x.dim() == 4);
-  TORCH_CHECK(x.is_contiguous(torch::MemoryFormat::ChannelsLast),
-              "instnorm_cl: channels_last input required");
-  const bool bf16 = x.scalar_type() == torch::kBFloat16;
-  TORCH_CHECK(bf16 || x.scalar_type() == torch::kFloat32);
-  const int N = x.size(0), C = x.size(1);
-  const long P = (long)x.size(2) * x.size(3);
-  auto y = torch::empty(x.sizes(), x.options(),
-                        torch::MemoryFormat::ChannelsLast);
-  auto mean = torch::empty({(long)N, (long)C},
-                           x.options().dtype(torch::kFloat32));
-  auto rstd = torch::empty_like(mean);
-  const c10::cuda::CUDAGuard guard(x.device());
-  hipStream_t stream = at::cuda::getCurrentCUDAStream().stream();
-  auto partials = torch::empty({(long)flowhip_instnorm_partial_rows(N, C, P),
-                                64},
-                               x.options().dtype(torch::kFloat32));
-  flowhip_instnorm_cl_fwd_launch(x.data_ptr(), y.data_ptr(),
-                                 mean.data_ptr<float>(),
-                                 rstd.data_ptr<float>(),
-                                 partials.data_ptr<float>(), N, C, P,
-                                 (float)eps, bf16 ? 1 : 0, stream);
-  return {y, mean, rstd};
-}
-
-torch::Tensor instnorm_cl_bwd(torch::Tensor x, torch::Tensor dy,
-                              torch::Tensor mean, torch::Tensor rstd) {
-  const bool bf16 = x.scalar_type() == torch::kBFloat16;
-  const int N = x.size(0), C = x.size(1);
-  const long P = (long)x.size(2) * x.size(3);
-  auto dx = torch::empty(x.sizes(), x.options(),
-                         torch::MemoryFormat::ChannelsLast);
-  auto gmean = torch::empty_like(mean);
-  auto gxmean = torch::empty_like(mean);
-  auto partials = torch::empty({(long)flowhip_instnorm_partial_rows(N, C, P),
-                                64},
-                               x.options().dtype(torch::kFloat32));
-  const c10::cuda::CUDAGuard guard(x.device());
-  hipStream_t stream = at::cuda::getCurrentCUDAStream().stream();
-  flowhip_instnorm_cl_bwd_launch(x.data_ptr(), dy.data_ptr(),
-                                 mean.data_ptr<float>(),
-                                 rstd.data_ptr<float>(),
-                                 gmean.data_ptr<float>(),
-                                 gxmean.data_ptr<float>(),
-                                 partials.data_ptr<float>(), dx.data_ptr(),
-                                 N, C, P, bf16 ? 1 : 0, stream);
-  return dx;
-}
-
-static const void* cg_zero_page() {
-  static void* p = nullptr;
-  if (p == nullptr) {
-    hipMalloc(&p, 256);
-    hipMemset(p, 0, 256);
-  }
-  return p;
-}
-
-// x: (N, C, H, W) with channels-last-like strides (stride(1)==1); a
-// channel-narrowed view is allowed (ld_x = stride at dim 3 >= C).
-static void cg_check_x(const torch::Tensor& x, int& ld) {
-  TORCH_CHECK(x.is_cuda() && x.dim() == 4 &&
-              x.scalar_type() == torch::kBFloat16);
-  TORCH_CHECK(x.stride(1) == 1, "conv_gemm: channels-last layout required");
-  ld = (int)x.stride(3);
-  TORCH_CHECK(ld >= x.size(1) && x.stride(2) == (long)ld * x.size(3) &&
-              x.stride(0) == (long)ld * x.size(3) * x.size(2));
-}
-
-torch::Tensor conv_gemm_fwd(torch::Tensor x, torch::Tensor wpk,
-                            c10::optional<torch::Tensor> bias, int64_t Cout,
-                            int64_t KH, int64_t KW, int64_t act) {
-  int ld_x;
-  cg_check_x(x, ld_x);
-  TORCH_CHECK(wpk.is_cuda() && wpk.is_contiguous() &&
-              wpk.scalar_type() == torch::kBFloat16 && wpk.dim() == 3);
-  const int N = x.size(0), Cin = x.size(1), H = x.size(2), W = x.size(3);
-  const int cpad = wpk.size(2);
-  TORCH_CHECK(wpk.size(0) == KH * KW && wpk.size(1) == Cout);
-  const long Mtot = (long)N * H * W;
-  auto out = torch::empty({(long)N, Cout, (long)H, (long)W},
-                          x.options(), torch::MemoryFormat::ChannelsLast);
-  const float* bptr = nullptr;
-  if (bias.has_value()) {
-    TORCH_CHECK(bias->is_contiguous() &&
-                bias->scalar_type() == torch::kFloat32 &&
-                bias->numel() == Cout);
-    bptr = bias->data_ptr<float>();
-  }
-  const c10::cuda::CUDAGuard guard(x.device());
-  hipStream_t stream = at::cuda::getCurrentCUDAStream().stream();
-  flowhip_conv_gemm_fwd_launch(x.data_ptr(), wpk.data_ptr(), bptr,
-                               out.data_ptr(), cg_zero_page(), Mtot, H, W,
-                               ld_x, Cin, (int)Cout, cpad, (int)KH, (int)KW,
-                               (int)KH / 2, (int)KW / 2, (int)act, stream);
-  return out;
-}
-
-torch::Tensor conv_gemm_wrw(torch::Tensor dy, torch::Tensor x, int64_t KH,
-                            int64_t KW) {
-  int ld_x;
-  cg_check_x(x, ld_x);
-  TORCH_CHECK(dy.is_cuda() && dy.scalar_type() == torch::kBFloat16 &&
-              dy.stride(1) == 1 && dy.stride(3) == dy.size(1),
-              "conv_gemm_wrw: dy must be channels-last contiguous");
-  const int N = x.size(0), Cin = x.size(1), H = x.size(2), W = x.size(3);
-  const int Cout = dy.size(1);
-  const long Mtot = (long)N * H * W;
-  const int cpad = (int)((Cin + 63) / 64) * 64;
-  const int tiles_o = (Cout + 63) / 64;
-  const int nchunk = 8;
-  auto partials = torch::empty(
-      {(long)nchunk * KH * KW * tiles_o * 64 * cpad},
-      x.options().dtype(torch::kFloat32));
-  auto dw = torch::empty({(long)Cout, (long)Cin, KH, KW},
-                         x.options().dtype(torch::kFloat32));
-  const c10::cuda::CUDAGuard guard(x.device());
-  hipStream_t stream = at::cuda::getCurrentCUDAStream().stream();
-  flowhip_conv_gemm_wrw_launch(dy.data_ptr(), x.data_ptr(),
-                               partials.data_ptr<float>(),
-                               dw.data_ptr<float>(), cg_zero_page(), Mtot, H,
-                               W, ld_x, Cin, Cout, cpad, (int)KH, (int)KW,
-                               (int)KH / 2, (int)KW / 2, nchunk, stream);
-  return dw;
-}
-
-torch::Tensor transpose_cast_bf16(torch::Tensor in) {
-  TORCH_CHECK(in.is_cuda() && in.dim() == 3 && in.is_contiguous() &&
-              in.dtype() == torch::kFloat32);
-  const int B = in.size(0), M = in.size(1), N = in.size(2);
-  auto out = torch::empty({(long)B, (long)N, (long)M},
-                          in.options().dtype(torch::kBFloat16));
-  const c10::cuda::CUDAGuard guard(in.device());
-  hipStream_t stream = at::cuda::getCurrentCUDAStream().stream();
-  flowhip_transpose_cast_launch(in.data_ptr<float>(), out.data_ptr(), B, M,
-                                N, stream);
-  return out;
-}
-
-std::vector<torch::Tensor> conf_pool_fwd(torch::Tensor data,
-                                         torch::Tensor conf) {
-  TORCH_CHECK(data.is_cuda() && data.is_contiguous() &&
-              data.dtype() == torch::kFloat32);
-  TORCH_CHECK(conf.is_cuda() && conf.is_contiguous() &&
-              conf.sizes() == data.sizes());
-  const long N = data.size(0), C = data.size(1);
-  const int H = data.size(2), W = data.size(3);
-  const int OH = H / 2, OW = W / 2;
-  auto dds = torch::empty({N, C, (long)OH, (long)OW}, data.options());
-  auto cds = torch::empty_like(dds);
-  auto code = torch::empty({N, C, (long)OH, (long)OW},
-                           data.options().dtype(torch::kUInt8));
-  const c10::cuda::CUDAGuard guard(data.device());
-  hipStream_t stream = at::cuda::getCurrentCUDAStream().stream();
-  flowhip_conf_pool_fwd_launch(data.data_ptr<float>(),
-                               conf.data_ptr<float>(), dds.data_ptr<float>(),
-                               cds.data_ptr<float>(),
-                               code.data_ptr<unsigned char>(), dds.numel(),
-                               H, W, OH, OW, stream);
-  return {dds, cds, code};
-}
-
-std::vector<torch::Tensor> conf_pool_bwd(c10::optional<torch::Tensor> gdds,
-                                         c10::optional<torch::Tensor> gcds,
-                                         torch::Tensor code,
-                                         std::vector<int64_t> in_shape) {
-  TORCH_CHECK(gdds.has_value() || gcds.has_value());
-  auto& any = gdds.has_value() ? gdds.value() : gcds.value();
-  const int H = in_shape[2], W = in_shape[3];
-  const int OH = code.size(2), OW = code.size(3);
-  auto gdata = torch::empty(in_shape, any.options());
-  auto gconf = torch::empty(in_shape, any.options());
-  const c10::cuda::CUDAGuard guard(any.device());
-  hipStream_t stream = at::cuda::getCurrentCUDAStream().stream();
-  flowhip_conf_pool_bwd_launch(
-      gdds.has_value() ? gdds->data_ptr<float>() : nullptr,
-      gcds.has_value() ? gcds->data_ptr<float>() : nullptr,
-      code.data_ptr<unsigned char>(), gdata.data_ptr<float>(),
-      gconf.data_ptr<float>(), gdata.numel(), H, W, OH, OW, stream);
-  return {gdata, gconf};
-}
-
 }  // namespace gg
 
 std::vector<torch::Tensor> gru_gate1_fwd(torch::Tensor zr, torch::Tensor h) {
@@ -827,11 +618,11 @@ std::vector<torch::Tensor> instnorm_cl_fwd(torch::Tensor x, double eps) {
   auto mean = torch::empty({(long)N, (long)C},
                            x.options().dtype(torch::kFloat32));
   auto rstd = torch::empty_like(mean);
-  const c10::cuda::CUDAGuard guard(x.device());
-  hipStream_t stream = at::cuda::getCurrentCUDAStream().stream();
   auto partials = torch::empty({(long)flowhip_instnorm_partial_rows(N, C, P),
                                 64},
                                x.options().dtype(torch::kFloat32));
+  const c10::cuda::CUDAGuard guard(x.device());
+  hipStream_t stream = at::cuda::getCurrentCUDAStream().stream();
   flowhip_instnorm_cl_fwd_launch(x.data_ptr(), y.data_ptr(),
                                  mean.data_ptr<float>(),
                                  rstd.data_ptr<float>(),
@@ -862,96 +653,6 @@ torch::Tensor instnorm_cl_bwd(torch::Tensor x, torch::Tensor dy,
                                  partials.data_ptr<float>(), dx.data_ptr(),
                                  N, C, P, bf16 ? 1 : 0, stream);
   return dx;
-}
-
-static const void* cg_zero_page() {
-  static void* p = nullptr;
-  if (p == nullptr) {
-    hipMalloc(&p, 256);
-    hipMemset(p, 0, 256);
-  }
-  return p;
-}
-
-// x: (N, C, H, W) with channels-last-like strides (stride(1)==1); a
-// channel-narrowed view is allowed (ld_x = stride at dim 3 >= C).
-static void cg_check_x(const torch::Tensor& x, int& ld) {
-  TORCH_CHECK(x.is_cuda() && x.dim() == 4 &&
-              x.scalar_type() == torch::kBFloat16);
-  TORCH_CHECK(x.stride(1) == 1, "conv_gemm: channels-last layout required");
-  ld = (int)x.stride(3);
-  TORCH_CHECK(ld >= x.size(1) && x.stride(2) == (long)ld * x.size(3) &&
-              x.stride(0) == (long)ld * x.size(3) * x.size(2));
-}
-
-torch::Tensor conv_gemm_fwd(torch::Tensor x, torch::Tensor wpk,
-                            c10::optional<torch::Tensor> bias, int64_t Cout,
-                            int64_t KH, int64_t KW, int64_t act) {
-  int ld_x;
-  cg_check_x(x, ld_x);
-  TORCH_CHECK(wpk.is_cuda() && wpk.is_contiguous() &&
-              wpk.scalar_type() == torch::kBFloat16 && wpk.dim() == 3);
-  const int N = x.size(0), Cin = x.size(1), H = x.size(2), W = x.size(3);
-  const int cpad = wpk.size(2);
-  TORCH_CHECK(wpk.size(0) == KH * KW && wpk.size(1) == Cout);
-  const long Mtot = (long)N * H * W;
-  auto out = torch::empty({(long)N, Cout, (long)H, (long)W},
-                          x.options(), torch::MemoryFormat::ChannelsLast);
-  const float* bptr = nullptr;
-  if (bias.has_value()) {
-    TORCH_CHECK(bias->is_contiguous() &&
-                bias->scalar_type() == torch::kFloat32 &&
-                bias->numel() == Cout);
-    bptr = bias->data_ptr<float>();
-  }
-  const c10::cuda::CUDAGuard guard(x.device());
-  hipStream_t stream = at::cuda::getCurrentCUDAStream().stream();
-  flowhip_conv_gemm_fwd_launch(x.data_ptr(), wpk.data_ptr(), bptr,
-                               out.data_ptr(), cg_zero_page(), Mtot, H, W,
-                               ld_x, Cin, (int)Cout, cpad, (int)KH, (int)KW,
-                               (int)KH / 2, (int)KW / 2, (int)act, stream);
-  return out;
-}
-
-torch::Tensor conv_gemm_wrw(torch::Tensor dy, torch::Tensor x, int64_t KH,
-                            int64_t KW) {
-  int ld_x;
-  cg_check_x(x, ld_x);
-  TORCH_CHECK(dy.is_cuda() && dy.scalar_type() == torch::kBFloat16 &&
-              dy.stride(1) == 1 && dy.stride(3) == dy.size(1),
-              "conv_gemm_wrw: dy must be channels-last contiguous");
-  const int N = x.size(0), Cin = x.size(1), H = x.size(2), W = x.size(3);
-  const int Cout = dy.size(1);
-  const long Mtot = (long)N * H * W;
-  const int cpad = (int)((Cin + 63) / 64) * 64;
-  const int tiles_o = (Cout + 63) / 64;
-  const int nchunk = 8;
-  auto partials = torch::empty(
-      {(long)nchunk * KH * KW * tiles_o * 64 * cpad},
-      x.options().dtype(torch::kFloat32));
-  auto dw = torch::empty({(long)Cout, (long)Cin, KH, KW},
-                         x.options().dtype(torch::kFloat32));
-  const c10::cuda::CUDAGuard guard(x.device());
-  hipStream_t stream = at::cuda::getCurrentCUDAStream().stream();
-  flowhip_conv_gemm_wrw_launch(dy.data_ptr(), x.data_ptr(),
-                               partials.data_ptr<float>(),
-                               dw.data_ptr<float>(), cg_zero_page(), Mtot, H,
-                               W, ld_x, Cin, Cout, cpad, (int)KH, (int)KW,
-                               (int)KH / 2, (int)KW / 2, nchunk, stream);
-  return dw;
-}
-
-torch::Tensor transpose_cast_bf16(torch::Tensor in) {
-  TORCH_CHECK(in.is_cuda() && in.dim() == 3 && in.is_contiguous() &&
-              in.dtype() == torch::kFloat32);
-  const int B = in.size(0), M = in.size(1), N = in.size(2);
-  auto out = torch::empty({(long)B, (long)N, (long)M},
-                          in.options().dtype(torch::kBFloat16));
-  const c10::cuda::CUDAGuard guard(in.device());
-  hipStream_t stream = at::cuda::getCurrentCUDAStream().stream();
-  flowhip_transpose_cast_launch(in.data_ptr<float>(), out.data_ptr(), B, M,
-                                N, stream);
-  return out;
 }
 
 std::vector<torch::Tensor> conf_pool_fwd(torch::Tensor data,
@@ -997,6 +698,141 @@ std::vector<torch::Tensor> conf_pool_bwd(c10::optional<torch::Tensor> gdds,
   return {gdata, gconf};
 }
 
+torch::Tensor transpose_cast_bf16(torch::Tensor in) {
+  TORCH_CHECK(in.is_cuda() && in.dim() == 3 && in.is_contiguous() &&
+              in.dtype() == torch::kFloat32);
+  const int B = in.size(0), M = in.size(1), N = in.size(2);
+  auto out = torch::empty({(long)B, (long)N, (long)M},
+                          in.options().dtype(torch::kBFloat16));
+  const c10::cuda::CUDAGuard guard(in.device());
+  hipStream_t stream = at::cuda::getCurrentCUDAStream().stream();
+  flowhip_transpose_cast_launch(in.data_ptr<float>(), out.data_ptr(), B, M,
+                                N, stream);
+  return out;
+}
+
+static const void* cg_zero_page() {
+  static void* p = nullptr;
+  if (p == nullptr) {
+    hipMalloc(&p, 256);
+    hipMemset(p, 0, 256);
+  }
+  return p;
+}
+
+// x: (N, C, H, W) with channels-last-like strides (stride(1)==1); a
+// channel-narrowed view is allowed (ld_x = stride at dim 3 >= C).
+static void cg_check_x(const torch::Tensor& x, int& ld) {
+  TORCH_CHECK(x.is_cuda() && x.dim() == 4 &&
+              x.scalar_type() == torch::kBFloat16);
+  TORCH_CHECK(x.stride(1) == 1, "conv_gemm: channels-last layout required");
+  ld = (int)x.stride(3);
+  TORCH_CHECK(ld >= x.size(1) && x.stride(2) == (long)ld * x.size(3) &&
+              x.stride(0) == (long)ld * x.size(3) * x.size(2));
+}
+
+std::vector<torch::Tensor> conv_gemm_fwd2(
+    torch::Tensor x, c10::optional<torch::Tensor> x2, torch::Tensor wpk,
+    c10::optional<torch::Tensor> bias, int64_t Cout, int64_t KH, int64_t KW,
+    int64_t osplit, int64_t act) {
+  int ld_x, ld_x2 = 0;
+  cg_check_x(x, ld_x);
+  TORCH_CHECK(wpk.is_cuda() && wpk.is_contiguous() &&
+              wpk.scalar_type() == torch::kBFloat16 && wpk.dim() == 3);
+  const int N = x.size(0), C1 = x.size(1), H = x.size(2), W = x.size(3);
+  int Cin = C1;
+  const void* x2p = nullptr;
+  if (x2.has_value()) {
+    int l2;
+    cg_check_x(x2.value(), l2);
+    TORCH_CHECK(x2->size(0) == N && x2->size(2) == H && x2->size(3) == W);
+    TORCH_CHECK(C1 % 64 == 0, "conv_gemm cat2: first input must be a "
+                "multiple of 64 channels");
+    ld_x2 = l2;
+    Cin = C1 + (int)x2->size(1);
+    x2p = x2->data_ptr();
+  }
+  const int cpad = wpk.size(2);
+  TORCH_CHECK(wpk.size(0) == KH * KW && wpk.size(1) == Cout);
+  const long Mtot = (long)N * H * W;
+  const float* bptr = nullptr;
+  if (bias.has_value()) {
+    TORCH_CHECK(bias->is_contiguous() &&
+                bias->scalar_type() == torch::kFloat32 &&
+                bias->numel() == Cout);
+    bptr = bias->data_ptr<float>();
+  }
+  const c10::cuda::CUDAGuard guard(x.device());
+  hipStream_t stream = at::cuda::getCurrentCUDAStream().stream();
+  if (osplit <= 0 || osplit >= Cout) {
+    auto out = torch::empty({(long)N, Cout, (long)H, (long)W}, x.options(),
+                            torch::MemoryFormat::ChannelsLast);
+    flowhip_conv_gemm_fwd_launch(x.data_ptr(), x2p, wpk.data_ptr(), bptr,
+                                 out.data_ptr(), nullptr, cg_zero_page(),
+                                 Mtot, H, W, ld_x, ld_x2, C1, Cin, (int)Cout,
+                                 cpad, (int)KH, (int)KW, (int)KH / 2,
+                                 (int)KW / 2, (int)Cout, (int)act, stream);
+    return {out};
+  }
+  auto out = torch::empty({(long)N, osplit, (long)H, (long)W}, x.options(),
+                          torch::MemoryFormat::ChannelsLast);
+  auto out2 = torch::empty({(long)N, Cout - osplit, (long)H, (long)W},
+                           x.options(), torch::MemoryFormat::ChannelsLast);
+  flowhip_conv_gemm_fwd_launch(x.data_ptr(), x2p, wpk.data_ptr(), bptr,
+                               out.data_ptr(), out2.data_ptr(),
+                               cg_zero_page(), Mtot, H, W, ld_x, ld_x2, C1,
+                               Cin, (int)Cout, cpad, (int)KH, (int)KW,
+                               (int)KH / 2, (int)KW / 2, (int)osplit,
+                               (int)act, stream);
+  return {out, out2};
+}
+
+torch::Tensor conv_gemm_fwd(torch::Tensor x, torch::Tensor wpk,
+                            c10::optional<torch::Tensor> bias, int64_t Cout,
+                            int64_t KH, int64_t KW, int64_t act) {
+  return conv_gemm_fwd2(x, c10::nullopt, wpk, bias, Cout, KH, KW, 0,
+                        act)[0];
+}
+
+torch::Tensor conv_gemm_wrw(torch::Tensor dy, torch::Tensor x,
+                            c10::optional<torch::Tensor> x2, int64_t KH,
+                            int64_t KW) {
+  int ld_x, ld_x2 = 0;
+  cg_check_x(x, ld_x);
+  TORCH_CHECK(dy.is_cuda() && dy.scalar_type() == torch::kBFloat16 &&
+              dy.stride(1) == 1 && dy.stride(3) == dy.size(1),
+              "conv_gemm_wrw: dy must be channels-last contiguous");
+  const int N = x.size(0), C1 = x.size(1), H = x.size(2), W = x.size(3);
+  int Cin = C1;
+  const void* x2p = nullptr;
+  if (x2.has_value()) {
+    int l2;
+    cg_check_x(x2.value(), l2);
+    ld_x2 = l2;
+    Cin = C1 + (int)x2->size(1);
+    x2p = x2->data_ptr();
+  }
+  const int Cout = dy.size(1);
+  const long Mtot = (long)N * H * W;
+  const int cpad = (int)((Cin + 63) / 64) * 64;
+  const int tiles_o = (Cout + 63) / 64;
+  const int nchunk = 8;
+  auto partials = torch::empty(
+      {(long)nchunk * KH * KW * tiles_o * 64 * cpad},
+      x.options().dtype(torch::kFloat32));
+  auto dw = torch::empty({(long)Cout, (long)Cin, KH, KW},
+                         x.options().dtype(torch::kFloat32));
+  const c10::cuda::CUDAGuard guard(x.device());
+  hipStream_t stream = at::cuda::getCurrentCUDAStream().stream();
+  flowhip_conv_gemm_wrw_launch(dy.data_ptr(), x.data_ptr(), x2p,
+                               partials.data_ptr<float>(),
+                               dw.data_ptr<float>(), cg_zero_page(), Mtot, H,
+                               W, ld_x, ld_x2, C1, Cin, Cout, cpad, (int)KH,
+                               (int)KW, (int)KH / 2, (int)KW / 2, nchunk,
+                               stream);
+  return dw;
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -1023,6 +859,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv_gemm_fwd", &conv_gemm_fwd,
         "implicit-GEMM NHWC bf16 conv forward (also bwd-data with flipped "
         "packed weights)");
+  m.def("conv_gemm_fwd2", &conv_gemm_fwd2,
+        "conv forward over a virtually-concatenated pair of inputs and/or "
+        "with a split output (bwd-data of a cat input)");
   m.def("conv_gemm_wrw", &conv_gemm_wrw,
         "implicit-GEMM conv weight gradient (split-M + reduce)");
   m.def("instnorm_cl_fwd", &instnorm_cl_fwd,

@@ -41,10 +41,14 @@ __device__ __forceinline__ unsigned cg_swz(unsigned row, unsigned slot) {
 
 // Stage the A (im2col) tile: rows = 64 consecutive output pixels, k = one
 // 64-channel slab of one (ky,kx) section. OOB rows read the zero page.
+// Two-source form: channels [0, C1) come from x, [C1, Cin) from x2 — the
+// virtually-concatenated GRU input cat([h, x]) without materializing the
+// cat (C1 must be a multiple of 64 so a slab never straddles sources).
 __device__ __forceinline__ void cg_stage_a(
-    const __bf16* __restrict__ x, const __bf16* __restrict__ zpage,
+    const __bf16* __restrict__ x, const __bf16* __restrict__ x2,
+    const __bf16* __restrict__ zpage,
     char* lds_buf, int wave, int lane, long m0, long Mtot, int HH, int WW,
-    int ld_x, int Cin, int dy, int dx, int c0) {
+    int ld_x, int ld_x2, int C1, int Cin, int dy, int dx, int c0) {
 #pragma unroll
   for (int j = 0; j < CG_PPT; ++j) {
     const int piece0 = wave * 64 + CG_THREADS * j;
@@ -59,11 +63,13 @@ __device__ __forceinline__ void cg_stage_a(
     const long n = m / ((long)WW * HH);
     const int sy = yy + dy, sx = xx + dx;
     const int c = c0 + sslot * 8;
-    const __bf16* src;
-    if (sy >= 0 && sy < HH && sx >= 0 && sx < WW && c < Cin)
-      src = x + (((long)n * HH + sy) * WW + sx) * ld_x + c;
-    else
-      src = zpage;
+    const __bf16* src = zpage;
+    if (sy >= 0 && sy < HH && sx >= 0 && sx < WW && c < Cin) {
+      if (x2 == nullptr || c < C1)
+        src = x + (((long)n * HH + sy) * WW + sx) * ld_x + c;
+      else
+        src = x2 + (((long)n * HH + sy) * WW + sx) * ld_x2 + (c - C1);
+    }
     __builtin_amdgcn_global_load_lds(
         (const __attribute__((address_space(1))) void*)src,
         (__attribute__((address_space(3))) void*)(lds_buf + piece0 * 16),
@@ -98,12 +104,15 @@ __device__ __forceinline__ void cg_stage_b(const __bf16* __restrict__ wsec,
 template <int ACT>
 __global__ __launch_bounds__(CG_THREADS, 2) void conv_gemm_fwd_kernel(
     const __bf16* __restrict__ x,     // (Mtot, ld_x) NHWC rows
+    const __bf16* __restrict__ x2,    // second input source or nullptr
     const __bf16* __restrict__ wpk,   // (KYX, Cout, cpad)
     const float* __restrict__ bias,   // (Cout) or nullptr
-    __bf16* __restrict__ out,         // (Mtot, Cout)
+    __bf16* __restrict__ out,         // (Mtot, Cout) — or (Mtot, osplit)
+    __bf16* __restrict__ out2,        // (Mtot, Cout-osplit) or nullptr
     const __bf16* __restrict__ zpage,
-    long Mtot, int HH, int WW, int ld_x, int Cin, int Cout, int cpad,
-    int KH, int KW, int padH, int padW, int tiles_m) {
+    long Mtot, int HH, int WW, int ld_x, int ld_x2, int C1, int Cin,
+    int Cout, int cpad, int KH, int KW, int padH, int padW, int osplit,
+    int tiles_m) {
   __shared__ __attribute__((aligned(16))) char lds[2 * 2 * CG_BM * CG_BK * 2];
 
   const int tid = threadIdx.x;
@@ -129,14 +138,9 @@ __global__ __launch_bounds__(CG_THREADS, 2) void conv_gemm_fwd_kernel(
   const unsigned TS = CG_BM * CG_BK * 2;  // 8 KB
 
   // subtile s -> (kyx = s / cslabs, c0 = (s % cslabs) * 64)
-  {
-    const int kyx = 0, cs = 0;
-    const int dy = -padH, dx = -padW;
-    cg_stage_a(x, zpage, lds, wave, lane, m0, Mtot, HH, WW, ld_x, Cin, dy, dx,
-               cs * CG_BK);
-    cg_stage_b(wpk, lds + TS, wave, lane, n0, Cout, cpad, cs * CG_BK);
-    (void)kyx;
-  }
+  cg_stage_a(x, x2, zpage, lds, wave, lane, m0, Mtot, HH, WW, ld_x, ld_x2,
+             C1, Cin, -padH, -padW, 0);
+  cg_stage_b(wpk, lds + TS, wave, lane, n0, Cout, cpad, 0);
   asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
   __syncthreads();
 
@@ -146,8 +150,9 @@ __global__ __launch_bounds__(CG_THREADS, 2) void conv_gemm_fwd_kernel(
       const int kyx = (s + 1) / cslabs;
       const int cs = (s + 1) - kyx * cslabs;
       const int ky = kyx / KW, kx = kyx - ky * KW;
-      cg_stage_a(x, zpage, lds + (cur ^ 1) * 2 * TS, wave, lane, m0, Mtot, HH,
-                 WW, ld_x, Cin, ky - padH, kx - padW, cs * CG_BK);
+      cg_stage_a(x, x2, zpage, lds + (cur ^ 1) * 2 * TS, wave, lane, m0,
+                 Mtot, HH, WW, ld_x, ld_x2, C1, Cin, ky - padH, kx - padW,
+                 cs * CG_BK);
       cg_stage_b(wpk + (long)kyx * Cout * cpad, lds + (cur ^ 1) * 2 * TS + TS,
                  wave, lane, n0, Cout, cpad, cs * CG_BK);
     }
@@ -198,13 +203,22 @@ __global__ __launch_bounds__(CG_THREADS, 2) void conv_gemm_fwd_kernel(
       const int gco = n0 + wc + j * 16 + fcol;
       if (gco >= Cout) continue;
       const float b = bias ? bias[gco] : 0.f;
+      // split store: channels [0, osplit) -> out, rest -> out2 (the
+      // backward-data of a virtually-concatenated input)
+      __bf16* dst = out;
+      int co = gco, ldo = out2 ? osplit : Cout;
+      if (out2 != nullptr && gco >= osplit) {
+        dst = out2;
+        co = gco - osplit;
+        ldo = Cout - osplit;
+      }
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         const long gm = m0 + wr + i * 16 + frow0 + r;
         if (gm >= Mtot) continue;
         float v = acc[i][j][r] + b;
         if (ACT == 1) v = fmaxf(v, 0.f);
-        out[gm * Cout + gco] = (__bf16)v;
+        dst[gm * ldo + co] = (__bf16)v;
       }
     }
   }
@@ -220,11 +234,12 @@ __global__ __launch_bounds__(CG_THREADS, 2) void conv_gemm_fwd_kernel(
 __global__ __launch_bounds__(CG_THREADS, 2) void conv_gemm_wrw_kernel(
     const __bf16* __restrict__ dy,  // (Mtot, Cout)
     const __bf16* __restrict__ x,   // (Mtot, ld_x)
+    const __bf16* __restrict__ x2,  // second source or nullptr
     float* __restrict__ partials,
     const __bf16* __restrict__ zpage,
-    long Mtot, int HH, int WW, int ld_x, int Cin, int Cout, int cpad,
-    int KH, int KW, int padH, int padW, int tiles_o, int tiles_c,
-    int nchunk) {
+    long Mtot, int HH, int WW, int ld_x, int ld_x2, int C1, int Cin,
+    int Cout, int cpad, int KH, int KW, int padH, int padW, int tiles_o,
+    int tiles_c, int nchunk) {
   __shared__ __attribute__((aligned(16))) char lds[2 * 2 * CG_BM * CG_BK * 2];
 
   const int tid = threadIdx.x;
@@ -292,8 +307,12 @@ __global__ __launch_bounds__(CG_THREADS, 2) void conv_gemm_wrw_kernel(
         const long n = m / ((long)WW * HH);
         const int sy = yy + dyo, sx = xx + dxo;
         const int c = c0 + sslot * 8;
-        if (sy >= 0 && sy < HH && sx >= 0 && sx < WW && c < Cin)
-          src = x + (((long)n * HH + sy) * WW + sx) * ld_x + c;
+        if (sy >= 0 && sy < HH && sx >= 0 && sx < WW && c < Cin) {
+          if (x2 == nullptr || c < C1)
+            src = x + (((long)n * HH + sy) * WW + sx) * ld_x + c;
+          else
+            src = x2 + (((long)n * HH + sy) * WW + sx) * ld_x2 + (c - C1);
+        }
       }
       __builtin_amdgcn_global_load_lds(
           (const __attribute__((address_space(1))) void*)src,
@@ -410,40 +429,46 @@ __global__ __launch_bounds__(CG_THREADS) void conv_gemm_wrw_reduce_kernel(
 
 // ---------------------------------------------------------------------------
 
-void flowhip_conv_gemm_fwd_launch(const void* x, const void* wpk,
-                                  const float* bias, void* out,
-                                  const void* zpage, long Mtot, int HH,
-                                  int WW, int ld_x, int Cin, int Cout,
+void flowhip_conv_gemm_fwd_launch(const void* x, const void* x2,
+                                  const void* wpk, const float* bias,
+                                  void* out, void* out2, const void* zpage,
+                                  long Mtot, int HH, int WW, int ld_x,
+                                  int ld_x2, int C1, int Cin, int Cout,
                                   int cpad, int KH, int KW, int padH,
-                                  int padW, int act, hipStream_t stream) {
+                                  int padW, int osplit, int act,
+                                  hipStream_t stream) {
   const int tiles_m = (int)((Mtot + CG_BM - 1) / CG_BM);
   const int tiles_n = fh_cdiv(Cout, CG_BN);
   dim3 grid(tiles_m * tiles_n), block(CG_THREADS);
   if (act == 1)
     hipLaunchKernelGGL((conv_gemm_fwd_kernel<1>), grid, block, 0, stream,
-                       (const __bf16*)x, (const __bf16*)wpk, bias,
-                       (__bf16*)out, (const __bf16*)zpage, Mtot, HH, WW,
-                       ld_x, Cin, Cout, cpad, KH, KW, padH, padW, tiles_m);
+                       (const __bf16*)x, (const __bf16*)x2,
+                       (const __bf16*)wpk, bias, (__bf16*)out, (__bf16*)out2,
+                       (const __bf16*)zpage, Mtot, HH, WW, ld_x, ld_x2, C1,
+                       Cin, Cout, cpad, KH, KW, padH, padW, osplit, tiles_m);
   else
     hipLaunchKernelGGL((conv_gemm_fwd_kernel<0>), grid, block, 0, stream,
-                       (const __bf16*)x, (const __bf16*)wpk, bias,
-                       (__bf16*)out, (const __bf16*)zpage, Mtot, HH, WW,
-                       ld_x, Cin, Cout, cpad, KH, KW, padH, padW, tiles_m);
+                       (const __bf16*)x, (const __bf16*)x2,
+                       (const __bf16*)wpk, bias, (__bf16*)out, (__bf16*)out2,
+                       (const __bf16*)zpage, Mtot, HH, WW, ld_x, ld_x2, C1,
+                       Cin, Cout, cpad, KH, KW, padH, padW, osplit, tiles_m);
 }
 
 void flowhip_conv_gemm_wrw_launch(const void* dy, const void* x,
-                                  float* partials, float* dw,
+                                  const void* x2, float* partials, float* dw,
                                   const void* zpage, long Mtot, int HH,
-                                  int WW, int ld_x, int Cin, int Cout,
-                                  int cpad, int KH, int KW, int padH,
-                                  int padW, int nchunk, hipStream_t stream) {
+                                  int WW, int ld_x, int ld_x2, int C1,
+                                  int Cin, int Cout, int cpad, int KH,
+                                  int KW, int padH, int padW, int nchunk,
+                                  hipStream_t stream) {
   const int tiles_o = fh_cdiv(Cout, 64);
   const int tiles_c = fh_cdiv(cpad, 64);
   dim3 grid(tiles_o * tiles_c, KH * KW, nchunk), block(CG_THREADS);
   hipLaunchKernelGGL(conv_gemm_wrw_kernel, grid, block, 0, stream,
-                     (const __bf16*)dy, (const __bf16*)x, partials,
-                     (const __bf16*)zpage, Mtot, HH, WW, ld_x, Cin, Cout,
-                     cpad, KH, KW, padH, padW, tiles_o, tiles_c, nchunk);
+                     (const __bf16*)dy, (const __bf16*)x, (const __bf16*)x2,
+                     partials, (const __bf16*)zpage, Mtot, HH, WW, ld_x,
+                     ld_x2, C1, Cin, Cout, cpad, KH, KW, padH, padW, tiles_o,
+                     tiles_c, nchunk);
   const long total = (long)Cout * Cin * KH * KW;
   long rblocks = (total + CG_THREADS - 1) / CG_THREADS;
   if (rblocks > 4096) rblocks = 4096;

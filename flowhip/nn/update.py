@@ -15,7 +15,7 @@ import torch
 import torch.nn as nn
 import torch.nn.functional as F
 
-from ..ops.functional_conv import fused_conv2d
+from ..ops.functional_conv import fused_conv2d, fused_conv2d_cat2
 
 
 class FusedConv2d(nn.Conv2d):
@@ -85,20 +85,21 @@ class SepConvGRU(nn.Module):
         self.convr2 = FusedConv2d(hidden_dim + input_dim, hidden_dim, (5, 1), padding=(2, 0))
         self.convq2 = FusedConv2d(hidden_dim + input_dim, hidden_dim, (5, 1), padding=(2, 0))
 
-    def _pass(self, h, x, convz, convr, convq, padding, zr_cache):
-        hx = torch.cat([h, x], dim=1)
+    def _pass(self, h, x, convz, convr, convq, padding, zr_cache, q_cache):
         zr_w = torch.cat([convz.weight, convr.weight])
         zr_b = torch.cat([convz.bias, convr.bias])
-        # the packed-weight cache is keyed by the source params' versions
-        # (the cat tensor itself is fresh every call)
-        zr = fused_conv2d(hx, zr_w, zr_b, 1, padding, 1, 1, zr_cache,
-                          key=(convz.weight._version, convr.weight._version))
+        # virtually-concatenated input (no cat materialization on GPU); the
+        # packed-weight cache is keyed by the source params' versions
+        zr = fused_conv2d_cat2(h, x, zr_w, zr_b, padding, zr_cache,
+                               key=(convz.weight._version,
+                                    convr.weight._version))
         if h.is_cuda:
             # fused gate kernels (ops/functional_gru): one kernel for
             # sigmoid/chunk/r*h, one for tanh + lerp, fused backwards
             from ..ops.functional_gru import GruGate1Fn, GruGate2Fn
             z, rh = GruGate1Fn.apply(zr, h)
-            qp = convq(torch.cat([rh, x], dim=1))
+            qp = fused_conv2d_cat2(rh, x, convq.weight, convq.bias, padding,
+                                   q_cache, key=(convq.weight._version,))
             return GruGate2Fn.apply(qp, z, h)
         z, r = torch.sigmoid(zr).chunk(2, dim=1)
         q = torch.tanh(convq(torch.cat([r * h, x], dim=1)))
@@ -107,10 +108,11 @@ class SepConvGRU(nn.Module):
     def forward(self, h, x):
         if not hasattr(self, "_zr1_cache"):
             self._zr1_cache, self._zr2_cache = {}, {}
+            self._q1_cache, self._q2_cache = {}, {}
         h = self._pass(h, x, self.convz1, self.convr1, self.convq1, (0, 2),
-                       self._zr1_cache)  # horizontal
+                       self._zr1_cache, self._q1_cache)  # horizontal
         h = self._pass(h, x, self.convz2, self.convr2, self.convq2, (2, 0),
-                       self._zr2_cache)  # vertical
+                       self._zr2_cache, self._q2_cache)  # vertical
         return h
 
 

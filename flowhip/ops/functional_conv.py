@@ -66,7 +66,7 @@ class ConvGemmFn(torch.autograd.Function):
         if dy.dtype != torch.bfloat16:
             dy = dy.to(torch.bfloat16)
         dx = _ext.ext().conv_gemm_fwd(dy, wpk_bwd, None, I, KH, KW, 0)
-        dw = _ext.ext().conv_gemm_wrw(dy, x, KH, KW)
+        dw = _ext.ext().conv_gemm_wrw(dy, x, None, KH, KW)
         dbias = dy.float().sum(dim=(0, 2, 3)) if has_bias else None
         return dx, dw, dbias, None, None
 
@@ -99,3 +99,50 @@ def fused_conv2d(x, weight, bias, stride, padding, dilation, groups, cache,
         wf, wb = _packs(weight, cache, key)
         return ConvGemmFn.apply(x, weight, bias, wf, wb)
     return F.conv2d(x, weight, bias, stride, padding, dilation, groups)
+
+
+
+class ConvGemmCat2Fn(torch.autograd.Function):
+    """conv over a virtually-concatenated input pair cat([x1, x2], 1) —
+    removes the per-call cat materialization and its backward narrows
+    (4 per GRU iteration). x1's channel count must be a multiple of 64."""
+
+    @staticmethod
+    def forward(ctx, x1, x2, weight, bias, wpk_fwd, wpk_bwd):
+        O, I, KH, KW = weight.shape
+        b = bias.detach().float().contiguous() if bias is not None else None
+        out = _ext.ext().conv_gemm_fwd2(x1, x2, wpk_fwd, b, O, KH, KW, 0,
+                                        0)[0]
+        ctx.save_for_backward(x1, x2, wpk_bwd)
+        ctx.meta = (O, I, KH, KW, bias is not None, x1.shape[1])
+        return out
+
+    @staticmethod
+    def backward(ctx, dy):
+        x1, x2, wpk_bwd = ctx.saved_tensors
+        O, I, KH, KW, has_bias, C1 = ctx.meta
+        dy = dy.contiguous(memory_format=torch.channels_last)
+        if dy.dtype != torch.bfloat16:
+            dy = dy.to(torch.bfloat16)
+        dx1, dx2 = _ext.ext().conv_gemm_fwd2(dy, None, wpk_bwd, None, I, KH,
+                                             KW, C1, 0)
+        dw = _ext.ext().conv_gemm_wrw(dy, x1, x2, KH, KW)
+        dbias = dy.float().sum(dim=(0, 2, 3)) if has_bias else None
+        return dx1, dx2, dw, dbias, None, None
+
+
+def fused_conv2d_cat2(x1, x2, weight, bias, padding, cache, key):
+    """conv2d(cat([x1, x2], 1), weight) without materializing the cat.
+    Falls back to the eager cat + F.conv2d outside the fused envelope."""
+    O, I, KH, KW = weight.shape
+    fusable = (x1.is_cuda and x1.dtype == torch.bfloat16
+               and x2.dtype == torch.bfloat16
+               and x1.shape[1] % 64 == 0
+               and x1.stride(1) == 1 and x2.stride(1) == 1
+               and _ext.ext() is not None)
+    from torch.nn.modules.utils import _pair
+    if fusable and _pair(padding) == (KH // 2, KW // 2):
+        wf, wb = _packs(weight, cache, key)
+        return ConvGemmCat2Fn.apply(x1, x2, weight, bias, wf, wb)
+    hx = torch.cat([x1, x2], dim=1)
+    return F.conv2d(hx, weight, bias, padding=padding)

@@ -542,3 +542,37 @@ class TestConfPoolFused:
         (rdds * g1 + rcds * g2).sum().backward()
         torch.testing.assert_close(data.grad.cpu(), d2.grad)
         torch.testing.assert_close(conf.grad.cpu(), c2.grad)
+
+
+class TestConvGemmCat2:
+    def test_matches_cat_conv(self):
+        from flowhip.ops.functional_conv import fused_conv2d_cat2
+        torch.manual_seed(23)
+        B, H, W = 2, 20, 36
+        x1 = (torch.randn(B, 128, H, W, device=_dev()) / 8) \
+            .to(torch.bfloat16).contiguous(memory_format=torch.channels_last) \
+            .requires_grad_(True)
+        x2 = (torch.randn(B, 256, H, W, device=_dev()) / 8) \
+            .to(torch.bfloat16).contiguous(memory_format=torch.channels_last) \
+            .requires_grad_(True)
+        w = (torch.randn(256, 384, 1, 5, device=_dev()) / 44).requires_grad_(True)
+        b = torch.randn(256, device=_dev()).requires_grad_(True)
+        out = fused_conv2d_cat2(x1, x2, w, b, (0, 2), {}, key=(0,))
+
+        x1r = x1.detach().float().requires_grad_(True)
+        x2r = x2.detach().float().requires_grad_(True)
+        wr = w.detach().clone().requires_grad_(True)
+        br = b.detach().clone().requires_grad_(True)
+        ref = torch.nn.functional.conv2d(torch.cat([x1r, x2r], 1), wr, br,
+                                         padding=(0, 2))
+        tol = dict(atol=5e-2, rtol=5e-2)
+        torch.testing.assert_close(out.float(), ref, **tol)
+
+        g = torch.randn_like(ref).to(torch.bfloat16) \
+            .contiguous(memory_format=torch.channels_last)
+        out.backward(g)
+        ref.backward(g.float())
+        torch.testing.assert_close(x1.grad.float(), x1r.grad, **tol)
+        torch.testing.assert_close(x2.grad.float(), x2r.grad, **tol)
+        torch.testing.assert_close(w.grad.float(), wr.grad, atol=1e-1,
+                                   rtol=5e-2)
