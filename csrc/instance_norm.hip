@@ -21,12 +21,20 @@
 
 #include "common.h"
 
+#include <type_traits>
+
 #define IN_THREADS 256
 #define IN_CB 64      // channels per block column
-#define IN_PCHUNK 4096  // pixels per partial-reduce block
+#define IN_PCHUNK 1024  // pixels per partial-reduce block
 
 // mode 0: s1 = x, s2 = x*x          (fwd statistics)
 // mode 1: s1 = dy, s2 = dy*(x-mu)   (bwd reductions; a = dy)
+//
+// Thread layout (v3): each thread owns EIGHT consecutive channels and one
+// p-stream — per-lane 16-byte bf16x8 loads instead of one 2-byte element
+// (the v2 per-lane-per-channel form was load-issue-bound at 427 us for a
+// 44 M-element plane; ~40 us is the bandwidth bound). Requires C % 8 == 0
+// (all extractor widths); the module falls back to torch otherwise.
 template <typename T, int MODE>
 __global__ __launch_bounds__(IN_THREADS) void instnorm_partial_kernel(
     const T* __restrict__ x, const T* __restrict__ a,
@@ -37,44 +45,84 @@ __global__ __launch_bounds__(IN_THREADS) void instnorm_partial_kernel(
   const int chunk = b % nchunk; b /= nchunk;
   const int cb = b % ncb; b /= ncb;
   const int n = b;
-  const int c = cb * IN_CB + (threadIdx.x & 63);
-  const int wave = threadIdx.x >> 6;
-  const bool cv = c < C;
+  const int CBW = min(IN_CB, C - cb * IN_CB);  // channels in this block
+  const int CG = CBW / 8;                      // 8-channel groups (<= 8)
+  const int g = threadIdx.x % CG;              // this thread's group
+  const int s = threadIdx.x / CG;              // p-stream
+  const int S = IN_THREADS / CG;               // streams
+  const int c0 = cb * IN_CB + g * 8;
 
   const long p0 = (long)chunk * IN_PCHUNK;
   const long p1 = min(p0 + IN_PCHUNK, P);
-  const float mu =
-      (MODE == 1 && cv) ? mean[(long)n * C + c] : 0.f;
 
-  float s1 = 0.f, s2 = 0.f;
-  if (cv) {
+  float s1[8], s2[8], mu[8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) { s1[j] = 0.f; s2[j] = 0.f; mu[j] = 0.f; }
+  if (MODE == 1 && s < S) {
+#pragma unroll
+    for (int j = 0; j < 8; ++j) mu[j] = mean[(long)n * C + c0 + j];
+  }
+
+  if (s < S) {
     const T* xb = x + (long)n * P * C;
     const T* ab = (MODE == 1) ? a + (long)n * P * C : nullptr;
-    for (long p = p0 + wave; p < p1; p += 4) {
-      if (MODE == 0) {
-        const float v = (float)xb[p * C + c];
-        s1 += v;
-        s2 += v * v;
-      } else {
-        const float g = (float)ab[p * C + c];
-        s1 += g;
-        s2 += g * ((float)xb[p * C + c] - mu);
+    // one 16-B load per tensor (bf16x8 / f32x4 pair)
+    using Tv = typename std::conditional<sizeof(T) == 2, bf16x8, f32x4>::type;
+    constexpr int VW = sizeof(T) == 2 ? 8 : 4;
+    for (long p = p0 + s; p < p1; p += S) {
+      Tv xv0 = *(const Tv*)(xb + p * C + c0);
+      Tv xv1 = xv0;
+      if (VW == 4) xv1 = *(const Tv*)(xb + p * C + c0 + 4);
+      Tv av0 = xv0, av1 = xv1;
+      if (MODE == 1) {
+        av0 = *(const Tv*)(ab + p * C + c0);
+        if (VW == 4) av1 = *(const Tv*)(ab + p * C + c0 + 4);
+      }
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const float v = (float)(j < VW ? xv0[j % VW] : xv1[j % VW]);
+        if (MODE == 0) {
+          s1[j] += v;
+          s2[j] += v * v;
+        } else {
+          const float gv = (float)(j < VW ? av0[j % VW] : av1[j % VW]);
+          s1[j] += gv;
+          s2[j] += gv * (v - mu[j]);
+        }
       }
     }
   }
-  __shared__ float red1[4][IN_CB];
-  __shared__ float red2[4][IN_CB];
-  red1[wave][threadIdx.x & 63] = s1;
-  red2[wave][threadIdx.x & 63] = s2;
+
+  // LDS reduce over the p-streams: lds[2][64 channels][stride]
+  __shared__ float red[2 * IN_CB * 33];
+  const int stride = 33;  // odd stride: no bank conflicts across s
+  for (int j = 0; j < 8; ++j) {
+    const int c = g * 8 + j;
+    if (s < S && s < 32) {
+      red[(c)*stride + s] = s1[j];
+      red[(IN_CB + c) * stride + s] = s2[j];
+    }
+  }
   __syncthreads();
-  if (threadIdx.x < IN_CB) {
+  // streams beyond 32 fold in (S can be 42 for CG=6... cap: accumulate)
+  if (s >= 32 && s < S) {
+    for (int j = 0; j < 8; ++j) {
+      const int c = g * 8 + j;
+      atomicAdd(&red[(c)*stride + (s & 31)], s1[j]);
+      atomicAdd(&red[(IN_CB + c) * stride + (s & 31)], s2[j]);
+    }
+  }
+  __syncthreads();
+  if (threadIdx.x < (unsigned)CBW) {
+    float t1 = 0.f, t2 = 0.f;
+    const int smax = S < 32 ? S : 32;
+    for (int ss = 0; ss < smax; ++ss) {
+      t1 += red[threadIdx.x * stride + ss];
+      t2 += red[(IN_CB + threadIdx.x) * stride + ss];
+    }
     const long row = (((long)n * ncb + cb) * nchunk + chunk) * 2;
-    partials[(row + 0) * IN_CB + threadIdx.x] =
-        red1[0][threadIdx.x] + red1[1][threadIdx.x] + red1[2][threadIdx.x] +
-        red1[3][threadIdx.x];
-    partials[(row + 1) * IN_CB + threadIdx.x] =
-        red2[0][threadIdx.x] + red2[1][threadIdx.x] + red2[2][threadIdx.x] +
-        red2[3][threadIdx.x];
+    partials[(row + 0) * IN_CB + threadIdx.x] = t1;
+    partials[(row + 1) * IN_CB + threadIdx.x] = t2;
   }
 }
 
