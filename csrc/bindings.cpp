@@ -128,6 +128,11 @@ void flowhip_nconv_bwd_prep_launch(const float* gout, const float* gcout,
                                    float* dnomin, float* ddenom, long total,
                                    long plane, int Co, float eps,
                                    hipStream_t stream);
+void flowhip_area_up2x_fwd_launch(const float* in, float* out, long total,
+                                  int H, int W, hipStream_t stream);
+void flowhip_area_up2x_bwd_launch(const float* gout, float* gin,
+                                  long total_in, int H, int W,
+                                  hipStream_t stream);
 bool flowhip_corr_pyramid_fwd_launch(const float* corr, float* l1, float* l2,
                                      float* l3, int BP, int H0, int W0,
                                      int nlev, hipStream_t stream);
@@ -833,6 +838,31 @@ torch::Tensor conv_gemm_wrw(torch::Tensor dy, torch::Tensor x,
   return dw;
 }
 
+torch::Tensor area_up2x_fwd(torch::Tensor in) {
+  TORCH_CHECK(in.is_cuda() && in.is_contiguous() && in.dim() == 4 &&
+              in.dtype() == torch::kFloat32);
+  const long N = in.size(0), C = in.size(1);
+  const int H = in.size(2), W = in.size(3);
+  auto out = torch::empty({N, C, (long)2 * H, (long)2 * W}, in.options());
+  const c10::cuda::CUDAGuard guard(in.device());
+  hipStream_t stream = at::cuda::getCurrentCUDAStream().stream();
+  flowhip_area_up2x_fwd_launch(in.data_ptr<float>(), out.data_ptr<float>(),
+                               out.numel(), H, W, stream);
+  return out;
+}
+
+torch::Tensor area_up2x_bwd(torch::Tensor gout) {
+  auto g = gout.contiguous();
+  const long N = g.size(0), C = g.size(1);
+  const int H = g.size(2) / 2, W = g.size(3) / 2;
+  auto gin = torch::empty({N, C, (long)H, (long)W}, g.options());
+  const c10::cuda::CUDAGuard guard(g.device());
+  hipStream_t stream = at::cuda::getCurrentCUDAStream().stream();
+  flowhip_area_up2x_bwd_launch(g.data_ptr<float>(), gin.data_ptr<float>(),
+                               gin.numel(), H, W, stream);
+  return gin;
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -851,6 +881,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("convex_up_bwd", &convex_up_bwd, "backward of convex_up_fwd");
   m.def("nconv_fwd", &nconv_fwd,
         "fused normalized convolution forward (out, cout)");
+  m.def("area_up2x_fwd", &area_up2x_fwd, "exact-2x area upsample");
+  m.def("area_up2x_bwd", &area_up2x_bwd, "backward of area_up2x");
   m.def("conf_pool_fwd", &conf_pool_fwd,
         "confidence-based 2x pooling forward (data_ds, conf_ds, argmax code)");
   m.def("conf_pool_bwd", &conf_pool_bwd, "backward of conf_pool");

@@ -167,7 +167,7 @@ class NConvUpsampler(nn.Module):
         if self.est_on_high_res:
             x_data_for_guidance = x_highres
         else:
-            x_guidance = F.interpolate(x_guidance, x_lowres.size()[2:], mode="area")
+            x_guidance = ops.area_resize(x_guidance, x_lowres.size()[2:])
             x_data_for_guidance = x_lowres
 
         if self.use_data_for_guidance:

@@ -84,6 +84,18 @@ def conf_pool(data, conf, ds_factor=2, pooling_type="conf_based"):
     return torch_ref.conf_pool(data, conf, ds_factor, pooling_type)
 
 
+def area_resize(x, size):
+    """F.interpolate(x, size, mode='area') with a fused kernel for the
+    exact-2x-upsample case (the per-iteration guidance resize)."""
+    import torch.nn.functional as F
+    ih, iw = x.shape[-2:]
+    if (_ext.use_hip(x) and x.dtype == torch.float32
+            and size[0] == 2 * ih and size[1] == 2 * iw):
+        from .functional_upsample import AreaUp2xFn
+        return AreaUp2xFn.apply(x)
+    return F.interpolate(x, size, mode="area")
+
+
 def zero_inject(inp, scale_h, scale_w, out_h=None, out_w=None):
     if _ext.use_hip(inp) and inp.dtype == torch.float32:
         from .functional_upsample import ZeroInjectFn

@@ -67,7 +67,7 @@ class ConvGemmFn(torch.autograd.Function):
             dy = dy.to(torch.bfloat16)
         dx = _ext.ext().conv_gemm_fwd(dy, wpk_bwd, None, I, KH, KW, 0)
         dw = _ext.ext().conv_gemm_wrw(dy, x, None, KH, KW)
-        dbias = dy.float().sum(dim=(0, 2, 3)) if has_bias else None
+        dbias = dy.sum(dim=(0, 2, 3), dtype=torch.float32) if has_bias else None
         return dx, dw, dbias, None, None
 
 
@@ -127,7 +127,7 @@ class ConvGemmCat2Fn(torch.autograd.Function):
         dx1, dx2 = _ext.ext().conv_gemm_fwd2(dy, None, wpk_bwd, None, I, KH,
                                              KW, C1, 0)
         dw = _ext.ext().conv_gemm_wrw(dy, x1, x2, KH, KW)
-        dbias = dy.float().sum(dim=(0, 2, 3)) if has_bias else None
+        dbias = dy.sum(dim=(0, 2, 3), dtype=torch.float32) if has_bias else None
         return dx1, dx2, dw, dbias, None, None
 
 

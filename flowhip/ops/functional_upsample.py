@@ -38,3 +38,17 @@ class ZeroInjectFn(torch.autograd.Function):
     def backward(ctx, gout):
         sH, sW, ih, iw = ctx.meta
         return _ext.ext().zero_inject_bwd(gout, sH, sW, ih, iw), None, None, None, None
+
+
+
+class AreaUp2xFn(torch.autograd.Function):
+    """Exact-2x area interpolation (== nearest duplication) with a
+    gather-only backward (torch's is an atomic adaptive-pool scatter)."""
+
+    @staticmethod
+    def forward(ctx, x):
+        return _ext.ext().area_up2x_fwd(x.contiguous())
+
+    @staticmethod
+    def backward(ctx, gout):
+        return _ext.ext().area_up2x_bwd(gout)

@@ -576,3 +576,18 @@ class TestConvGemmCat2:
         torch.testing.assert_close(x2.grad.float(), x2r.grad, **tol)
         torch.testing.assert_close(w.grad.float(), wr.grad, atol=1e-1,
                                    rtol=5e-2)
+
+
+class TestAreaUp2x:
+    def test_matches_interpolate(self):
+        from flowhip import ops
+        torch.manual_seed(25)
+        x = torch.randn(2, 8, 14, 22, device=_dev(), requires_grad=True)
+        out = ops.area_resize(x, (28, 44))
+        x2 = x.detach().cpu().requires_grad_(True)
+        ref = torch.nn.functional.interpolate(x2, (28, 44), mode="area")
+        torch.testing.assert_close(out.cpu(), ref)
+        g = torch.randn_like(ref)
+        (out * g.to(_dev())).sum().backward()
+        (ref * g).sum().backward()
+        torch.testing.assert_close(x.grad.cpu(), x2.grad)
