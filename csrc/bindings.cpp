@@ -821,7 +821,12 @@ torch::Tensor conv_gemm_wrw(torch::Tensor dy, torch::Tensor x,
   const long Mtot = (long)N * H * W;
   const int cpad = (int)((Cin + 63) / 64) * 64;
   const int tiles_o = (Cout + 63) / 64;
-  const int nchunk = 8;
+  // split M only as much as needed to fill the chip (~2 blocks/CU);
+  // more chunks = more fp32 partial traffic + a longer reduce
+  const int base_tiles = tiles_o * ((cpad + 63) / 64) * (int)(KH * KW);
+  int nchunk = (512 + base_tiles - 1) / base_tiles;
+  if (nchunk < 1) nchunk = 1;
+  if (nchunk > 8) nchunk = 8;
   auto partials = torch::empty(
       {(long)nchunk * KH * KW * tiles_o * 64 * cpad},
       x.options().dtype(torch::kFloat32));

@@ -144,8 +144,10 @@ __global__ __launch_bounds__(LK_THREADS) void corr_lookup_bwd_kernel(
           float v = 0.0f;
           if (j < K) v += wy0 * tx[j];
           if (j > 0) v += wy1 * tx[j - 1];
-          scalar_t* p = gmap + (long)yy * Wl + xx;
-          *p = (scalar_t)((float)*p + v);
+          // each patch position of this thread's PRIVATE map is written
+          // exactly once per call (buffer is fresh-zeroed): plain store,
+          // no read-modify-write
+          gmap[(long)yy * Wl + xx] = (scalar_t)v;
         }
       }
     }
