@@ -13,6 +13,8 @@ import torch.nn as nn
 import torch.nn.functional as F
 from torch.nn.modules.utils import _pair
 
+from .update import FusedConv2d
+
 
 class Simple(nn.Module):
     """Plain conv stack: num_ch[0] -> ... -> num_ch[-1] -> out_ch, sigmoid.
@@ -40,17 +42,17 @@ class Simple(nn.Module):
 
         self.conv = nn.ModuleList()
         for i in range(self.num_layers):
-            layers = [nn.Conv2d(num_ch[i], num_ch[i + 1], filter_sz[i],
-                                padding=pad_for(filter_sz[i], dilation[i]),
-                                dilation=dilation[i], stride=1)]
+            layers = [FusedConv2d(num_ch[i], num_ch[i + 1], filter_sz[i],
+                                  padding=pad_for(filter_sz[i], dilation[i]),
+                                  dilation=dilation[i], stride=1)]
             if use_bn:
                 layers.append(nn.BatchNorm2d(num_ch[i + 1]))
             layers.append(nn.ReLU(inplace=True))
             self.conv.append(nn.Sequential(*layers))
 
-        self.out = nn.Conv2d(num_ch[-1], out_ch, filter_sz[-1],
-                             padding=pad_for(filter_sz[-1], dilation[-1]),
-                             dilation=dilation[-1], stride=1)
+        self.out = FusedConv2d(num_ch[-1], out_ch, filter_sz[-1],
+                               padding=pad_for(filter_sz[-1], dilation[-1]),
+                               dilation=dilation[-1], stride=1)
 
         self.final_act = nn.Sequential() if final_act is None else final_act
 

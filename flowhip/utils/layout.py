@@ -43,6 +43,11 @@ def apply_channels_last(model):
         ups = getattr(model, "upsampler", None)
         if ups is not None:
             ups.to(memory_format=torch.contiguous_format)
+            # ...except the confidence net, which runs bf16/NHWC on the
+            # MFMA conv kernel (see NConvUpsampler.forward)
+            wen = getattr(ups, "weights_est_net", None)
+            if isinstance(wen, torch.nn.Module):
+                wen.to(memory_format=torch.channels_last)
     return model
 
 

@@ -591,3 +591,21 @@ class TestAreaUp2x:
         (out * g.to(_dev())).sum().backward()
         (ref * g).sum().backward()
         torch.testing.assert_close(x.grad.cpu(), x2.grad)
+
+
+class TestWeightsEstBf16:
+    def test_conf_close_to_fp32_oracle(self):
+        """The confidence net runs bf16/NHWC on GPU (documented deviation —
+        the reference ran it fp32 only because its autocast region ended
+        before the upsampler). Bound the deviation of the sigmoid output."""
+        from flowhip.nn.interp_weights_est import Simple
+        torch.manual_seed(27)
+        net = Simple(num_ch=[130, 64, 32], out_ch=2, filter_sz=[3, 3, 1]) \
+            .to(_dev()).to(memory_format=torch.channels_last).eval()
+        x = torch.randn(2, 130, 28, 64, device=_dev())
+        with torch.no_grad():
+            out_bf = net(x.to(torch.bfloat16)
+                         .contiguous(memory_format=torch.channels_last))
+            out_fp = net(x)
+        err = (out_bf.float() - out_fp).abs().max().item()
+        assert err < 0.03, err  # sigmoid-domain absolute deviation
