@@ -128,6 +128,8 @@ void flowhip_nconv_bwd_prep_launch(const float* gout, const float* gcout,
                                    float* dnomin, float* ddenom, long total,
                                    long plane, int Co, float eps,
                                    hipStream_t stream);
+bool flowhip_col_sum_launch(const void* dy, float* partials, float* out,
+                            long M, int C, int nchunk, hipStream_t stream);
 void flowhip_area_up2x_fwd_launch(const float* in, float* out, long total,
                                   int H, int W, hipStream_t stream);
 void flowhip_area_up2x_bwd_launch(const float* gout, float* gin,
@@ -843,6 +845,27 @@ torch::Tensor conv_gemm_wrw(torch::Tensor dy, torch::Tensor x,
   return dw;
 }
 
+torch::Tensor col_sum_bf16(torch::Tensor dy) {
+  TORCH_CHECK(dy.is_cuda() && dy.dim() == 4 &&
+              dy.scalar_type() == torch::kBFloat16 && dy.stride(1) == 1 &&
+              dy.stride(3) == dy.size(1));
+  const int C = dy.size(1);
+  const long M = dy.numel() / C;
+  const int nchunk = (int)std::min<long>((M + 8191) / 8192, 64);
+  const int ncb = (C + 63) / 64;
+  auto partials = torch::empty({(long)nchunk * ncb * 64},
+                               dy.options().dtype(torch::kFloat32));
+  auto out = torch::empty({(long)C}, dy.options().dtype(torch::kFloat32));
+  const c10::cuda::CUDAGuard guard(dy.device());
+  hipStream_t stream = at::cuda::getCurrentCUDAStream().stream();
+  bool ok = flowhip_col_sum_launch(dy.data_ptr(),
+                                   partials.data_ptr<float>(),
+                                   out.data_ptr<float>(), M, C, nchunk,
+                                   stream);
+  TORCH_CHECK(ok, "col_sum_bf16: C must be a multiple of 8");
+  return out;
+}
+
 torch::Tensor area_up2x_fwd(torch::Tensor in) {
   TORCH_CHECK(in.is_cuda() && in.is_contiguous() && in.dim() == 4 &&
               in.dtype() == torch::kFloat32);
@@ -886,6 +909,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("convex_up_bwd", &convex_up_bwd, "backward of convex_up_fwd");
   m.def("nconv_fwd", &nconv_fwd,
         "fused normalized convolution forward (out, cout)");
+  m.def("col_sum_bf16", &col_sum_bf16,
+        "(N,C,H,W) channels-last bf16 -> (C) fp32 bias-grad column sum");
   m.def("area_up2x_fwd", &area_up2x_fwd, "exact-2x area upsample");
   m.def("area_up2x_bwd", &area_up2x_bwd, "backward of area_up2x");
   m.def("conf_pool_fwd", &conf_pool_fwd,

@@ -1,0 +1,78 @@
+// Column sum for conv bias gradients: (M, C) bf16 channels-last rows ->
+// (C) fp32, dbias[c] = sum_m dy[m, c]. torch's strided reduce over the
+// pixel dims of a channels-last tensor runs ~22 us for 5.5 MB (~100 calls
+// per step); this is a chunked 16-B-load partial reduction + tiny finalize.
+
+#include "common.h"
+
+#define CS_THREADS 256
+#define CS_MCHUNK 8192
+
+__global__ __launch_bounds__(CS_THREADS) void col_sum_partial_kernel(
+    const __bf16* __restrict__ dy, float* __restrict__ partials, long M,
+    int C, int nchunk) {
+  const int ncb = (C + 63) / 64;
+  int b = blockIdx.x;
+  const int chunk = b % nchunk; b /= nchunk;
+  const int cb = b;
+  const int CBW = min(64, C - cb * 64);
+  const int CG = CBW / 8;            // 8-channel groups
+  const int g = threadIdx.x % CG;
+  const int s = threadIdx.x / CG;
+  const int S = CS_THREADS / CG;
+  const int c0 = cb * 64 + g * 8;
+
+  const long m0 = (long)chunk * CS_MCHUNK;
+  const long m1 = min(m0 + CS_MCHUNK, M);
+
+  float acc[8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) acc[j] = 0.f;
+  if (s < S) {
+    for (long m = m0 + s; m < m1; m += S) {
+      const bf16x8 v = *(const bf16x8*)(dy + m * C + c0);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) acc[j] += (float)v[j];
+    }
+  }
+
+  __shared__ float red[64 * 33];
+  for (int j = 0; j < 8; ++j)
+    if (s < S && s < 32) red[(g * 8 + j) * 33 + s] = acc[j];
+  __syncthreads();
+  if (s >= 32 && s < S)
+    for (int j = 0; j < 8; ++j)
+      atomicAdd(&red[(g * 8 + j) * 33 + (s & 31)], acc[j]);
+  __syncthreads();
+  if (threadIdx.x < (unsigned)CBW) {
+    float t = 0.f;
+    const int smax = S < 32 ? S : 32;
+    for (int ss = 0; ss < smax; ++ss) t += red[threadIdx.x * 33 + ss];
+    partials[((long)chunk * ncb + cb) * 64 + threadIdx.x] = t;
+  }
+}
+
+__global__ __launch_bounds__(64) void col_sum_finalize_kernel(
+    const float* __restrict__ partials, float* __restrict__ out, int C,
+    int nchunk) {
+  const int ncb = (C + 63) / 64;
+  const int cb = blockIdx.x;
+  const int c = cb * 64 + threadIdx.x;
+  if (c >= C) return;
+  float t = 0.f;
+  for (int ch = 0; ch < nchunk; ++ch)
+    t += partials[((long)ch * ncb + cb) * 64 + threadIdx.x];
+  out[c] = t;
+}
+
+bool flowhip_col_sum_launch(const void* dy, float* partials, float* out,
+                            long M, int C, int nchunk, hipStream_t stream) {
+  if (C % 8 != 0 || C < 8) return false;
+  const int ncb = (C + 63) / 64;
+  hipLaunchKernelGGL(col_sum_partial_kernel, dim3(ncb * nchunk),
+                     dim3(CS_THREADS), 0, stream, (const __bf16*)dy,
+                     partials, M, C, nchunk);
+  hipLaunchKernelGGL(col_sum_finalize_kernel, dim3(ncb), dim3(64), 0, stream,
+                     partials, out, C, nchunk);
+  return true;
+}

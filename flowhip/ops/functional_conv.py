@@ -22,6 +22,14 @@ def _pad64(n):
     return (n + 63) // 64 * 64
 
 
+def _col_sum(dy):
+    """bias grad: one chunked column-sum kernel (torch's strided reduce on a
+    channels-last tensor is ~4x slower)."""
+    if dy.shape[1] % 8 == 0:
+        return _ext.ext().col_sum_bf16(dy)
+    return dy.sum(dim=(0, 2, 3), dtype=torch.float32)
+
+
 def _pack_fwd(weight):
     """(O, I, KH, KW) fp32 -> (KYX, O, pad64(I)) bf16 contiguous."""
     O, I, KH, KW = weight.shape
@@ -67,7 +75,7 @@ class ConvGemmFn(torch.autograd.Function):
             dy = dy.to(torch.bfloat16)
         dx = _ext.ext().conv_gemm_fwd(dy, wpk_bwd, None, I, KH, KW, 0)
         dw = _ext.ext().conv_gemm_wrw(dy, x, None, KH, KW)
-        dbias = dy.sum(dim=(0, 2, 3), dtype=torch.float32) if has_bias else None
+        dbias = _col_sum(dy) if has_bias else None
         return dx, dw, dbias, None, None
 
 
@@ -127,7 +135,7 @@ class ConvGemmCat2Fn(torch.autograd.Function):
         dx1, dx2 = _ext.ext().conv_gemm_fwd2(dy, None, wpk_bwd, None, I, KH,
                                              KW, C1, 0)
         dw = _ext.ext().conv_gemm_wrw(dy, x1, x2, KH, KW)
-        dbias = dy.sum(dim=(0, 2, 3), dtype=torch.float32) if has_bias else None
+        dbias = _col_sum(dy) if has_bias else None
         return dx1, dx2, dw, dbias, None, None
 
 
