@@ -14,6 +14,8 @@ slots in underneath without touching this module.
 import torch
 import torch.nn as nn
 
+from .update import FusedConv2d
+
 
 def _norm(norm_fn, planes, groups_planes=None):
     """Build one normalization module. `group` uses planes//8 groups unless an
@@ -48,8 +50,8 @@ class ResidualBlock(nn.Module):
 
     def __init__(self, in_planes, planes, norm_fn="group", stride=1):
         super().__init__()
-        self.conv1 = nn.Conv2d(in_planes, planes, kernel_size=3, padding=1, stride=stride)
-        self.conv2 = nn.Conv2d(planes, planes, kernel_size=3, padding=1)
+        self.conv1 = FusedConv2d(in_planes, planes, kernel_size=3, padding=1, stride=stride)
+        self.conv2 = FusedConv2d(planes, planes, kernel_size=3, padding=1)
         self.relu = nn.ReLU(inplace=True)
 
         self.norm1 = _norm(norm_fn, planes)
@@ -78,9 +80,9 @@ class BottleneckBlock(nn.Module):
 
     def __init__(self, in_planes, planes, norm_fn="group", stride=1):
         super().__init__()
-        self.conv1 = nn.Conv2d(in_planes, planes // 4, kernel_size=1, padding=0)
-        self.conv2 = nn.Conv2d(planes // 4, planes // 4, kernel_size=3, padding=1, stride=stride)
-        self.conv3 = nn.Conv2d(planes // 4, planes, kernel_size=1, padding=0)
+        self.conv1 = FusedConv2d(in_planes, planes // 4, kernel_size=1, padding=0)
+        self.conv2 = FusedConv2d(planes // 4, planes // 4, kernel_size=3, padding=1, stride=stride)
+        self.conv3 = FusedConv2d(planes // 4, planes, kernel_size=1, padding=0)
         self.relu = nn.ReLU(inplace=True)
 
         ng = planes // 8
@@ -124,7 +126,7 @@ class _Encoder(nn.Module):
         self.layer2 = self._make_layer(self.stage_planes[1], stride=2)
         self.layer3 = self._make_layer(self.stage_planes[2], stride=2)
 
-        self.conv2 = nn.Conv2d(self.stage_planes[2], output_dim, kernel_size=1)
+        self.conv2 = FusedConv2d(self.stage_planes[2], output_dim, kernel_size=1)
 
         self.dropout = nn.Dropout2d(p=dropout) if dropout > 0 else None
         _kaiming_init(self)
