@@ -828,7 +828,10 @@ torch::Tensor conv_gemm_wrw(torch::Tensor dy, torch::Tensor x,
   const int base_tiles = tiles_o * ((cpad + 63) / 64) * (int)(KH * KW);
   int nchunk = (512 + base_tiles - 1) / base_tiles;
   if (nchunk < 1) nchunk = 1;
-  if (nchunk > 8) nchunk = 8;
+  if (nchunk > 64) nchunk = 64;  // encoder shapes: few tiles, huge M
+  // never more chunks than 64-row m-tiles
+  const long mtiles = (Mtot + 63) / 64;
+  if (nchunk > mtiles) nchunk = (int)mtiles;
   auto partials = torch::empty(
       {(long)nchunk * KH * KW * tiles_o * 64 * cpad},
       x.options().dtype(torch::kFloat32));

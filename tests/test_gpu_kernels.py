@@ -609,3 +609,44 @@ class TestWeightsEstBf16:
             out_fp = net(x)
         err = (out_bf.float() - out_fp).abs().max().item()
         assert err < 0.03, err  # sigmoid-domain absolute deviation
+
+
+class TestTrainingLearns:
+    def test_loss_decreases_on_synthetic(self):
+        """End-to-end on GPU: 25 optimizer steps on a fixed synthetic pair
+        reduce the sequence loss substantially (all HIP kernels in the
+        loop: corr GEMM/pyramid/lookup, conv_gemm, GRU gates, nconv,
+        injection, loss)."""
+        from flowhip.config.args import default_ncup_args
+        from flowhip.models import build_model
+        from flowhip import ops
+
+        torch.manual_seed(1234)
+        args = default_ncup_args(model="raft_nc_dbl", small=False,
+                                 mixed_precision=True, dataset="sintel")
+        model = build_model(args).to(_dev())
+        model.train()
+        opt = torch.optim.AdamW(model.parameters(), lr=2e-4)
+
+        h, w = 128, 256
+        g = torch.Generator().manual_seed(7)
+        img1 = (torch.rand(2, 3, h, w, generator=g) * 255).to(_dev())
+        img2 = (torch.rand(2, 3, h, w, generator=g) * 255).to(_dev())
+        flow_gt = torch.nn.functional.interpolate(
+            torch.randn(2, 2, 8, 16, generator=g) * 4, size=(h, w),
+            mode="bilinear", align_corners=False).to(_dev())
+        valid = torch.ones(2, h, w, device=_dev())
+
+        losses = []
+        for _ in range(25):
+            opt.zero_grad(set_to_none=True)
+            preds = model(img1, img2, iters=4)
+            loss, _ = ops.sequence_loss(preds, flow_gt, valid, 0.85)
+            loss.backward()
+            torch.nn.utils.clip_grad_norm_(model.parameters(), 1.0)
+            opt.step()
+            losses.append(loss.item())
+        assert all(torch.isfinite(torch.tensor(losses)))
+        first = sum(losses[:5]) / 5
+        last = sum(losses[-5:]) / 5
+        assert last < 0.7 * first, (first, last)
