@@ -22,6 +22,23 @@ def _pad64(n):
     return (n + 63) // 64 * 64
 
 
+def _pad_c8(x, weight):
+    """Channel-pad input AND weight to a multiple of 8 via differentiable
+    cats. The staging kernel reads whole 16-B pieces; with Cin % 8 != 0 the
+    tail piece of the last pixel would read past the tensor allocation
+    (fault, allocator-layout dependent). Zero input channels x zero weight
+    columns = identical math, all reads in-bounds."""
+    O, I, KH, KW = weight.shape
+    pad = (-I) % 8
+    if pad == 0:
+        return x, weight
+    z = x.new_zeros(x.shape[0], pad, x.shape[2], x.shape[3])         .contiguous(memory_format=torch.channels_last)
+    xp = torch.cat([x, z], dim=1)
+    wz = weight.new_zeros(O, pad, KH, KW)
+    wp = torch.cat([weight, wz], dim=1)
+    return xp, wp
+
+
 def _col_sum(dy):
     """bias grad: one chunked column-sum kernel (torch's strided reduce on a
     channels-last tensor is ~4x slower)."""
@@ -94,9 +111,13 @@ def can_fuse_conv(x, weight, stride, padding, dilation, groups):
         return False
     if I != x.shape[1] or O > 512 or KH * KW > 49:
         return False
-    # channel-narrowed views are fine; anything else must be cl-contiguous
     if x.stride(3) < x.size(1):
         return False
+    if I % 8 != 0 and x.stride(3) != x.size(1):
+        # narrowed view with ragged Cin: the 16-B tail piece must stay
+        # inside the row (the padded-allocation case has ld % 8 == 0)
+        if x.stride(3) % 8 != 0 or x.stride(3) < I + ((-I) % 8):
+            return False
     return True
 
 
@@ -104,6 +125,10 @@ def fused_conv2d(x, weight, bias, stride, padding, dilation, groups, cache,
                  key=None):
     """F.conv2d drop-in that routes supported shapes to the MFMA kernel."""
     if can_fuse_conv(x, weight, stride, padding, dilation, groups):
+        if key is None:
+            key = (weight._version, weight.data_ptr())
+        if weight.shape[1] % 8 != 0:
+            x, weight = _pad_c8(x, weight)
         wf, wb = _packs(weight, cache, key)
         return ConvGemmFn.apply(x, weight, bias, wf, wb)
     return F.conv2d(x, weight, bias, stride, padding, dilation, groups)
