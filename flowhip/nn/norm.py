@@ -33,6 +33,7 @@ class InstanceNorm2d(nn.InstanceNorm2d):
         if (x.is_cuda and not self.affine and not self.track_running_stats
                 and x.is_contiguous(memory_format=torch.channels_last)
                 and x.dtype in (torch.float32, torch.bfloat16)
+                and x.shape[1] % 8 == 0 and x.shape[1] >= 8
                 and _ext.ext() is not None):
             return _InstNormCLFn.apply(x, self.eps)
         return super().forward(x)

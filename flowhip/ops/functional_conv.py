@@ -39,6 +39,18 @@ def _pad_c8(x, weight):
     return xp, wp
 
 
+def _pad_dy8(dy):
+    """Channel-pad a backward gradient to a multiple of 8 (same OOB-tail
+    argument as _pad_c8; the packed-bwd weight k-columns beyond Cout are
+    zero, so padded dy channels contribute nothing)."""
+    O = dy.shape[1]
+    pad = (-O) % 8
+    if pad == 0:
+        return dy, O
+    z = dy.new_zeros(dy.shape[0], pad, dy.shape[2], dy.shape[3])         .contiguous(memory_format=torch.channels_last)
+    return torch.cat([dy, z], dim=1), O
+
+
 def _col_sum(dy):
     """bias grad: one chunked column-sum kernel (torch's strided reduce on a
     channels-last tensor is ~4x slower)."""
@@ -90,8 +102,11 @@ class ConvGemmFn(torch.autograd.Function):
         dy = dy.contiguous(memory_format=torch.channels_last)
         if dy.dtype != torch.bfloat16:
             dy = dy.to(torch.bfloat16)
-        dx = _ext.ext().conv_gemm_fwd(dy, wpk_bwd, None, I, KH, KW, 0)
-        dw = _ext.ext().conv_gemm_wrw(dy, x, None, KH, KW)
+        dyp, O_real = _pad_dy8(dy)
+        dx = _ext.ext().conv_gemm_fwd(dyp, wpk_bwd, None, I, KH, KW, 0)
+        dw = _ext.ext().conv_gemm_wrw(dyp, x, None, KH, KW)
+        if dw.shape[0] != O_real:
+            dw = dw[:O_real].contiguous()
         dbias = _col_sum(dy) if has_bias else None
         return dx, dw, dbias, None, None
 
@@ -157,9 +172,12 @@ class ConvGemmCat2Fn(torch.autograd.Function):
         dy = dy.contiguous(memory_format=torch.channels_last)
         if dy.dtype != torch.bfloat16:
             dy = dy.to(torch.bfloat16)
-        dx1, dx2 = _ext.ext().conv_gemm_fwd2(dy, None, wpk_bwd, None, I, KH,
+        dyp, O_real = _pad_dy8(dy)
+        dx1, dx2 = _ext.ext().conv_gemm_fwd2(dyp, None, wpk_bwd, None, I, KH,
                                              KW, C1, 0)
-        dw = _ext.ext().conv_gemm_wrw(dy, x1, x2, KH, KW)
+        dw = _ext.ext().conv_gemm_wrw(dyp, x1, x2, KH, KW)
+        if dw.shape[0] != O_real:
+            dw = dw[:O_real].contiguous()
         dbias = _col_sum(dy) if has_bias else None
         return dx1, dx2, dw, dbias, None, None
 
