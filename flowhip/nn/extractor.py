@@ -11,10 +11,21 @@ GPU; per-shape fused HIP conv kernels are an ops-level optimization that
 slots in underneath without touching this module.
 """
 
+import os
+
 import torch
 import torch.nn as nn
 
 from .update import FusedConv2d
+
+
+def _enc_conv(*a, **k):
+    """Encoder convs default to the fused MFMA kernel; FLOWHIP_FUSED_ENCODER=0
+    keeps them on MIOpen (A/B lever — the encoder runs once per step and its
+    huge-M shapes are closer to library territory than the update block's)."""
+    if os.environ.get("FLOWHIP_FUSED_ENCODER", "1") == "0":
+        return nn.Conv2d(*a, **k)
+    return FusedConv2d(*a, **k)
 
 
 def _norm(norm_fn, planes, groups_planes=None):
@@ -50,8 +61,8 @@ class ResidualBlock(nn.Module):
 
     def __init__(self, in_planes, planes, norm_fn="group", stride=1):
         super().__init__()
-        self.conv1 = FusedConv2d(in_planes, planes, kernel_size=3, padding=1, stride=stride)
-        self.conv2 = FusedConv2d(planes, planes, kernel_size=3, padding=1)
+        self.conv1 = _enc_conv(in_planes, planes, kernel_size=3, padding=1, stride=stride)
+        self.conv2 = _enc_conv(planes, planes, kernel_size=3, padding=1)
         self.relu = nn.ReLU(inplace=True)
 
         self.norm1 = _norm(norm_fn, planes)
@@ -80,9 +91,9 @@ class BottleneckBlock(nn.Module):
 
     def __init__(self, in_planes, planes, norm_fn="group", stride=1):
         super().__init__()
-        self.conv1 = FusedConv2d(in_planes, planes // 4, kernel_size=1, padding=0)
-        self.conv2 = FusedConv2d(planes // 4, planes // 4, kernel_size=3, padding=1, stride=stride)
-        self.conv3 = FusedConv2d(planes // 4, planes, kernel_size=1, padding=0)
+        self.conv1 = _enc_conv(in_planes, planes // 4, kernel_size=1, padding=0)
+        self.conv2 = _enc_conv(planes // 4, planes // 4, kernel_size=3, padding=1, stride=stride)
+        self.conv3 = _enc_conv(planes // 4, planes, kernel_size=1, padding=0)
         self.relu = nn.ReLU(inplace=True)
 
         ng = planes // 8
@@ -126,7 +137,7 @@ class _Encoder(nn.Module):
         self.layer2 = self._make_layer(self.stage_planes[1], stride=2)
         self.layer3 = self._make_layer(self.stage_planes[2], stride=2)
 
-        self.conv2 = FusedConv2d(self.stage_planes[2], output_dim, kernel_size=1)
+        self.conv2 = _enc_conv(self.stage_planes[2], output_dim, kernel_size=1)
 
         self.dropout = nn.Dropout2d(p=dropout) if dropout > 0 else None
         _kaiming_init(self)
