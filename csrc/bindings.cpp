@@ -14,7 +14,7 @@ void flowhip_bgemm_nt_launch(const void* A, const void* B, void* C,
                              hipStream_t stream);
 void flowhip_corr_lookup_fwd_launch(const float* level, const float* coords,
                                     float* out, int BP, int P, int Hl, int Wl,
-                                    int l, int L, int radius, int cl,
+                                    int l, int L, int radius, int cl, int ldc,
                                     hipStream_t stream);
 void flowhip_corr_lookup_bwd_launch(const float* gout, const float* coords,
                                     float* glevel, int BP, int P, int Hl,
@@ -176,12 +176,17 @@ torch::Tensor corr_lookup_fwd(std::vector<torch::Tensor> pyramid,
   const int L = (int)pyramid.size();
   const int K = 2 * (int)radius + 1;
 
-  auto out = channels_last
-                 ? torch::empty({B, (long)L * K * K, H, W},
-                                coords.options().dtype(torch::kFloat32)
-                                    .memory_format(torch::MemoryFormat::ChannelsLast))
-                 : torch::empty({B, (long)L * K * K, H, W},
-                                coords.options().dtype(torch::kFloat32));
+  // channels-last: allocate with the channel count rounded up to 8 and
+  // return a narrow view — the NHWC conv consumer reads whole 16-B pieces
+  const long C = (long)L * K * K;
+  const long C8 = channels_last ? (C + 7) / 8 * 8 : C;
+  auto full = channels_last
+                  ? torch::empty({B, C8, H, W},
+                                 coords.options().dtype(torch::kFloat32)
+                                     .memory_format(torch::MemoryFormat::ChannelsLast))
+                  : torch::empty({B, C, H, W},
+                                 coords.options().dtype(torch::kFloat32));
+  auto out = (C8 != C) ? full.narrow(1, 0, C) : full;
   const c10::cuda::CUDAGuard guard(coords.device());
   hipStream_t stream = at::cuda::getCurrentCUDAStream().stream();
 
@@ -193,8 +198,9 @@ torch::Tensor corr_lookup_fwd(std::vector<torch::Tensor> pyramid,
     TORCH_CHECK(lvl.size(0) == (long)B * P, "corr_lookup: level batch mismatch");
     const int Hl = lvl.size(-2), Wl = lvl.size(-1);
     flowhip_corr_lookup_fwd_launch(
-        lvl.data_ptr<float>(), coords.data_ptr<float>(), out.data_ptr<float>(),
-        B * P, P, Hl, Wl, l, L, (int)radius, channels_last ? 1 : 0, stream);
+        lvl.data_ptr<float>(), coords.data_ptr<float>(),
+        full.data_ptr<float>(), B * P, P, Hl, Wl, l, L, (int)radius,
+        channels_last ? 1 : 0, (int)C8, stream);
   }
   return out;
 }

@@ -41,7 +41,7 @@ __global__ __launch_bounds__(LK_THREADS) void corr_lookup_fwd_kernel(
     const scalar_t* __restrict__ level,  // (B*P, Hl, Wl)
     const float* __restrict__ coords,    // (B, 2, H, W)
     float* __restrict__ out,             // (B, L*K2, H, W) NCHW or NHWC
-    int BP, int P, int Hl, int Wl, int l, int L, int cl) {
+    int BP, int P, int Hl, int Wl, int l, int L, int cl, int ldc) {
   constexpr int K = 2 * R + 1;
   constexpr int K2 = K * K;
 
@@ -62,8 +62,15 @@ __global__ __launch_bounds__(LK_THREADS) void corr_lookup_fwd_kernel(
 
   const scalar_t* map = level + (long)pix * Hl * Wl;
   const long tap_stride = cl ? 1 : (long)P;
-  float* outb = cl ? out + ((long)b * P + i) * (L * K2) + (long)l * K2
+  float* outb = cl ? out + ((long)b * P + i) * ldc + (long)l * K2
                    : out + ((long)b * L * K2 + (long)l * K2) * P + i;
+  if (cl && l == 0) {
+    // zero the channel-pad tail once (ldc > L*K2: the emitted tensor is a
+    // narrow view of an 8-channel-aligned allocation so the NHWC conv
+    // consumer's 16-B staging reads stay in-row)
+    float* rowb = out + ((long)b * P + i) * ldc;
+    for (int c = L * K2; c < ldc; ++c) rowb[c] = 0.0f;
+  }
 
   float tprev[K];   // wy-blended column a-1
   float tcur[K];
@@ -159,11 +166,11 @@ __global__ __launch_bounds__(LK_THREADS) void corr_lookup_bwd_kernel(
 template <int R>
 static void lookup_fwd_level(const float* level, const float* coords,
                              float* out, int BP, int P, int Hl, int Wl, int l,
-                             int L, int cl, hipStream_t stream) {
+                             int L, int cl, int ldc, hipStream_t stream) {
   dim3 grid(fh_cdiv(BP, LK_THREADS));
   hipLaunchKernelGGL((corr_lookup_fwd_kernel<R, float>), grid,
                      dim3(LK_THREADS), 0, stream, level, coords, out, BP, P,
-                     Hl, Wl, l, L, cl);
+                     Hl, Wl, l, L, cl, ldc);
 }
 
 template <int R>
@@ -178,11 +185,11 @@ static void lookup_bwd_level(const float* gout, const float* coords,
 
 void flowhip_corr_lookup_fwd_launch(const float* level, const float* coords,
                                     float* out, int BP, int P, int Hl, int Wl,
-                                    int l, int L, int radius, int cl,
+                                    int l, int L, int radius, int cl, int ldc,
                                     hipStream_t stream) {
   switch (radius) {
-    case 3: lookup_fwd_level<3>(level, coords, out, BP, P, Hl, Wl, l, L, cl, stream); break;
-    case 4: lookup_fwd_level<4>(level, coords, out, BP, P, Hl, Wl, l, L, cl, stream); break;
+    case 3: lookup_fwd_level<3>(level, coords, out, BP, P, Hl, Wl, l, L, cl, ldc, stream); break;
+    case 4: lookup_fwd_level<4>(level, coords, out, BP, P, Hl, Wl, l, L, cl, ldc, stream); break;
     default: abort();
   }
 }
