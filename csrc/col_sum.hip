@@ -6,7 +6,7 @@
 #include "common.h"
 
 #define CS_THREADS 256
-#define CS_MCHUNK 8192
+#define CS_MCHUNK 256  // small chunks: enough blocks to fill 256 CUs at M~21k
 
 __global__ __launch_bounds__(CS_THREADS) void col_sum_partial_kernel(
     const __bf16* __restrict__ dy, float* __restrict__ partials, long M,
