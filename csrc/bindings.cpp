@@ -860,7 +860,7 @@ torch::Tensor col_sum_bf16(torch::Tensor dy) {
               dy.stride(3) == dy.size(1));
   const int C = dy.size(1);
   const long M = dy.numel() / C;
-  const int nchunk = (int)std::min<long>((M + 255) / 256, 256);
+  const int nchunk = (int)((M + 255) / 256);  // must cover ALL rows (kernel bounds come from the chunk index)
   const int ncb = (C + 63) / 64;
   auto partials = torch::empty({(long)nchunk * ncb * 64},
                                dy.options().dtype(torch::kFloat32));
