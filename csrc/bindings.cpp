@@ -135,6 +135,24 @@ void flowhip_area_up2x_fwd_launch(const float* in, float* out, long total,
 void flowhip_area_up2x_bwd_launch(const float* gout, float* gin,
                                   long total_in, int H, int W,
                                   hipStream_t stream);
+bool flowhip_packernel_fwd_launch(const float* f, float* k, int B, int C,
+                                  int H, int W, int K, int dil, int norm,
+                                  hipStream_t stream);
+bool flowhip_packernel_bwd_launch(const float* f, const float* k,
+                                  const float* dk, float* df, int B, int C,
+                                  int H, int W, int K, int dil, int norm,
+                                  hipStream_t stream);
+bool flowhip_pacconv_fwd_launch(const float* x, const float* kr,
+                                const float* w, const float* bias, float* out,
+                                int B, int Ci, int Co, int H, int W, int OH,
+                                int OW, int pH, int pW, int K, int dil,
+                                int shared, hipStream_t stream);
+bool flowhip_pacconv_bwd_launch(const float* dy, const float* x,
+                                const float* kr, const float* w, float* dx,
+                                float* dk, float* partials, float* dw,
+                                int nchunk, int B, int Ci, int Co, int H,
+                                int W, int OH, int OW, int pH, int pW, int K,
+                                int dil, int shared, hipStream_t stream);
 bool flowhip_corr_pyramid_fwd_launch(const float* corr, float* l1, float* l2,
                                      float* l3, int BP, int H0, int W0,
                                      int nlev, hipStream_t stream);
@@ -900,6 +918,92 @@ torch::Tensor area_up2x_bwd(torch::Tensor gout) {
   return gin;
 }
 
+torch::Tensor packernel_fwd(torch::Tensor f, int64_t K, int64_t dil,
+                            bool normalize) {
+  TORCH_CHECK(f.is_cuda() && f.is_contiguous() && f.dim() == 4 &&
+              f.dtype() == torch::kFloat32);
+  const int B = f.size(0), C = f.size(1), H = f.size(2), W = f.size(3);
+  auto k = torch::empty({(long)B, K * K, (long)H, (long)W}, f.options());
+  const c10::cuda::CUDAGuard guard(f.device());
+  hipStream_t stream = at::cuda::getCurrentCUDAStream().stream();
+  bool ok = flowhip_packernel_fwd_launch(f.data_ptr<float>(),
+                                         k.data_ptr<float>(), B, C, H, W,
+                                         (int)K, (int)dil, normalize ? 1 : 0,
+                                         stream);
+  TORCH_CHECK(ok, "packernel: unsupported K");
+  return k;
+}
+
+torch::Tensor packernel_bwd(torch::Tensor f, torch::Tensor k,
+                            torch::Tensor dk, int64_t K, int64_t dil,
+                            bool normalize) {
+  const int B = f.size(0), C = f.size(1), H = f.size(2), W = f.size(3);
+  auto df = torch::empty_like(f);
+  const c10::cuda::CUDAGuard guard(f.device());
+  hipStream_t stream = at::cuda::getCurrentCUDAStream().stream();
+  bool ok = flowhip_packernel_bwd_launch(
+      f.data_ptr<float>(), k.data_ptr<float>(), dk.data_ptr<float>(),
+      df.data_ptr<float>(), B, C, H, W, (int)K, (int)dil, normalize ? 1 : 0,
+      stream);
+  TORCH_CHECK(ok);
+  return df;
+}
+
+torch::Tensor pacconv_fwd(torch::Tensor x, torch::Tensor kr,
+                          torch::Tensor w, c10::optional<torch::Tensor> bias,
+                          int64_t pH, int64_t pW, int64_t dil, bool shared) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous() &&
+              x.dtype() == torch::kFloat32);
+  TORCH_CHECK(kr.is_contiguous() && w.is_contiguous());
+  const int B = x.size(0), Ci = x.size(1), H = x.size(2), W = x.size(3);
+  const int K = w.size(-1);
+  const int Co = shared ? Ci : (int)w.size(0);
+  const int OH = H + 2 * (int)pH - ((K - 1) * (int)dil);
+  const int OW = W + 2 * (int)pW - ((K - 1) * (int)dil);
+  TORCH_CHECK(kr.size(-2) == OH && kr.size(-1) == OW,
+              "pacconv: kernel grid must match the output grid");
+  auto out = torch::empty({(long)B, (long)Co, (long)OH, (long)OW},
+                          x.options());
+  const c10::cuda::CUDAGuard guard(x.device());
+  hipStream_t stream = at::cuda::getCurrentCUDAStream().stream();
+  bool ok = flowhip_pacconv_fwd_launch(
+      x.data_ptr<float>(), kr.data_ptr<float>(), w.data_ptr<float>(),
+      bias.has_value() ? bias->data_ptr<float>() : nullptr,
+      out.data_ptr<float>(), B, Ci, Co, H, W, OH, OW, (int)pH, (int)pW, K,
+      (int)dil, shared ? 1 : 0, stream);
+  TORCH_CHECK(ok, "pacconv: unsupported K");
+  return out;
+}
+
+std::vector<torch::Tensor> pacconv_bwd(torch::Tensor dy, torch::Tensor x,
+                                       torch::Tensor kr, torch::Tensor w,
+                                       int64_t pH, int64_t pW, int64_t dil,
+                                       bool shared) {
+  const int B = x.size(0), Ci = x.size(1), H = x.size(2), W = x.size(3);
+  const int K = w.size(-1);
+  const int Co = shared ? Ci : (int)w.size(0);
+  const int OH = dy.size(2), OW = dy.size(3);
+  const int K2 = K * K;
+  const int nw = shared ? K2 : Co * Ci * K2;
+  TORCH_CHECK((long)nw * 4 <= 65536, "pacconv dw: weight too large for the "
+              "LDS-accumulated backward");
+  const int nchunk = 512;
+  auto dx = torch::empty_like(x);
+  auto dk = torch::empty_like(kr);
+  auto partials = torch::empty({(long)nchunk * nw},
+                               x.options().dtype(torch::kFloat32));
+  auto dw = torch::empty_like(w);
+  const c10::cuda::CUDAGuard guard(x.device());
+  hipStream_t stream = at::cuda::getCurrentCUDAStream().stream();
+  bool ok = flowhip_pacconv_bwd_launch(
+      dy.data_ptr<float>(), x.data_ptr<float>(), kr.data_ptr<float>(),
+      w.data_ptr<float>(), dx.data_ptr<float>(), dk.data_ptr<float>(),
+      partials.data_ptr<float>(), dw.data_ptr<float>(), nchunk, B, Ci, Co, H,
+      W, OH, OW, (int)pH, (int)pW, K, (int)dil, shared ? 1 : 0, stream);
+  TORCH_CHECK(ok);
+  return {dx, dk, dw};
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -910,6 +1014,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "fused multi-level correlation window lookup");
   m.def("corr_lookup_bwd", &corr_lookup_bwd,
         "backward of corr_lookup_fwd (pyramid grads)");
+  m.def("packernel_fwd", &packernel_fwd, "PAC gaussian adapting kernel");
+  m.def("packernel_bwd", &packernel_bwd, "backward of packernel_fwd");
+  m.def("pacconv_fwd", &pacconv_fwd, "pixel-adaptive convolution forward");
+  m.def("pacconv_bwd", &pacconv_bwd, "pixel-adaptive conv backward (dx,dk,dw)");
   m.def("corr_pyramid_fwd", &corr_pyramid_fwd,
         "fused avg-pool pyramid build (levels 1..n-1)");
   m.def("corr_pyramid_bwd", &corr_pyramid_bwd,

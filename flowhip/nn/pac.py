@@ -114,6 +114,18 @@ def packernel2d(input, mask=None, kernel_size=0, stride=1, padding=0,
                             for (k, d) in zip(kernel_size, dilation_))
     else:
         eff_stride, eff_padding = stride_, padding_
+
+    from ..ops.functional_pac import PacKernelGaussFn, pac_kernel_fusable
+    if pac_kernel_fusable(input, mask, kernel_type, smooth_kernel_type,
+                          channel_wise, kernel_size, dilation_, eff_stride,
+                          eff_padding):
+        k = PacKernelGaussFn.apply(input, kernel_size[0], dilation_[0],
+                                   bool(normalize_kernel))
+        bs = input.shape[0]
+        out = k.view(bs, 1, kernel_size[0], kernel_size[1],
+                     input.shape[-2], input.shape[-1])
+        return out, None
+
     feat = nd2col(input, kernel_size, stride=eff_stride,
                   padding=eff_padding, dilation=dilation_)
 
@@ -166,6 +178,17 @@ def pacconv2d(input, kernel, weight, bias=None, stride=1, padding=0,
               dilation=1, shared_filters=False):
     """out = sum over window of (unfold(input) * kernel) . weight."""
     kernel_size = tuple(weight.shape[-2:])
+
+    from ..ops.functional_pac import PacConv2dFn, pac_conv_fusable
+    if pac_conv_fusable(input, kernel, weight, stride, padding, dilation):
+        p = _pair(padding)
+        d = _pair(dilation)
+        bs = input.shape[0]
+        kr = kernel.reshape(bs, kernel_size[0] * kernel_size[1],
+                            *kernel.shape[-2:])
+        return PacConv2dFn.apply(input, kr, weight, bias, p[0], p[1], d[0],
+                                 bool(shared_filters))
+
     cols = nd2col(input, kernel_size, stride=stride, padding=padding,
                   dilation=dilation)
     if shared_filters:

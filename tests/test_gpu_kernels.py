@@ -650,3 +650,67 @@ class TestTrainingLearns:
         first = sum(losses[:5]) / 5
         last = sum(losses[-5:]) / 5
         assert last < 0.7 * first, (first, last)
+
+
+class TestPacHIP:
+    @pytest.mark.parametrize("K,dil,norm", [(3, 1, False), (5, 1, True),
+                                            (5, 2, False)])
+    def test_packernel_matches_torch(self, K, dil, norm):
+        from flowhip.nn import pac
+        torch.manual_seed(29)
+        f = torch.randn(2, 6, 18, 24, device=_dev(), requires_grad=True)
+        pad = (K - 1) * dil // 2
+        k_hip, _ = pac.packernel2d(f, kernel_size=K, stride=1, padding=pad,
+                                   dilation=dil, normalize_kernel=norm)
+        f2 = f.detach().cpu().requires_grad_(True)
+        k_ref, _ = pac.packernel2d(f2, kernel_size=K, stride=1, padding=pad,
+                                   dilation=dil, normalize_kernel=norm)
+        torch.testing.assert_close(k_hip.cpu(), k_ref, atol=1e-5, rtol=1e-5)
+        g = torch.randn_like(k_ref)
+        (k_hip * g.to(_dev())).sum().backward()
+        (k_ref * g).sum().backward()
+        torch.testing.assert_close(f.grad.cpu(), f2.grad, atol=1e-4,
+                                   rtol=1e-4)
+
+    @pytest.mark.parametrize("shared,pad", [(False, 2), (True, 2),
+                                            (False, 0)])
+    def test_pacconv_matches_torch(self, shared, pad):
+        from flowhip.nn import pac
+        torch.manual_seed(30)
+        B, Ci, H, W, K = 2, 4, 16, 20, 5
+        Co = Ci if shared else 3
+        oh, ow = H + 2 * pad - (K - 1), W + 2 * pad - (K - 1)
+        x = torch.randn(B, Ci, H, W, device=_dev(), requires_grad=True)
+        kr = torch.rand(B, 1, K, K, oh, ow, device=_dev(),
+                        requires_grad=True)
+        w = (torch.randn(1, 1, K, K) if shared
+             else torch.randn(Co, Ci, K, K)).to(_dev()).requires_grad_(True)
+        b = torch.randn(Co, device=_dev(), requires_grad=True)
+
+        out = pac.pacconv2d(x, kr, w, b, stride=1, padding=pad,
+                            shared_filters=shared)
+        x2 = x.detach().cpu().requires_grad_(True)
+        kr2 = kr.detach().cpu().requires_grad_(True)
+        w2 = w.detach().cpu().requires_grad_(True)
+        b2 = b.detach().cpu().requires_grad_(True)
+        ref = pac.pacconv2d(x2, kr2, w2, b2, stride=1, padding=pad,
+                            shared_filters=shared)
+        torch.testing.assert_close(out.cpu(), ref, atol=1e-4, rtol=1e-4)
+        g = torch.randn_like(ref)
+        (out * g.to(_dev())).sum().backward()
+        (ref * g).sum().backward()
+        torch.testing.assert_close(x.grad.cpu(), x2.grad, atol=1e-4, rtol=1e-4)
+        torch.testing.assert_close(kr.grad.cpu(), kr2.grad, atol=1e-4, rtol=1e-4)
+        torch.testing.assert_close(w.grad.cpu(), w2.grad, atol=1e-3, rtol=1e-3)
+        torch.testing.assert_close(b.grad.cpu(), b2.grad, atol=1e-3, rtol=1e-3)
+
+    def test_pac_upsampler_head_runs_on_gpu(self):
+        from flowhip.nn.pac_upsampler import PacJointUpsample
+        torch.manual_seed(31)
+        net = PacJointUpsample(factor=4, channels=2).to(_dev())
+        lr = torch.randn(1, 2, 12, 16, device=_dev(), requires_grad=True)
+        guide = torch.randn(1, 3, 48, 64, device=_dev())
+        out = net(lr, guide)
+        assert out.shape == (1, 2, 48, 64)
+        out.sum().backward()
+        assert lr.grad is not None and torch.isfinite(lr.grad).all()
