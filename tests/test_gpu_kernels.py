@@ -707,7 +707,9 @@ class TestPacHIP:
     def test_pac_upsampler_head_runs_on_gpu(self):
         from flowhip.nn.pac_upsampler import PacJointUpsample
         torch.manual_seed(31)
-        net = PacJointUpsample(factor=4, channels=2).to(_dev())
+        # channels=1: the head folds multi-channel inputs to single-channel
+        # batches (convert_to_single_channel, ref pac_upsampler.py:16)
+        net = PacJointUpsample(factor=4, channels=1).to(_dev())
         lr = torch.randn(1, 2, 12, 16, device=_dev(), requires_grad=True)
         guide = torch.randn(1, 3, 48, 64, device=_dev())
         out = net(lr, guide)
