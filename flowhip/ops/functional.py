@@ -81,9 +81,9 @@ class CorrVolumeFn(torch.autograd.Function):
         # tiled transpose+cast kernel (eager transpose().to(bf16) is an
         # uncoalesced ~300us elementwise op at P=7168)
         dct = _ext.ext().transpose_cast_bf16(g3)
-        # dF1t[i,d] = sum_j dC[i,j] * F2t[j,d] -> A=dC (M=P,K=P), B=f2 with
-        # (N=D, K=P) k-major == f2t^T; i.e. B operand is f2 (D,P) row-major? No:
-        # gemm_nt wants B (N,K) row-major = (D,P) with P contiguous == f2 view.
+        # dF1t[i,d] = sum_j dC[i,j] * F2t[j,d]: A = dC (M=P, K=P); the B
+        # operand must be (N=D, K=P) row-major = f2 in (D, P) layout with P
+        # contiguous.
         f2_kn = f2t.transpose(1, 2).contiguous()  # (B,D,P) bf16, P contiguous
         f1_kn = f1t.transpose(1, 2).contiguous()
         df1t = _bgemm_nt(dc, f2_kn, alpha)   # (B,P,D) fp32
