@@ -42,8 +42,19 @@ __global__ __launch_bounds__(PAC_THREADS) void packernel_gauss_fwd_kernel(
     const float* fb = f + b * C * plane;
 
     float u[K2];
+    bool valid[K2];
 #pragma unroll
     for (int i = 0; i < K2; ++i) u[i] = 0.f;
+#pragma unroll
+    for (int ky = 0; ky < K; ++ky) {
+      const int yy = y + dil * (ky - R);
+#pragma unroll
+      for (int kx = 0; kx < K; ++kx) {
+        const int xx = x + dil * (kx - R);
+        valid[ky * K + kx] =
+            (yy >= 0) & (yy < H) & (xx >= 0) & (xx < W);
+      }
+    }
 
     for (int c = 0; c < C; ++c) {
       const float fc = fb[c * plane + (long)y * W + x];
@@ -67,6 +78,9 @@ __global__ __launch_bounds__(PAC_THREADS) void packernel_gauss_fwd_kernel(
 #pragma unroll
     for (int i = 0; i < K2; ++i) {
       u[i] = expf(-0.5f * u[i]);
+      // normalize path: the reference multiplies by the unfolded ones
+      // mask first, zeroing out-of-range taps (packernel2d :384-389)
+      if (norm && !valid[i]) u[i] = 0.f;
       s += u[i];
     }
     const float inv = (norm && s > 0.f) ? 1.0f / s : 1.0f;
