@@ -96,23 +96,22 @@ def packernel2d(input, mask=None, kernel_size=0, stride=1, padding=0,
     # and the transposed case). Must dispatch BEFORE the ones-mask below
     # reassigns `mask`; the kernel replicates the mask's border-tap zeroing
     # for normalize_kernel.
-    if True:
-        if transposed:
-            _eff_stride = _pair(1)
-            _eff_padding = tuple((k - 1) * d // 2
-                                 for (k, d) in zip(kernel_size, dilation_))
-        else:
-            _eff_stride, _eff_padding = stride_, padding_
-        from ..ops.functional_pac import PacKernelGaussFn, pac_kernel_fusable
-        if pac_kernel_fusable(input, mask, kernel_type, smooth_kernel_type,
-                              channel_wise, kernel_size, dilation_,
-                              _eff_stride, _eff_padding):
-            k = PacKernelGaussFn.apply(input, kernel_size[0], dilation_[0],
-                                       bool(normalize_kernel))
-            bs_ = input.shape[0]
-            out = k.view(bs_, 1, kernel_size[0], kernel_size[1],
-                         input.shape[-2], input.shape[-1])
-            return out, None
+    if transposed:
+        _eff_stride = _pair(1)
+        _eff_padding = tuple((k - 1) * d // 2
+                             for (k, d) in zip(kernel_size, dilation_))
+    else:
+        _eff_stride, _eff_padding = stride_, padding_
+    from ..ops.functional_pac import PacKernelGaussFn, pac_kernel_fusable
+    if pac_kernel_fusable(input, mask, kernel_type, smooth_kernel_type,
+                          channel_wise, kernel_size, dilation_,
+                          _eff_stride, _eff_padding):
+        k = PacKernelGaussFn.apply(input, kernel_size[0], dilation_[0],
+                                   bool(normalize_kernel))
+        bs_ = input.shape[0]
+        out = k.view(bs_, 1, kernel_size[0], kernel_size[1],
+                     input.shape[-2], input.shape[-1])
+        return out, None
 
     if mask is not None or normalize_kernel:
         mask_pattern = input.new_ones(1, 1, *in_sz)
