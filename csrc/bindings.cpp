@@ -56,6 +56,9 @@ bool flowhip_nconv_wrw_tiled_launch(const float* dnomin, const float* ddenom,
                                     int Ci, int Co, int H, int W, int K,
                                     hipStream_t stream);
 int flowhip_instnorm_partial_rows(int N, int C, long P);
+void flowhip_conv_gemm_pack_launch(const float* w, void* wpk, long total,
+                                   int O, int I, int KH, int KW, int cpad,
+                                   int flip, hipStream_t stream);
 void flowhip_transpose_cast_launch(const float* in, void* out, int B, int M,
                                    int N, hipStream_t stream);
 void flowhip_conf_pool_fwd_launch(const float* data, const float* conf,
@@ -762,6 +765,22 @@ static void cg_check_x(const torch::Tensor& x, int& ld) {
               x.stride(0) == (long)ld * x.size(3) * x.size(2));
 }
 
+torch::Tensor conv_gemm_pack(torch::Tensor w, bool flip) {
+  TORCH_CHECK(w.is_cuda() && w.is_contiguous() &&
+              w.scalar_type() == torch::kFloat32 && w.dim() == 4);
+  const int O = w.size(0), I = w.size(1), KH = w.size(2), KW = w.size(3);
+  const int R0 = flip ? I : O;
+  const int cpad = ((flip ? O : I) + 63) / 64 * 64;
+  auto wpk = torch::empty({(long)KH * KW, (long)R0, (long)cpad},
+                          w.options().dtype(torch::kBFloat16));
+  const c10::cuda::CUDAGuard guard(w.device());
+  hipStream_t stream = at::cuda::getCurrentCUDAStream().stream();
+  flowhip_conv_gemm_pack_launch(w.data_ptr<float>(), wpk.data_ptr(),
+                                wpk.numel(), O, I, KH, KW, cpad,
+                                flip ? 1 : 0, stream);
+  return wpk;
+}
+
 std::vector<torch::Tensor> conv_gemm_fwd2(
     torch::Tensor x, c10::optional<torch::Tensor> x2, torch::Tensor wpk,
     c10::optional<torch::Tensor> bias, int64_t Cout, int64_t KH, int64_t KW,
@@ -1035,6 +1054,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conf_pool_bwd", &conf_pool_bwd, "backward of conf_pool");
   m.def("transpose_cast_bf16", &transpose_cast_bf16,
         "(B,M,N) fp32 -> (B,N,M) bf16 tiled transpose");
+  m.def("conv_gemm_pack", &conv_gemm_pack,
+        "one-kernel conv weight packing (fwd or flipped-bwd layout)");
   m.def("conv_gemm_fwd", &conv_gemm_fwd,
         "implicit-GEMM NHWC bf16 conv forward (also bwd-data with flipped "
         "packed weights)");

@@ -476,3 +476,44 @@ void flowhip_conv_gemm_wrw_launch(const void* dy, const void* x,
                      0, stream, partials, dw, nchunk, KH * KW, tiles_o * 64,
                      cpad, Cout, Cin);
 }
+
+// ---------------------------------------------------------------------------
+// Weight packing (host-side cache refill after every optimizer step): one
+// gather kernel instead of the ~5-op torch chain (permute/reshape/pad/cast/
+// contiguous) per pack, ~52 packs per training step.
+//   fwd pack:  wpk[kyx][o][c]  = c < I ? w[o][c][ky][kx] : 0     (bf16)
+//   bwd pack:  wpk[kyx][i][oc] = oc < O ? w[oc][i][KH-1-ky][KW-1-kx] : 0
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(256) void conv_gemm_pack_kernel(
+    const float* __restrict__ w,  // (O, I, KH, KW)
+    __bf16* __restrict__ wpk,     // (KYX, R0, cpad)
+    long total, int O, int I, int KH, int KW, int cpad, int flip) {
+  const int KYX = KH * KW;
+  for (long idx = (long)blockIdx.x * 256 + threadIdx.x; idx < total;
+       idx += (long)gridDim.x * 256) {
+    long t = idx;
+    const int c = t % cpad; t /= cpad;
+    const int r = t % (flip ? I : O); t /= (flip ? I : O);
+    const int kyx = (int)t;
+    float v = 0.f;
+    if (!flip) {
+      if (c < I) v = w[(((long)r * I + c) * KYX) + kyx];
+    } else {
+      // r = input channel, c = output channel, spatially flipped
+      const int ky = kyx / KW, kx = kyx - (kyx / KW) * KW;
+      const int fk = (KH - 1 - ky) * KW + (KW - 1 - kx);
+      if (c < O) v = w[(((long)c * I + r) * KYX) + fk];
+    }
+    wpk[idx] = (__bf16)v;
+  }
+}
+
+void flowhip_conv_gemm_pack_launch(const float* w, void* wpk, long total,
+                                   int O, int I, int KH, int KW, int cpad,
+                                   int flip, hipStream_t stream) {
+  long blocks = (total + 255) / 256;
+  if (blocks > 4096) blocks = 4096;
+  hipLaunchKernelGGL(conv_gemm_pack_kernel, dim3((int)blocks), dim3(256), 0,
+                     stream, w, (__bf16*)wpk, total, O, I, KH, KW, cpad,
+                     flip);
+}

@@ -61,16 +61,21 @@ def _col_sum(dy):
 
 def _pack_fwd(weight):
     """(O, I, KH, KW) fp32 -> (KYX, O, pad64(I)) bf16 contiguous."""
-    O, I, KH, KW = weight.shape
-    w = weight.detach().to(torch.bfloat16).permute(2, 3, 0, 1).reshape(
-        KH * KW, O, I)
+    w = weight.detach()
+    if w.is_cuda and w.dtype == torch.float32:
+        return _ext.ext().conv_gemm_pack(w.contiguous(), False)
+    O, I, KH, KW = w.shape
+    w = w.to(torch.bfloat16).permute(2, 3, 0, 1).reshape(KH * KW, O, I)
     return F.pad(w, (0, _pad64(I) - I)).contiguous()
 
 
 def _pack_bwd(weight):
     """flip + transpose: (O, I, KH, KW) -> (KYX, I, pad64(O)) bf16."""
-    O, I, KH, KW = weight.shape
-    w = weight.detach().to(torch.bfloat16).flip(2, 3).permute(2, 3, 1, 0)
+    w = weight.detach()
+    if w.is_cuda and w.dtype == torch.float32:
+        return _ext.ext().conv_gemm_pack(w.contiguous(), True)
+    O, I, KH, KW = w.shape
+    w = w.to(torch.bfloat16).flip(2, 3).permute(2, 3, 1, 0)
     w = w.reshape(KH * KW, I, O)
     return F.pad(w, (0, _pad64(O) - O)).contiguous()
 
