@@ -55,3 +55,46 @@ void flowhip_area_up2x_bwd_launch(const float* gout, float* gin,
   hipLaunchKernelGGL(area_up2x_bwd_kernel, dim3((int)blocks),
                      dim3(AU_THREADS), 0, stream, gout, gin, total_in, H, W);
 }
+
+// Fused nearest-2x upsample + channel concat for the NConvUNet decoder
+// skips (nconv_modules.py:128-134: F.interpolate(nearest) + cat, done for
+// both data and conf every iteration):
+//   out[:, :C1]  = low[n, c, y/2, x/2]   (exact 2x nearest)
+//   out[:, C1:]  = skip[n, c-C1, y, x]
+// Backward: dlow = 2x2 sum gather (one kernel); dskip = channel-slice view
+// (zero-copy on the Python side).
+__global__ __launch_bounds__(AU_THREADS) void up2x_cat_fwd_kernel(
+    const float* __restrict__ low,   // (N, C1, H/2, W/2)
+    const float* __restrict__ skip,  // (N, C2, H, W)
+    float* __restrict__ out,         // (N, C1+C2, H, W)
+    long total, int C1, int C2, int H, int W) {
+  const int HL = H / 2, WL = W / 2;
+  for (long idx = (long)blockIdx.x * AU_THREADS + threadIdx.x; idx < total;
+       idx += (long)gridDim.x * AU_THREADS) {
+    long t = idx;
+    const int x = t % W; t /= W;
+    const int y = t % H; t /= H;
+    const int c = t % (C1 + C2); t /= (C1 + C2);
+    const long n = t;
+    float v;
+    if (c < C1) {
+      int yl = y >> 1, xl = x >> 1;
+      if (yl >= HL) yl = HL - 1;  // odd-size guard (nearest floor clamp)
+      if (xl >= WL) xl = WL - 1;
+      v = low[((n * C1 + c) * HL + yl) * WL + xl];
+    } else {
+      v = skip[((n * C2 + (c - C1)) * H + y) * W + x];
+    }
+    out[idx] = v;
+  }
+}
+
+void flowhip_up2x_cat_fwd_launch(const float* low, const float* skip,
+                                 float* out, long total, int C1, int C2,
+                                 int H, int W, hipStream_t stream) {
+  long blocks = (total + AU_THREADS - 1) / AU_THREADS;
+  if (blocks > 16384) blocks = 16384;
+  hipLaunchKernelGGL(up2x_cat_fwd_kernel, dim3((int)blocks),
+                     dim3(AU_THREADS), 0, stream, low, skip, out, total, C1,
+                     C2, H, W);
+}

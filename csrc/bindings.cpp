@@ -133,6 +133,9 @@ void flowhip_nconv_bwd_prep_launch(const float* gout, const float* gcout,
                                    hipStream_t stream);
 bool flowhip_col_sum_launch(const void* dy, float* partials, float* out,
                             long M, int C, int nchunk, hipStream_t stream);
+void flowhip_up2x_cat_fwd_launch(const float* low, const float* skip,
+                                 float* out, long total, int C1, int C2,
+                                 int H, int W, hipStream_t stream);
 void flowhip_area_up2x_fwd_launch(const float* in, float* out, long total,
                                   int H, int W, hipStream_t stream);
 void flowhip_area_up2x_bwd_launch(const float* gout, float* gin,
@@ -912,6 +915,22 @@ torch::Tensor col_sum_bf16(torch::Tensor dy) {
   return out;
 }
 
+torch::Tensor up2x_cat_fwd(torch::Tensor low, torch::Tensor skip) {
+  TORCH_CHECK(low.is_cuda() && low.is_contiguous() &&
+              low.dtype() == torch::kFloat32);
+  TORCH_CHECK(skip.is_cuda() && skip.is_contiguous() &&
+              skip.sizes()[0] == low.sizes()[0]);
+  const long N = skip.size(0), C1 = low.size(1), C2 = skip.size(1);
+  const int H = skip.size(2), W = skip.size(3);
+  auto out = torch::empty({N, C1 + C2, (long)H, (long)W}, low.options());
+  const c10::cuda::CUDAGuard guard(low.device());
+  hipStream_t stream = at::cuda::getCurrentCUDAStream().stream();
+  flowhip_up2x_cat_fwd_launch(low.data_ptr<float>(), skip.data_ptr<float>(),
+                              out.data_ptr<float>(), out.numel(), (int)C1,
+                              (int)C2, H, W, stream);
+  return out;
+}
+
 torch::Tensor area_up2x_fwd(torch::Tensor in) {
   TORCH_CHECK(in.is_cuda() && in.is_contiguous() && in.dim() == 4 &&
               in.dtype() == torch::kFloat32);
@@ -1047,6 +1066,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "fused normalized convolution forward (out, cout)");
   m.def("col_sum_bf16", &col_sum_bf16,
         "(N,C,H,W) channels-last bf16 -> (C) fp32 bias-grad column sum");
+  m.def("up2x_cat_fwd", &up2x_cat_fwd,
+        "fused nearest-2x upsample + channel concat");
   m.def("area_up2x_fwd", &area_up2x_fwd, "exact-2x area upsample");
   m.def("area_up2x_bwd", &area_up2x_bwd, "backward of area_up2x");
   m.def("conf_pool_fwd", &conf_pool_fwd,

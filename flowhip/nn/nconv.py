@@ -192,9 +192,10 @@ class NConvUNet(nn.Module):
         # Decoder (reference index arithmetic: stage 0 pairs x[nds] with
         # itself at the same scale; later stages upsample and concat skips).
         for i in range(nds):
-            x_up = F.interpolate(x[i + nds], size=c[nds - i].shape[2:], mode="nearest")
-            c_up = F.interpolate(c[i + nds], size=c[nds - i].shape[2:], mode="nearest")
-            x[i + nds + 1], c[i + nds + 1] = self.decoder[i]((
-                torch.cat((x_up, x[nds - i]), 1), torch.cat((c_up, c[nds - i]), 1)))
+            # fused nearest-2x upsample + concat (ops.up2x_cat; falls back
+            # to interpolate+cat off-GPU or at non-2x scales)
+            xin = ops.up2x_cat(x[i + nds], x[nds - i])
+            cin = ops.up2x_cat(c[i + nds], c[nds - i])
+            x[i + nds + 1], c[i + nds + 1] = self.decoder[i]((xin, cin))
 
         return self.nconv_out((x[-1], c[-1]))

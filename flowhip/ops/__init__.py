@@ -96,6 +96,20 @@ def area_resize(x, size):
     return F.interpolate(x, size, mode="area")
 
 
+def up2x_cat(low, skip):
+    """cat([nearest_2x(low), skip], dim=1) in one kernel (NConvUNet decoder
+    skip path). Exact-2x only; callers fall back otherwise."""
+    import torch.nn.functional as F
+    import torch as _t
+    if (_ext.use_hip(low) and low.dtype == _t.float32
+            and skip.shape[2] == 2 * low.shape[2]
+            and skip.shape[3] == 2 * low.shape[3]):
+        from .functional_upsample import Up2xCatFn
+        return Up2xCatFn.apply(low, skip)
+    up = F.interpolate(low, size=skip.shape[2:], mode="nearest")
+    return _t.cat((up, skip), 1)
+
+
 def zero_inject(inp, scale_h, scale_w, out_h=None, out_w=None):
     if _ext.use_hip(inp) and inp.dtype == torch.float32:
         from .functional_upsample import ZeroInjectFn

@@ -52,3 +52,23 @@ class AreaUp2xFn(torch.autograd.Function):
     @staticmethod
     def backward(ctx, gout):
         return _ext.ext().area_up2x_bwd(gout)
+
+
+
+class Up2xCatFn(torch.autograd.Function):
+    """Fused nearest-2x upsample + channel concat (NConvUNet decoder skip;
+    nconv_modules.py:128-134). Backward: the upsample grad is the area-2x
+    gather kernel; the skip grad is a zero-copy channel slice."""
+
+    @staticmethod
+    def forward(ctx, low, skip):
+        ctx.c1 = low.shape[1]
+        ctx.low_hw = (low.shape[2], low.shape[3])
+        return _ext.ext().up2x_cat_fwd(low.contiguous(), skip.contiguous())
+
+    @staticmethod
+    def backward(ctx, gout):
+        c1 = ctx.c1
+        g = gout.contiguous()
+        dlow = _ext.ext().area_up2x_bwd(g[:, :c1].contiguous())
+        return dlow, g[:, c1:]
