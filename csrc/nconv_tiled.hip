@@ -25,7 +25,7 @@
 // Shared tile loader: stages conf and data*conf (with K/2 halo, zero-padded)
 // for all CI channels into LDS.  LW = TW+K-1 row stride.
 // ---------------------------------------------------------------------------
-template <int K, int CI>
+template <int K, int CI, int TH = NCT_TH>
 __device__ inline void nct_stage_tile(const float* __restrict__ data,
                                       const float* __restrict__ conf,
                                       float* __restrict__ lds_c,
@@ -33,7 +33,7 @@ __device__ inline void nct_stage_tile(const float* __restrict__ data,
                                       int n, int x0, int y0, int Ci_stride_n,
                                       int H, int W) {
   constexpr int LW = NCT_TW + K - 1;
-  constexpr int LH = NCT_TH + K - 1;
+  constexpr int LH = TH + K - 1;
   const long plane = (long)H * W;
   for (int idx = threadIdx.x; idx < LH * LW; idx += NCT_THREADS) {
     const int row = idx / LW, col = idx - row * LW;
@@ -232,6 +232,8 @@ __global__ __launch_bounds__(NCT_THREADS) void nconv_bwd_data_tiled_kernel(
 // (wave shuffle -> LDS across waves) and writes one row of partials.
 // partials layout: (nblocks, Co*CI*K*K).
 // ---------------------------------------------------------------------------
+#define NCT_TH_WRW 32
+
 template <int K, int CI>
 __global__ __launch_bounds__(NCT_THREADS) void nconv_wrw_tiled_kernel(
     const float* __restrict__ dnomin, const float* __restrict__ ddenom,
@@ -239,7 +241,7 @@ __global__ __launch_bounds__(NCT_THREADS) void nconv_wrw_tiled_kernel(
     float* __restrict__ partials,
     int N, int Co, int H, int W, int ntx, int nty) {
   constexpr int LW = NCT_TW + K - 1;
-  constexpr int LH = NCT_TH + K - 1;
+  constexpr int LH = NCT_TH_WRW + K - 1;
   constexpr int NW = CI * K * K;  // weights per co
   __shared__ float lds_c[CI * LH * LW];
   __shared__ float lds_dc[CI * LH * LW];
@@ -249,9 +251,10 @@ __global__ __launch_bounds__(NCT_THREADS) void nconv_wrw_tiled_kernel(
   const int tx = t % ntx; t /= ntx;
   const int ty = t % nty; t /= nty;
   const int n = t;
-  const int x0 = tx * NCT_TW, y0 = ty * NCT_TH;
+  const int x0 = tx * NCT_TW, y0 = ty * NCT_TH_WRW;
 
-  nct_stage_tile<K, CI>(data, conf, lds_c, lds_dc, n, x0, y0, CI, H, W);
+  nct_stage_tile<K, CI, NCT_TH_WRW>(data, conf, lds_c, lds_dc, n, x0, y0,
+                                    CI, H, W);
   __syncthreads();
 
   const long plane = (long)H * W;
@@ -267,7 +270,7 @@ __global__ __launch_bounds__(NCT_THREADS) void nconv_wrw_tiled_kernel(
     for (int i = 0; i < NW; ++i) acc[i] = 0.f;
 
 #pragma unroll
-    for (int j = 0; j < NCT_TH / 4; ++j) {
+    for (int j = 0; j < NCT_TH_WRW / 4; ++j) {
       const int lyy = ly0 + 4 * j;
       const int y = y0 + lyy;
       if (x >= W || y >= H) continue;
@@ -389,7 +392,7 @@ bool flowhip_nconv_bwd_data_tiled_launch(
   }
 
 int flowhip_nconv_tiled_nblocks(int N, int H, int W) {
-  return fh_cdiv(W, NCT_TW) * fh_cdiv(H, NCT_TH) * N;
+  return fh_cdiv(W, NCT_TW) * fh_cdiv(H, NCT_TH_WRW) * N;
 }
 
 bool flowhip_nconv_wrw_tiled_launch(const float* dnomin, const float* ddenom,
@@ -397,7 +400,7 @@ bool flowhip_nconv_wrw_tiled_launch(const float* dnomin, const float* ddenom,
                                     float* partials, float* dweight, int N,
                                     int Ci, int Co, int H, int W, int K,
                                     hipStream_t stream) {
-  const int ntx = fh_cdiv(W, NCT_TW), nty = fh_cdiv(H, NCT_TH);
+  const int ntx = fh_cdiv(W, NCT_TW), nty = fh_cdiv(H, NCT_TH_WRW);
   const int nblocks = ntx * nty * N;
   const int nw = Co * Ci * K * K;
   dim3 grid(nblocks), block(NCT_THREADS);
