@@ -72,6 +72,33 @@ __global__ __launch_bounds__(LK_THREADS) void corr_lookup_fwd_kernel(
     for (int c = L * K2; c < ldc; ++c) rowb[c] = 0.0f;
   }
 
+  const bool interior = (x0 >= 0) & (y0 >= 0) & (x0 + K < Wl) &
+                        (y0 + K < Hl);
+  if (interior) {
+    // Fast path: stream patch ROWS (K+1 contiguous loads each — one base
+    // address per row instead of per-element strided address math), blend
+    // x within the row, then blend y across consecutive rows.
+    float rprev[K], rcur[K];
+#pragma unroll
+    for (int j = 0; j <= K; ++j) {
+      const float* row = map + (long)(y0 + j) * Wl + x0;
+      float rv[K + 1];
+#pragma unroll
+      for (int a = 0; a <= K; ++a) rv[a] = (float)row[a];
+#pragma unroll
+      for (int a = 0; a < K; ++a) rcur[a] = wx0 * rv[a] + wx1 * rv[a + 1];
+      if (j > 0) {
+#pragma unroll
+        for (int a = 0; a < K; ++a)
+          outb[(long)(a * K + (j - 1)) * tap_stride] =
+              wy0 * rprev[a] + wy1 * rcur[a];
+      }
+#pragma unroll
+      for (int a = 0; a < K; ++a) rprev[a] = rcur[a];
+    }
+    return;
+  }
+
   float tprev[K];   // wy-blended column a-1
   float tcur[K];
 #pragma unroll
