@@ -160,3 +160,53 @@ def test_sequence_loss_excludes_large_flow():
     loss, metrics = torch_ref.sequence_loss([pred], gt, valid)
     # excluded pixel contributes 0 despite |pred-gt|=500
     assert loss.item() < 1e-6
+
+
+def test_sequence_loss_all_invalid():
+    """Zero valid pixels: the masked loss is 0. Metrics over an empty valid
+    set are NaN in the reference (train.py:61-69 .mean() on empty) and this
+    torch path mirrors that; the fused GPU kernel clamps the count to 1
+    (metrics 0) — a benign deviation on a degenerate input."""
+    import torch
+    from flowhip.ops import torch_ref
+    preds = [torch.randn(1, 2, 8, 8) for _ in range(3)]
+    gt = torch.randn(1, 2, 8, 8)
+    valid = torch.zeros(1, 8, 8)
+    loss, metrics = torch_ref.sequence_loss(preds, gt, valid, 0.8)
+    assert loss.item() == 0.0
+
+
+def test_sequence_loss_max_flow_exclusion():
+    import torch
+    from flowhip.ops import torch_ref
+    preds = [torch.zeros(1, 2, 4, 4)]
+    gt = torch.zeros(1, 2, 4, 4)
+    gt[0, 0, 0, 0] = 1000.0  # ||gt|| >= 400 -> excluded
+    valid = torch.ones(1, 4, 4)
+    loss, metrics = torch_ref.sequence_loss(preds, gt, valid, 0.8)
+    # excluded pixel contributes nothing to the (masked) loss
+    assert loss.item() == 0.0
+
+
+def test_up2x_cat_fallback_matches_eager():
+    import torch
+    import torch.nn.functional as F
+    from flowhip import ops
+    low = torch.randn(2, 3, 5, 7)   # odd sizes: CPU fallback path
+    skip = torch.randn(2, 2, 10, 14)
+    got = ops.up2x_cat(low, skip)
+    up = F.interpolate(low, size=(10, 14), mode="nearest")
+    ref = torch.cat((up, skip), 1)
+    torch.testing.assert_close(got, ref)
+
+
+def test_zero_inject_explicit_out_size():
+    import torch
+    from flowhip.ops import torch_ref
+    x = torch.arange(6.0).view(1, 1, 2, 3)
+    out = torch_ref.zero_inject(x, 2, 2, out_h=5, out_w=7)
+    assert out.shape == (1, 1, 5, 7)
+    # samples at stride 2, offset 1
+    assert out[0, 0, 1, 1] == x[0, 0, 0, 0]
+    assert out[0, 0, 3, 5] == x[0, 0, 1, 2]
+    assert out.sum() == x.sum()
