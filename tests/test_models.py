@@ -115,3 +115,32 @@ def test_model_iter_count_affects_predictions():
     with torch.no_grad():
         preds = model(img1, img2, iters=3)
     assert len(preds) == 3
+
+
+def test_ncup_unet_weights_est_builds_and_runs():
+    """--weights_est_net UNet variant (reference interp_weights_est.py:50)."""
+    import torch
+    from flowhip.config.args import default_ncup_args
+    from flowhip.models import build_model
+
+    args = default_ncup_args(model="raft_nc_dbl", small=True)
+    args.weights_est_net = "UNet"
+    torch.manual_seed(0)
+    model = build_model(args)
+    img = torch.rand(1, 3, 64, 64) * 255
+    preds = model(img, img, iters=2)
+    assert preds[-1].shape == (1, 2, 64, 64)
+
+
+def test_final_upsampling_choice_respected_optin():
+    """The factory reproduces the reference's hardcoded NConv choice by
+    default (upsampler.py:12 quirk) but honors --final_upsampling with
+    respect_choice=True."""
+    from flowhip.config.args import default_ncup_args
+    from flowhip.nn.upsampler import get_upsampler, NConvUpsampler, Bilinear
+
+    args = default_ncup_args(model="raft_nc_dbl", small=True)
+    args.final_upsampling = "Bilinear"
+    assert isinstance(get_upsampler(2, 96, args), NConvUpsampler)
+    assert isinstance(get_upsampler(2, 96, args, respect_choice=True),
+                      Bilinear)
