@@ -155,6 +155,44 @@ __global__ __launch_bounds__(LK_THREADS) void corr_lookup_bwd_kernel(
   const float* gin = cl ? gout + ((long)b * P + i) * (L * K2) + (long)l * K2
                         : gout + ((long)b * L * K2 + (long)l * K2) * P + i;
 
+  const bool interior = (x0 >= 0) & (y0 >= 0) & (x0 + K < Wl) &
+                        (y0 + K < Hl);
+  if (interior) {
+    // Fast path (mirror of the forward's): stream patch ROWS —
+    // A_j(u) = wx0*g[u][j] + wx1*g[u-1][j]; patch row j = wy0*A_j + wy1*A_{j-1}
+    // — the (K+1) stores per row are CONTIGUOUS in the map.
+    float aprev[K + 1], acur[K + 1];
+#pragma unroll
+    for (int j = 0; j <= K; ++j) {
+      if (j < K) {
+        float grow[K];
+#pragma unroll
+        for (int a = 0; a < K; ++a)
+          grow[a] = gin[(long)(a * K + j) * tap_stride];
+#pragma unroll
+        for (int u = 0; u <= K; ++u) {
+          float v = 0.0f;
+          if (u < K) v += wx0 * grow[u];
+          if (u > 0) v += wx1 * grow[u - 1];
+          acur[u] = v;
+        }
+      } else {
+#pragma unroll
+        for (int u = 0; u <= K; ++u) acur[u] = 0.0f;
+      }
+      scalar_t* dst = gmap + (long)(y0 + j) * Wl + x0;
+#pragma unroll
+      for (int u = 0; u <= K; ++u) {
+        float v = wy0 * acur[u];
+        if (j > 0) v += wy1 * aprev[u];
+        dst[u] = (scalar_t)v;
+      }
+#pragma unroll
+      for (int u = 0; u <= K; ++u) aprev[u] = acur[u];
+    }
+    return;
+  }
+
   // patch[u][j] = sum_{a,c} wx_{u-a} wy_{j-c} g[a][c]; stream over patch
   // columns u holding g columns a=u-1 (gprev) and a=u (gcur) in registers.
   float gprev[K], gcur[K];
