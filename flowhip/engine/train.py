@@ -16,14 +16,13 @@ torch.distributed:
   resume, which the reference lacks.
 """
 
+import copy
 import os
 
 import numpy as np
 import torch
 
-import os as _os
-
-if torch.cuda.is_available() and _os.environ.get("FLOWHIP_MIOPEN_FIND", "1") != "0":
+if torch.cuda.is_available() and os.environ.get("FLOWHIP_MIOPEN_FIND", "1") != "0":
     # MIOpen find mode: benchmark=True lets MIOpen search its perf-db for
     # the fastest conv solver per (static) shape during warmup
     torch.backends.cudnn.benchmark = True
@@ -88,7 +87,6 @@ def train(args):
         f"global batch {args.batch_size} must divide world size {world_size}"
     per_rank = args.batch_size // world_size
 
-    import copy
     loader_args = copy.copy(args)
     loader_args.batch_size = per_rank
     train_loader = fetch_dataloader(loader_args, distributed=world_size > 1,
