@@ -62,3 +62,64 @@ def test_logger_files(tmp_path):
     assert any("loss" in r for r in jsonl)
     assert any("validation" in r for r in jsonl)
     assert (tmp_path / "run" / "log.txt").exists()
+
+
+def test_sintel_submission_writer(tmp_path, monkeypatch):
+    """End-to-end submission path on CPU: fake test split -> .flo files,
+    warm-started second frame (reference evaluate.py:23-57)."""
+    import numpy as np
+    from PIL import Image
+
+    from flowhip.config.args import default_ncup_args
+    from flowhip.data import frame_utils
+    from flowhip.engine.evaluate import create_sintel_submission
+    from flowhip.models import build_model
+
+    for dstype in ("clean", "final"):
+        scene = tmp_path / "datasets" / "Sintel" / "test" / dstype / "seq_1"
+        scene.mkdir(parents=True)
+        for i in range(3):
+            arr = (np.random.rand(64, 96, 3) * 255).astype(np.uint8)
+            Image.fromarray(arr).save(scene / f"frame_{i:04d}.png")
+
+    monkeypatch.chdir(tmp_path)
+    torch.manual_seed(0)
+    args = default_ncup_args(model="raft_nc_dbl", small=True)
+    model = build_model(args)
+    create_sintel_submission(model, iters=2, warm_start=True,
+                             output_path=str(tmp_path / "out"))
+
+    flo = tmp_path / "out" / "clean" / "seq_1" / "frame0001.flo"
+    assert flo.exists()
+    flow = frame_utils.readFlow(str(flo))
+    assert flow.shape == (64, 96, 2)
+    assert (tmp_path / "out" / "final" / "seq_1" / "frame0002.flo").exists()
+
+
+def test_kitti_submission_writer(tmp_path, monkeypatch):
+    import numpy as np
+    from PIL import Image
+
+    from flowhip.config.args import default_ncup_args
+    from flowhip.data import frame_utils
+    from flowhip.engine.evaluate import create_kitti_submission
+    from flowhip.models import build_model
+
+    img_dir = tmp_path / "datasets" / "KITTI" / "testing" / "image_2"
+    img_dir.mkdir(parents=True)
+    for suffix in ("10", "11"):
+        arr = (np.random.rand(64, 96, 3) * 255).astype(np.uint8)
+        Image.fromarray(arr).save(img_dir / f"000000_{suffix}.png")
+
+    monkeypatch.chdir(tmp_path)
+    torch.manual_seed(0)
+    args = default_ncup_args(model="raft_nc_dbl", small=True)
+    model = build_model(args)
+    create_kitti_submission(model, iters=2,
+                            output_path=str(tmp_path / "kout"))
+
+    out = tmp_path / "kout" / "000000_10.png"
+    assert out.exists()
+    flow, valid = frame_utils.readFlowKITTI(str(out))
+    assert flow.shape == (64, 96, 2)
+    assert valid.min() == 1  # submission marks everything valid
