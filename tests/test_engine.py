@@ -123,3 +123,38 @@ def test_kitti_submission_writer(tmp_path, monkeypatch):
     flow, valid = frame_utils.readFlowKITTI(str(out))
     assert flow.shape == (64, 96, 2)
     assert valid.min() == 1  # submission marks everything valid
+
+
+def test_demo_cli_end_to_end(tmp_path):
+    """demo.py over a directory of frames writes flow visualizations."""
+    import subprocess
+    import sys
+
+    import numpy as np
+    from PIL import Image
+
+    from flowhip.config.args import default_ncup_args
+    from flowhip.engine import checkpoints
+    from flowhip.models import build_model
+
+    frames = tmp_path / "frames"
+    frames.mkdir()
+    for i in range(3):
+        arr = (np.random.rand(64, 96, 3) * 255).astype(np.uint8)
+        Image.fromarray(arr).save(frames / f"f{i:03d}.png")
+
+    torch.manual_seed(0)
+    args = default_ncup_args(model="raft_nc_dbl", small=True)
+    model = build_model(args)
+    ckpt = tmp_path / "model.pth"
+    checkpoints.save_weights(model, str(ckpt))
+
+    out = tmp_path / "viz"
+    r = subprocess.run(
+        [sys.executable, "demo.py", "--model", str(ckpt), "--path",
+         str(frames), "--out", str(out), "--arch", "raft_nc_dbl",
+         "--small"],
+        capture_output=True, text=True, timeout=500)
+    assert r.returncode == 0, r.stderr[-2000:]
+    pngs = list(out.glob("*.png"))
+    assert len(pngs) == 2  # consecutive pairs
