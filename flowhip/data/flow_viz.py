@@ -88,9 +88,13 @@ def flow_to_image(flow):
     u = flow[:, :, 0]
     v = flow[:, :, 1]
 
-    idxUnknown = (np.abs(u) > UNKNOWN_FLOW_THRESH) | (np.abs(v) > UNKNOWN_FLOW_THRESH)
-    u[idxUnknown] = 0
-    v[idxUnknown] = 0
+    # non-finite values count as unknown flow (NaN would otherwise poison
+    # the color-wheel index cast)
+    idxUnknown = ((np.abs(u) > UNKNOWN_FLOW_THRESH)
+                  | (np.abs(v) > UNKNOWN_FLOW_THRESH)
+                  | ~np.isfinite(u) | ~np.isfinite(v))
+    u = np.where(idxUnknown, 0.0, u)
+    v = np.where(idxUnknown, 0.0, v)
 
     rad = np.sqrt(u ** 2 + v ** 2)
     maxrad = max(-1, np.max(rad))

@@ -139,8 +139,10 @@ def test_demo_cli_end_to_end(tmp_path):
 
     frames = tmp_path / "frames"
     frames.mkdir()
+    # >=128 px: the 4-level pyramid needs level 3 >= 2 px (the reference's
+    # grid_sample(align_corners=True) divides by size-1 on a 1x1 level)
     for i in range(3):
-        arr = (np.random.rand(64, 96, 3) * 255).astype(np.uint8)
+        arr = (np.random.rand(128, 128, 3) * 255).astype(np.uint8)
         Image.fromarray(arr).save(frames / f"f{i:03d}.png")
 
     torch.manual_seed(0)
