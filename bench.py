@@ -25,9 +25,11 @@ import torch
 
 import os as _os
 
-if torch.cuda.is_available() and _os.environ.get("FLOWHIP_MIOPEN_FIND", "1") != "0":
-    # MIOpen find mode: benchmark=True lets MIOpen search its perf-db for
-    # the fastest conv solver per (static) shape during warmup
+if torch.cuda.is_available() and _os.environ.get("FLOWHIP_MIOPEN_FIND", "0") == "1":
+    # MIOpen benchmark/find mode, off by default: the A/B measured no
+    # steady-state difference (bench10: 30.80 vs 30.74 pairs/s), and with
+    # one process per GPU the concurrent find's user-db file locking only
+    # adds warmup wall time and trace noise
     torch.backends.cudnn.benchmark = True
 
 from flowhip.config.args import default_ncup_args
