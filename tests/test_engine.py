@@ -160,3 +160,24 @@ def test_demo_cli_end_to_end(tmp_path):
     assert r.returncode == 0, r.stderr[-2000:]
     pngs = list(out.glob("*.png"))
     assert len(pngs) == 2  # consecutive pairs
+
+
+def test_pmc_summarize_tool(tmp_path):
+    """tools/pmc_summarize.py aggregates rocprofv3 counter rows per kernel."""
+    import subprocess
+    import sys
+
+    csv_path = tmp_path / "counters.csv"
+    csv_path.write_text(
+        "Kernel_Name,Counter_Name,Counter_Value,Dispatch_Id\n"
+        "kA,SQ_WAVE_CYCLES,100,1\n"
+        "kA,SQ_WAVE_CYCLES,50,2\n"
+        "kA,SQ_WAIT_ANY,30,1\n"
+        "kB,SQ_WAVE_CYCLES,10,3\n")
+    r = subprocess.run([sys.executable, "tools/pmc_summarize.py",
+                        str(csv_path)], capture_output=True, text=True,
+                       timeout=120)
+    assert r.returncode == 0, r.stderr
+    lines = r.stdout.strip().splitlines()
+    assert lines[0].startswith("Kernel,Dispatches")
+    assert any(row.startswith("kA,2,") and ",150" in row for row in lines)
