@@ -1,0 +1,140 @@
+"""Cross-checks of the torch reference ops (flowhip/ops/torch_ref.py)
+against independent loop-based NumPy implementations on fuzzed small
+shapes. torch_ref is the numerical oracle for every HIP kernel, so it gets
+its own independent verification here (SURVEY.md §4.2 item 2 contracts,
+written from the math, not from either implementation).
+"""
+
+import numpy as np
+import pytest
+import torch
+from hypothesis import given, settings, strategies as st
+
+from flowhip.ops import torch_ref
+
+
+def naive_corr_volume(f1, f2):
+    """C[b, i, j] = <f1[b,:,i], f2[b,:,j]> / sqrt(D), loops only."""
+    B, D, H, W = f1.shape
+    P = H * W
+    out = np.zeros((B, P, P), dtype=np.float64)
+    f1f = f1.reshape(B, D, P)
+    f2f = f2.reshape(B, D, P)
+    for b in range(B):
+        for i in range(P):
+            for j in range(P):
+                out[b, i, j] = np.dot(f1f[b, :, i], f2f[b, :, j])
+    return out / np.sqrt(D)
+
+
+def naive_bilinear(map2d, x, y):
+    """align_corners=True grid_sample semantics on pixel coords with zero
+    padding, scalar implementation."""
+    H, W = map2d.shape
+    x0, y0 = int(np.floor(x)), int(np.floor(y))
+    wx1, wy1 = x - x0, y - y0
+    v = 0.0
+    for (xi, wx) in ((x0, 1 - wx1), (x0 + 1, wx1)):
+        for (yi, wy) in ((y0, 1 - wy1), (y0 + 1, wy1)):
+            if 0 <= xi < W and 0 <= yi < H:
+                v += wx * wy * map2d[yi, xi]
+    return v
+
+
+@settings(max_examples=10, deadline=None)
+@given(st.integers(2, 4), st.integers(2, 4), st.integers(2, 5),
+       st.integers(0, 2 ** 31 - 1))
+def test_corr_volume_matches_naive(h, w, d, seed):
+    rng = np.random.default_rng(seed)
+    f1 = rng.standard_normal((1, d, h, w)).astype(np.float32)
+    f2 = rng.standard_normal((1, d, h, w)).astype(np.float32)
+    got = torch_ref.corr_volume(torch.from_numpy(f1), torch.from_numpy(f2))
+    # torch_ref returns (B*P, 1, H, W): volume C[i] viewed as image-2 maps
+    got = got.reshape(1, h * w, h * w).numpy()
+    ref = naive_corr_volume(f1, f2)
+    np.testing.assert_allclose(got, ref, atol=1e-5)
+
+
+@settings(max_examples=10, deadline=None)
+@given(st.integers(6, 10), st.integers(6, 12), st.integers(0, 2 ** 31 - 1))
+def test_corr_lookup_matches_naive_taps(h, w, seed):
+    """Every output channel of the fused lookup equals a scalar bilinear
+    sample at coords/2^l + (dx, dy) — x-offset-major channel order."""
+    rng = np.random.default_rng(seed)
+    radius, levels = 2, 2
+    P = h * w
+    pyramid = [torch.from_numpy(
+        rng.standard_normal((P, 1, h >> l, w >> l)).astype(np.float32))
+        for l in range(levels)]
+    coords = torch.from_numpy(
+        (rng.random((1, 2, h, w)) * [[[[w]], [[h]]]]).astype(np.float32))
+
+    out = torch_ref.corr_lookup(pyramid, coords, radius).numpy()
+    K = 2 * radius + 1
+
+    for _ in range(12):  # spot-check random taps
+        i = rng.integers(P)
+        l = rng.integers(levels)
+        a = rng.integers(K)
+        c = rng.integers(K)
+        y, x = divmod(int(i), w)
+        cx = coords[0, 0, y, x].item() / (1 << l) + (a - radius)
+        cy = coords[0, 1, y, x].item() / (1 << l) + (c - radius)
+        ref = naive_bilinear(pyramid[l][i, 0].numpy(), cx, cy)
+        ch = l * K * K + a * K + c
+        np.testing.assert_allclose(out[0, ch, y, x], ref, atol=1e-5)
+
+
+@settings(max_examples=10, deadline=None)
+@given(st.integers(3, 8), st.integers(3, 8), st.integers(1, 2),
+       st.integers(1, 2), st.sampled_from([1, 3]),
+       st.integers(0, 2 ** 31 - 1))
+def test_nconv2d_matches_naive(h, w, ci, co, k, seed):
+    rng = np.random.default_rng(seed)
+    data = rng.standard_normal((1, ci, h, w)).astype(np.float32)
+    conf = rng.random((1, ci, h, w)).astype(np.float32)
+    wt = (rng.random((co, ci, k, k)) + 0.1).astype(np.float32)
+    pad = k // 2
+
+    out, cout = torch_ref.nconv2d(
+        torch.from_numpy(data), torch.from_numpy(conf), torch.from_numpy(wt),
+        padding=pad, prop_conf=True)
+
+    for _ in range(8):
+        o = rng.integers(co)
+        y = rng.integers(h)
+        x = rng.integers(w)
+        nomin = denom = 0.0
+        for c in range(ci):
+            for ky in range(k):
+                for kx in range(k):
+                    yy, xx = y + ky - pad, x + kx - pad
+                    if 0 <= yy < h and 0 <= xx < w:
+                        cv = conf[0, c, yy, xx]
+                        nomin += wt[o, c, ky, kx] * data[0, c, yy, xx] * cv
+                        denom += wt[o, c, ky, kx] * cv
+        np.testing.assert_allclose(out[0, o, y, x].item(),
+                                   nomin / (denom + 1e-20), atol=1e-4)
+        np.testing.assert_allclose(cout[0, o, y, x].item(),
+                                   denom / wt[o].sum(), atol=1e-4)
+
+
+@settings(max_examples=8, deadline=None)
+@given(st.integers(2, 4), st.integers(2, 4), st.integers(0, 2 ** 31 - 1))
+def test_convex_upsample_partition_of_unity(h, w, seed):
+    """Constant flow fields are reproduced exactly up to the x`factor`
+    magnitude scale (reference raft.py:79 multiplies the flow by the
+    upsampling factor inside the op; softmax weights form a convex
+    combination, so a constant field stays constant)."""
+    rng = np.random.default_rng(seed)
+    flow = np.broadcast_to(
+        rng.standard_normal((1, 2, 1, 1)).astype(np.float32),
+        (1, 2, h, w)).copy()
+    mask = rng.standard_normal((1, 576, h, w)).astype(np.float32)
+    out = torch_ref.convex_upsample(torch.from_numpy(flow),
+                                    torch.from_numpy(mask), 8)
+    assert out.shape == (1, 2, 8 * h, 8 * w)
+    np.testing.assert_allclose(out.numpy(),
+                               np.broadcast_to(8.0 * flow[:, :, :1, :1],
+                                               (1, 2, 8 * h, 8 * w)),
+                               atol=1e-4)
