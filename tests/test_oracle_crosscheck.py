@@ -120,12 +120,13 @@ def test_nconv2d_matches_naive(h, w, ci, co, k, seed):
 
 
 @settings(max_examples=8, deadline=None)
-@given(st.integers(2, 4), st.integers(2, 4), st.integers(0, 2 ** 31 - 1))
+@given(st.integers(3, 5), st.integers(3, 5), st.integers(0, 2 ** 31 - 1))
 def test_convex_upsample_partition_of_unity(h, w, seed):
-    """Constant flow fields are reproduced exactly up to the x`factor`
-    magnitude scale (reference raft.py:79 multiplies the flow by the
-    upsampling factor inside the op; softmax weights form a convex
-    combination, so a constant field stays constant)."""
+    """INTERIOR coarse pixels of a constant field reproduce factor*flow
+    exactly: the softmax weights are a convex combination over the 3x3
+    neighborhood and the op scales flow by `factor` internally
+    (reference raft.py:79). Border pixels mix the unfold's zero padding —
+    same as the reference — so only the interior is checked."""
     rng = np.random.default_rng(seed)
     flow = np.broadcast_to(
         rng.standard_normal((1, 2, 1, 1)).astype(np.float32),
@@ -134,7 +135,7 @@ def test_convex_upsample_partition_of_unity(h, w, seed):
     out = torch_ref.convex_upsample(torch.from_numpy(flow),
                                     torch.from_numpy(mask), 8)
     assert out.shape == (1, 2, 8 * h, 8 * w)
-    np.testing.assert_allclose(out.numpy(),
-                               np.broadcast_to(8.0 * flow[:, :, :1, :1],
-                                               (1, 2, 8 * h, 8 * w)),
-                               atol=1e-4)
+    interior = out[:, :, 8:-8, 8:-8].numpy()
+    np.testing.assert_allclose(
+        interior, np.broadcast_to(8.0 * flow[:, :, :1, :1], interior.shape),
+        atol=1e-4)
