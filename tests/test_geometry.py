@@ -47,3 +47,27 @@ def test_upflow8_shape_and_scale():
     up = upflow8(flow)
     assert up.shape == (1, 2, 32, 40)
     assert torch.allclose(up, 8 * torch.ones_like(up))
+
+
+def test_forward_interpolate_zero_flow_identity():
+    """Warm-start splat: zero flow maps every pixel onto itself, so the
+    interpolated field is (numerically) zero (reference utils.py:28-56)."""
+    import torch
+    from flowhip.utils.geometry import forward_interpolate
+    flow = torch.zeros(2, 12, 16)
+    out = forward_interpolate(flow)
+    assert out.shape == (2, 12, 16)
+    assert torch.allclose(out, torch.zeros_like(out), atol=1e-5)
+
+
+def test_forward_interpolate_constant_shift():
+    """A constant integer shift moves the field; in-range target pixels keep
+    the constant value (nearest-neighbor griddata fill elsewhere)."""
+    import torch
+    from flowhip.utils.geometry import forward_interpolate
+    flow = torch.zeros(2, 10, 14)
+    flow[0] = 3.0  # shift x by +3
+    out = forward_interpolate(flow)
+    assert out.shape == (2, 10, 14)
+    # splatted positions carry the same constant flow
+    assert torch.allclose(out[0], torch.full_like(out[0], 3.0), atol=1e-4)
