@@ -40,8 +40,11 @@ def module_classes_to_dict(module, include_classes="*", exclude_classes=()):
     if isinstance(exclude_classes, str):
         exclude_classes = [exclude_classes]
 
+    # only classes DEFINED in the module: imported helpers (e.g. conv
+    # wrappers) must not leak into the CLI choice surface
     items = {name: getattr(module, name) for name in dir(module)
-             if inspect.isclass(getattr(module, name))}
+             if inspect.isclass(getattr(module, name))
+             and getattr(module, name).__module__ == module.__name__}
 
     matched = set(itertools.chain.from_iterable(
         fnmatch.filter(items.keys(), pat) for pat in include_classes))

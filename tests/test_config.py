@@ -82,3 +82,52 @@ def test_eval_parser():
     args = parser.parse_args(argv)
     assert args.dataset == "sintel"
     assert args.weights_est_net_num_ch == [64, 32]
+
+
+def test_eval_parser_accepts_reference_eval_script():
+    """The canonical eval invocation (eval_raft_nc_sintel.sh flags)."""
+    import shlex
+    argv = shlex.split("""
+    --model checkpoints/raft_nc_sintel/final_model.pth
+    --dataset sintel
+    --final_upsampling=NConvUpsampler --final_upsampling_scale=4
+    --final_upsampling_use_data_for_guidance=True
+    --final_upsampling_channels_to_batch=True
+    --interp_net=NConvUNet --interp_net_channels_multiplier=2
+    --interp_net_num_downsampling=1
+    --weights_est_net=Simple --weights_est_net_num_ch=[64,32]
+    --weights_est_net_filter_sz=[3,3,1]
+    """)
+    parser = build_eval_parser(argv=argv)
+    args = finalize_args(parser.parse_args(argv))
+    assert args.dataset == "sintel"
+    assert args.interp_net_channels_multiplier == 2
+    assert args.weights_est_net_num_ch == [64, 32]
+
+
+def test_weights_est_choices_are_module_classes_only():
+    """The reflective --weights_est_net choices expose only classes defined
+    in interp_weights_est (imported helpers like the conv wrapper must not
+    leak into the CLI surface)."""
+    parser = build_train_parser(argv=[])
+    for action in parser._actions:
+        if action.dest == "weights_est_net":
+            assert "FusedConv2d" not in (action.choices or [])
+            assert "Simple" in action.choices and "UNet" in action.choices
+            break
+    else:
+        raise AssertionError("--weights_est_net flag missing")
+
+
+def test_intlist_and_bool_flag_parsing():
+    """str2intlist / str2bool semantics of the reflective system
+    (reference utils/args.py:166-175)."""
+    argv = ["--weights_est_net_num_ch", "[8,4]",
+            "--interp_net_use_bias", "False",
+            "--interp_net_shared_encoder", "true"]
+    parser = build_train_parser(argv=argv)
+    args = finalize_args(parser.parse_args(
+        ["--name", "t", "--model", "raft_nc_dbl"] + argv))
+    assert args.weights_est_net_num_ch == [8, 4]
+    assert args.interp_net_use_bias is False
+    assert args.interp_net_shared_encoder is True
