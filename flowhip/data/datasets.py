@@ -149,11 +149,15 @@ class FlyingThings3D(FlowDataset):
                  dstype="frames_cleanpass", load_compressed=False):
         super().__init__(aug_params)
 
+        # Reference defect fixed (core/datasets.py:144-146): the reference
+        # appends "_webp" to dstype INSIDE the direction loop, so the second
+        # pass globs "<dstype>_webp_webp" and the into_past pairs silently
+        # vanish in compressed mode. Appended once here.
+        if load_compressed:
+            dstype += "_webp"
+
         for cam in ["left"]:
             for direction in ["into_future", "into_past"]:
-                if load_compressed:
-                    dstype += "_webp"
-
                 image_dirs = sorted(glob(osp.join(root, dstype, "TRAIN/*/*")))
                 image_dirs = sorted([osp.join(f, cam) for f in image_dirs])
 

@@ -143,3 +143,36 @@ def test_kitti_dataset_sparse_from_disk(tmp_path):
     assert v.shape == (60, 80)
     assert 0 < v.sum() < 60 * 80
     assert abs(v.numpy().sum() - valid.sum()) == 0
+
+
+def test_flyingthings_compressed_from_disk(tmp_path):
+    """Compressed FT3D (webp images + npz flow; datasets.py:138-167).
+    Also covers the reference defect fix: BOTH flow directions must be
+    listed in compressed mode (the reference's in-loop `dstype += '_webp'`
+    emptied the into_past glob)."""
+    from PIL import Image
+    from flowhip.data.datasets import FlyingThings3D
+
+    root = tmp_path / "FT3D"
+    img_dir = root / "frames_cleanpass_webp" / "TRAIN" / "A" / "0000" / "left"
+    img_dir.mkdir(parents=True)
+    for i in range(3):
+        arr = (np.random.rand(32, 48, 3) * 255).astype(np.uint8)
+        Image.fromarray(arr).save(img_dir / f"{i:04d}.webp")
+    for direction in ("into_future", "into_past"):
+        fdir = (root / "optical_flow" / "TRAIN" / "A" / "0000" / direction
+                / "left")
+        fdir.mkdir(parents=True)
+        for i in range(3):
+            # compressed-FT3D npz layout: key 'optical_flow', CHW
+            # (frame_utils read_gen transposes to HWC — ref :137-139)
+            np.savez(fdir / f"{i:04d}.npz",
+                     optical_flow=np.random.randn(2, 32, 48)
+                     .astype(np.float32))
+
+    ds = FlyingThings3D(aug_params=None, root=str(root),
+                        load_compressed=True)
+    # 2 consecutive pairs per direction
+    assert len(ds) == 4
+    img1, img2, flow, valid = ds[0]
+    assert img1.shape == (3, 32, 48) and flow.shape == (2, 32, 48)
