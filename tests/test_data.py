@@ -176,3 +176,28 @@ def test_flyingthings_compressed_from_disk(tmp_path):
     assert len(ds) == 4
     img1, img2, flow, valid = ds[0]
     assert img1.shape == (3, 32, 48) and flow.shape == (2, 32, 48)
+
+
+def test_flying_chairs_split_table(tmp_path):
+    """chairs_split.txt drives the canonical train/val split
+    (datasets.py:121-137; the table ships with the repo)."""
+    import os
+    from PIL import Image
+    from flowhip.data import frame_utils
+    from flowhip.data.datasets import FlyingChairs
+
+    assert os.path.exists("chairs_split.txt")
+    root = tmp_path / "chairs"
+    root.mkdir()
+    # first 4 entries of the real table are 1 1 1 1 (training)
+    for i in range(1, 5):
+        for k in (1, 2):
+            arr = (np.random.rand(24, 32, 3) * 255).astype(np.uint8)
+            Image.fromarray(arr).save(root / f"{i:05d}_img{k}.png")
+        frame_utils.writeFlow(str(root / f"{i:05d}_flow.flo"),
+                              np.random.randn(24, 32, 2).astype(np.float32))
+
+    tr = FlyingChairs(aug_params=None, split="training", root=str(root))
+    va = FlyingChairs(aug_params=None, split="validation", root=str(root))
+    assert len(tr) + len(va) == 4
+    assert len(tr) == 4  # table rows 0-3 are all split 1
