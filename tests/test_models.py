@@ -144,3 +144,29 @@ def test_final_upsampling_choice_respected_optin():
     assert isinstance(get_upsampler(2, 96, args), NConvUpsampler)
     assert isinstance(get_upsampler(2, 96, args, respect_choice=True),
                       Bilinear)
+
+
+def test_enforce_pos_weight_p_semantics():
+    """EnforcePos: the registered parameter is `weight_p`; the effective
+    weight is softplus(weight_p, beta=10) recomputed per forward
+    (nconv_modules.py:218-264 — checkpoint key contract)."""
+    import torch
+    import torch.nn.functional as F
+    from flowhip.nn.nconv import NConv2d
+
+    m = NConv2d(1, 2, (3, 3), stride=(1, 1), pos_fn="SoftPlus", groups=1,
+                bias=False)
+    names = dict(m.named_parameters())
+    assert any(n.endswith("weight_p") for n in names), list(names)
+    wp = [p for n, p in names.items() if n.endswith("weight_p")][0]
+
+    data = torch.randn(1, 1, 8, 8)
+    conf = torch.rand(1, 1, 8, 8)
+    out, cout = m((data, conf))
+    # conf propagation stays within (0, 1] for conf in [0, 1]
+    assert torch.isfinite(out).all()
+    assert (cout >= 0).all() and (cout <= 1.0 + 1e-5).all()
+
+    # the effective weight is positive everywhere
+    eff = F.softplus(wp, beta=10)
+    assert (eff > 0).all()
