@@ -139,3 +139,61 @@ def test_convex_upsample_partition_of_unity(h, w, seed):
     np.testing.assert_allclose(
         interior, np.broadcast_to(8.0 * flow[:, :, :1, :1], interior.shape),
         atol=1e-4)
+
+
+@settings(max_examples=10, deadline=None)
+@given(st.integers(4, 9), st.integers(4, 9), st.integers(0, 2 ** 31 - 1))
+def test_corr_pyramid_matches_naive(h, w, seed):
+    """Each level is a 2x2 mean of the previous (floor semantics: trailing
+    odd row/col dropped, as F.avg_pool2d(2,2))."""
+    rng = np.random.default_rng(seed)
+    corr = rng.standard_normal((3, 1, h, w)).astype(np.float32)
+    levels = torch_ref.corr_pyramid(torch.from_numpy(corr), 3)
+    assert len(levels) == 3
+    cur = corr
+    for lvl in levels[1:]:
+        hh, ww = cur.shape[-2] // 2, cur.shape[-1] // 2
+        ref = np.zeros((3, 1, hh, ww), dtype=np.float64)
+        for y in range(hh):
+            for x in range(ww):
+                ref[:, :, y, x] = cur[:, :, 2 * y:2 * y + 2,
+                                      2 * x:2 * x + 2].mean(axis=(-1, -2))
+        np.testing.assert_allclose(lvl.numpy(), ref, atol=1e-5)
+        cur = lvl.numpy()
+
+
+@settings(max_examples=10, deadline=None)
+@given(st.integers(4, 10), st.integers(4, 10), st.integers(0, 2 ** 31 - 1))
+def test_conf_pool_matches_naive(h, w, seed):
+    """data_ds keeps the value at the argmax-confidence position of each
+    2x2 block (row-major first-max ties); conf_ds = max/4."""
+    rng = np.random.default_rng(seed)
+    data = rng.standard_normal((1, 2, h, w)).astype(np.float32)
+    conf = rng.random((1, 2, h, w)).astype(np.float32)
+    dds, cds = torch_ref.conf_pool(torch.from_numpy(data),
+                                   torch.from_numpy(conf))
+    for c in range(2):
+        for y in range(h // 2):
+            for x in range(w // 2):
+                block_c = conf[0, c, 2 * y:2 * y + 2, 2 * x:2 * x + 2]
+                block_d = data[0, c, 2 * y:2 * y + 2, 2 * x:2 * x + 2]
+                flat = block_c.reshape(-1)
+                arg = int(np.argmax(flat))  # first max, row-major
+                np.testing.assert_allclose(cds[0, c, y, x].item(),
+                                           flat[arg] / 4, atol=1e-6)
+                np.testing.assert_allclose(dds[0, c, y, x].item(),
+                                           block_d.reshape(-1)[arg],
+                                           atol=1e-6)
+
+
+@settings(max_examples=10, deadline=None)
+@given(st.integers(2, 6), st.integers(2, 6), st.sampled_from([2, 4]),
+       st.integers(0, 2 ** 31 - 1))
+def test_zero_inject_matches_naive(h, w, s, seed):
+    rng = np.random.default_rng(seed)
+    x = rng.standard_normal((1, 2, h, w)).astype(np.float32)
+    out = torch_ref.zero_inject(torch.from_numpy(x), s, s).numpy()
+    assert out.shape == (1, 2, h * s, w * s)
+    ref = np.zeros_like(out)
+    ref[:, :, s // 2::s, s // 2::s] = x
+    np.testing.assert_allclose(out, ref)
