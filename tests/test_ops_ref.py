@@ -210,3 +210,26 @@ def test_zero_inject_explicit_out_size():
     assert out[0, 0, 1, 1] == x[0, 0, 0, 0]
     assert out[0, 0, 3, 5] == x[0, 0, 1, 2]
     assert out.sum() == x.sum()
+
+
+def test_extension_symbol_surface():
+    """flowhip._C must expose every kernel entry the op layer dispatches to
+    (catches bindings drift at CPU-test time, before any GPU run)."""
+    import flowhip._C as C
+    expected = [
+        "bgemm_nt", "corr_lookup_fwd", "corr_lookup_bwd",
+        "corr_pyramid_fwd", "corr_pyramid_bwd",
+        "convex_up_fwd", "convex_up_bwd",
+        "nconv_fwd", "nconv_bwd", "nconv_bwd_prep",
+        "conf_pool_fwd", "conf_pool_bwd",
+        "zero_inject_fwd", "zero_inject_bwd",
+        "gru_gate1_fwd", "gru_gate1_bwd", "gru_gate2_fwd", "gru_gate2_bwd",
+        "seq_loss_fwd", "seq_loss_bwd",
+        "conv_gemm_fwd", "conv_gemm_fwd2", "conv_gemm_wrw", "conv_gemm_pack",
+        "instnorm_cl_fwd", "instnorm_cl_bwd",
+        "transpose_cast_bf16", "col_sum_bf16",
+        "area_up2x_fwd", "area_up2x_bwd", "up2x_cat_fwd",
+        "packernel_fwd", "packernel_bwd", "pacconv_fwd", "pacconv_bwd",
+    ]
+    missing = [name for name in expected if not hasattr(C, name)]
+    assert not missing, missing
