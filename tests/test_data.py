@@ -201,3 +201,84 @@ def test_flying_chairs_split_table(tmp_path):
     va = FlyingChairs(aug_params=None, split="validation", root=str(root))
     assert len(tr) + len(va) == 4
     assert len(tr) == 4  # table rows 0-3 are all split 1
+
+
+def test_augmentor_hflip_negates_u():
+    """h-flip mirrors columns and negates the u component (reference
+    augmentor.py:91-95). Probabilities pinned to make the path
+    deterministic; input = crop+1 so the crop offset draw is always 0."""
+    np.random.seed(3)
+    aug = FlowAugmentor(crop_size=(32, 32), do_flip=True)
+    aug.spatial_aug_prob = 0.0   # no resize
+    aug.stretch_prob = 0.0
+    aug.h_flip_prob = 1.0
+    aug.v_flip_prob = 0.0
+    h, w = 33, 33
+    img1 = np.random.randint(0, 255, (h, w, 3), dtype=np.uint8)
+    img2 = np.random.randint(0, 255, (h, w, 3), dtype=np.uint8)
+    flow = np.random.randn(h, w, 2).astype(np.float32)
+    _, _, f = aug.spatial_transform(img1.copy(), img2.copy(), flow.copy())
+    exp = (flow[:, ::-1] * [-1.0, 1.0])[:32, :32]
+    np.testing.assert_allclose(f, exp)
+
+
+def test_augmentor_vflip_negates_v():
+    np.random.seed(4)
+    aug = FlowAugmentor(crop_size=(32, 32), do_flip=True)
+    aug.spatial_aug_prob = 0.0
+    aug.stretch_prob = 0.0
+    aug.h_flip_prob = 0.0
+    aug.v_flip_prob = 1.0
+    h, w = 33, 33
+    img1 = np.random.randint(0, 255, (h, w, 3), dtype=np.uint8)
+    img2 = np.random.randint(0, 255, (h, w, 3), dtype=np.uint8)
+    flow = np.random.randn(h, w, 2).astype(np.float32)
+    _, _, f = aug.spatial_transform(img1.copy(), img2.copy(), flow.copy())
+    exp = (flow[::-1, :] * [1.0, -1.0])[:32, :32]
+    np.testing.assert_allclose(f, exp)
+
+
+def test_sparse_resize_identity_keeps_valid_flow():
+    """resize_sparse_flow_map at scale 1 re-rasterizes losslessly except
+    row/col 0, which the validity test (x>0, y>0 — reference
+    augmentor.py:180) intentionally drops."""
+    np.random.seed(5)
+    aug = SparseFlowAugmentor(crop_size=(8, 8))
+    h, w = 20, 24
+    flow = np.random.randn(h, w, 2).astype(np.float32)
+    valid = (np.random.rand(h, w) > 0.4).astype(np.float32)
+    f1, v1 = aug.resize_sparse_flow_map(flow, valid, fx=1.0, fy=1.0)
+    assert f1.shape == (h, w, 2) and v1.shape == (h, w)
+    np.testing.assert_array_equal(v1[0, :], 0)
+    np.testing.assert_array_equal(v1[:, 0], 0)
+    inner_valid = valid[1:, 1:].astype(bool)
+    np.testing.assert_array_equal(v1[1:, 1:].astype(bool), inner_valid)
+    np.testing.assert_allclose(f1[1:, 1:][inner_valid], flow[1:, 1:][inner_valid])
+
+
+def test_sparse_resize_scales_flow_values():
+    """Scaling the grid by f scales flow vectors by f (pixel units)."""
+    aug = SparseFlowAugmentor(crop_size=(8, 8))
+    h, w = 10, 12
+    flow = np.ones((h, w, 2), dtype=np.float32) * [3.0, -2.0]
+    valid = np.ones((h, w), dtype=np.float32)
+    f2, v2 = aug.resize_sparse_flow_map(flow, valid, fx=2.0, fy=2.0)
+    assert f2.shape == (2 * h, 2 * w, 2)
+    got = f2[v2.astype(bool)]
+    np.testing.assert_allclose(got, np.broadcast_to([6.0, -4.0], got.shape))
+
+
+def test_eraser_touches_only_img2():
+    np.random.seed(6)
+    aug = FlowAugmentor(crop_size=(32, 32))
+    aug.eraser_aug_prob = 1.0
+    h, w = 120, 140
+    img1 = np.random.randint(0, 255, (h, w, 3), dtype=np.uint8)
+    img2 = np.random.randint(0, 255, (h, w, 3), dtype=np.uint8)
+    a, b = aug.eraser_transform(img1.copy(), img2.copy())
+    np.testing.assert_array_equal(a, img1)          # img1 untouched
+    changed = (b != img2).any(axis=-1)
+    assert changed.any()                            # a rectangle was erased
+    mean_color = np.mean(img2.reshape(-1, 3), axis=0)
+    np.testing.assert_allclose(
+        b[changed], np.broadcast_to(mean_color, b[changed].shape), atol=1.0)
