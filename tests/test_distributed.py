@@ -98,3 +98,32 @@ def test_distributed_helpers_single_process():
     assert rank == 0 and world == 1
     assert distributed.is_main()
     distributed.barrier()  # no-op
+
+
+@pytest.mark.timeout(900)
+def test_bench_contract_world_size_2(tmp_path):
+    """Launch bench.py exactly the way the driver does (torchrun, nnodes=1,
+    2 ranks — gloo on CPU) and validate the one-JSON-line contract: whole-job
+    value, max-over-ranks timing, n_gpus/steps/warmup/config fields."""
+    import json
+    import subprocess
+    import sys
+
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    cmd = [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+           "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+           "--master-port", "29713", os.path.join(repo, "bench.py"),
+           "--gpus", "2", "--steps", "2", "--warmup", "1"]
+    out = subprocess.run(cmd, capture_output=True, text=True, timeout=840,
+                         cwd=repo).stdout
+    lines = [l for l in out.splitlines() if l.startswith("{")]
+    assert len(lines) == 1, f"expected exactly one JSON line, got: {out!r}"
+    rec = json.loads(lines[0])
+    assert rec["metric"] == "train_image_pairs_per_sec"
+    assert rec["steps"] == 2 and rec["warmup"] == 1
+    assert rec["value"] > 0 and rec["ms_per_step"] > 0
+    assert rec["higher_is_better"] is True and rec["scaling"] == "weak"
+    assert rec["data"] == "synthetic"
+    # CPU plumbing config: n_gpus 0, world-size-2 global batch
+    assert rec["n_gpus"] == 0
+    assert rec["config"]["global_batch"] == 2
