@@ -1,0 +1,126 @@
+"""Direct numerical parity against the reference implementation.
+
+These tests import the PUBLIC reference snapshot mounted at /root/reference
+(read-only) and compare it to this framework end-to-end on CPU: identical
+state-dict surface (keys + shapes), identical forward outputs under shared
+weights, and identical gradients — the strongest form of the SURVEY.md §4.2
+"vs reference" contract. Skipped wherever the reference isn't mounted (e.g.
+on GPU boxes, which only receive this repo).
+
+No reference code is copied or adapted here; it is imported as an external
+oracle, exactly like the plain-PyTorch oracles in test_oracle_crosscheck.py.
+"""
+
+import os
+import sys
+
+import pytest
+import torch
+
+REF_CORE = "/root/reference/core"
+
+pytestmark = pytest.mark.skipif(
+    not os.path.isdir(REF_CORE), reason="reference snapshot not mounted")
+
+
+@pytest.fixture(scope="module")
+def ref_modules():
+    sys.path.insert(0, REF_CORE)
+    try:
+        import raft as ref_raft
+        import raft_nc_dbl as ref_nc_dbl
+        yield ref_raft, ref_nc_dbl
+    finally:
+        sys.path.remove(REF_CORE)
+
+
+def _ncup_args(**kw):
+    from flowhip.config.args import default_ncup_args
+    args = default_ncup_args(model="raft_nc_dbl", **kw)
+    args.mixed_precision = False
+    return args
+
+
+def test_ncup_state_dict_surface_matches_reference(ref_modules):
+    """Every parameter/buffer key and shape is identical to the reference
+    RAFT_NC_DBL (checkpoints interchange 1:1)."""
+    from flowhip.models import build_model
+    _, ref_nc_dbl = ref_modules
+
+    args = _ncup_args()
+    ours = {k: tuple(v.shape) for k, v in build_model(args).state_dict().items()}
+    ref = {k: tuple(v.shape) for k, v in ref_nc_dbl.RAFT(args).state_dict().items()}
+    assert ours == ref
+    assert len(ours) > 150  # full model, not a stub
+
+
+@pytest.mark.parametrize("small", [True, False])
+def test_raft_state_dict_surface_matches_reference(ref_modules, small):
+    from flowhip.config.args import default_ncup_args
+    from flowhip.models import build_model
+    ref_raft, _ = ref_modules
+
+    args = default_ncup_args(model="raft", small=small)
+    args.mixed_precision = False
+    ours = {k: tuple(v.shape) for k, v in build_model(args).state_dict().items()}
+    ref = {k: tuple(v.shape) for k, v in ref_raft.RAFT(args).state_dict().items()}
+    assert ours == ref
+
+
+@pytest.mark.timeout(900)
+def test_ncup_forward_and_grads_match_reference(ref_modules):
+    """Shared weights => same flow predictions (<=1e-4) and same gradients,
+    over 3 refinement iterations at the minimum supported size. Also the
+    direct check that skipping the reference's dead deepest encoder stage
+    (flowhip/nn/nconv.py docstring) preserves outputs AND gradients."""
+    from flowhip.models import build_model
+    _, ref_nc_dbl = ref_modules
+
+    torch.manual_seed(0)
+    args = _ncup_args()
+    ours = build_model(args)
+    ref = ref_nc_dbl.RAFT(args)
+    ref.load_state_dict(ours.state_dict())
+    ours.train()
+    ref.train()
+
+    img1 = torch.rand(1, 3, 128, 128) * 255
+    img2 = torch.rand(1, 3, 128, 128) * 255
+
+    preds_ref = ref(img1, img2, iters=3)
+    preds_ours = ours(img1, img2, iters=3)
+    assert len(preds_ref) == len(preds_ours) == 3
+    for a, b in zip(preds_ref, preds_ours):
+        assert (a - b).abs().max().item() < 1e-4
+
+    # identical scalar objective => comparable grads
+    sum(p.abs().mean() for p in preds_ref).backward()
+    sum(p.abs().mean() for p in preds_ours).backward()
+
+    ref_named = dict(ref.named_parameters())
+    rels = []
+    for name, p in ours.named_parameters():
+        g_ours, g_ref = p.grad, ref_named[name].grad
+        if g_ours is None and g_ref is None:
+            continue
+        assert g_ours is not None and g_ref is not None, name
+        ref_norm = g_ref.norm().item()
+        if ref_norm < 1e-6:
+            # analytically-zero grads (e.g. conv bias followed by a
+            # normalization): both sides emit ~1e-9 fp32 noise whose ratio
+            # is meaningless — just require ours to be equally negligible
+            assert g_ours.norm().item() < 1e-6, name
+            continue
+        rel = (g_ours - g_ref).norm().item() / ref_norm
+        # fnet gradients flow exclusively through the corr lookup, whose
+        # bilinear-tap derivative is discontinuous at integer coords: ~1e-5
+        # forward coordinate differences flip a few tap cells, giving a
+        # uniform ~5e-3 rel diff on fnet.* (measured; the reference's own
+        # thread-order noise on those layers is ~1e-6, and cnet — not
+        # behind the lookup — matches to <5e-3). Bound the tail at 2e-2
+        # and the bulk (median) at 1e-4 to still catch systematic errors.
+        assert rel < 2e-2, f"{name}: rel grad diff {rel:.2e}"
+        rels.append(rel)
+    assert len(rels) > 100
+    rels.sort()
+    assert rels[len(rels) // 2] < 1e-4, f"median rel grad diff {rels[len(rels)//2]:.2e}"
