@@ -124,3 +124,129 @@ def test_ncup_forward_and_grads_match_reference(ref_modules):
     assert len(rels) > 100
     rels.sort()
     assert rels[len(rels) // 2] < 1e-4, f"median rel grad diff {rels[len(rels)//2]:.2e}"
+
+
+@pytest.mark.timeout(900)
+def test_raft_basic_forward_matches_reference(ref_modules):
+    """RAFT-basic (convex-upsample path): shared weights => same outputs in
+    train and test mode."""
+    from flowhip.config.args import default_ncup_args
+    from flowhip.models import build_model
+    ref_raft, _ = ref_modules
+
+    torch.manual_seed(0)
+    args = default_ncup_args(model="raft", small=False)
+    args.mixed_precision = False
+    args.align_corners = False  # reference train.py:294 flag (accessed by ref)
+    ours = build_model(args)
+    ref = ref_raft.RAFT(args)
+    ref.load_state_dict(ours.state_dict())
+
+    img1 = torch.rand(1, 3, 128, 128) * 255
+    img2 = torch.rand(1, 3, 128, 128) * 255
+    preds_ref = ref(img1, img2, iters=2)
+    preds_ours = ours(img1, img2, iters=2)
+    for a, b in zip(preds_ref, preds_ours):
+        assert (a - b).abs().max().item() < 1e-4
+
+    low_r, up_r = ref(img1, img2, iters=2, test_mode=True)
+    low_o, up_o = ours(img1, img2, iters=2, test_mode=True)
+    assert (low_r - low_o).abs().max().item() < 1e-4
+    assert (up_r - up_o).abs().max().item() < 1e-4
+
+
+def test_raft_small_reference_is_broken_ours_works(ref_modules):
+    """Reference defect: raft.py:134 passes align_corners= to upflow8, whose
+    only definition (utils/utils.py:82) doesn't accept it — the reference's
+    RAFT-small (mask-less) upsample path always raises TypeError as shipped.
+    We keep the working utils.py default (align_corners=True, original-RAFT
+    behavior). Documented in PARITY.md."""
+    from flowhip.config.args import default_ncup_args
+    from flowhip.models import build_model
+    ref_raft, _ = ref_modules
+
+    torch.manual_seed(0)
+    args = default_ncup_args(model="raft", small=True)
+    args.mixed_precision = False
+    args.align_corners = False
+    img1 = torch.rand(1, 3, 128, 128) * 255
+    img2 = torch.rand(1, 3, 128, 128) * 255
+
+    ours = build_model(args)
+    out = ours(img1, img2, iters=2)
+    assert len(out) == 2 and out[-1].shape == (1, 2, 128, 128)
+
+    ref = ref_raft.RAFT(args)
+    with pytest.raises(TypeError):
+        ref(img1, img2, iters=2)
+
+
+@pytest.mark.parametrize("shared", [True, False])
+@pytest.mark.parametrize("pool", ["conf_based", "max_pooling"])
+def test_nconv_unet_bitexact_vs_reference(ref_modules, shared, pool):
+    """NConvUNet forward is BIT-exact vs the reference in every
+    shared-encoder / pooling configuration — direct proof that eliding the
+    reference's dead deepest encoder stage (see flowhip/nn/nconv.py) changes
+    nothing."""
+    sys.path.insert(0, REF_CORE)
+    try:
+        import nconv_modules as ref_nconv
+    finally:
+        sys.path.remove(REF_CORE)
+    from flowhip.nn.nconv import NConvUNet
+
+    torch.manual_seed(1)
+    ours = NConvUNet(in_ch=1, channels_multiplier=2, num_downsampling=3,
+                     shared_encoder=shared, data_pooling=pool)
+    ref = ref_nconv.NConvUNet(in_ch=1, channels_multiplier=2,
+                              num_downsampling=3, shared_encoder=shared,
+                              data_pooling=pool)
+    ref.load_state_dict(ours.state_dict())
+
+    data = torch.randn(2, 1, 64, 64)
+    conf = torch.rand(2, 1, 64, 64) * (torch.rand(2, 1, 64, 64) > 0.5)
+    x_o, c_o = ours((data, conf))
+    x_r, c_r = ref((data, conf))
+    assert torch.equal(x_o, x_r)
+    assert torch.equal(c_o, c_r)
+
+
+def test_pac_ops_bitexact_vs_reference(ref_modules):
+    """packernel2d / pacconv2d and both guided-upsampling heads are bit-exact
+    vs the reference on CPU under shared weights."""
+    sys.path.insert(0, REF_CORE)
+    try:
+        import pac_modules as ref_pac
+        import pac_upsampler as ref_pup
+    finally:
+        sys.path.remove(REF_CORE)
+    from flowhip.nn import pac as our_pac
+    from flowhip.nn import pac_upsampler as our_pup
+
+    torch.manual_seed(0)
+    guide = torch.randn(2, 4, 16, 16)
+    x = torch.randn(2, 6, 16, 16)
+    w = torch.randn(8, 6, 3, 3)
+    b = torch.randn(8)
+    k_o, _ = our_pac.packernel2d(guide, kernel_size=3, stride=1, padding=1,
+                                 dilation=1)
+    k_r, _ = ref_pac.packernel2d(guide, kernel_size=3, stride=1, padding=1,
+                                 dilation=1)
+    assert torch.equal(k_o, k_r)
+    y_o = our_pac.pacconv2d(x, k_o, w, b, stride=1, padding=1, dilation=1)
+    y_r = ref_pac.pacconv2d(x, k_r, w, b, stride=1, padding=1, dilation=1)
+    assert torch.equal(y_o, y_r)
+
+    torch.manual_seed(2)
+    up_o = our_pup.PacJointUpsample(factor=4, channels=1, guide_channels=3)
+    up_r = ref_pup.PacJointUpsample(factor=4, channels=1, guide_channels=3)
+    up_r.load_state_dict(up_o.state_dict())
+    lr = torch.randn(2, 1, 8, 8)
+    hr_guide = torch.randn(2, 3, 32, 32)
+    assert torch.equal(up_o(lr, hr_guide), up_r(lr, hr_guide))
+
+    torch.manual_seed(3)
+    dj_o = our_pup.DJIF(factor=4, channels=1, guide_channels=3)
+    dj_r = ref_pup.DJIF(factor=4, channels=1, guide_channels=3)
+    dj_r.load_state_dict(dj_o.state_dict())
+    assert torch.equal(dj_o(lr, hr_guide), dj_r(lr, hr_guide))
