@@ -37,8 +37,17 @@ def make_colorwheel():
     return colorwheel
 
 
-def flow_compute_color(u, v, convert_to_bgr=False):
-    """Map normalized (u, v) to wheel colors."""
+def flow_compute_color(u, v, convert_to_bgr=False, legacy_offset=False):
+    """Map normalized (u, v) to wheel colors.
+
+    The reference ships two disagreeing Matlab ports: flow_to_color's
+    (core/utils/flow_viz.py:88-92) keeps the 1-based `+1` offset and wraps
+    k1 at ncols -> 1, rotating the wheel one slot; flow_to_image's
+    compute_color (:216-227) is the faithful port (fk+1, wrap at ncols+1,
+    indexed with k-1 — equivalent to the plain 0-based form). Both are
+    reproduced byte-for-byte: `legacy_offset=True` selects the rotated
+    flow_to_color variant (in range because callers normalize by
+    rad_max + eps, so |a| < 1 strictly)."""
     flow_image = np.zeros((u.shape[0], u.shape[1], 3), np.uint8)
     colorwheel = make_colorwheel()
     ncols = colorwheel.shape[0]
@@ -47,10 +56,15 @@ def flow_compute_color(u, v, convert_to_bgr=False):
     a = np.arctan2(-v, -u) / np.pi
 
     fk = (a + 1) / 2 * (ncols - 1)
+    if legacy_offset:
+        fk = fk + 1
     k0 = np.floor(fk).astype(np.int32)
     k1 = k0 + 1
-    k1[k1 == ncols] = 0
+    k1[k1 == ncols] = 1 if legacy_offset else 0
     f = fk - k0
+    if legacy_offset:
+        k0 = k0 % ncols
+        k1 = k1 % ncols
 
     for i in range(colorwheel.shape[1]):
         tmp = colorwheel[:, i]
@@ -80,7 +94,7 @@ def flow_to_color(flow_uv, clip_flow=None, convert_to_bgr=False):
     epsilon = 1e-5
     u = u / (rad_max + epsilon)
     v = v / (rad_max + epsilon)
-    return flow_compute_color(u, v, convert_to_bgr)
+    return flow_compute_color(u, v, convert_to_bgr, legacy_offset=True)
 
 
 def flow_to_image(flow):

@@ -250,3 +250,87 @@ def test_pac_ops_bitexact_vs_reference(ref_modules):
     dj_r = ref_pup.DJIF(factor=4, channels=1, guide_channels=3)
     dj_r.load_state_dict(dj_o.state_dict())
     assert torch.equal(dj_o(lr, hr_guide), dj_r(lr, hr_guide))
+
+
+def test_reflective_flag_surface_matches_reference():
+    """The reflective-CLI flag surface for the three module families is
+    name-for-name identical to what the reference's introspection machinery
+    generates (core/utils/args.py driven exactly as train.py:299-342 drives
+    it, with the NCUP classes selected). Defaults deliberately differ: ours
+    are the SHIPPED NCUP configuration (SURVEY §2.5 / PARITY.md), the
+    reference's are the raw constructor defaults."""
+    import argparse
+
+    saved_argv = sys.argv
+    sys.path.insert(0, REF_CORE)
+    try:
+        sys.argv = ["x", "--final_upsampling", "NConvUpsampler",
+                    "--interp_net", "NConvUNet",
+                    "--weights_est_net", "Simple"]
+        from utils.args import (_add_arguments_for_module, str2bool,
+                                str2intlist)
+        import interp_weights_est as ref_iwe
+        import nconv_modules as ref_nconv
+        import upsampler as ref_up
+
+        parser = argparse.ArgumentParser()
+        _add_arguments_for_module(
+            parser, ref_up, name="final_upsampling", default_class=None,
+            exclude_classes=["_*"],
+            exclude_params=["self", "args", "interpolation_net",
+                            "weights_est_net", "size"],
+            forced_default_types={"scale": int,
+                                  "use_data_for_guidance": str2bool,
+                                  "channels_to_batch": str2bool,
+                                  "use_residuals": str2bool,
+                                  "est_on_high_res": str2bool})
+        _add_arguments_for_module(
+            parser, ref_nconv, name="interp_net", default_class=None,
+            exclude_classes=["_*"], exclude_params=["self", "args"],
+            forced_default_types={"encoder_fiter_sz": int,
+                                  "decoder_fiter_sz": int,
+                                  "out_filter_size": int,
+                                  "use_double_conv": str2bool,
+                                  "use_bias": str2bool})
+        _add_arguments_for_module(
+            parser, ref_iwe, name="weights_est_net", default_class=None,
+            exclude_classes=["_*"],
+            exclude_params=["self", "args", "out_ch", "final_act"],
+            unknown_default_types={"num_ch": str2intlist,
+                                   "filter_sz": str2intlist},
+            forced_default_types={"dilation": str2intlist})
+    finally:
+        sys.argv = saved_argv
+        sys.path.remove(REF_CORE)
+
+    ref_opts = sorted(o for a in parser._actions for o in a.option_strings
+                      if o.startswith("--") and o != "--help")
+
+    from flowhip.config import build_train_parser
+    ours = build_train_parser(argv=[])
+    our_opts = sorted(o for a in ours._actions for o in a.option_strings
+                      if o.startswith(("--final_upsampling", "--interp_net",
+                                       "--weights_est_net")))
+    assert our_opts == ref_opts
+    assert len(ref_opts) == 24
+
+
+def test_flow_viz_bitexact_vs_reference():
+    """flow_to_color / flow_to_image match the reference byte-for-byte on
+    finite flows (our only change is defined behavior for non-finite
+    values, which crash/garble the reference)."""
+    import numpy as np
+
+    sys.path.insert(0, REF_CORE)
+    try:
+        from utils import flow_viz as ref_viz
+    finally:
+        sys.path.remove(REF_CORE)
+    from flowhip.data import flow_viz as our_viz
+
+    rng = np.random.default_rng(0)
+    flow = (rng.standard_normal((16, 20, 2)) * 10).astype(np.float32)
+    assert np.array_equal(our_viz.flow_to_color(flow.copy()),
+                          ref_viz.flow_to_color(flow.copy()))
+    assert np.array_equal(our_viz.flow_to_image(flow.copy()),
+                          ref_viz.flow_to_image(flow.copy()))
