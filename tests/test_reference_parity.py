@@ -334,3 +334,103 @@ def test_flow_viz_bitexact_vs_reference():
                           ref_viz.flow_to_color(flow.copy()))
     assert np.array_equal(our_viz.flow_to_image(flow.copy()),
                           ref_viz.flow_to_image(flow.copy()))
+
+
+def test_geometry_utils_bitexact_vs_reference():
+    """InputPadder (both modes), bilinear_sampler, coords_grid, upflow8 and
+    the scipy forward_interpolate warm start are bit-exact vs
+    core/utils/utils.py."""
+    sys.path.insert(0, REF_CORE)
+    try:
+        from utils import utils as ref_utils
+    finally:
+        sys.path.remove(REF_CORE)
+    from flowhip.utils import geometry as ours
+
+    torch.manual_seed(0)
+    for mode in ("sintel", "kitti"):
+        x = torch.randn(2, 3, 437, 1021)
+        pr = ref_utils.InputPadder(x.shape, mode=mode)
+        po = ours.InputPadder(x.shape, mode=mode)
+        a = pr.pad(x.clone())[0]
+        b = po.pad(x.clone())[0]
+        assert torch.equal(a, b)
+        assert torch.equal(pr.unpad(a), po.unpad(b))
+
+    img = torch.randn(4, 2, 17, 23)
+    coords = torch.rand(4, 9, 11, 2) * 25 - 1  # includes out-of-bounds
+    assert torch.equal(ref_utils.bilinear_sampler(img, coords),
+                       ours.bilinear_sampler(img, coords))
+    assert torch.equal(ref_utils.coords_grid(2, 7, 9), ours.coords_grid(2, 7, 9))
+    f = torch.randn(1, 2, 6, 8)
+    assert torch.equal(ref_utils.upflow8(f), ours.upflow8(f))
+    fl = torch.randn(2, 12, 14) * 3
+    assert torch.equal(ref_utils.forward_interpolate(fl.clone()),
+                       ours.forward_interpolate(fl.clone()))
+
+
+@pytest.mark.parametrize("norm", ["batch", "group", "instance", "none"])
+def test_extractors_bitexact_vs_reference(ref_modules, norm):
+    """Basic/Small encoders match the reference bit-for-bit under shared
+    weights for every norm_fn."""
+    sys.path.insert(0, REF_CORE)
+    try:
+        import extractor as ref_ext
+    finally:
+        sys.path.remove(REF_CORE)
+    from flowhip.nn import extractor as our_ext
+
+    x = torch.randn(2, 3, 64, 64)
+    for our_cls, ref_cls, dim in ((our_ext.BasicEncoder, ref_ext.BasicEncoder, 64),
+                                  (our_ext.SmallEncoder, ref_ext.SmallEncoder, 32)):
+        torch.manual_seed(1)
+        o = our_cls(output_dim=dim, norm_fn=norm, dropout=0.0)
+        r = ref_cls(output_dim=dim, norm_fn=norm, dropout=0.0)
+        r.load_state_dict(o.state_dict())
+        o.eval()
+        r.eval()
+        assert torch.equal(o(x), r(x))
+
+
+def test_update_blocks_bitexact_vs_reference(ref_modules):
+    """Small/Basic update blocks (motion encoders + GRUs + flow/mask heads)
+    match bit-for-bit under shared weights."""
+    sys.path.insert(0, REF_CORE)
+    try:
+        import update as ref_upd
+    finally:
+        sys.path.remove(REF_CORE)
+    from flowhip.config.args import default_ncup_args
+    from flowhip.nn import update as our_upd
+
+    torch.manual_seed(0)
+
+    args = default_ncup_args(model="raft", small=True)
+    args.corr_levels, args.corr_radius = 4, 3  # set by RAFT.__init__ (raft.py:28-29)
+    cor_planes = args.corr_levels * (2 * args.corr_radius + 1) ** 2
+    o = our_upd.SmallUpdateBlock(args, hidden_dim=96)
+    r = ref_upd.SmallUpdateBlock(args, hidden_dim=96)
+    r.load_state_dict(o.state_dict())
+    net = torch.randn(2, 96, 8, 8)
+    inp = torch.randn(2, 64, 8, 8)
+    corr = torch.randn(2, cor_planes, 8, 8)
+    flow = torch.randn(2, 2, 8, 8)
+    net_o, mask_o, df_o = o(net, inp, corr, flow)
+    net_r, mask_r, df_r = r(net, inp, corr, flow)
+    assert mask_o is None and mask_r is None
+    assert torch.equal(net_o, net_r) and torch.equal(df_o, df_r)
+
+    args = default_ncup_args(model="raft", small=False)
+    args.corr_levels, args.corr_radius = 4, 4  # raft.py:33-34
+    cor_planes = args.corr_levels * (2 * args.corr_radius + 1) ** 2
+    o = our_upd.BasicUpdateBlock(args, hidden_dim=128)
+    r = ref_upd.BasicUpdateBlock(args, hidden_dim=128)
+    r.load_state_dict(o.state_dict())
+    net = torch.randn(2, 128, 8, 8)
+    inp = torch.randn(2, 128, 8, 8)
+    corr = torch.randn(2, cor_planes, 8, 8)
+    net_o, mask_o, df_o = o(net, inp, corr, flow)
+    net_r, mask_r, df_r = r(net, inp, corr, flow)
+    assert torch.equal(net_o, net_r)
+    assert torch.equal(mask_o, mask_r)
+    assert torch.equal(df_o, df_r)
