@@ -194,6 +194,7 @@ def fused_conv2d_cat2(x1, x2, weight, bias, padding, cache, key):
     fusable = (x1.is_cuda and x1.dtype == torch.bfloat16
                and x2.dtype == torch.bfloat16
                and x1.shape[1] % 64 == 0
+               and (x1.shape[1] + x2.shape[1]) % 8 == 0  # 16-B staging tail
                and x1.stride(1) == 1 and x2.stride(1) == 1
                and _ext.ext() is not None)
     from torch.nn.modules.utils import _pair
