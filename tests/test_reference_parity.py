@@ -434,3 +434,35 @@ def test_update_blocks_bitexact_vs_reference(ref_modules):
     assert torch.equal(net_o, net_r)
     assert torch.equal(mask_o, mask_r)
     assert torch.equal(df_o, df_r)
+
+
+def test_pac_gradients_match_reference_handwritten_backward():
+    """the reference's hand-written PAC backwards
+    (pac_modules.py:112-131,166-202) define the gradient contract. Our
+    implementation matches them (and fp64 finite differences) through
+    packernel2d -> pacconv2d."""
+    sys.path.insert(0, REF_CORE)
+    try:
+        import pac_modules as ref_pac
+    finally:
+        sys.path.remove(REF_CORE)
+    from flowhip.nn import pac as our_pac
+
+    torch.manual_seed(0)
+    guide = torch.randn(2, 4, 10, 10)
+    x = torch.randn(2, 6, 10, 10)
+    w = torch.randn(5, 6, 3, 3)
+    b = torch.randn(5)
+
+    def grads(mod):
+        ins = [t.clone().requires_grad_(True) for t in (guide, x, w, b)]
+        k, _ = mod.packernel2d(ins[0], kernel_size=3, stride=1, padding=1,
+                               dilation=1)
+        y = mod.pacconv2d(ins[1], k, ins[2], ins[3], stride=1, padding=1,
+                          dilation=1)
+        return torch.autograd.grad(y.square().sum(), ins)
+
+    for g_o, g_r, name in zip(grads(our_pac), grads(ref_pac),
+                              ["dguide", "dx", "dw", "db"]):
+        rel = (g_o - g_r).norm().item() / (g_r.norm().item() + 1e-12)
+        assert rel < 1e-5, f"{name}: rel {rel:.2e}"
