@@ -466,3 +466,100 @@ def test_pac_gradients_match_reference_handwritten_backward():
                               ["dguide", "dx", "dw", "db"]):
         rel = (g_o - g_r).norm().item() / (g_r.norm().item() + 1e-12)
         assert rel < 1e-5, f"{name}: rel {rel:.2e}"
+
+
+def test_pac_modules_transpose_pool_bitexact_vs_reference(ref_modules):
+    """PacConvTranspose2d and PacPool2d module classes match the reference
+    bit-for-bit under shared weights."""
+    sys.path.insert(0, REF_CORE)
+    try:
+        import pac_modules as ref_pac
+    finally:
+        sys.path.remove(REF_CORE)
+    from flowhip.nn import pac as our_pac
+
+    torch.manual_seed(0)
+    o = our_pac.PacConvTranspose2d(4, 5, kernel_size=5, stride=2, padding=2,
+                                   output_padding=1)
+    r = ref_pac.PacConvTranspose2d(4, 5, kernel_size=5, stride=2, padding=2,
+                                   output_padding=1)
+    r.load_state_dict(o.state_dict())
+    x = torch.randn(2, 4, 8, 8)
+    guide = torch.randn(2, 3, 16, 16)
+    assert torch.equal(o(x, guide), r(x, guide))
+
+    o2 = our_pac.PacPool2d(kernel_size=3, stride=2, padding=1)
+    r2 = ref_pac.PacPool2d(kernel_size=3, stride=2, padding=1)
+    xp = torch.randn(2, 4, 16, 16)
+    gp = torch.randn(2, 4, 16, 16)
+    assert torch.equal(o2(xp, gp), r2(xp, gp))
+
+
+def test_unet_weights_est_bitexact_vs_reference(ref_modules):
+    sys.path.insert(0, REF_CORE)
+    try:
+        import interp_weights_est as ref_iwe
+    finally:
+        sys.path.remove(REF_CORE)
+    from flowhip.nn import interp_weights_est as our_iwe
+
+    torch.manual_seed(1)
+    o = our_iwe.UNet(num_ch=[5, 8, 12, 16], out_ch=1)
+    r = ref_iwe.UNet(num_ch=[5, 8, 12, 16], out_ch=1)
+    r.load_state_dict(o.state_dict())
+    o.eval()
+    r.eval()
+    x = torch.randn(2, 5, 32, 32)
+    assert torch.equal(o(x), r(x))
+
+
+@pytest.mark.parametrize("weights_in,kw", [
+    (5, dict(use_data_for_guidance=True, channels_to_batch=True,
+             use_residuals=False)),
+    (5, dict(use_data_for_guidance=True, channels_to_batch=True,
+             use_residuals=True)),
+    (4, dict(use_data_for_guidance=False, channels_to_batch=False,
+             use_residuals=False)),
+    (0, dict(use_data_for_guidance=False)),  # binary-weights fallback
+])
+def test_nconv_upsampler_head_bitexact_vs_reference(ref_modules, weights_in, kw):
+    """The full NCUP head (zero-injection + weights estimation + NConvUNet)
+    matches the reference bit-for-bit across the guidance / folding /
+    residual flag combinations, including the binary-weights fallback."""
+    sys.path.insert(0, REF_CORE)
+    try:
+        import interp_weights_est as ref_iwe
+        import nconv_modules as ref_nconv
+        import upsampler as ref_up
+    finally:
+        sys.path.remove(REF_CORE)
+    from flowhip.nn import upsampler as our_up
+    from flowhip.nn.interp_weights_est import Simple
+    from flowhip.nn.nconv import NConvUNet
+
+    torch.manual_seed(3)
+    inet_o = NConvUNet(in_ch=1, channels_multiplier=2, num_downsampling=1)
+    inet_r = ref_nconv.NConvUNet(in_ch=1, channels_multiplier=2,
+                                 num_downsampling=1)
+    inet_r.load_state_dict(inet_o.state_dict())
+    if weights_in:
+        w_o = Simple(num_ch=[weights_in, 8], out_ch=1, filter_sz=[3, 1],
+                     dilation=[1, 1], final_act=torch.sigmoid)
+        w_r = ref_iwe.Simple(num_ch=[weights_in, 8], out_ch=1,
+                             filter_sz=[3, 1], dilation=[1, 1],
+                             final_act=torch.sigmoid)
+        w_r.load_state_dict(w_o.state_dict())
+    else:
+        w_o = w_r = None
+
+    o = our_up.NConvUpsampler(scale=4, interpolation_net=inet_o,
+                              weights_est_net=w_o, **kw)
+    r = ref_up.NConvUpsampler(scale=4, interpolation_net=inet_r,
+                              weights_est_net=w_r, **kw)
+
+    torch.manual_seed(4)
+    lr = torch.rand(2, 1, 8, 8) * (torch.rand(2, 1, 8, 8) > 0.6)
+    guide = (lr.new_ones(2, 1, 32, 32) if weights_in == 0
+             else torch.randn(2, 4, 32, 32))
+    assert torch.equal(o(lr.clone(), guide.clone()),
+                       r(lr.clone(), guide.clone()))
