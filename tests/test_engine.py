@@ -222,3 +222,69 @@ def test_train_resume_full_state(tmp_path, monkeypatch):
     # proving the counter (and opt/sched state with it) was restored
     assert state2["total_steps"] > first_ckpt_step
     assert state2["scheduler"]["last_epoch"] > state["scheduler"]["last_epoch"]
+
+
+@pytest.mark.timeout(900)
+def test_validators_on_fake_datasets(tmp_path, monkeypatch):
+    """validate_chairs / validate_sintel / validate_kitti run end-to-end on
+    CPU against tiny fake dataset trees and return the reference's metric
+    dict keys (reference evaluate.py:91-181)."""
+    import numpy as np
+    from PIL import Image
+
+    from flowhip.data import frame_utils
+    from flowhip.engine import evaluate as ev
+
+    monkeypatch.chdir(tmp_path)
+    h, w = 160, 160  # >=128: 4-level pyramid constraint (PARITY.md)
+
+    def put_img(path):
+        path.parent.mkdir(parents=True, exist_ok=True)
+        arr = (np.random.rand(h, w, 3) * 255).astype(np.uint8)
+        Image.fromarray(arr).save(path)
+
+    # FlyingChairs: validation rows (xid==2) + split table in cwd
+    chairs = tmp_path / "datasets" / "FlyingChairs_release" / "data"
+    chairs.mkdir(parents=True)
+    for i in (1, 2):
+        put_img(chairs / f"{i:05d}_img1.png")
+        put_img(chairs / f"{i:05d}_img2.png")
+        frame_utils.writeFlow(str(chairs / f"{i:05d}_flow.flo"),
+                              np.random.randn(h, w, 2).astype(np.float32))
+    (tmp_path / "chairs_split.txt").write_text("2\n2\n")
+
+    # Sintel training: clean/final scenes + flow
+    for dstype in ("clean", "final"):
+        scene = tmp_path / "datasets" / "Sintel" / "training" / dstype / "s1"
+        for i in range(3):
+            put_img(scene / f"frame_{i:04d}.png")
+    fdir = tmp_path / "datasets" / "Sintel" / "training" / "flow" / "s1"
+    fdir.mkdir(parents=True)
+    for i in range(2):
+        frame_utils.writeFlow(str(fdir / f"frame_{i:04d}.flo"),
+                              np.random.randn(h, w, 2).astype(np.float32))
+
+    # KITTI training: image pairs + sparse flow_occ 16-bit pngs
+    kroot = tmp_path / "datasets" / "KITTI" / "training"
+    for i in range(2):
+        put_img(kroot / "image_2" / f"{i:06d}_10.png")
+        put_img(kroot / "image_2" / f"{i:06d}_11.png")
+    (kroot / "flow_occ").mkdir(parents=True)
+    for i in range(2):
+        frame_utils.writeFlowKITTI(
+            str(kroot / "flow_occ" / f"{i:06d}_10.png"),
+            np.random.randn(h, w, 2).astype(np.float32) * 4)
+
+    args = default_ncup_args(model="raft_nc_dbl", small=True)
+    model = build_model(args)
+
+    res = ev.validate_chairs(model, iters=2)
+    assert set(res) == {"chairs"} and np.isfinite(res["chairs"])
+
+    res = ev.validate_sintel(model, iters=2)
+    assert set(res) == {"clean", "final"}
+    assert all(np.isfinite(v) for v in res.values())
+
+    res = ev.validate_kitti(model, iters=2)
+    assert set(res) == {"kitti-epe", "kitti-f1"}
+    assert np.isfinite(res["kitti-epe"]) and 0 <= res["kitti-f1"] <= 100
