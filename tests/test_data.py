@@ -282,3 +282,56 @@ def test_eraser_touches_only_img2():
     mean_color = np.mean(img2.reshape(-1, 3), axis=0)
     np.testing.assert_allclose(
         b[changed], np.broadcast_to(mean_color, b[changed].shape), atol=1.0)
+
+
+def test_hd1k_dataset(tmp_path):
+    """HD1K sequence discovery (reference datasets.py:181-197): per-sequence
+    %06d_* globs, sparse 16-bit flow, last frame of each sequence unpaired."""
+    from flowhip.data.datasets import HD1K
+
+    root = tmp_path / "hd1k"
+    img_dir = root / "hd1k_input" / "image_2"
+    flo_dir = root / "hd1k_flow_gt" / "flow_occ"
+    img_dir.mkdir(parents=True)
+    flo_dir.mkdir(parents=True)
+    from PIL import Image
+    for seq in range(2):
+        for i in range(3):
+            arr = (np.random.rand(32, 48, 3) * 255).astype(np.uint8)
+            Image.fromarray(arr).save(img_dir / f"{seq:06d}_{i:04d}.png")
+            frame_utils.writeFlowKITTI(
+                str(flo_dir / f"{seq:06d}_{i:04d}.png"),
+                np.random.randn(32, 48, 2).astype(np.float32))
+
+    ds = HD1K(aug_params=None, root=str(root))
+    assert len(ds) == 4  # 2 sequences x (3 flows - 1)
+    img1, img2, flow, valid = ds[0]
+    assert img1.shape == (3, 32, 48) and flow.shape == (2, 32, 48)
+    assert set(valid.unique().tolist()).issubset({0.0, 1.0})
+
+
+def test_fetch_dataloader_chairs_stage(tmp_path, monkeypatch):
+    """The chairs stage recipe builds a crop-augmented loader from the split
+    table (reference datasets.py:210-213)."""
+    import argparse
+
+    from PIL import Image
+    from flowhip.data.datasets import fetch_dataloader
+
+    monkeypatch.chdir(tmp_path)
+    root = tmp_path / "datasets" / "FlyingChairs_release" / "data"
+    root.mkdir(parents=True)
+    for i in (1, 2):
+        for k in (1, 2):
+            arr = (np.random.rand(96, 128, 3) * 255).astype(np.uint8)
+            Image.fromarray(arr).save(root / f"{i:05d}_img{k}.png")
+        frame_utils.writeFlow(str(root / f"{i:05d}_flow.flo"),
+                              np.random.randn(96, 128, 2).astype(np.float32))
+    (tmp_path / "chairs_split.txt").write_text("1\n1\n")
+
+    args = argparse.Namespace(stage="chairs", image_size=[64, 64],
+                              batch_size=2, num_workers=0)
+    loader = fetch_dataloader(args)
+    img1, img2, flow, valid = next(iter(loader))
+    assert img1.shape == (2, 3, 64, 64)
+    assert flow.shape == (2, 2, 64, 64) and valid.shape == (2, 64, 64)

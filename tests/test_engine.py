@@ -288,3 +288,29 @@ def test_validators_on_fake_datasets(tmp_path, monkeypatch):
     res = ev.validate_kitti(model, iters=2)
     assert set(res) == {"kitti-epe", "kitti-f1"}
     assert np.isfinite(res["kitti-epe"]) and 0 <= res["kitti-f1"] <= 100
+
+
+@pytest.mark.timeout(600)
+def test_evaluate_cli_synthetic(tmp_path):
+    """Top-level evaluate.py CLI end-to-end on the synthetic validator."""
+    import subprocess
+    import sys
+
+    from flowhip.engine import checkpoints
+
+    torch.manual_seed(0)
+    # dataset controls BatchNorm in the weights-est net (BN only for
+    # sintel): the saved model must match the CLI's --dataset synthetic
+    args = default_ncup_args(model="raft_nc_dbl", small=True,
+                             dataset="synthetic")
+    model = build_model(args)
+    ckpt = tmp_path / "m.pth"
+    checkpoints.save_weights(model, str(ckpt))
+
+    r = subprocess.run(
+        [sys.executable, "evaluate.py", "--model", "raft_nc_dbl", "--small",
+         "--restore_ckpt", str(ckpt), "--dataset", "synthetic",
+         "--iters", "2"],
+        capture_output=True, text=True, timeout=500)
+    assert r.returncode == 0, r.stderr[-2000:]
+    assert "synthetic" in r.stdout.lower()
