@@ -335,3 +335,83 @@ def test_fetch_dataloader_chairs_stage(tmp_path, monkeypatch):
     img1, img2, flow, valid = next(iter(loader))
     assert img1.shape == (2, 3, 64, 64)
     assert flow.shape == (2, 2, 64, 64) and valid.shape == (2, 64, 64)
+
+
+def _write_pfm(path, arr):
+    """Minimal color-PFM writer for fixtures (matches frame_utils.readPFM:
+    'PF', dims, negative scale = little-endian, rows bottom-to-top)."""
+    h, w, c = arr.shape
+    assert c == 3
+    with open(path, "wb") as f:
+        f.write(b"PF\n")
+        f.write(f"{w} {h}\n".encode())
+        f.write(b"-1.0\n")
+        np.flipud(arr).astype("<f4").tofile(f)
+
+
+def test_sintel_stage_recipe_composition(tmp_path, monkeypatch):
+    """The sintel fine-tune stage mixes 100*clean + 100*final + 200*kitti +
+    5*hd1k + things (reference datasets.py:222-232) — dataset lengths
+    compose exactly and a batch loads through the augmentors."""
+    import argparse
+
+    from PIL import Image
+    from flowhip.data.datasets import fetch_dataloader
+
+    monkeypatch.chdir(tmp_path)
+    h, w = 96, 128
+
+    def put_img(path):
+        path.parent.mkdir(parents=True, exist_ok=True)
+        arr = (np.random.rand(h, w, 3) * 255).astype(np.uint8)
+        Image.fromarray(arr).save(path)
+
+    # Sintel training: 1 scene, 3 frames -> 2 pairs per dstype
+    for dstype in ("clean", "final"):
+        for i in range(3):
+            put_img(tmp_path / "datasets" / "Sintel" / "training" / dstype
+                    / "s1" / f"frame_{i:04d}.png")
+    fdir = tmp_path / "datasets" / "Sintel" / "training" / "flow" / "s1"
+    fdir.mkdir(parents=True)
+    for i in range(2):
+        frame_utils.writeFlow(str(fdir / f"frame_{i:04d}.flo"),
+                              np.random.randn(h, w, 2).astype(np.float32))
+
+    # KITTI training: 1 pair
+    kroot = tmp_path / "datasets" / "KITTI" / "training"
+    put_img(kroot / "image_2" / "000000_10.png")
+    put_img(kroot / "image_2" / "000000_11.png")
+    (kroot / "flow_occ").mkdir(parents=True)
+    frame_utils.writeFlowKITTI(str(kroot / "flow_occ" / "000000_10.png"),
+                               np.random.randn(h, w, 2).astype(np.float32))
+
+    # HD1K: 1 sequence, 2 frames -> 1 pair
+    hroot = tmp_path / "datasets" / "HD1k"
+    for i in range(2):
+        put_img(hroot / "hd1k_input" / "image_2" / f"000000_{i:04d}.png")
+        (hroot / "hd1k_flow_gt" / "flow_occ").mkdir(parents=True, exist_ok=True)
+        frame_utils.writeFlowKITTI(
+            str(hroot / "hd1k_flow_gt" / "flow_occ" / f"000000_{i:04d}.png"),
+            np.random.randn(h, w, 2).astype(np.float32))
+
+    # FlyingThings3D cleanpass: 1 dir, 3 frames/pfm -> 2+2 pairs (2 dirs)
+    idir = (tmp_path / "datasets" / "FlyingThings3D" / "frames_cleanpass"
+            / "TRAIN" / "A" / "0000" / "left")
+    fdir3 = (tmp_path / "datasets" / "FlyingThings3D" / "optical_flow"
+             / "TRAIN" / "A" / "0000")
+    for i in range(3):
+        put_img(idir / f"{i:04d}.png")
+    for direction in ("into_future", "into_past"):
+        d = fdir3 / direction / "left"
+        d.mkdir(parents=True)
+        for i in range(3):
+            _write_pfm(d / f"{i:04d}.pfm",
+                       np.random.randn(h, w, 3).astype(np.float32))
+
+    args = argparse.Namespace(stage="sintel", image_size=[64, 64],
+                              batch_size=2, num_workers=0)
+    loader = fetch_dataloader(args)
+    # 100*2 + 100*2 + 200*1 + 5*1 + (2 into_future + 2 into_past)
+    assert len(loader.dataset) == 100 * 2 + 100 * 2 + 200 * 1 + 5 * 1 + 4
+    img1, img2, flow, valid = next(iter(loader))
+    assert img1.shape == (2, 3, 64, 64) and flow.shape == (2, 2, 64, 64)
