@@ -415,3 +415,59 @@ def test_sintel_stage_recipe_composition(tmp_path, monkeypatch):
     assert len(loader.dataset) == 100 * 2 + 100 * 2 + 200 * 1 + 5 * 1 + 4
     img1, img2, flow, valid = next(iter(loader))
     assert img1.shape == (2, 3, 64, 64) and flow.shape == (2, 2, 64, 64)
+
+
+def test_things_and_kitti_stage_recipes(tmp_path, monkeypatch):
+    """things stage = cleanpass + finalpass concat; kitti stage = sparse
+    KITTI with no flips (reference datasets.py:214-236)."""
+    import argparse
+
+    from PIL import Image
+    from flowhip.data.datasets import fetch_dataloader
+
+    monkeypatch.chdir(tmp_path)
+    h, w = 96, 128
+
+    def put_img(path):
+        path.parent.mkdir(parents=True, exist_ok=True)
+        arr = (np.random.rand(h, w, 3) * 255).astype(np.uint8)
+        Image.fromarray(arr).save(path)
+
+    for dstype in ("frames_cleanpass", "frames_finalpass"):
+        idir = (tmp_path / "datasets" / "FlyingThings3D" / dstype
+                / "TRAIN" / "A" / "0000" / "left")
+        for i in range(3):
+            put_img(idir / f"{i:04d}.png")
+    for direction in ("into_future", "into_past"):
+        d = (tmp_path / "datasets" / "FlyingThings3D" / "optical_flow"
+             / "TRAIN" / "A" / "0000" / direction / "left")
+        d.mkdir(parents=True)
+        for i in range(3):
+            _write_pfm(d / f"{i:04d}.pfm",
+                       np.random.randn(h, w, 3).astype(np.float32))
+
+    args = argparse.Namespace(stage="things", image_size=[64, 64],
+                              batch_size=2, num_workers=0,
+                              compressed_ft=False)
+    loader = fetch_dataloader(args)
+    # (2 into_future + 2 into_past) per pass, both passes share the flow tree
+    assert len(loader.dataset) == 8
+    img1, _, flow, _ = next(iter(loader))
+    assert img1.shape == (2, 3, 64, 64) and flow.shape == (2, 2, 64, 64)
+
+    kroot = tmp_path / "datasets" / "KITTI" / "training"
+    for i in range(2):
+        put_img(kroot / "image_2" / f"{i:06d}_10.png")
+        put_img(kroot / "image_2" / f"{i:06d}_11.png")
+    (kroot / "flow_occ").mkdir(parents=True, exist_ok=True)
+    for i in range(2):
+        frame_utils.writeFlowKITTI(
+            str(kroot / "flow_occ" / f"{i:06d}_10.png"),
+            np.random.randn(h, w, 2).astype(np.float32))
+
+    args = argparse.Namespace(stage="kitti", image_size=[64, 64],
+                              batch_size=2, num_workers=0)
+    loader = fetch_dataloader(args)
+    assert len(loader.dataset) == 2
+    img1, _, flow, valid = next(iter(loader))
+    assert valid.shape == (2, 64, 64)

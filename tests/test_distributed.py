@@ -127,3 +127,24 @@ def test_bench_contract_world_size_2(tmp_path):
     # CPU plumbing config: n_gpus 0, world-size-2 global batch
     assert rec["n_gpus"] == 0
     assert rec["config"]["global_batch"] == 2
+
+
+@pytest.mark.timeout(900)
+def test_bench_default_invocation(tmp_path):
+    """`python bench.py` with no flags (the driver's fallback invocation)
+    emits exactly one valid JSON line and finishes quickly on CPU."""
+    import json
+    import subprocess
+    import sys
+
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    out = subprocess.run(
+        [sys.executable, os.path.join(repo, "bench.py"),
+         "--steps", "2", "--warmup", "1"],
+        capture_output=True, text=True, timeout=840, cwd=repo).stdout
+    lines = [l for l in out.splitlines() if l.startswith("{")]
+    assert len(lines) == 1
+    rec = json.loads(lines[0])
+    assert rec["metric"] == "train_image_pairs_per_sec"
+    assert rec["value"] > 0
+    assert rec["config"]["global_batch"] == 1
