@@ -197,3 +197,85 @@ def test_zero_inject_matches_naive(h, w, s, seed):
     ref = np.zeros_like(out)
     ref[:, :, s // 2::s, s // 2::s] = x
     np.testing.assert_allclose(out, ref)
+
+
+@settings(max_examples=6, deadline=None)
+@given(st.integers(16, 24), st.integers(16, 24), st.integers(0, 2 ** 31 - 1))
+def test_corr_lookup_flagship_radius_levels(h, w, seed):
+    """Flagship configuration (radius 4, 4 levels = 324 channels): taps
+    still match the scalar bilinear oracle at every level."""
+    rng = np.random.default_rng(seed)
+    radius, levels = 4, 4
+    P = h * w
+    pyramid = [torch.from_numpy(
+        rng.standard_normal((P, 1, max(h >> l, 1), max(w >> l, 1)))
+        .astype(np.float32)) for l in range(levels)]
+    coords = torch.from_numpy(
+        (rng.random((1, 2, h, w)) * [[[[w]], [[h]]]]).astype(np.float32))
+
+    out = torch_ref.corr_lookup(pyramid, coords, radius).numpy()
+    K = 2 * radius + 1
+    assert out.shape == (1, levels * K * K, h, w)
+
+    for _ in range(10):
+        i = rng.integers(P)
+        l = rng.integers(levels)
+        a = rng.integers(K)
+        c = rng.integers(K)
+        y, x = divmod(int(i), w)
+        cx = coords[0, 0, y, x].item() / (1 << l) + (a - radius)
+        cy = coords[0, 1, y, x].item() / (1 << l) + (c - radius)
+        ref = naive_bilinear(pyramid[l][i, 0].numpy(), cx, cy)
+        ch = l * K * K + a * K + c
+        np.testing.assert_allclose(out[0, ch, y, x], ref, atol=1e-5)
+
+
+@settings(max_examples=10, deadline=None)
+@given(st.integers(1, 5), st.integers(4, 8), st.integers(4, 8),
+       st.integers(0, 2 ** 31 - 1))
+def test_sequence_loss_matches_naive(n_preds, h, w, seed):
+    """gamma-weighted L1 + final-pred metrics vs an independent NumPy
+    implementation of the reference formula (train.py:46-71)."""
+    rng = np.random.default_rng(seed)
+    gamma = 0.85
+    gt = rng.standard_normal((2, 2, h, w)).astype(np.float32)
+    gt[0, :, 0, 0] = 300.0  # ||gt|| ~ 424 > 400: excluded by max_flow
+    valid = (rng.random((2, h, w)) > 0.3).astype(np.float32)
+    preds = [rng.standard_normal((2, 2, h, w)).astype(np.float32)
+             for _ in range(n_preds)]
+
+    loss, metrics = torch_ref.sequence_loss(
+        [torch.from_numpy(p) for p in preds], torch.from_numpy(gt),
+        torch.from_numpy(valid), gamma)
+
+    mag = np.sqrt((gt ** 2).sum(axis=1))
+    v = (valid >= 0.5) & (mag < 400.0)
+    expect = 0.0
+    for i, p in enumerate(preds):
+        w_i = gamma ** (n_preds - i - 1)
+        expect += w_i * (v[:, None] * np.abs(p - gt)).mean()
+    np.testing.assert_allclose(loss.item(), expect, rtol=1e-5)
+
+    epe = np.sqrt(((preds[-1] - gt) ** 2).sum(axis=1)).reshape(-1)[v.reshape(-1)]
+    np.testing.assert_allclose(metrics["epe"], epe.mean(), rtol=1e-5)
+    np.testing.assert_allclose(metrics["3px"], (epe < 3).mean(), rtol=1e-6)
+
+
+def test_nconv2d_bias_and_no_prop():
+    """bias branch adds after normalization; prop_conf=False returns
+    cout=None (nconv_modules.py:176-199)."""
+    rng = np.random.default_rng(0)
+    data = torch.from_numpy(rng.standard_normal((1, 2, 6, 6)).astype(np.float32))
+    conf = torch.from_numpy(rng.random((1, 2, 6, 6)).astype(np.float32))
+    wt = torch.from_numpy((rng.random((3, 2, 3, 3)) + 0.1).astype(np.float32))
+    bias = torch.from_numpy(rng.standard_normal(3).astype(np.float32))
+
+    out_nb, cout = torch_ref.nconv2d(data, conf, wt, None, padding=1)
+    out_b, cout_none = torch_ref.nconv2d(data, conf, wt, bias, padding=1,
+                                         prop_conf=False)
+    assert cout_none is None
+    np.testing.assert_allclose(
+        (out_b - out_nb).numpy(),
+        np.broadcast_to(bias.numpy()[None, :, None, None], out_b.shape),
+        atol=1e-6)
+    assert cout.shape == out_nb.shape
