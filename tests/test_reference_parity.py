@@ -563,3 +563,39 @@ def test_nconv_upsampler_head_bitexact_vs_reference(ref_modules, weights_in, kw)
              else torch.randn(2, 4, 32, 32))
     assert torch.equal(o(lr.clone(), guide.clone()),
                        r(lr.clone(), guide.clone()))
+
+
+def test_pac_upsampler_baseline_heads_bitexact_vs_reference(ref_modules):
+    """The remaining guided-upsampling baseline heads (Lite/Wide/bilateral/
+    bilinear) and the th_epe metric helper match the reference bit-for-bit."""
+    sys.path.insert(0, REF_CORE)
+    try:
+        import pac_upsampler as ref_pup
+    finally:
+        sys.path.remove(REF_CORE)
+    from flowhip.nn import pac_upsampler as our_pup
+
+    torch.manual_seed(0)
+    lr = torch.randn(2, 1, 8, 8)
+    guide = torch.randn(2, 3, 32, 32)
+
+    for name in ("PacJointUpsampleLite", "DJIFWide"):
+        torch.manual_seed(5)
+        o = getattr(our_pup, name)(factor=4, channels=1, guide_channels=3)
+        r = getattr(ref_pup, name)(factor=4, channels=1, guide_channels=3)
+        r.load_state_dict(o.state_dict())
+        assert torch.equal(o(lr, guide), r(lr, guide)), name
+
+    o = our_pup.JointBilateral(factor=4, channels=1, kernel_size=5,
+                               scale_space=0.2, scale_color=5.0)
+    r = ref_pup.JointBilateral(factor=4, channels=1, kernel_size=5,
+                               scale_space=0.2, scale_color=5.0)
+    r.load_state_dict(o.state_dict())
+    assert torch.equal(o(lr, guide), r(lr, guide))
+
+    assert torch.equal(our_pup.Bilinear(factor=4)(lr, guide),
+                       ref_pup.Bilinear(factor=4)(lr, guide))
+
+    a = torch.randn(2, 2, 16, 16)
+    b = torch.randn(2, 2, 16, 16)
+    assert torch.equal(our_pup.th_epe(a, b), ref_pup.th_epe(a, b))
