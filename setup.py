@@ -15,7 +15,10 @@ os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
 
 from torch.utils.cpp_extension import BuildExtension, CUDAExtension  # noqa: E402
 
-sources = sorted(glob.glob("csrc/*.hip")) + sorted(glob.glob("csrc/*.cpp"))
+# exclude the *_hip.* copies torch's hipify writes next to the sources on
+# a previous build — globbing them too would compile every TU twice
+sources = [s for s in sorted(glob.glob("csrc/*.hip")) + sorted(glob.glob("csrc/*.cpp"))
+           if not s.endswith(("_hip.hip", "_hip.cpp"))]
 
 setup(
     name="flowhip",
