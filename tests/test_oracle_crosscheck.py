@@ -279,3 +279,39 @@ def test_nconv2d_bias_and_no_prop():
         np.broadcast_to(bias.numpy()[None, :, None, None], out_b.shape),
         atol=1e-6)
     assert cout.shape == out_nb.shape
+
+
+@settings(max_examples=8, deadline=None)
+@given(st.integers(2, 4), st.integers(2, 4), st.sampled_from([2, 8]),
+       st.integers(0, 2 ** 31 - 1))
+def test_convex_upsample_matches_naive(h, w, factor, seed):
+    """Full scalar oracle for kernel #11 (reference raft.py:73-84): output
+    subpixel (dy,dx) of coarse pixel (y,x) = softmax-weighted sum of
+    factor*flow over the zero-padded 3x3 neighborhood."""
+    rng = np.random.default_rng(seed)
+    flow = rng.standard_normal((1, 2, h, w)).astype(np.float32)
+    mask = rng.standard_normal((1, 9 * factor * factor, h, w)).astype(np.float32)
+
+    out = torch_ref.convex_upsample(torch.from_numpy(flow),
+                                    torch.from_numpy(mask), factor).numpy()
+    assert out.shape == (1, 2, factor * h, factor * w)
+
+    m = mask.reshape(1, 1, 9, factor, factor, h, w)
+    for _ in range(10):
+        y = rng.integers(h)
+        x = rng.integers(w)
+        dy = rng.integers(factor)
+        dx = rng.integers(factor)
+        c = rng.integers(2)
+        logits = m[0, 0, :, dy, dx, y, x]
+        weights = np.exp(logits - logits.max())
+        weights /= weights.sum()
+        val = 0.0
+        # unfold(3x3, pad 1) tap order: k = ky*3 + kx, tap = (y+ky-1, x+kx-1)
+        for k in range(9):
+            yy = y + k // 3 - 1
+            xx = x + k % 3 - 1
+            if 0 <= yy < h and 0 <= xx < w:
+                val += weights[k] * factor * flow[0, c, yy, xx]
+        np.testing.assert_allclose(out[0, c, y * factor + dy, x * factor + dx],
+                                   val, atol=1e-5)
