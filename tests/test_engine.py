@@ -314,3 +314,29 @@ def test_evaluate_cli_synthetic(tmp_path):
         capture_output=True, text=True, timeout=500)
     assert r.returncode == 0, r.stderr[-2000:]
     assert "synthetic" in r.stdout.lower()
+
+
+@pytest.mark.timeout(600)
+def test_train_determinism_same_seed(tmp_path, monkeypatch):
+    """Two identical-seed runs produce byte-identical final weights
+    (SURVEY §2.8 RNG surface: torch/np seeding + deterministic synthetic
+    loader with num_workers=0)."""
+    weights = []
+    for run in ("a", "b"):
+        rundir = tmp_path / run
+        rundir.mkdir()
+        monkeypatch.chdir(rundir)
+        argv = ["--name", "det", "--model", "raft_nc_dbl", "--stage",
+                "synthetic", "--small", "--num_steps", "2", "--batch_size",
+                "1", "--image_size", "128", "128", "--iters", "2",
+                "--lr", "1e-4", "--num_workers", "0", "--seed", "7"]
+        parser = build_train_parser(argv=argv)
+        args = finalize_args(parser.parse_args(argv))
+        from flowhip.engine.train import train
+        path = train(args)
+        weights.append(torch.load(path, weights_only=True))
+
+    a, b = weights
+    assert a.keys() == b.keys()
+    for k in a:
+        assert torch.equal(a[k], b[k]), k

@@ -599,3 +599,33 @@ def test_pac_upsampler_baseline_heads_bitexact_vs_reference(ref_modules):
     a = torch.randn(2, 2, 16, 16)
     b = torch.randn(2, 2, 16, 16)
     assert torch.equal(our_pup.th_epe(a, b), ref_pup.th_epe(a, b))
+
+
+def test_checkpoint_file_interchange_with_reference(ref_modules, tmp_path):
+    """The literal user workflow: a checkpoint file written by this framework
+    loads strict into the reference model (after its own module. strip), and
+    a reference-style file (module.-prefixed raw state_dict, as its
+    DataParallel training saves) loads into ours."""
+    from flowhip.engine import checkpoints
+    from flowhip.models import build_model
+    _, ref_nc_dbl = ref_modules
+
+    args = _ncup_args()
+    ours = build_model(args)
+
+    # ours -> reference
+    path = os.path.join(tmp_path, "ours.pth")
+    checkpoints.save_weights(ours, path)
+    sd = torch.load(path, weights_only=True)
+    ref = ref_nc_dbl.RAFT(args)
+    ref.load_state_dict({k[len("module."):]: v for k, v in sd.items()},
+                        strict=True)
+
+    # reference-style file -> ours
+    ref_path = os.path.join(tmp_path, "ref.pth")
+    torch.save({"module." + k: v for k, v in ref.state_dict().items()},
+               ref_path)
+    ours2 = build_model(args)
+    checkpoints.load_weights(ours2, ref_path, strict=True)
+    assert torch.equal(ours2.state_dict()["fnet.conv1.weight"],
+                       ours.state_dict()["fnet.conv1.weight"])
