@@ -471,3 +471,19 @@ def test_things_and_kitti_stage_recipes(tmp_path, monkeypatch):
     assert len(loader.dataset) == 2
     img1, _, flow, valid = next(iter(loader))
     assert valid.shape == (2, 64, 64)
+
+
+def test_read_disp_kitti(tmp_path):
+    """16-bit grayscale disparity png: disp = px/256, valid = disp>0, flow =
+    (-disp, 0) (reference frame_utils.py:109-113; PIL mode 'I' replaces
+    cv2.IMREAD_ANYDEPTH)."""
+    from PIL import Image
+
+    disp = (np.arange(12, dtype=np.uint16).reshape(3, 4)) * 128
+    Image.fromarray(disp, mode="I;16").save(tmp_path / "d.png")
+
+    flow, valid = frame_utils.readDispKITTI(str(tmp_path / "d.png"))
+    assert flow.shape == (3, 4, 2)
+    np.testing.assert_allclose(flow[..., 0], -disp.astype(np.float32) / 256.0)
+    np.testing.assert_array_equal(flow[..., 1], 0)
+    np.testing.assert_array_equal(valid, disp > 0)
