@@ -246,6 +246,59 @@ class SyntheticFlowDataset(data.Dataset):
         return img1, img2, flow, valid
 
 
+# Per-stage augmentation envelopes (scale range is log2; flips per stage).
+# Values are the reference's published training schedules (datasets.py:210-236).
+_STAGE_AUG = {
+    "chairs": dict(min_scale=-0.1, max_scale=1.0, do_flip=True),
+    "things": dict(min_scale=-0.4, max_scale=0.8, do_flip=True),
+    "sintel": dict(min_scale=-0.2, max_scale=0.6, do_flip=True),
+    "sintel_kitti": dict(min_scale=-0.3, max_scale=0.5, do_flip=True),
+    "sintel_hd1k": dict(min_scale=-0.5, max_scale=0.2, do_flip=True),
+    "kitti": dict(min_scale=-0.2, max_scale=0.4, do_flip=False),
+}
+
+
+def _stage_dataset(args, TRAIN_DS):
+    """Assemble the training mixture for a stage.
+
+    Oversampling weights for the sintel fine-tune mixture follow the
+    reference exactly: 100x each Sintel pass + 200x KITTI + 5x HD1K + one
+    pass of FlyingThings (datasets.py:231).
+    """
+    def aug(key):
+        return {"crop_size": args.image_size, **_STAGE_AUG[key]}
+
+    stage = args.stage
+    if stage == "chairs":
+        return FlyingChairs(aug("chairs"), split="training")
+
+    if stage == "things":
+        clean, final = (FlyingThings3D(aug("things"), dstype=pas,
+                                       load_compressed=args.compressed_ft)
+                        for pas in ("frames_cleanpass", "frames_finalpass"))
+        return clean + final
+
+    if stage == "sintel":
+        base = (100 * MpiSintel(aug("sintel"), split="training", dstype="clean")
+                + 100 * MpiSintel(aug("sintel"), split="training",
+                                  dstype="final"))
+        things = FlyingThings3D(aug("sintel"), dstype="frames_cleanpass")
+        if TRAIN_DS == "C+T+K+S+H":
+            return (base + 200 * KITTI(aug("sintel_kitti"))
+                    + 5 * HD1K(aug("sintel_hd1k")) + things)
+        if TRAIN_DS == "C+T+K/S":
+            return base + things
+        raise ValueError(f"unknown TRAIN_DS {TRAIN_DS!r}")
+
+    if stage == "kitti":
+        return KITTI(aug("kitti"), split="training")
+
+    if stage == "synthetic":
+        return SyntheticFlowDataset(image_size=args.image_size)
+
+    raise ValueError(f"unknown stage {args.stage!r}")
+
+
 def fetch_dataloader(args, TRAIN_DS="C+T+K+S+H", distributed=False, rank=0,
                      world_size=1):
     """Create the training loader for a stage (reference datasets.py:207-243).
@@ -255,42 +308,7 @@ def fetch_dataloader(args, TRAIN_DS="C+T+K+S+H", distributed=False, rank=0,
     GPUs by DataParallel — keep global batch = batch_size * world_size in
     mind when reproducing reference schedules).
     """
-    if args.stage == "chairs":
-        aug_params = {"crop_size": args.image_size, "min_scale": -0.1,
-                      "max_scale": 1.0, "do_flip": True}
-        train_dataset = FlyingChairs(aug_params, split="training")
-    elif args.stage == "things":
-        aug_params = {"crop_size": args.image_size, "min_scale": -0.4,
-                      "max_scale": 0.8, "do_flip": True}
-        clean = FlyingThings3D(aug_params, dstype="frames_cleanpass",
-                               load_compressed=args.compressed_ft)
-        final = FlyingThings3D(aug_params, dstype="frames_finalpass",
-                               load_compressed=args.compressed_ft)
-        train_dataset = clean + final
-    elif args.stage == "sintel":
-        aug_params = {"crop_size": args.image_size, "min_scale": -0.2,
-                      "max_scale": 0.6, "do_flip": True}
-        things = FlyingThings3D(aug_params, dstype="frames_cleanpass")
-        sintel_clean = MpiSintel(aug_params, split="training", dstype="clean")
-        sintel_final = MpiSintel(aug_params, split="training", dstype="final")
-
-        if TRAIN_DS == "C+T+K+S+H":
-            kitti = KITTI({"crop_size": args.image_size, "min_scale": -0.3,
-                           "max_scale": 0.5, "do_flip": True})
-            hd1k = HD1K({"crop_size": args.image_size, "min_scale": -0.5,
-                         "max_scale": 0.2, "do_flip": True})
-            train_dataset = (100 * sintel_clean + 100 * sintel_final
-                             + 200 * kitti + 5 * hd1k + things)
-        elif TRAIN_DS == "C+T+K/S":
-            train_dataset = 100 * sintel_clean + 100 * sintel_final + things
-    elif args.stage == "kitti":
-        aug_params = {"crop_size": args.image_size, "min_scale": -0.2,
-                      "max_scale": 0.4, "do_flip": False}
-        train_dataset = KITTI(aug_params, split="training")
-    elif args.stage == "synthetic":
-        train_dataset = SyntheticFlowDataset(image_size=args.image_size)
-    else:
-        raise ValueError(f"unknown stage {args.stage!r}")
+    train_dataset = _stage_dataset(args, TRAIN_DS)
 
     sampler = None
     shuffle = True
