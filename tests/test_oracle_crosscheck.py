@@ -315,3 +315,37 @@ def test_convex_upsample_matches_naive(h, w, factor, seed):
                 val += weights[k] * factor * flow[0, c, yy, xx]
         np.testing.assert_allclose(out[0, c, y * factor + dy, x * factor + dx],
                                    val, atol=1e-5)
+
+
+@settings(max_examples=8, deadline=None)
+@given(st.integers(8, 16), st.integers(8, 16), st.integers(0, 2 ** 31 - 1))
+def test_corr_lookup_adversarial_coords(h, w, seed):
+    """Boundary-hostile coords — exact integers, negatives, far out of
+    range — where bilinear implementations typically diverge (floor vs
+    trunc, edge-tap weighting). All levels kept >=2 px per side (the
+    documented input constraint, PARITY.md deviation 6)."""
+    rng = np.random.default_rng(seed)
+    radius, levels = 2, 2
+    P = h * w
+    pyramid = [torch.from_numpy(
+        rng.standard_normal((P, 1, h >> l, w >> l)).astype(np.float32))
+        for l in range(levels)]
+    base = rng.random((1, 2, h, w)) * [[[[2 * w]], [[2 * h]]]] \
+        - [[[[w / 2]], [[h / 2]]]]
+    mask = rng.random((1, 2, h, w)) < 0.3
+    base[mask] = np.round(base[mask])
+    coords = torch.from_numpy(base.astype(np.float32))
+
+    out = torch_ref.corr_lookup(pyramid, coords, radius).numpy()
+    K = 2 * radius + 1
+    for _ in range(12):
+        i = rng.integers(P)
+        l = rng.integers(levels)
+        a = rng.integers(K)
+        c = rng.integers(K)
+        y, x = divmod(int(i), w)
+        cx = coords[0, 0, y, x].item() / (1 << l) + (a - radius)
+        cy = coords[0, 1, y, x].item() / (1 << l) + (c - radius)
+        ref = naive_bilinear(pyramid[l][i, 0].numpy(), cx, cy)
+        ch = l * K * K + a * K + c
+        np.testing.assert_allclose(out[0, ch, y, x], ref, atol=2e-5)
