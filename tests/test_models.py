@@ -170,3 +170,30 @@ def test_enforce_pos_weight_p_semantics():
     # the effective weight is positive everywhere
     eff = F.softplus(wp, beta=10)
     assert (eff > 0).all()
+
+
+def test_channels_last_policy():
+    """Layout policy (flowhip/utils/layout.py): conv weights channels_last
+    everywhere EXCEPT the NCUP upsampler subtree (NCHW for the nconv
+    kernels + fp32 Winograd), with the weights-est confidence net back on
+    channels_last (bf16/NHWC island). Applied by build_model on every
+    device, so it is testable on CPU."""
+    import torch
+
+    args = default_ncup_args(model="raft_nc_dbl")
+    model = build_model(args)
+
+    def is_cl(conv):
+        return conv.weight.is_contiguous(memory_format=torch.channels_last)
+
+    # encoder + update-block convs: channels_last
+    assert is_cl(model.fnet.conv1)
+    assert is_cl(model.update_block.encoder.convc1)
+    # upsampler interpolation net: NCHW (the nconv_in weight has Cin=1,
+    # where both layouts coincide — check a multi-channel decoder weight)
+    wp = model.upsampler.interpolation_net.decoder[0].weight_p
+    assert wp.shape[1] > 1
+    assert wp.is_contiguous() and not wp.is_contiguous(
+        memory_format=torch.channels_last)
+    # weights-est confidence net: back to channels_last
+    assert is_cl(model.upsampler.weights_est_net.conv[0][0])
