@@ -340,3 +340,33 @@ def test_train_determinism_same_seed(tmp_path, monkeypatch):
     assert a.keys() == b.keys()
     for k in a:
         assert torch.equal(a[k], b[k]), k
+
+
+def test_graphed_inference_requires_gpu():
+    """GraphedInference fails loudly off-GPU (capture is hipGraph-only)."""
+    from flowhip.engine.graph import GraphedInference
+
+    args = default_ncup_args(model="raft_nc_dbl", small=True)
+    model = build_model(args)
+    if not torch.cuda.is_available():
+        with pytest.raises(AssertionError):
+            GraphedInference(model, (1, 3, 128, 128), iters=2)
+
+
+def test_logger_accepts_numpy_scalars(tmp_path):
+    """Validators return np.float64 means — the jsonl stream must accept
+    them (np.float64 subclasses float; a regression to np.float32 would
+    break json serialization)."""
+    import argparse
+
+    import numpy as np
+
+    from flowhip.engine.logger import Logger
+
+    logger = Logger(None, argparse.Namespace(name="nplog"),
+                    run_dir=str(tmp_path / "run"))
+    logger.write_dict({"kitti-epe": np.float64(1.25),
+                       "kitti-f1": np.float64(7.5)})
+    logger.close()
+    rec = json.loads(open(tmp_path / "run" / "metrics.jsonl").read())
+    assert rec["validation"]["kitti-epe"] == 1.25
