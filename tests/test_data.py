@@ -475,12 +475,23 @@ def test_things_and_kitti_stage_recipes(tmp_path, monkeypatch):
 
 def test_read_disp_kitti(tmp_path):
     """16-bit grayscale disparity png: disp = px/256, valid = disp>0, flow =
-    (-disp, 0) (reference frame_utils.py:109-113; PIL mode 'I' replaces
-    cv2.IMREAD_ANYDEPTH)."""
-    from PIL import Image
+    (-disp, 0) (reference frame_utils.py:109-113; PIL's 16-bit read
+    replaces cv2.IMREAD_ANYDEPTH)."""
+    import struct
+    import zlib
 
     disp = (np.arange(12, dtype=np.uint16).reshape(3, 4)) * 128
-    Image.fromarray(disp, mode="I;16").save(tmp_path / "d.png")
+    # write the 16-bit grayscale png manually (PIL deprecated I-mode saves)
+    raw = b"".join(b"\x00" + row.astype(">u2").tobytes() for row in disp)
+
+    def chunk(tag, payload):
+        return (struct.pack(">I", len(payload)) + tag + payload
+                + struct.pack(">I", zlib.crc32(tag + payload)))
+
+    png = (b"\x89PNG\r\n\x1a\n"
+           + chunk(b"IHDR", struct.pack(">IIBBBBB", 4, 3, 16, 0, 0, 0, 0))
+           + chunk(b"IDAT", zlib.compress(raw)) + chunk(b"IEND", b""))
+    (tmp_path / "d.png").write_bytes(png)
 
     flow, valid = frame_utils.readDispKITTI(str(tmp_path / "d.png"))
     assert flow.shape == (3, 4, 2)

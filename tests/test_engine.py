@@ -79,7 +79,10 @@ def test_sintel_submission_writer(tmp_path, monkeypatch):
         scene = tmp_path / "datasets" / "Sintel" / "test" / dstype / "seq_1"
         scene.mkdir(parents=True)
         for i in range(3):
-            arr = (np.random.rand(64, 96, 3) * 255).astype(np.uint8)
+            # >=128 px per side: below that the 4-level pyramid has a 1-px
+            # level and the reference-parity lookup path produces NaNs
+            # (PARITY.md deviation 6)
+            arr = (np.random.rand(128, 136, 3) * 255).astype(np.uint8)
             Image.fromarray(arr).save(scene / f"frame_{i:04d}.png")
 
     monkeypatch.chdir(tmp_path)
@@ -92,7 +95,7 @@ def test_sintel_submission_writer(tmp_path, monkeypatch):
     flo = tmp_path / "out" / "clean" / "seq_1" / "frame0001.flo"
     assert flo.exists()
     flow = frame_utils.readFlow(str(flo))
-    assert flow.shape == (64, 96, 2)
+    assert flow.shape == (128, 136, 2)
     assert (tmp_path / "out" / "final" / "seq_1" / "frame0002.flo").exists()
 
 
@@ -108,7 +111,7 @@ def test_kitti_submission_writer(tmp_path, monkeypatch):
     img_dir = tmp_path / "datasets" / "KITTI" / "testing" / "image_2"
     img_dir.mkdir(parents=True)
     for suffix in ("10", "11"):
-        arr = (np.random.rand(64, 96, 3) * 255).astype(np.uint8)
+        arr = (np.random.rand(128, 136, 3) * 255).astype(np.uint8)
         Image.fromarray(arr).save(img_dir / f"000000_{suffix}.png")
 
     monkeypatch.chdir(tmp_path)
@@ -121,7 +124,7 @@ def test_kitti_submission_writer(tmp_path, monkeypatch):
     out = tmp_path / "kout" / "000000_10.png"
     assert out.exists()
     flow, valid = frame_utils.readFlowKITTI(str(out))
-    assert flow.shape == (64, 96, 2)
+    assert flow.shape == (128, 136, 2)
     assert valid.min() == 1  # submission marks everything valid
 
 
