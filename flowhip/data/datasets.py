@@ -319,7 +319,8 @@ def fetch_dataloader(args, TRAIN_DS="C+T+K+S+H", distributed=False, rank=0,
         shuffle = False
 
     train_loader = data.DataLoader(
-        train_dataset, batch_size=args.batch_size, pin_memory=True,
+        train_dataset, batch_size=args.batch_size,
+        pin_memory=torch.cuda.is_available(),
         shuffle=shuffle, sampler=sampler,
         num_workers=getattr(args, "num_workers", 4), drop_last=True,
         persistent_workers=getattr(args, "num_workers", 4) > 0)
