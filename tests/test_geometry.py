@@ -71,3 +71,25 @@ def test_forward_interpolate_constant_shift():
     assert out.shape == (2, 10, 14)
     # splatted positions carry the same constant flow
     assert torch.allclose(out[0], torch.full_like(out[0], 3.0), atol=1e-4)
+
+
+def test_bilinear_sampler_mask_branch():
+    """mask=True returns strict-interior validity in normalized coords
+    (reference utils.py:69-71)."""
+    img = torch.randn(1, 1, 5, 7)
+    coords = torch.tensor([[[[0.0, 0.0], [6.0, 4.0], [-1.0, 2.0],
+                             [3.0, 2.0]]]])
+    out, mask = bilinear_sampler(img, coords, mask=True)
+    # corners sit exactly on the +-1 normalized boundary -> excluded (strict)
+    assert mask.flatten().tolist() == [0.0, 0.0, 0.0, 1.0]
+    assert out.shape[:2] == (1, 1)
+
+
+def test_input_padder_noop_when_divisible():
+    """Dims already divisible by 8 pad by zero in both modes."""
+    x = torch.randn(1, 3, 64, 96)
+    for mode in ("sintel", "kitti"):
+        p = InputPadder(x.shape, mode=mode)
+        (y,) = p.pad(x.clone())
+        assert y.shape == x.shape
+        assert torch.equal(p.unpad(y), x)
