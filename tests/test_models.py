@@ -197,3 +197,20 @@ def test_channels_last_policy():
         memory_format=torch.channels_last)
     # weights-est confidence net: back to channels_last
     assert is_cl(model.upsampler.weights_est_net.conv[0][0])
+
+
+def test_raft_nc_dbl_non_square_shape():
+    """Forward/backward at a non-square, non-power-of-two /8 shape
+    (136x152) — exercises the pyramid with odd level sizes (17x19 -> 8x9
+    -> 4x4 -> 2x2) and the NCUP scale chain off the benchmark shapes."""
+    torch.manual_seed(0)
+    args = default_ncup_args(model="raft_nc_dbl", small=True)
+    model = build_model(args)
+    img1 = torch.rand(1, 3, 136, 152) * 255
+    img2 = torch.rand(1, 3, 136, 152) * 255
+    preds = model(img1, img2, iters=2)
+    assert preds[-1].shape == (1, 2, 136, 152)
+    assert all(torch.isfinite(p).all() for p in preds)
+    preds[-1].abs().mean().backward()
+    g = model.fnet.conv1.weight.grad
+    assert g is not None and torch.isfinite(g).all()
