@@ -1,8 +1,12 @@
 """Flow datasets + the training dataloader recipe.
 
-Behavioral parity with the reference `core/datasets.py` (file-list layouts,
-per-worker RNG reseeding, test mode, oversampling via __rmul__, stage
-mixture recipes). Additions for the MI355X framework:
+Matches the observable behavior of the reference `core/datasets.py`
+(directory layouts, train/val splits, test mode, oversampling, per-worker
+RNG reseeding, stage mixture recipes) with a record-based design: each
+dataset is a discovery function producing a flat list of `FlowSample`
+records, consumed by one generic loading pipeline.
+
+Additions for the MI355X framework:
 
 - `SyntheticFlowDataset`: random image pairs + random smooth flow of a given
   shape — powers bench.py and GPU tests (no network for real datasets).
@@ -11,10 +15,13 @@ mixture recipes). Additions for the MI355X framework:
   drop_last — datasets.py:240-241).
 """
 
-import os
 import os.path as osp
 import random
+from dataclasses import dataclass, field
 from glob import glob
+from itertools import groupby
+from pathlib import Path
+from typing import Optional, Tuple
 
 import numpy as np
 import torch
@@ -24,195 +31,225 @@ from . import frame_utils
 from .augmentor import FlowAugmentor, SparseFlowAugmentor
 
 
-class FlowDataset(data.Dataset):
-    def __init__(self, aug_params=None, sparse=False):
-        self.augmentor = None
-        self.sparse = sparse
-        if aug_params is not None:
-            if sparse:
-                self.augmentor = SparseFlowAugmentor(**aug_params)
-            else:
-                self.augmentor = FlowAugmentor(**aug_params)
+@dataclass(frozen=True)
+class FlowSample:
+    """One training/eval example: a frame pair, its GT flow (if any), and
+    the metadata the submission writers need (scene/frame id)."""
 
-        self.is_test = False
-        self.init_seed = False
-        self.flow_list = []
-        self.image_list = []
-        self.extra_info = []
+    frame_a: str
+    frame_b: str
+    flow: Optional[str] = None
+    meta: Tuple = field(default_factory=tuple)
+
+
+def _seed_worker_once():
+    """First-touch per-worker RNG reseed (reference datasets.py:45-51):
+    each DataLoader worker reseeds all three RNG streams from its worker id
+    so augmentation draws differ across workers but are reproducible."""
+    info = torch.utils.data.get_worker_info()
+    if info is not None:
+        torch.manual_seed(info.id)
+        np.random.seed(info.id)
+        random.seed(info.id)
+    return info is not None
+
+
+def _load_frame(path):
+    """Decode one image to uint8 HxWx3: alpha dropped, grayscale
+    replicated across channels (reference datasets.py:63-73)."""
+    arr = np.asarray(frame_utils.read_gen(path)).astype(np.uint8)
+    if arr.ndim == 2:
+        arr = np.repeat(arr[:, :, None], 3, axis=2)
+    return arr[:, :, :3]
+
+
+def _chw_float(arr):
+    return torch.from_numpy(np.ascontiguousarray(arr.transpose(2, 0, 1))).float()
+
+
+class FlowDataset(data.Dataset):
+    """Generic pipeline over a list of FlowSample records.
+
+    `sparse` selects the KITTI 16-bit GT reader + the sparse augmentor;
+    `test_mode` (set by test splits) makes __getitem__ return
+    (img1, img2, meta) instead of (img1, img2, flow, valid).
+    """
+
+    def __init__(self, aug_params=None, sparse=False):
+        self.sparse = sparse
+        self.test_mode = False
+        self.samples = []
+        self._worker_seeded = False
+        cls = SparseFlowAugmentor if sparse else FlowAugmentor
+        self.augmentor = cls(**aug_params) if aug_params is not None else None
+
+    def __len__(self):
+        return len(self.samples)
+
+    def __rmul__(self, times):
+        """`k * ds` oversamples by repeating the record list (reference
+        datasets.py:93-96; used by the sintel mixture weights)."""
+        self.samples = times * self.samples
+        return self
+
+    def _read_gt(self, sample):
+        """Returns (flow HxWx2 float32, valid HxW or None)."""
+        if self.sparse:
+            return frame_utils.readFlowKITTI(sample.flow)
+        return np.asarray(frame_utils.read_gen(sample.flow),
+                          dtype=np.float32), None
 
     def __getitem__(self, index):
-        if self.is_test:
-            img1 = frame_utils.read_gen(self.image_list[index][0])
-            img2 = frame_utils.read_gen(self.image_list[index][1])
-            img1 = np.array(img1).astype(np.uint8)[..., :3]
-            img2 = np.array(img2).astype(np.uint8)[..., :3]
-            img1 = torch.from_numpy(img1).permute(2, 0, 1).float()
-            img2 = torch.from_numpy(img2).permute(2, 0, 1).float()
-            return img1, img2, self.extra_info[index]
+        sample = self.samples[index % len(self.samples)]
+        img1 = _load_frame(sample.frame_a)
+        img2 = _load_frame(sample.frame_b)
 
-        if not self.init_seed:
-            worker_info = torch.utils.data.get_worker_info()
-            if worker_info is not None:
-                torch.manual_seed(worker_info.id)
-                np.random.seed(worker_info.id)
-                random.seed(worker_info.id)
-                self.init_seed = True
+        if self.test_mode:
+            return _chw_float(img1), _chw_float(img2), sample.meta
 
-        index = index % len(self.image_list)
-        valid = None
-        if self.sparse:
-            flow, valid = frame_utils.readFlowKITTI(self.flow_list[index])
-        else:
-            flow = frame_utils.read_gen(self.flow_list[index])
+        if not self._worker_seeded:
+            self._worker_seeded = _seed_worker_once()
 
-        img1 = frame_utils.read_gen(self.image_list[index][0])
-        img2 = frame_utils.read_gen(self.image_list[index][1])
-
-        flow = np.array(flow).astype(np.float32)
-        img1 = np.array(img1).astype(np.uint8)
-        img2 = np.array(img2).astype(np.uint8)
-
-        if len(img1.shape) == 2:  # grayscale
-            img1 = np.tile(img1[..., None], (1, 1, 3))
-            img2 = np.tile(img2[..., None], (1, 1, 3))
-        else:
-            img1 = img1[..., :3]
-            img2 = img2[..., :3]
+        flow, valid = self._read_gt(sample)
 
         if self.augmentor is not None:
             if self.sparse:
-                img1, img2, flow, valid = self.augmentor(img1, img2, flow, valid)
+                img1, img2, flow, valid = self.augmentor(img1, img2, flow,
+                                                         valid)
             else:
                 img1, img2, flow = self.augmentor(img1, img2, flow)
 
-        img1 = torch.from_numpy(img1).permute(2, 0, 1).float()
-        img2 = torch.from_numpy(img2).permute(2, 0, 1).float()
-        flow = torch.from_numpy(flow).permute(2, 0, 1).float()
-
-        if valid is not None:
-            valid = torch.from_numpy(np.ascontiguousarray(valid))
+        flow = _chw_float(flow)
+        if valid is None:
+            # dense GT: anything below the reference's sentinel magnitude
+            # counts as valid (datasets.py:88)
+            valid = (flow.abs() < 1000).all(dim=0)
         else:
-            valid = (flow[0].abs() < 1000) & (flow[1].abs() < 1000)
-
-        return img1, img2, flow, valid.float()
-
-    def __rmul__(self, v):
-        self.flow_list = v * self.flow_list
-        self.image_list = v * self.image_list
-        return self
-
-    def __len__(self):
-        return len(self.image_list)
+            valid = torch.from_numpy(np.ascontiguousarray(valid))
+        return _chw_float(img1), _chw_float(img2), flow, valid.float()
 
 
 class MpiSintel(FlowDataset):
-    def __init__(self, aug_params=None, split="training", root="datasets/Sintel",
-                 dstype="clean"):
+    """Scene-structured Sintel layout: <root>/<split>/<dstype>/<scene>/*.png
+    with per-scene flow dirs (reference datasets.py:102-119)."""
+
+    def __init__(self, aug_params=None, split="training",
+                 root="datasets/Sintel", dstype="clean"):
         super().__init__(aug_params)
-        flow_root = osp.join(root, split, "flow")
-        image_root = osp.join(root, split, dstype)
+        self.test_mode = split == "test"
 
-        if split == "test":
-            self.is_test = True
-
-        for scene in sorted(os.listdir(image_root)) if osp.isdir(image_root) else []:
-            image_list = sorted(glob(osp.join(image_root, scene, "*.png")))
-            for i in range(len(image_list) - 1):
-                self.image_list += [[image_list[i], image_list[i + 1]]]
-                self.extra_info += [(scene, i)]
-            if split != "test":
-                self.flow_list += sorted(glob(osp.join(flow_root, scene, "*.flo")))
+        scene_root = Path(root) / split / dstype
+        scenes = sorted(p for p in scene_root.iterdir() if p.is_dir()) \
+            if scene_root.is_dir() else []
+        for scene in scenes:
+            frames = sorted(str(p) for p in scene.glob("*.png"))
+            flows = [] if self.test_mode else sorted(
+                str(p) for p in (Path(root) / split / "flow"
+                                 / scene.name).glob("*.flo"))
+            for i, (a, b) in enumerate(zip(frames, frames[1:])):
+                self.samples.append(FlowSample(
+                    a, b, flows[i] if flows else None,
+                    meta=(scene.name, i)))
 
 
 class FlyingChairs(FlowDataset):
+    """Flat-directory chairs layout: NNNNN_img{1,2}.png + NNNNN_flow.flo,
+    train/val membership from the shipped split table (reference
+    datasets.py:121-137; chairs_split.txt rows: 1=train, 2=val)."""
+
     def __init__(self, aug_params=None, split="train",
                  root="datasets/FlyingChairs_release/data",
                  split_file="chairs_split.txt"):
         super().__init__(aug_params)
 
-        images = sorted(glob(osp.join(root, "*_img*.png")))
         flows = sorted(glob(osp.join(root, "*_flow.flo")))
         if not flows:
             return
-        assert len(images) // 2 == len(flows)
-
-        split_list = np.loadtxt(split_file, dtype=np.int32)
-        for i in range(len(flows)):
-            xid = split_list[i]
-            if (split == "training" and xid == 1) or (split == "validation" and xid == 2):
-                self.flow_list += [flows[i]]
-                self.image_list += [[images[2 * i], images[2 * i + 1]]]
+        table = np.loadtxt(split_file, dtype=np.int32)
+        want = 1 if split == "training" else 2
+        for i, flow_path in enumerate(flows):
+            if table[i] != want:
+                continue
+            stem = flow_path[:-len("_flow.flo")]
+            self.samples.append(FlowSample(
+                stem + "_img1.png", stem + "_img2.png", flow_path))
 
 
 class FlyingThings3D(FlowDataset):
+    """FlyingThings3D: per-scene frame dirs paired with per-direction flow
+    dirs; each scene contributes forward pairs (into_future) and reversed
+    pairs (into_past) (reference datasets.py:138-167).
+
+    Reference defect fixed (core/datasets.py:144-146): the reference appends
+    "_webp" to dstype INSIDE the direction loop, so the second pass globs
+    "<dstype>_webp_webp" and the into_past pairs silently vanish in
+    compressed mode. Appended once here (divergence noted in PARITY.md).
+    """
+
     def __init__(self, aug_params=None, root="datasets/FlyingThings3D",
                  dstype="frames_cleanpass", load_compressed=False):
         super().__init__(aug_params)
 
-        # Reference defect fixed (core/datasets.py:144-146): the reference
-        # appends "_webp" to dstype INSIDE the direction loop, so the second
-        # pass globs "<dstype>_webp_webp" and the into_past pairs silently
-        # vanish in compressed mode. Appended once here.
+        img_ext, flo_ext = ((".webp", ".npz") if load_compressed
+                            else (".png", ".pfm"))
         if load_compressed:
-            dstype += "_webp"
+            dstype = dstype + "_webp"
 
-        for cam in ["left"]:
-            for direction in ["into_future", "into_past"]:
-                image_dirs = sorted(glob(osp.join(root, dstype, "TRAIN/*/*")))
-                image_dirs = sorted([osp.join(f, cam) for f in image_dirs])
-
-                flow_dirs = sorted(glob(osp.join(root, "optical_flow/TRAIN/*/*")))
-                flow_dirs = sorted([osp.join(f, direction, cam) for f in flow_dirs])
-
-                for idir, fdir in zip(image_dirs, flow_dirs):
-                    if load_compressed:
-                        images = sorted(glob(osp.join(idir, "*.webp")))
-                        flows = sorted(glob(osp.join(fdir, "*.npz")))
-                    else:
-                        images = sorted(glob(osp.join(idir, "*.png")))
-                        flows = sorted(glob(osp.join(fdir, "*.pfm")))
-                    for i in range(len(flows) - 1):
-                        if direction == "into_future":
-                            self.image_list += [[images[i], images[i + 1]]]
-                            self.flow_list += [flows[i]]
-                        elif direction == "into_past":
-                            self.image_list += [[images[i + 1], images[i]]]
-                            self.flow_list += [flows[i + 1]]
+        scene_dirs = sorted(glob(osp.join(root, dstype, "TRAIN/*/*")))
+        for scene in scene_dirs:
+            rel = osp.join(*scene.split("/")[-2:])  # e.g. A/0000
+            frames = sorted(glob(osp.join(scene, "left", "*" + img_ext)))
+            for direction in ("into_future", "into_past"):
+                flows = sorted(glob(osp.join(
+                    root, "optical_flow/TRAIN", rel, direction, "left",
+                    "*" + flo_ext)))
+                if direction == "into_future":
+                    for a, b, fl in zip(frames, frames[1:], flows):
+                        self.samples.append(FlowSample(a, b, fl))
+                else:
+                    # into_past: the pair is reversed and the flow belongs
+                    # to the LATER frame
+                    for a, b, fl in zip(frames, frames[1:], flows[1:]):
+                        self.samples.append(FlowSample(b, a, fl))
 
 
 class KITTI(FlowDataset):
-    def __init__(self, aug_params=None, split="training", root="datasets/KITTI"):
+    """KITTI-2015 layout: image_2/<id>_10.png + <id>_11.png frame pairs with
+    sparse flow_occ 16-bit GT named after the first frame (reference
+    datasets.py:169-186)."""
+
+    def __init__(self, aug_params=None, split="training",
+                 root="datasets/KITTI"):
         super().__init__(aug_params, sparse=True)
-        if split == "testing":
-            self.is_test = True
+        self.test_mode = split == "testing"
 
-        root = osp.join(root, split)
-        images1 = sorted(glob(osp.join(root, "image_2/*_10.png")))
-        images2 = sorted(glob(osp.join(root, "image_2/*_11.png")))
-
-        for img1, img2 in zip(images1, images2):
-            frame_id = img1.split("/")[-1]
-            self.extra_info += [[frame_id]]
-            self.image_list += [[img1, img2]]
-
-        if split == "training":
-            self.flow_list = sorted(glob(osp.join(root, "flow_occ/*_10.png")))
+        base = Path(root) / split
+        for first in sorted((base / "image_2").glob("*_10.png")):
+            second = first.with_name(first.name.replace("_10.", "_11."))
+            gt = base / "flow_occ" / first.name
+            self.samples.append(FlowSample(
+                str(first), str(second),
+                None if self.test_mode else str(gt),
+                meta=(first.name,)))
 
 
 class HD1K(FlowDataset):
+    """HD1K: frames named <seq>_<frame>.png; consecutive frames within each
+    sequence form pairs, the last frame of a sequence is unpaired
+    (reference datasets.py:188-204)."""
+
     def __init__(self, aug_params=None, root="datasets/HD1k"):
         super().__init__(aug_params, sparse=True)
 
-        seq_ix = 0
-        while 1:
-            flows = sorted(glob(osp.join(root, "hd1k_flow_gt", "flow_occ/%06d_*.png" % seq_ix)))
-            images = sorted(glob(osp.join(root, "hd1k_input", "image_2/%06d_*.png" % seq_ix)))
-            if len(flows) == 0:
-                break
-            for i in range(len(flows) - 1):
-                self.flow_list += [flows[i]]
-                self.image_list += [[images[i], images[i + 1]]]
-            seq_ix += 1
+        def frame_of(flow_path):
+            return osp.join(root, "hd1k_input/image_2",
+                            osp.basename(flow_path))
+
+        flows = sorted(glob(osp.join(root, "hd1k_flow_gt/flow_occ/*.png")))
+        for _, group in groupby(flows, key=lambda p: osp.basename(p).split("_")[0]):
+            group = list(group)
+            for fl, nxt in zip(group, group[1:]):
+                self.samples.append(FlowSample(frame_of(fl), frame_of(nxt), fl))
 
 
 class SyntheticFlowDataset(data.Dataset):

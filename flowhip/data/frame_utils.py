@@ -3,70 +3,60 @@
 FlyingThings webp/npz), implemented on numpy + PIL (no OpenCV dependency).
 """
 
-import re
 from os.path import splitext
 
 import numpy as np
 from PIL import Image
 
-TAG_CHAR = np.array([202021.25], np.float32)
+# Middlebury .flo layout: 4-byte float sanity tag, then little-endian
+# int32 width, int32 height, then h*w (u, v) float32 pairs row-major.
+FLO_TAG = 202021.25
 
 
 def readFlow(fn):
-    """Read a Middlebury .flo file (magic 202021.25, little-endian)."""
+    """Read a Middlebury .flo file into an HxWx2 float32 array."""
     with open(fn, "rb") as f:
-        magic = np.fromfile(f, np.float32, count=1)
-        if magic.size == 0 or magic[0] != 202021.25:
-            print("Magic number incorrect. Invalid .flo file")
-            return None
-        w = int(np.fromfile(f, np.int32, count=1)[0])
-        h = int(np.fromfile(f, np.int32, count=1)[0])
-        data = np.fromfile(f, np.float32, count=2 * w * h)
-        return np.resize(data, (h, w, 2))
+        head = f.read(12)
+    if len(head) < 12 or np.frombuffer(head, "<f4", 1)[0] != FLO_TAG:
+        print("Magic number incorrect. Invalid .flo file")
+        return None
+    w, h = np.frombuffer(head, "<i4", 2, offset=4)
+    body = np.fromfile(fn, "<f4", offset=12)
+    return body[:2 * w * h].reshape(int(h), int(w), 2)
 
 
 def writeFlow(filename, uv, v=None):
-    """Write a Middlebury .flo file (parity: frame_utils.py:70-99)."""
-    if v is None:
-        assert uv.ndim == 3 and uv.shape[2] == 2
-        u = uv[:, :, 0]
-        v = uv[:, :, 1]
-    else:
-        u = uv
-    assert u.shape == v.shape
-    height, width = u.shape
+    """Write a Middlebury .flo file. Accepts either one HxWx2 array or
+    separate u, v planes (reference API — frame_utils.py:70-99)."""
+    if v is not None:
+        uv = np.stack([uv, v], axis=-1)
+    assert uv.ndim == 3 and uv.shape[2] == 2
+    h, w = uv.shape[:2]
     with open(filename, "wb") as f:
-        f.write(TAG_CHAR.tobytes())
-        np.array(width, np.int32).tofile(f)
-        np.array(height, np.int32).tofile(f)
-        tmp = np.zeros((height, width * 2), np.float32)
-        tmp[:, 0::2] = u
-        tmp[:, 1::2] = v
-        tmp.tofile(f)
+        f.write(np.float32(FLO_TAG).tobytes())
+        f.write(np.asarray([w, h], "<i4").tobytes())
+        # HWC float32 is already the interleaved (u, v)-per-pixel layout
+        f.write(np.ascontiguousarray(uv, "<f4").tobytes())
 
 
 def readPFM(file):
-    """Read a PFM file (FlyingThings3D flow)."""
+    """Read a PFM image (FlyingThings3D flow GT). Header = type line
+    ('PF' color / 'Pf' gray), dimensions line, scale line (sign encodes
+    endianness); pixel rows are stored bottom-to-top."""
     with open(file, "rb") as f:
-        header = f.readline().rstrip()
-        if header == b"PF":
-            color = True
-        elif header == b"Pf":
-            color = False
-        else:
-            raise Exception("Not a PFM file.")
-
-        dim_match = re.match(rb"^(\d+)\s(\d+)\s$", f.readline())
-        if not dim_match:
-            raise Exception("Malformed PFM header.")
-        width, height = map(int, dim_match.groups())
-
-        scale = float(f.readline().rstrip())
-        endian = "<" if scale < 0 else ">"
-
-        data = np.fromfile(f, endian + "f")
-    shape = (height, width, 3) if color else (height, width)
-    return np.flipud(np.reshape(data, shape))
+        kind = f.readline().strip()
+        if kind not in (b"PF", b"Pf"):
+            raise ValueError(f"{file}: not a PFM file")
+        dims = f.readline().split()
+        if len(dims) != 2:
+            raise ValueError(f"{file}: malformed PFM dimensions")
+        w, h = (int(d) for d in dims)
+        scale = float(f.readline())
+        dtype = np.dtype("<f4" if scale < 0 else ">f4")
+        channels = 3 if kind == b"PF" else 1
+        pixels = np.fromfile(f, dtype, w * h * channels)
+    img = pixels.reshape((h, w, 3) if channels == 3 else (h, w))
+    return img[::-1]
 
 
 def _read_png16_bgr(filename):
