@@ -73,8 +73,16 @@ def cleanup():
         dist.destroy_process_group()
 
 
-def wrap_ddp(model, device):
-    """Wrap for data-parallel training (one bucket, see module docstring)."""
+def wrap_ddp(model, device, static_iters=True):
+    """Wrap for data-parallel training (one bucket, see module docstring).
+
+    CONTRACT: static_graph=True assumes the unrolled autograd
+    graph is IDENTICAL every step — fixed --iters, no activation
+    checkpointing, no conditionally-skipped branches. The training driver
+    satisfies this (iters is constant for a run). Pass static_iters=False
+    if a variable-schedule caller is ever added, or DDP will fail loudly
+    mid-training.
+    """
     if not is_initialized():
         return model
     # ~21 MB of fp32 grads (5.3M params): one flat bucket — a ring stage is
@@ -82,7 +90,7 @@ def wrap_ddp(model, device):
     # latency. static_graph: the unrolled iteration graph is identical every
     # step (fixed --iters), letting DDP skip graph re-discovery.
     kwargs = dict(bucket_cap_mb=64, gradient_as_bucket_view=True,
-                  static_graph=True)
+                  static_graph=static_iters)
     if device.type == "cuda":
         kwargs["device_ids"] = [device.index]
     return torch.nn.parallel.DistributedDataParallel(model, **kwargs)

@@ -61,7 +61,8 @@ class ConvGRU(nn.Module):
             hx, torch.cat([self.convz.weight, self.convr.weight]),
             torch.cat([self.convz.bias, self.convr.bias]), 1, 1, 1, 1,
             self._zr_cache,
-            key=(self.convz.weight._version, self.convr.weight._version))
+            key=(self.convz.weight.data_ptr(), self.convz.weight._version,
+                 self.convr.weight.data_ptr(), self.convr.weight._version))
         if h.is_cuda:
             from ..ops.functional_gru import GruGate1Fn, GruGate2Fn
             z, rh = GruGate1Fn.apply(zr, h)
@@ -91,7 +92,9 @@ class SepConvGRU(nn.Module):
         # virtually-concatenated input (no cat materialization on GPU); the
         # packed-weight cache is keyed by the source params' versions
         zr = fused_conv2d_cat2(h, x, zr_w, zr_b, padding, zr_cache,
-                               key=(convz.weight._version,
+                               key=(convz.weight.data_ptr(),
+                                    convz.weight._version,
+                                    convr.weight.data_ptr(),
                                     convr.weight._version))
         if h.is_cuda:
             # fused gate kernels (ops/functional_gru): one kernel for
@@ -99,7 +102,8 @@ class SepConvGRU(nn.Module):
             from ..ops.functional_gru import GruGate1Fn, GruGate2Fn
             z, rh = GruGate1Fn.apply(zr, h)
             qp = fused_conv2d_cat2(rh, x, convq.weight, convq.bias, padding,
-                                   q_cache, key=(convq.weight._version,))
+                                   q_cache, key=(convq.weight.data_ptr(),
+                                                 convq.weight._version))
             return GruGate2Fn.apply(qp, z, h)
         z, r = torch.sigmoid(zr).chunk(2, dim=1)
         q = torch.tanh(convq(torch.cat([r * h, x], dim=1)))

@@ -69,7 +69,7 @@ def get_upsampler(in_ch, guidance_ch, args, respect_choice=False):
         else:
             raise NotImplementedError(f"weights_est_net {name!r}")
 
-        return NConvUpsampler(
+        ups = NConvUpsampler(
             scale=args.final_upsampling_scale,
             interpolation_net=interpolation_net,
             weights_est_net=weights_est_net,
@@ -77,6 +77,11 @@ def get_upsampler(in_ch, guidance_ch, args, respect_choice=False):
             channels_to_batch=args.final_upsampling_channels_to_batch,
             use_residuals=args.final_upsampling_use_residuals,
             est_on_high_res=args.final_upsampling_est_on_high_res)
+        # plain attribute, NOT a constructor kwarg: the reflective config
+        # mirrors __init__ signatures into CLI flags and the flag surface
+        # must stay identical to the reference's
+        ups.conf_net_bf16 = getattr(args, "mixed_precision", False)
+        return ups
 
     if upsampler_name == "bilinear":
         return Bilinear(args.final_upsampling_scale)
@@ -104,6 +109,10 @@ class NConvUpsampler(nn.Module):
                  channels_to_batch=True, use_residuals=False,
                  est_on_high_res=False):
         super().__init__()
+        # bf16/NHWC confidence net only under mixed_precision (the factory
+        # flips this from args): fp32 runs stay bit-comparable to the
+        # reference.
+        self.conf_net_bf16 = False
         self.__name__ = "NConvUpsampler"
 
         if scale is None and size is None:
@@ -172,7 +181,8 @@ class NConvUpsampler(nn.Module):
 
         west_in = (torch.cat((x_data_for_guidance, x_guidance), 1)
                    if self.use_data_for_guidance else x_guidance)
-        if west_in.is_cuda and isinstance(self.weights_est_net, nn.Module):
+        if (west_in.is_cuda and self.conf_net_bf16
+                and isinstance(self.weights_est_net, nn.Module)):
             # The confidence net runs bf16/NHWC on the MFMA conv kernel.
             # Documented deviation from the reference (which leaves the
             # upsampler outside its AMP region, an artifact of train.py's

@@ -23,7 +23,14 @@ from .torch_ref import MAX_FLOW  # re-export  # noqa: F401
 
 def sequence_loss(flow_preds, flow_gt, valid, gamma=0.8, max_flow=MAX_FLOW):
     """gamma-weighted L1 over the prediction sequence + final-pred metrics
-    (kernel #12; reference train.py:46-71)."""
+    (kernel #12; reference train.py:46-71).
+
+    Degenerate-batch behavior: with ZERO valid pixels the fused
+    kernel clamps the divisor to 1 and reports 0 loss/metrics, whereas the
+    reference's `epe.mean()` over an empty selection yields NaN and poisons
+    the step. 0 is the defined behavior here (a no-op step instead of a
+    NaN-corrupted model); the torch_ref path mirrors it.
+    """
     if (_ext.use_hip(flow_gt) and 1 <= len(flow_preds) <= 32
             and all(p.dtype == torch.float32 for p in flow_preds)):
         from .functional_loss import sequence_loss_fused
