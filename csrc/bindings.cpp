@@ -73,15 +73,17 @@ void flowhip_conf_pool_bwd_launch(const float* gdata_ds,
 void flowhip_conv_gemm_fwd_launch(const void* x, const void* x2,
                                   const void* wpk, const float* bias,
                                   void* out, void* out2, const void* zpage,
-                                  long Mtot, int HH, int WW, int ld_x,
+                                  long Mtot, int HH, int WW, int srcH,
+                                  int srcW, int sH, int sW, int ld_x,
                                   int ld_x2, int C1, int Cin, int Cout,
                                   int cpad, int KH, int KW, int padH,
-                                  int padW, int osplit, int act,
+                                  int padW, int osplit, int act, int smode,
                                   hipStream_t stream);
 void flowhip_conv_gemm_wrw_launch(const void* dy, const void* x,
                                   const void* x2, float* partials, float* dw,
                                   const void* zpage, long Mtot, int HH,
-                                  int WW, int ld_x, int ld_x2, int C1,
+                                  int WW, int srcH, int srcW, int sH, int sW,
+                                  int ld_x, int ld_x2, int C1,
                                   int Cin, int Cout, int cpad, int KH,
                                   int KW, int padH, int padW, int nchunk,
                                   hipStream_t stream);
@@ -787,7 +789,8 @@ torch::Tensor conv_gemm_pack(torch::Tensor w, bool flip) {
 std::vector<torch::Tensor> conv_gemm_fwd2(
     torch::Tensor x, c10::optional<torch::Tensor> x2, torch::Tensor wpk,
     c10::optional<torch::Tensor> bias, int64_t Cout, int64_t KH, int64_t KW,
-    int64_t osplit, int64_t act) {
+    int64_t osplit, int64_t act, int64_t sH, int64_t sW, int64_t smode,
+    int64_t outH, int64_t outW) {
   int ld_x, ld_x2 = 0;
   cg_check_x(x, ld_x);
   TORCH_CHECK(wpk.is_cuda() && wpk.is_contiguous() &&
@@ -807,7 +810,18 @@ std::vector<torch::Tensor> conv_gemm_fwd2(
   }
   const int cpad = wpk.size(2);
   TORCH_CHECK(wpk.size(0) == KH * KW && wpk.size(1) == Cout);
-  const long Mtot = (long)N * H * W;
+  TORCH_CHECK(smode >= 0 && smode <= 2 && sH >= 1 && sW >= 1);
+  // output spatial dims: same-pad for smode 0/1; smode 2 (strided
+  // transposed conv, = backward-data of smode 1) takes them explicitly
+  int OH = H, OW = W;
+  if (smode == 1) {
+    OH = (int)((H + 2 * (KH / 2) - KH) / sH + 1);
+    OW = (int)((W + 2 * (KW / 2) - KW) / sW + 1);
+  } else if (smode == 2) {
+    TORCH_CHECK(outH > 0 && outW > 0, "conv_gemm smode=2 needs outH/outW");
+    OH = (int)outH; OW = (int)outW;
+  }
+  const long Mtot = (long)N * OH * OW;
   const float* bptr = nullptr;
   if (bias.has_value()) {
     TORCH_CHECK(bias->is_contiguous() &&
@@ -818,25 +832,29 @@ std::vector<torch::Tensor> conv_gemm_fwd2(
   const c10::cuda::CUDAGuard guard(x.device());
   hipStream_t stream = at::cuda::getCurrentCUDAStream().stream();
   if (osplit <= 0 || osplit >= Cout) {
-    auto out = torch::empty({(long)N, Cout, (long)H, (long)W}, x.options(),
+    auto out = torch::empty({(long)N, Cout, (long)OH, (long)OW}, x.options(),
                             torch::MemoryFormat::ChannelsLast);
     flowhip_conv_gemm_fwd_launch(x.data_ptr(), x2p, wpk.data_ptr(), bptr,
                                  out.data_ptr(), nullptr, cg_zero_page(),
-                                 Mtot, H, W, ld_x, ld_x2, C1, Cin, (int)Cout,
+                                 Mtot, OH, OW, H, W, (int)sH, (int)sW, ld_x,
+                                 ld_x2, C1, Cin, (int)Cout,
                                  cpad, (int)KH, (int)KW, (int)KH / 2,
-                                 (int)KW / 2, (int)Cout, (int)act, stream);
+                                 (int)KW / 2, (int)Cout, (int)act,
+                                 (int)smode, stream);
     return {out};
   }
+  TORCH_CHECK(smode == 0, "conv_gemm: osplit only with smode 0");
   auto out = torch::empty({(long)N, osplit, (long)H, (long)W}, x.options(),
                           torch::MemoryFormat::ChannelsLast);
   auto out2 = torch::empty({(long)N, Cout - osplit, (long)H, (long)W},
                            x.options(), torch::MemoryFormat::ChannelsLast);
   flowhip_conv_gemm_fwd_launch(x.data_ptr(), x2p, wpk.data_ptr(), bptr,
                                out.data_ptr(), out2.data_ptr(),
-                               cg_zero_page(), Mtot, H, W, ld_x, ld_x2, C1,
+                               cg_zero_page(), Mtot, H, W, H, W, 1, 1, ld_x,
+                               ld_x2, C1,
                                Cin, (int)Cout, cpad, (int)KH, (int)KW,
                                (int)KH / 2, (int)KW / 2, (int)osplit,
-                               (int)act, stream);
+                               (int)act, 0, stream);
   return {out, out2};
 }
 
@@ -844,18 +862,20 @@ torch::Tensor conv_gemm_fwd(torch::Tensor x, torch::Tensor wpk,
                             c10::optional<torch::Tensor> bias, int64_t Cout,
                             int64_t KH, int64_t KW, int64_t act) {
   return conv_gemm_fwd2(x, c10::nullopt, wpk, bias, Cout, KH, KW, 0,
-                        act)[0];
+                        act, 1, 1, 0, 0, 0)[0];
 }
 
 torch::Tensor conv_gemm_wrw(torch::Tensor dy, torch::Tensor x,
                             c10::optional<torch::Tensor> x2, int64_t KH,
-                            int64_t KW) {
+                            int64_t KW, int64_t sH, int64_t sW) {
   int ld_x, ld_x2 = 0;
   cg_check_x(x, ld_x);
   TORCH_CHECK(dy.is_cuda() && dy.scalar_type() == torch::kBFloat16 &&
               dy.stride(1) == 1 && dy.stride(3) == dy.size(1),
               "conv_gemm_wrw: dy must be channels-last contiguous");
   const int N = x.size(0), C1 = x.size(1), H = x.size(2), W = x.size(3);
+  const int OH = dy.size(2), OW = dy.size(3);
+  TORCH_CHECK(dy.size(0) == N);
   int Cin = C1;
   const void* x2p = nullptr;
   if (x2.has_value()) {
@@ -866,7 +886,7 @@ torch::Tensor conv_gemm_wrw(torch::Tensor dy, torch::Tensor x,
     x2p = x2->data_ptr();
   }
   const int Cout = dy.size(1);
-  const long Mtot = (long)N * H * W;
+  const long Mtot = (long)N * OH * OW;
   const int cpad = (int)((Cin + 63) / 64) * 64;
   const int tiles_o = (Cout + 63) / 64;
   // split M only as much as needed to fill the chip (~2 blocks/CU);
@@ -887,8 +907,9 @@ torch::Tensor conv_gemm_wrw(torch::Tensor dy, torch::Tensor x,
   hipStream_t stream = at::cuda::getCurrentCUDAStream().stream();
   flowhip_conv_gemm_wrw_launch(dy.data_ptr(), x.data_ptr(), x2p,
                                partials.data_ptr<float>(),
-                               dw.data_ptr<float>(), cg_zero_page(), Mtot, H,
-                               W, ld_x, ld_x2, C1, Cin, Cout, cpad, (int)KH,
+                               dw.data_ptr<float>(), cg_zero_page(), Mtot,
+                               OH, OW, H, W, (int)sH, (int)sW,
+                               ld_x, ld_x2, C1, Cin, Cout, cpad, (int)KH,
                                (int)KW, (int)KH / 2, (int)KW / 2, nchunk,
                                stream);
   return dw;
@@ -1082,9 +1103,16 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "packed weights)");
   m.def("conv_gemm_fwd2", &conv_gemm_fwd2,
         "conv forward over a virtually-concatenated pair of inputs and/or "
-        "with a split output (bwd-data of a cat input)");
+        "with a split output (bwd-data of a cat input); smode 1 = strided "
+        "direct conv, smode 2 = strided transposed conv (bwd-data)",
+        py::arg("x"), py::arg("x2"), py::arg("wpk"), py::arg("bias"),
+        py::arg("Cout"), py::arg("KH"), py::arg("KW"), py::arg("osplit"),
+        py::arg("act"), py::arg("sH") = 1, py::arg("sW") = 1,
+        py::arg("smode") = 0, py::arg("outH") = 0, py::arg("outW") = 0);
   m.def("conv_gemm_wrw", &conv_gemm_wrw,
-        "implicit-GEMM conv weight gradient (split-M + reduce)");
+        "implicit-GEMM conv weight gradient (split-M + reduce)",
+        py::arg("dy"), py::arg("x"), py::arg("x2"), py::arg("KH"),
+        py::arg("KW"), py::arg("sH") = 1, py::arg("sW") = 1);
   m.def("instnorm_cl_fwd", &instnorm_cl_fwd,
         "channels-last InstanceNorm2d forward (y, mean, rstd)");
   m.def("instnorm_cl_bwd", &instnorm_cl_bwd,

@@ -52,17 +52,30 @@ __global__ __launch_bounds__(CS_THREADS) void col_sum_partial_kernel(
   }
 }
 
-__global__ __launch_bounds__(64) void col_sum_finalize_kernel(
+// One block per 64-channel group; the chunk loop is split across 8 thread
+// slices (tid/64) so the dependent-load chain is 8x shorter and each
+// slice's loads pipeline independently (the old single-wave version was
+// latency-bound at ~43 us for a few hundred chunks).
+__global__ __launch_bounds__(512) void col_sum_finalize_kernel(
     const float* __restrict__ partials, float* __restrict__ out, int C,
     int nchunk) {
   const int ncb = (C + 63) / 64;
   const int cb = blockIdx.x;
-  const int c = cb * 64 + threadIdx.x;
-  if (c >= C) return;
+  const int lane = threadIdx.x & 63;
+  const int slice = threadIdx.x >> 6;  // 0..7
   float t = 0.f;
-  for (int ch = 0; ch < nchunk; ++ch)
-    t += partials[((long)ch * ncb + cb) * 64 + threadIdx.x];
-  out[c] = t;
+  for (int ch = slice; ch < nchunk; ch += 8)
+    t += partials[((long)ch * ncb + cb) * 64 + lane];
+  __shared__ float red[8][64];
+  red[slice][lane] = t;
+  __syncthreads();
+  const int c = cb * 64 + threadIdx.x;
+  if (threadIdx.x < 64 && c < C) {
+    float s = 0.f;
+#pragma unroll
+    for (int k = 0; k < 8; ++k) s += red[k][threadIdx.x];
+    out[c] = s;
+  }
 }
 
 bool flowhip_col_sum_launch(const void* dy, float* partials, float* out,
@@ -72,7 +85,7 @@ bool flowhip_col_sum_launch(const void* dy, float* partials, float* out,
   hipLaunchKernelGGL(col_sum_partial_kernel, dim3(ncb * nchunk),
                      dim3(CS_THREADS), 0, stream, (const __bf16*)dy,
                      partials, M, C, nchunk);
-  hipLaunchKernelGGL(col_sum_finalize_kernel, dim3(ncb), dim3(64), 0, stream,
+  hipLaunchKernelGGL(col_sum_finalize_kernel, dim3(ncb), dim3(512), 0, stream,
                      partials, out, C, nchunk);
   return true;
 }

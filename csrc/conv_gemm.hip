@@ -1,21 +1,33 @@
 // Implicit-GEMM NHWC bf16 convolution family (kernel #5 of SURVEY.md §2.2)
 // for the iterative update block (motion encoder, SepConvGRU, flow head —
-// reference update.py:6-146). Stride 1, dilation 1, "same" padding, odd or
-// 1xK/Kx1 kernels, Cin % 8 == 0 (callers pad the one 324-channel case).
+// reference update.py:6-146) and the feature/context encoders (reference
+// extractor.py:118-193). Dilation 1, odd or 1xK/Kx1 kernels, Cin % 8 == 0
+// (callers pad the one 324-channel case).
 //
-// Why not MIOpen here: at the RAFT working set (M = B*H/8*W/8 = 21504 at
-// batch 3) MIOpen's 128x128/256x128 igemm tiles launch ~170-340
-// workgroups on a 256-CU chip — underfilled — and every call drags
-// SubTensorOp cast/Set kernels with it (profiles/). This kernel uses a
-// 64x64 tile (>=1300 workgroups at Cout 256), stages the im2col gather
-// through LDS with 16-byte global_load_lds (lane-linear image, XOR bank
-// swizzle on the SOURCE address — guide §5 rule 21), double-buffered, and
-// fuses bias + ReLU into the epilogue.
+// Tile geometry is templated:
+//   BM=64  (4 waves, 256 thr)  — update-block shapes (M ~ 21k, fill first)
+//   BM=128 (8 waves, 512 thr)  — encoder shapes (M up to ~1M at 448x1024
+//          batch 3): halves the per-output barrier/staging overhead and the
+//          B-operand traffic; this is the profile-driven fix for the 19%
+//          MFMA-util reading of the round-1 64x64-only kernel
+//          (profiles/README.md headroom #2/#4).
+// Both stage the im2col gather through LDS with 16-byte global_load_lds
+// (lane-linear image, XOR bank swizzle on the SOURCE address — guide §5
+// rule 21), double-buffered, bias + ReLU fused into the epilogue.
+//
+// Stride support (SMODE template):
+//   0: stride 1, same padding (the original fast path; src dims == out dims)
+//   1: strided direct conv     sy = oy*sH + ky - padH    (encoder stems /
+//      downsample convs, stride 2)
+//   2: strided TRANSPOSED conv (the backward-data of mode 1): with the
+//      flipped/transposed weight pack, t = iy + ky + (padH - KH + 1) taps
+//      dy at oy = t/sH iff t >= 0 && t % sH == 0 && oy < srcH.
+//      For sH == 1 this degenerates to the classic flipped-conv identity
+//      used by the stride-1 backward.
 //
 //   fwd:       out[m, o] = act( sum_{ky,kx,c} x[m+off, c] * w[o, ky,kx, c] + b[o] )
 //   bwd-data:  SAME kernel with w' = flip(w).T packed as [kyx][ci][co]
-//              (stride-1 same-pad transposed conv == conv with flipped w)
-//   wrw:       dW[o, kyx, c] = sum_m dy[m, o] * x[m+off, c]
+//   wrw:       dW[o, kyx, c] = sum_m dy[m, o] * x[m*s+off, c]
 //              (split-M partials + small reduce; MFMA over the m axis with
 //              LDS-transposed fragment reads)
 //
@@ -31,29 +43,30 @@
 #define CG_BK 64
 #define CG_THREADS 256
 
-#define CG_SLOTS 8                       // 16-B pieces per row of a 64-k tile
-#define CG_PIECES (CG_BM * CG_SLOTS)     // 512 per tile
-#define CG_PPT (CG_PIECES / CG_THREADS)  // 2 per thread
-
 __device__ __forceinline__ unsigned cg_swz(unsigned row, unsigned slot) {
   return slot ^ ((row >> 1) & 7u);
 }
 
-// Stage the A (im2col) tile: rows = 64 consecutive output pixels, k = one
+// Stage the A (im2col) tile: rows = BM consecutive output pixels, k = one
 // 64-channel slab of one (ky,kx) section. OOB rows read the zero page.
 // Two-source form: channels [0, C1) come from x, [C1, Cin) from x2 — the
 // virtually-concatenated GRU input cat([h, x]) without materializing the
 // cat (C1 must be a multiple of 64 so a slab never straddles sources).
+// SMODE semantics in the header comment; (dy, dx) carry the per-mode tap
+// offset, (srcH, srcW) the source image dims, (HH, WW) the m-mapping dims.
+template <int BM, int THREADS, int SMODE>
 __device__ __forceinline__ void cg_stage_a(
     const __bf16* __restrict__ x, const __bf16* __restrict__ x2,
     const __bf16* __restrict__ zpage,
     char* lds_buf, int wave, int lane, long m0, long Mtot, int HH, int WW,
-    int ld_x, int ld_x2, int C1, int Cin, int dy, int dx, int c0) {
+    int srcH, int srcW, int sH, int sW, int ld_x, int ld_x2, int C1, int Cin,
+    int dy, int dx, int c0) {
+  constexpr int PPT = BM * 8 / THREADS;
 #pragma unroll
-  for (int j = 0; j < CG_PPT; ++j) {
-    const int piece0 = wave * 64 + CG_THREADS * j;
+  for (int j = 0; j < PPT; ++j) {
+    const int piece0 = wave * 64 + THREADS * j;
     const int piece = piece0 + lane;
-    const int row = piece >> 3;          // 0..63
+    const int row = piece >> 3;          // 0..BM-1
     const int slot = piece & 7;
     const int sslot = cg_swz(row, slot);
     long m = m0 + row;
@@ -61,14 +74,27 @@ __device__ __forceinline__ void cg_stage_a(
     const int xx = (int)(m % WW);
     const int yy = (int)((m / WW) % HH);
     const long n = m / ((long)WW * HH);
-    const int sy = yy + dy, sx = xx + dx;
+    int sy, sx;
+    bool ok;
+    if (SMODE == 0) {
+      sy = yy + dy; sx = xx + dx;
+      ok = sy >= 0 && sy < srcH && sx >= 0 && sx < srcW;
+    } else if (SMODE == 1) {
+      sy = yy * sH + dy; sx = xx * sW + dx;
+      ok = sy >= 0 && sy < srcH && sx >= 0 && sx < srcW;
+    } else {
+      const int ty = yy + dy, tx = xx + dx;
+      sy = ty / sH; sx = tx / sW;
+      ok = ty >= 0 && tx >= 0 && ty % sH == 0 && tx % sW == 0 &&
+           sy < srcH && sx < srcW;
+    }
     const int c = c0 + sslot * 8;
     const __bf16* src = zpage;
-    if (sy >= 0 && sy < HH && sx >= 0 && sx < WW && c < Cin) {
+    if (ok && c < Cin) {
       if (x2 == nullptr || c < C1)
-        src = x + (((long)n * HH + sy) * WW + sx) * ld_x + c;
+        src = x + (((long)n * srcH + sy) * srcW + sx) * ld_x + c;
       else
-        src = x2 + (((long)n * HH + sy) * WW + sx) * ld_x2 + (c - C1);
+        src = x2 + (((long)n * srcH + sy) * srcW + sx) * ld_x2 + (c - C1);
     }
     __builtin_amdgcn_global_load_lds(
         (const __attribute__((address_space(1))) void*)src,
@@ -79,13 +105,16 @@ __device__ __forceinline__ void cg_stage_a(
 
 // Stage the B (weight) tile: rows = 64 output channels (clamped), k = the
 // same 64-channel slab. wsec = wpk section base [kyx][.][.].
+template <int THREADS>
 __device__ __forceinline__ void cg_stage_b(const __bf16* __restrict__ wsec,
                                            char* lds_buf, int wave, int lane,
                                            int n0, int Cout, int cpad,
                                            int c0) {
+  constexpr int PPT = CG_BN * 8 / THREADS;
 #pragma unroll
-  for (int j = 0; j < CG_PPT; ++j) {
-    const int piece0 = wave * 64 + CG_THREADS * j;
+  for (int j = 0; j < PPT; ++j) {
+    const int piece0 = wave * 64 + THREADS * j;
+    if (piece0 >= CG_BN * 8) break;
     const int piece = piece0 + lane;
     const int row = piece >> 3;
     const int slot = piece & 7;
@@ -100,20 +129,26 @@ __device__ __forceinline__ void cg_stage_b(const __bf16* __restrict__ wsec,
   }
 }
 
-// ACT: 0 = none, 1 = ReLU
-template <int ACT>
-__global__ __launch_bounds__(CG_THREADS, 2) void conv_gemm_fwd_kernel(
-    const __bf16* __restrict__ x,     // (Mtot, ld_x) NHWC rows
+// ACT: 0 = none, 1 = ReLU. BM in {64, 128}: 64 -> 4 waves in a 2Mx2N
+// quadrant grid, 128 -> 8 waves as 4Mx2N.
+template <int ACT, int BM, int SMODE>
+__global__
+__launch_bounds__(BM == 64 ? 256 : 512, BM == 64 ? 2 : 4)
+void conv_gemm_fwd_kernel(
+    const __bf16* __restrict__ x,     // (N, srcH, srcW, ld_x) NHWC rows
     const __bf16* __restrict__ x2,    // second input source or nullptr
     const __bf16* __restrict__ wpk,   // (KYX, Cout, cpad)
     const float* __restrict__ bias,   // (Cout) or nullptr
     __bf16* __restrict__ out,         // (Mtot, Cout) — or (Mtot, osplit)
     __bf16* __restrict__ out2,        // (Mtot, Cout-osplit) or nullptr
     const __bf16* __restrict__ zpage,
-    long Mtot, int HH, int WW, int ld_x, int ld_x2, int C1, int Cin,
-    int Cout, int cpad, int KH, int KW, int padH, int padW, int osplit,
-    int tiles_m) {
-  __shared__ __attribute__((aligned(16))) char lds[2 * 2 * CG_BM * CG_BK * 2];
+    long Mtot, int HH, int WW, int srcH, int srcW, int sH, int sW, int ld_x,
+    int ld_x2, int C1, int Cin, int Cout, int cpad, int KH, int KW,
+    int offH, int offW, int osplit, int tiles_m) {
+  constexpr int THREADS = BM == 64 ? 256 : 512;
+  constexpr unsigned TSA = BM * CG_BK * 2;       // A tile bytes
+  constexpr unsigned TSB = CG_BN * CG_BK * 2;    // B tile bytes
+  __shared__ __attribute__((aligned(16))) char lds[2 * (TSA + TSB)];
 
   const int tid = threadIdx.x;
   const int wave = tid >> 6;
@@ -121,11 +156,12 @@ __global__ __launch_bounds__(CG_THREADS, 2) void conv_gemm_fwd_kernel(
 
   const int tm = blockIdx.x % tiles_m;
   const int tn = blockIdx.x / tiles_m;
-  const long m0 = (long)tm * CG_BM;
+  const long m0 = (long)tm * BM;
   const int n0 = tn * CG_BN;
 
-  const int wr = (wave >> 1) * 32;  // M offset of this wave's quadrant
-  const int wc = (wave & 1) * 32;   // N offset
+  // wave quadrant: BM=64 -> 2Mx2N of 32x32; BM=128 -> 4Mx2N
+  const int wr = (BM == 64 ? (wave >> 1) : (wave & 3)) * 32;
+  const int wc = (BM == 64 ? (wave & 1) : (wave >> 2)) * 32;
 
   f32x4 acc[2][2];
 #pragma unroll
@@ -135,12 +171,12 @@ __global__ __launch_bounds__(CG_THREADS, 2) void conv_gemm_fwd_kernel(
 
   const int cslabs = cpad / CG_BK;
   const int nsub = KH * KW * cslabs;
-  const unsigned TS = CG_BM * CG_BK * 2;  // 8 KB
 
   // subtile s -> (kyx = s / cslabs, c0 = (s % cslabs) * 64)
-  cg_stage_a(x, x2, zpage, lds, wave, lane, m0, Mtot, HH, WW, ld_x, ld_x2,
-             C1, Cin, -padH, -padW, 0);
-  cg_stage_b(wpk, lds + TS, wave, lane, n0, Cout, cpad, 0);
+  cg_stage_a<BM, THREADS, SMODE>(x, x2, zpage, lds, wave, lane, m0, Mtot, HH,
+                                 WW, srcH, srcW, sH, sW, ld_x, ld_x2, C1,
+                                 Cin, offH, offW, 0);
+  cg_stage_b<THREADS>(wpk, lds + TSA, wave, lane, n0, Cout, cpad, 0);
   asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
   __syncthreads();
 
@@ -150,15 +186,17 @@ __global__ __launch_bounds__(CG_THREADS, 2) void conv_gemm_fwd_kernel(
       const int kyx = (s + 1) / cslabs;
       const int cs = (s + 1) - kyx * cslabs;
       const int ky = kyx / KW, kx = kyx - ky * KW;
-      cg_stage_a(x, x2, zpage, lds + (cur ^ 1) * 2 * TS, wave, lane, m0,
-                 Mtot, HH, WW, ld_x, ld_x2, C1, Cin, ky - padH, kx - padW,
-                 cs * CG_BK);
-      cg_stage_b(wpk + (long)kyx * Cout * cpad, lds + (cur ^ 1) * 2 * TS + TS,
-                 wave, lane, n0, Cout, cpad, cs * CG_BK);
+      cg_stage_a<BM, THREADS, SMODE>(
+          x, x2, zpage, lds + (cur ^ 1) * (TSA + TSB), wave, lane, m0, Mtot,
+          HH, WW, srcH, srcW, sH, sW, ld_x, ld_x2, C1, Cin, ky + offH,
+          kx + offW, cs * CG_BK);
+      cg_stage_b<THREADS>(wpk + (long)kyx * Cout * cpad,
+                          lds + (cur ^ 1) * (TSA + TSB) + TSA, wave, lane,
+                          n0, Cout, cpad, cs * CG_BK);
     }
 
-    const char* abuf = lds + cur * 2 * TS;
-    const char* bbuf = abuf + TS;
+    const char* abuf = lds + cur * (TSA + TSB);
+    const char* bbuf = abuf + TSA;
     const int frow = lane & 15;
     const int fk = lane >> 4;
 
@@ -230,16 +268,18 @@ __global__ __launch_bounds__(CG_THREADS, 2) void conv_gemm_fwd_kernel(
 // lane-linear); fragments are read transposed element-wise (u16 loads) —
 // correctness-first; the m-contraction MFMA dominates.
 // partials layout: (CHUNKS, KYX, tiles_o*64, cpad) fp32.
+// Strided conv support: m walks dy's (OH, OW) grid, x is gathered at
+// sy = yy*sH + tap (x dims srcH, srcW).
 // ---------------------------------------------------------------------------
 __global__ __launch_bounds__(CG_THREADS, 2) void conv_gemm_wrw_kernel(
     const __bf16* __restrict__ dy,  // (Mtot, Cout)
-    const __bf16* __restrict__ x,   // (Mtot, ld_x)
+    const __bf16* __restrict__ x,   // (N, srcH, srcW, ld_x)
     const __bf16* __restrict__ x2,  // second source or nullptr
     float* __restrict__ partials,
     const __bf16* __restrict__ zpage,
-    long Mtot, int HH, int WW, int ld_x, int ld_x2, int C1, int Cin,
-    int Cout, int cpad, int KH, int KW, int padH, int padW, int tiles_o,
-    int tiles_c, int nchunk) {
+    long Mtot, int HH, int WW, int srcH, int srcW, int sH, int sW, int ld_x,
+    int ld_x2, int C1, int Cin, int Cout, int cpad, int KH, int KW, int padH,
+    int padW, int tiles_o, int tiles_c, int nchunk) {
   __shared__ __attribute__((aligned(16))) char lds[2 * 2 * CG_BM * CG_BK * 2];
 
   const int tid = threadIdx.x;
@@ -272,7 +312,7 @@ __global__ __launch_bounds__(CG_THREADS, 2) void conv_gemm_wrw_kernel(
   const unsigned TS = CG_BM * CG_BK * 2;
   auto stage_dy = [&](long mt, char* buf) {
 #pragma unroll
-    for (int j = 0; j < CG_PPT; ++j) {
+    for (int j = 0; j < 2; ++j) {
       const int piece0 = wave * 64 + CG_THREADS * j;
       const int piece = piece0 + lane;
       const int row = piece >> 3;
@@ -293,7 +333,7 @@ __global__ __launch_bounds__(CG_THREADS, 2) void conv_gemm_wrw_kernel(
   };
   auto stage_x = [&](long mt, char* buf) {
 #pragma unroll
-    for (int j = 0; j < CG_PPT; ++j) {
+    for (int j = 0; j < 2; ++j) {
       const int piece0 = wave * 64 + CG_THREADS * j;
       const int piece = piece0 + lane;
       const int row = piece >> 3;
@@ -305,13 +345,13 @@ __global__ __launch_bounds__(CG_THREADS, 2) void conv_gemm_wrw_kernel(
         const int xx = (int)(m % WW);
         const int yy = (int)((m / WW) % HH);
         const long n = m / ((long)WW * HH);
-        const int sy = yy + dyo, sx = xx + dxo;
+        const int sy = yy * sH + dyo, sx = xx * sW + dxo;
         const int c = c0 + sslot * 8;
-        if (sy >= 0 && sy < HH && sx >= 0 && sx < WW && c < Cin) {
+        if (sy >= 0 && sy < srcH && sx >= 0 && sx < srcW && c < Cin) {
           if (x2 == nullptr || c < C1)
-            src = x + (((long)n * HH + sy) * WW + sx) * ld_x + c;
+            src = x + (((long)n * srcH + sy) * srcW + sx) * ld_x + c;
           else
-            src = x2 + (((long)n * HH + sy) * WW + sx) * ld_x2 + (c - C1);
+            src = x2 + (((long)n * srcH + sy) * srcW + sx) * ld_x2 + (c - C1);
         }
       }
       __builtin_amdgcn_global_load_lds(
@@ -429,46 +469,77 @@ __global__ __launch_bounds__(CG_THREADS) void conv_gemm_wrw_reduce_kernel(
 
 // ---------------------------------------------------------------------------
 
+template <int BM>
+static void cg_fwd_dispatch(const __bf16* x, const __bf16* x2,
+                            const __bf16* wpk, const float* bias, __bf16* out,
+                            __bf16* out2, const __bf16* zpage, long Mtot,
+                            int HH, int WW, int srcH, int srcW, int sH,
+                            int sW, int ld_x, int ld_x2, int C1, int Cin,
+                            int Cout, int cpad, int KH, int KW, int offH,
+                            int offW, int osplit, int act, int smode,
+                            hipStream_t stream) {
+  const int tiles_m = (int)((Mtot + BM - 1) / BM);
+  const int tiles_n = fh_cdiv(Cout, CG_BN);
+  dim3 grid(tiles_m * tiles_n), block(BM == 64 ? 256 : 512);
+#define CG_LAUNCH(A, S)                                                      \
+  hipLaunchKernelGGL((conv_gemm_fwd_kernel<A, BM, S>), grid, block, 0,       \
+                     stream, x, x2, wpk, bias, out, out2, zpage, Mtot, HH,   \
+                     WW, srcH, srcW, sH, sW, ld_x, ld_x2, C1, Cin, Cout,     \
+                     cpad, KH, KW, offH, offW, osplit, tiles_m)
+  if (smode == 0) { if (act == 1) CG_LAUNCH(1, 0); else CG_LAUNCH(0, 0); }
+  else if (smode == 1) { if (act == 1) CG_LAUNCH(1, 1); else CG_LAUNCH(0, 1); }
+  else { if (act == 1) CG_LAUNCH(1, 2); else CG_LAUNCH(0, 2); }
+#undef CG_LAUNCH
+}
+
 void flowhip_conv_gemm_fwd_launch(const void* x, const void* x2,
                                   const void* wpk, const float* bias,
                                   void* out, void* out2, const void* zpage,
-                                  long Mtot, int HH, int WW, int ld_x,
+                                  long Mtot, int HH, int WW, int srcH,
+                                  int srcW, int sH, int sW, int ld_x,
                                   int ld_x2, int C1, int Cin, int Cout,
                                   int cpad, int KH, int KW, int padH,
-                                  int padW, int osplit, int act,
+                                  int padW, int osplit, int act, int smode,
                                   hipStream_t stream) {
-  const int tiles_m = (int)((Mtot + CG_BM - 1) / CG_BM);
-  const int tiles_n = fh_cdiv(Cout, CG_BN);
-  dim3 grid(tiles_m * tiles_n), block(CG_THREADS);
-  if (act == 1)
-    hipLaunchKernelGGL((conv_gemm_fwd_kernel<1>), grid, block, 0, stream,
-                       (const __bf16*)x, (const __bf16*)x2,
-                       (const __bf16*)wpk, bias, (__bf16*)out, (__bf16*)out2,
-                       (const __bf16*)zpage, Mtot, HH, WW, ld_x, ld_x2, C1,
-                       Cin, Cout, cpad, KH, KW, padH, padW, osplit, tiles_m);
+  // tap offsets: direct conv (smode 0/1) reads sy = oy*sH + ky - padH;
+  // strided-transposed (smode 2) taps t = iy + ky + (padH - KH + 1)
+  const int offH = smode == 2 ? padH - KH + 1 : -padH;
+  const int offW = smode == 2 ? padW - KW + 1 : -padW;
+  // Tile choice: BM=128 once the grid still fills the 256-CU chip; the
+  // encoder shapes (M >= 64k) take it, the update block (M ~ 21k, 168
+  // M-tiles at BM=128) keeps the fill-first 64x64 tile.
+  const long t128 = ((Mtot + 127) / 128) * fh_cdiv(Cout, CG_BN);
+  if (t128 >= 320)
+    cg_fwd_dispatch<128>((const __bf16*)x, (const __bf16*)x2,
+                         (const __bf16*)wpk, bias, (__bf16*)out,
+                         (__bf16*)out2, (const __bf16*)zpage, Mtot, HH, WW,
+                         srcH, srcW, sH, sW, ld_x, ld_x2, C1, Cin, Cout,
+                         cpad, KH, KW, offH, offW, osplit, act, smode,
+                         stream);
   else
-    hipLaunchKernelGGL((conv_gemm_fwd_kernel<0>), grid, block, 0, stream,
-                       (const __bf16*)x, (const __bf16*)x2,
-                       (const __bf16*)wpk, bias, (__bf16*)out, (__bf16*)out2,
-                       (const __bf16*)zpage, Mtot, HH, WW, ld_x, ld_x2, C1,
-                       Cin, Cout, cpad, KH, KW, padH, padW, osplit, tiles_m);
+    cg_fwd_dispatch<64>((const __bf16*)x, (const __bf16*)x2,
+                        (const __bf16*)wpk, bias, (__bf16*)out, (__bf16*)out2,
+                        (const __bf16*)zpage, Mtot, HH, WW, srcH, srcW, sH,
+                        sW, ld_x, ld_x2, C1, Cin, Cout, cpad, KH, KW, offH,
+                        offW, osplit, act, smode, stream);
 }
 
 void flowhip_conv_gemm_wrw_launch(const void* dy, const void* x,
                                   const void* x2, float* partials, float* dw,
                                   const void* zpage, long Mtot, int HH,
-                                  int WW, int ld_x, int ld_x2, int C1,
-                                  int Cin, int Cout, int cpad, int KH,
-                                  int KW, int padH, int padW, int nchunk,
+                                  int WW, int srcH, int srcW, int sH, int sW,
+                                  int ld_x, int ld_x2, int C1, int Cin,
+                                  int Cout, int cpad, int KH, int KW,
+                                  int padH, int padW, int nchunk,
                                   hipStream_t stream) {
   const int tiles_o = fh_cdiv(Cout, 64);
   const int tiles_c = fh_cdiv(cpad, 64);
   dim3 grid(tiles_o * tiles_c, KH * KW, nchunk), block(CG_THREADS);
   hipLaunchKernelGGL(conv_gemm_wrw_kernel, grid, block, 0, stream,
                      (const __bf16*)dy, (const __bf16*)x, (const __bf16*)x2,
-                     partials, (const __bf16*)zpage, Mtot, HH, WW, ld_x,
-                     ld_x2, C1, Cin, Cout, cpad, KH, KW, padH, padW, tiles_o,
-                     tiles_c, nchunk);
+                     partials, (const __bf16*)zpage, Mtot, HH, WW, srcH,
+                     srcW, sH, sW, ld_x, ld_x2, C1, Cin, Cout, cpad, KH, KW,
+                     padH, padW, tiles_o, tiles_c, nchunk);
   const long total = (long)Cout * Cin * KH * KW;
   long rblocks = (total + CG_THREADS - 1) / CG_THREADS;
   if (rblocks > 4096) rblocks = 4096;

@@ -73,7 +73,7 @@ class ResidualBlock(nn.Module):
         else:
             self.norm3 = _norm(norm_fn, planes)
             self.downsample = nn.Sequential(
-                nn.Conv2d(in_planes, planes, kernel_size=1, stride=stride), self.norm3)
+                _enc_conv(in_planes, planes, kernel_size=1, stride=stride), self.norm3)
 
     def forward(self, x):
         y = self.relu(self.norm1(self.conv1(x)))
@@ -106,7 +106,7 @@ class BottleneckBlock(nn.Module):
         else:
             self.norm4 = _norm(norm_fn, planes)
             self.downsample = nn.Sequential(
-                nn.Conv2d(in_planes, planes, kernel_size=1, stride=stride), self.norm4)
+                _enc_conv(in_planes, planes, kernel_size=1, stride=stride), self.norm4)
 
     def forward(self, x):
         y = self.relu(self.norm1(self.conv1(x)))

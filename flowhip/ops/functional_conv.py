@@ -91,29 +91,41 @@ def _packs(weight, cache, key=None):
 
 
 class ConvGemmFn(torch.autograd.Function):
+    """Stride-1 (sH=sW=1) or strided (smode 1 fwd / smode 2 bwd-data)
+    same-padding conv on the MFMA kernel."""
+
     @staticmethod
-    def forward(ctx, x, weight, bias, wpk_fwd, wpk_bwd):
+    def forward(ctx, x, weight, bias, wpk_fwd, wpk_bwd, sH, sW):
         O, I, KH, KW = weight.shape
         b = bias.detach().float().contiguous() if bias is not None else None
-        out = _ext.ext().conv_gemm_fwd(x, wpk_fwd, b, O, KH, KW, 0)
+        if sH == 1 and sW == 1:
+            out = _ext.ext().conv_gemm_fwd(x, wpk_fwd, b, O, KH, KW, 0)
+        else:
+            out = _ext.ext().conv_gemm_fwd2(x, None, wpk_fwd, b, O, KH, KW,
+                                            0, 0, sH, sW, 1)[0]
         ctx.save_for_backward(x, wpk_bwd)
-        ctx.meta = (O, I, KH, KW, bias is not None)
+        ctx.meta = (O, I, KH, KW, bias is not None, sH, sW)
         return out
 
     @staticmethod
     def backward(ctx, dy):
         x, wpk_bwd = ctx.saved_tensors
-        O, I, KH, KW, has_bias = ctx.meta
+        O, I, KH, KW, has_bias, sH, sW = ctx.meta
         dy = dy.contiguous(memory_format=torch.channels_last)
         if dy.dtype != torch.bfloat16:
             dy = dy.to(torch.bfloat16)
         dyp, O_real = _pad_dy8(dy)
-        dx = _ext.ext().conv_gemm_fwd(dyp, wpk_bwd, None, I, KH, KW, 0)
-        dw = _ext.ext().conv_gemm_wrw(dyp, x, None, KH, KW)
+        if sH == 1 and sW == 1:
+            dx = _ext.ext().conv_gemm_fwd(dyp, wpk_bwd, None, I, KH, KW, 0)
+        else:
+            dx = _ext.ext().conv_gemm_fwd2(dyp, None, wpk_bwd, None, I, KH,
+                                           KW, 0, 0, sH, sW, 2,
+                                           x.shape[2], x.shape[3])[0]
+        dw = _ext.ext().conv_gemm_wrw(dyp, x, None, KH, KW, sH, sW)
         if dw.shape[0] != O_real:
             dw = dw[:O_real].contiguous()
         dbias = _col_sum(dy) if has_bias else None
-        return dx, dw, dbias, None, None
+        return dx, dw, dbias, None, None, None, None
 
 
 def can_fuse_conv(x, weight, stride, padding, dilation, groups):
@@ -124,7 +136,8 @@ def can_fuse_conv(x, weight, stride, padding, dilation, groups):
     if _ext.ext() is None:
         return False
     from torch.nn.modules.utils import _pair
-    if (_pair(stride) != (1, 1) or _pair(dilation) != (1, 1) or groups != 1):
+    if (_pair(stride) not in ((1, 1), (2, 2))
+            or _pair(dilation) != (1, 1) or groups != 1):
         return False
     O, I, KH, KW = weight.shape
     if _pair(padding) != (KH // 2, KW // 2):
@@ -150,7 +163,9 @@ def fused_conv2d(x, weight, bias, stride, padding, dilation, groups, cache,
         if weight.shape[1] % 8 != 0:
             x, weight = _pad_c8(x, weight)
         wf, wb = _packs(weight, cache, key)
-        return ConvGemmFn.apply(x, weight, bias, wf, wb)
+        from torch.nn.modules.utils import _pair
+        sH, sW = _pair(stride)
+        return ConvGemmFn.apply(x, weight, bias, wf, wb, sH, sW)
     return F.conv2d(x, weight, bias, stride, padding, dilation, groups)
 
 

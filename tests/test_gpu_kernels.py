@@ -499,6 +499,102 @@ class TestConvGemm:
         torch.testing.assert_close(b.grad.float(), b2.grad,
                                    atol=1e-1, rtol=5e-2)
 
+    STRIDED = [
+        # (Cin, Cout, K, stride, H, W) — encoder inventory (extractor.py):
+        # stage-transition 3x3 s2, downsample 1x1 s2; odd/hostile dims too
+        (64, 96, 3, 2, 24, 40),
+        (96, 128, 3, 2, 23, 37),
+        (64, 96, 1, 2, 24, 40),
+        (96, 96, 3, 1, 24, 40),
+    ]
+
+    @pytest.mark.parametrize("ci,co,k,s,H,W", STRIDED)
+    def test_strided_fwd_bwd_matches_torch(self, ci, co, k, s, H, W):
+        """Strided direct conv (smode 1) + its transposed backward-data
+        (smode 2) + strided wrw vs the fp32 torch oracle."""
+        from flowhip.ops.functional_conv import fused_conv2d, can_fuse_conv
+        torch.manual_seed(23)
+        B = 2
+        x = (torch.randn(B, ci, H, W, device=_dev()) / 8).to(torch.bfloat16) \
+            .contiguous(memory_format=torch.channels_last).requires_grad_(True)
+        w = (torch.randn(co, ci, k, k, device=_dev()) /
+             (ci * k * k) ** 0.5).requires_grad_(True)
+        b = torch.randn(co, device=_dev()).requires_grad_(True)
+        pad = k // 2
+        assert can_fuse_conv(x, w, s, pad, 1, 1)
+
+        out = fused_conv2d(x, w, b, s, pad, 1, 1, {})
+        ref = torch.nn.functional.conv2d(
+            x.detach().float(), w.detach().float(), b.detach().float(),
+            stride=s, padding=pad)
+        tol = dict(atol=5e-2, rtol=5e-2)
+        torch.testing.assert_close(out.float(), ref, **tol)
+
+        g = torch.randn_like(ref).to(torch.bfloat16) \
+            .contiguous(memory_format=torch.channels_last)
+        out.backward(g)
+        x2 = x.detach().float().requires_grad_(True)
+        w2 = w.detach().float().requires_grad_(True)
+        b2 = b.detach().float().requires_grad_(True)
+        torch.nn.functional.conv2d(x2, w2, b2, stride=s,
+                                   padding=pad).backward(g.float())
+        torch.testing.assert_close(x.grad.float(), x2.grad, **tol)
+        torch.testing.assert_close(w.grad.float(), w2.grad,
+                                   atol=1e-1, rtol=5e-2)
+        torch.testing.assert_close(b.grad.float(), b2.grad,
+                                   atol=1e-1, rtol=5e-2)
+
+    def test_large_m_bm128_tile(self):
+        """Encoder-sized M routes to the BM=128 8-wave tile (the launcher
+        switches at >=320 tiles); oracle-check fwd + both backwards there,
+        including a non-128-aligned M tail."""
+        from flowhip.ops.functional_conv import fused_conv2d
+        torch.manual_seed(29)
+        B, ci, co, H, W = 2, 64, 64, 161, 161  # M = 51842 -> 406 tiles
+        x = (torch.randn(B, ci, H, W, device=_dev()) / 8).to(torch.bfloat16) \
+            .contiguous(memory_format=torch.channels_last).requires_grad_(True)
+        w = (torch.randn(co, ci, 3, 3, device=_dev()) / 24).requires_grad_(True)
+        out = fused_conv2d(x, w, None, 1, 1, 1, 1, {})
+        ref = torch.nn.functional.conv2d(x.detach().float(),
+                                         w.detach().float(), padding=1)
+        tol = dict(atol=5e-2, rtol=5e-2)
+        torch.testing.assert_close(out.float(), ref, **tol)
+        g = torch.randn_like(ref).to(torch.bfloat16) \
+            .contiguous(memory_format=torch.channels_last)
+        out.backward(g)
+        x2 = x.detach().float().requires_grad_(True)
+        w2 = w.detach().float().requires_grad_(True)
+        torch.nn.functional.conv2d(x2, w2, padding=1).backward(g.float())
+        torch.testing.assert_close(x.grad.float(), x2.grad, **tol)
+        torch.testing.assert_close(w.grad.float(), w2.grad,
+                                   atol=2e-1, rtol=5e-2)
+
+    def test_fused_encoder_matches_miopen_encoder(self):
+        """BasicEncoder with FLOWHIP_FUSED_ENCODER=1 (fused stride-1 AND
+        stride-2 convs) vs the MIOpen path, same weights."""
+        import importlib
+        import os
+        from flowhip.nn import extractor as ex
+        torch.manual_seed(31)
+        old = os.environ.get("FLOWHIP_FUSED_ENCODER")
+        try:
+            os.environ["FLOWHIP_FUSED_ENCODER"] = "1"
+            enc_f = ex.BasicEncoder(output_dim=128, norm_fn="instance").to(_dev())
+            os.environ["FLOWHIP_FUSED_ENCODER"] = "0"
+            enc_m = ex.BasicEncoder(output_dim=128, norm_fn="instance").to(_dev())
+        finally:
+            if old is None:
+                os.environ.pop("FLOWHIP_FUSED_ENCODER", None)
+            else:
+                os.environ["FLOWHIP_FUSED_ENCODER"] = old
+        enc_m.load_state_dict(enc_f.state_dict())
+        x = torch.randn(2, 3, 128, 160, device=_dev())
+        with torch.autocast("cuda", dtype=torch.bfloat16):
+            yf = enc_f(x.contiguous(memory_format=torch.channels_last))
+            ym = enc_m(x.contiguous(memory_format=torch.channels_last))
+        torch.testing.assert_close(yf.float(), ym.float(),
+                                   atol=1e-1, rtol=1e-1)
+
     def test_narrowed_input_view(self):
         # a channel-narrowed channels-last view (the 324-of-328 corr case)
         from flowhip.ops.functional_conv import fused_conv2d
