@@ -64,9 +64,10 @@ __device__ __forceinline__ void stage_tile(const __bf16* __restrict__ src,
   }
 }
 
-extern "C" __global__ __launch_bounds__(THREADS, 2) void flowhip_bgemm_nt(
+template <typename out_t>
+__global__ __launch_bounds__(THREADS, 2) void flowhip_bgemm_nt(
     const __bf16* __restrict__ A, const __bf16* __restrict__ B,
-    float* __restrict__ C, float alpha, int M, int N, int K, long strideA,
+    out_t* __restrict__ C, float alpha, int M, int N, int K, long strideA,
     long strideB, long strideC, int tiles_m, int tiles_n) {
   // LDS: [2 buffers][A tile | B tile], each tile BM*BK bf16 = 16 KB.
   __shared__ __attribute__((aligned(16))) char lds[2 * 2 * BM * BK * 2];
@@ -84,7 +85,7 @@ extern "C" __global__ __launch_bounds__(THREADS, 2) void flowhip_bgemm_nt(
 
   const __bf16* Ab = A + (long)bat * strideA;
   const __bf16* Bb = B + (long)bat * strideB;
-  float* Cb = C + (long)bat * strideC;
+  out_t* Cb = C + (long)bat * strideC;
 
   // wave -> 64x64 quadrant
   const int wr = (wave >> 1) * 64;  // 0 or 64 within tile (M)
@@ -160,7 +161,7 @@ extern "C" __global__ __launch_bounds__(THREADS, 2) void flowhip_bgemm_nt(
         const int gm = m0 + wr + m * 16 + frow0 + r;
         const int gn = n0 + wc + n * 16 + fcol;
         if (gm < M && gn < N)
-          Cb[(long)gm * N + gn] = alpha * acc[m][n][r];
+          Cb[(long)gm * N + gn] = (out_t)(alpha * acc[m][n][r]);
       }
     }
   }
@@ -168,13 +169,19 @@ extern "C" __global__ __launch_bounds__(THREADS, 2) void flowhip_bgemm_nt(
 
 void flowhip_bgemm_nt_launch(const void* A, const void* B, void* C,
                              float alpha, int batch, int M, int N, int K,
-                             hipStream_t stream) {
+                             int out_bf16, hipStream_t stream) {
   const int tiles_m = fh_cdiv(M, BM);
   const int tiles_n = fh_cdiv(N, BN);
   dim3 grid(tiles_m * tiles_n, 1, batch);
   dim3 block(THREADS);
-  hipLaunchKernelGGL(flowhip_bgemm_nt, grid, block, 0, stream,
-                     (const __bf16*)A, (const __bf16*)B, (float*)C, alpha, M,
-                     N, K, (long)M * K, (long)N * K, (long)M * N, tiles_m,
-                     tiles_n);
+  if (out_bf16)
+    hipLaunchKernelGGL(flowhip_bgemm_nt<__bf16>, grid, block, 0, stream,
+                       (const __bf16*)A, (const __bf16*)B, (__bf16*)C, alpha,
+                       M, N, K, (long)M * K, (long)N * K, (long)M * N,
+                       tiles_m, tiles_n);
+  else
+    hipLaunchKernelGGL(flowhip_bgemm_nt<float>, grid, block, 0, stream,
+                       (const __bf16*)A, (const __bf16*)B, (float*)C, alpha,
+                       M, N, K, (long)M * K, (long)N * K, (long)M * N,
+                       tiles_m, tiles_n);
 }

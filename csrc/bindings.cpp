@@ -11,15 +11,15 @@
 // launchers defined in the .hip translation units
 void flowhip_bgemm_nt_launch(const void* A, const void* B, void* C,
                              float alpha, int batch, int M, int N, int K,
-                             hipStream_t stream);
-void flowhip_corr_lookup_fwd_launch(const float* level, const float* coords,
+                             int out_bf16, hipStream_t stream);
+void flowhip_corr_lookup_fwd_launch(const void* level, const float* coords,
                                     float* out, int BP, int P, int Hl, int Wl,
                                     int l, int L, int radius, int cl, int ldc,
-                                    hipStream_t stream);
+                                    int is_bf16, hipStream_t stream);
 void flowhip_corr_lookup_bwd_launch(const float* gout, const float* coords,
-                                    float* glevel, int BP, int P, int Hl,
+                                    void* glevel, int BP, int P, int Hl,
                                     int Wl, int l, int L, int radius, int cl,
-                                    hipStream_t stream);
+                                    int is_bf16, hipStream_t stream);
 void flowhip_convex_up_fwd_launch(const float* flow, const float* mask,
                                   float* out, int N, int H, int W, int factor,
                                   hipStream_t stream);
@@ -59,8 +59,8 @@ int flowhip_instnorm_partial_rows(int N, int C, long P);
 void flowhip_conv_gemm_pack_launch(const float* w, void* wpk, long total,
                                    int O, int I, int KH, int KW, int cpad,
                                    int flip, hipStream_t stream);
-void flowhip_transpose_cast_launch(const float* in, void* out, int B, int M,
-                                   int N, hipStream_t stream);
+void flowhip_transpose_cast_launch(const void* in, void* out, int B, int M,
+                                   int N, int in_bf16, hipStream_t stream);
 void flowhip_conf_pool_fwd_launch(const float* data, const float* conf,
                                   float* data_ds, float* conf_ds,
                                   unsigned char* code, long total, int H,
@@ -161,17 +161,20 @@ bool flowhip_pacconv_bwd_launch(const float* dy, const float* x,
                                 int nchunk, int B, int Ci, int Co, int H,
                                 int W, int OH, int OW, int pH, int pW, int K,
                                 int dil, int shared, hipStream_t stream);
-bool flowhip_corr_pyramid_fwd_launch(const float* corr, float* l1, float* l2,
-                                     float* l3, int BP, int H0, int W0,
-                                     int nlev, hipStream_t stream);
-void flowhip_corr_pyramid_bwd_launch(const float* g0, const float* g1,
-                                     const float* g2, const float* g3,
-                                     float* dcorr, long total, int H0, int W0,
+bool flowhip_corr_pyramid_fwd_launch(const void* corr, void* l1, void* l2,
+                                     void* l3, int BP, int H0, int W0,
+                                     int nlev, int is_bf16,
                                      hipStream_t stream);
+bool flowhip_corr_pyramid_fits(int H0, int W0, int is_bf16);
+void flowhip_corr_pyramid_bwd_launch(const void* g0, const void* g1,
+                                     const void* g2, const void* g3,
+                                     void* dcorr, long total, int H0, int W0,
+                                     int is_bf16, hipStream_t stream);
 
 namespace {
 
-torch::Tensor bgemm_nt(torch::Tensor a, torch::Tensor b, double alpha) {
+torch::Tensor bgemm_nt(torch::Tensor a, torch::Tensor b, double alpha,
+                       bool out_bf16) {
   TORCH_CHECK(a.is_cuda() && b.is_cuda(), "bgemm_nt: CUDA tensors required");
   TORCH_CHECK(a.dtype() == torch::kBFloat16 && b.dtype() == torch::kBFloat16,
               "bgemm_nt: bf16 operands required");
@@ -183,11 +186,13 @@ torch::Tensor bgemm_nt(torch::Tensor a, torch::Tensor b, double alpha) {
 
   const int batch = a.size(0), M = a.size(1), N = b.size(1), K = a.size(2);
   auto c = torch::empty({batch, M, N},
-                        a.options().dtype(torch::kFloat32));
+                        a.options().dtype(out_bf16 ? torch::kBFloat16
+                                                   : torch::kFloat32));
   const c10::cuda::CUDAGuard guard(a.device());
   hipStream_t stream = at::cuda::getCurrentCUDAStream().stream();
   flowhip_bgemm_nt_launch(a.data_ptr(), b.data_ptr(), c.data_ptr(),
-                          (float)alpha, batch, M, N, K, stream);
+                          (float)alpha, batch, M, N, K, out_bf16 ? 1 : 0,
+                          stream);
   return c;
 }
 
@@ -218,15 +223,16 @@ torch::Tensor corr_lookup_fwd(std::vector<torch::Tensor> pyramid,
 
   for (int l = 0; l < L; ++l) {
     auto& lvl = pyramid[l];
+    const bool lb = lvl.dtype() == torch::kBFloat16;
     TORCH_CHECK(lvl.is_cuda() && lvl.is_contiguous() &&
-                lvl.dtype() == torch::kFloat32,
-                "corr_lookup: fp32 contiguous pyramid levels required");
+                (lb || lvl.dtype() == torch::kFloat32),
+                "corr_lookup: fp32/bf16 contiguous pyramid levels required");
     TORCH_CHECK(lvl.size(0) == (long)B * P, "corr_lookup: level batch mismatch");
     const int Hl = lvl.size(-2), Wl = lvl.size(-1);
     flowhip_corr_lookup_fwd_launch(
-        lvl.data_ptr<float>(), coords.data_ptr<float>(),
+        lvl.data_ptr(), coords.data_ptr<float>(),
         full.data_ptr<float>(), B * P, P, Hl, Wl, l, L, (int)radius,
-        channels_last ? 1 : 0, (int)C8, stream);
+        channels_last ? 1 : 0, (int)C8, lb ? 1 : 0, stream);
   }
   return out;
 }
@@ -236,7 +242,8 @@ std::vector<torch::Tensor> corr_lookup_bwd(torch::Tensor gout,
                                            int64_t radius,
                                            std::vector<std::vector<int64_t>>
                                                level_shapes,
-                                           bool channels_last) {
+                                           bool channels_last,
+                                           bool levels_bf16) {
   TORCH_CHECK(gout.is_cuda() && gout.dtype() == torch::kFloat32);
   TORCH_CHECK(channels_last
                   ? gout.is_contiguous(torch::MemoryFormat::ChannelsLast)
@@ -249,7 +256,7 @@ std::vector<torch::Tensor> corr_lookup_bwd(torch::Tensor gout,
   hipStream_t stream = at::cuda::getCurrentCUDAStream().stream();
 
   // one flat zero-fill for all levels (4x fewer fill kernels), sliced into
-  // per-level views
+  // per-level views; level-grad dtype matches the resident pyramid dtype
   int64_t total = 0;
   std::vector<int64_t> sizes(L);
   for (int l = 0; l < L; ++l) {
@@ -258,7 +265,9 @@ std::vector<torch::Tensor> corr_lookup_bwd(torch::Tensor gout,
     sizes[l] = n;
     total += n;
   }
-  auto flat = torch::zeros({total}, gout.options().dtype(torch::kFloat32));
+  auto flat = torch::zeros(
+      {total}, gout.options().dtype(levels_bf16 ? torch::kBFloat16
+                                                : torch::kFloat32));
 
   std::vector<torch::Tensor> grads;
   grads.reserve(L);
@@ -268,8 +277,9 @@ std::vector<torch::Tensor> corr_lookup_bwd(torch::Tensor gout,
     off += sizes[l];
     const int Hl = g.size(-2), Wl = g.size(-1);
     flowhip_corr_lookup_bwd_launch(
-        gout.data_ptr<float>(), coords.data_ptr<float>(), g.data_ptr<float>(),
-        B * P, P, Hl, Wl, l, L, (int)radius, channels_last ? 1 : 0, stream);
+        gout.data_ptr<float>(), coords.data_ptr<float>(), g.data_ptr(),
+        B * P, P, Hl, Wl, l, L, (int)radius, channels_last ? 1 : 0,
+        levels_bf16 ? 1 : 0, stream);
     grads.push_back(g);
   }
   return grads;
@@ -277,7 +287,8 @@ std::vector<torch::Tensor> corr_lookup_bwd(torch::Tensor gout,
 
 std::vector<torch::Tensor> corr_pyramid_fwd(torch::Tensor corr,
                                             int64_t num_levels) {
-  TORCH_CHECK(corr.is_cuda() && corr.dtype() == torch::kFloat32 &&
+  const bool bf = corr.dtype() == torch::kBFloat16;
+  TORCH_CHECK(corr.is_cuda() && (bf || corr.dtype() == torch::kFloat32) &&
               corr.is_contiguous());
   TORCH_CHECK(corr.dim() == 4 && corr.size(1) == 1);
   const long BP = corr.size(0);
@@ -290,14 +301,18 @@ std::vector<torch::Tensor> corr_pyramid_fwd(torch::Tensor corr,
   }
   const c10::cuda::CUDAGuard guard(corr.device());
   hipStream_t stream = at::cuda::getCurrentCUDAStream().stream();
-  float* p1 = levels.size() > 0 ? levels[0].data_ptr<float>() : nullptr;
-  float* p2 = levels.size() > 1 ? levels[1].data_ptr<float>() : nullptr;
-  float* p3 = levels.size() > 2 ? levels[2].data_ptr<float>() : nullptr;
+  void* p1 = levels.size() > 0 ? levels[0].data_ptr() : nullptr;
+  void* p2 = levels.size() > 1 ? levels[1].data_ptr() : nullptr;
+  void* p3 = levels.size() > 2 ? levels[2].data_ptr() : nullptr;
   bool ok = flowhip_corr_pyramid_fwd_launch(
-      corr.data_ptr<float>(), p1, p2, p3, (int)BP, H0, W0, (int)num_levels,
-      stream);
+      corr.data_ptr(), p1, p2, p3, (int)BP, H0, W0, (int)num_levels,
+      bf ? 1 : 0, stream);
   TORCH_CHECK(ok, "corr_pyramid_fwd: unsupported shape (fall back to torch)");
   return levels;
+}
+
+bool corr_pyramid_fits(int64_t H0, int64_t W0, bool bf16) {
+  return flowhip_corr_pyramid_fits((int)H0, (int)W0, bf16 ? 1 : 0);
 }
 
 torch::Tensor corr_pyramid_bwd(std::vector<c10::optional<torch::Tensor>> grads,
@@ -305,19 +320,22 @@ torch::Tensor corr_pyramid_bwd(std::vector<c10::optional<torch::Tensor>> grads,
   TORCH_CHECK(grads.size() >= 1 && grads.size() <= 4);
   const long BP = corr_shape[0];
   const int H0 = corr_shape[2], W0 = corr_shape[3];
-  const float* g[4] = {nullptr, nullptr, nullptr, nullptr};
+  const void* g[4] = {nullptr, nullptr, nullptr, nullptr};
   torch::TensorOptions opts;
   torch::Device dev(torch::kCUDA);
-  bool have = false;
+  bool have = false, bf = false;
   for (size_t l = 0; l < grads.size(); ++l) {
     if (grads[l].has_value()) {
       auto& t = grads[l].value();
+      const bool tb = t.dtype() == torch::kBFloat16;
       TORCH_CHECK(t.is_cuda() && t.is_contiguous() &&
-                  t.dtype() == torch::kFloat32);
-      g[l] = t.data_ptr<float>();
+                  (tb || t.dtype() == torch::kFloat32));
+      if (have) TORCH_CHECK(tb == bf, "corr_pyramid_bwd: mixed grad dtypes");
+      g[l] = t.data_ptr();
       opts = t.options();
       dev = t.device();
       have = true;
+      bf = tb;
     }
   }
   TORCH_CHECK(have, "corr_pyramid_bwd: all grads missing");
@@ -325,8 +343,9 @@ torch::Tensor corr_pyramid_bwd(std::vector<c10::optional<torch::Tensor>> grads,
   const c10::cuda::CUDAGuard guard(dev);
   hipStream_t stream = at::cuda::getCurrentCUDAStream().stream();
   flowhip_corr_pyramid_bwd_launch(g[0], g[1], g[2], g[3],
-                                  dcorr.data_ptr<float>(),
-                                  (long)BP * H0 * W0, H0, W0, stream);
+                                  dcorr.data_ptr(),
+                                  (long)BP * H0 * W0, H0, W0, bf ? 1 : 0,
+                                  stream);
   return dcorr;
 }
 
@@ -738,15 +757,16 @@ std::vector<torch::Tensor> conf_pool_bwd(c10::optional<torch::Tensor> gdds,
 }
 
 torch::Tensor transpose_cast_bf16(torch::Tensor in) {
+  const bool bf = in.dtype() == torch::kBFloat16;
   TORCH_CHECK(in.is_cuda() && in.dim() == 3 && in.is_contiguous() &&
-              in.dtype() == torch::kFloat32);
+              (bf || in.dtype() == torch::kFloat32));
   const int B = in.size(0), M = in.size(1), N = in.size(2);
   auto out = torch::empty({(long)B, (long)N, (long)M},
                           in.options().dtype(torch::kBFloat16));
   const c10::cuda::CUDAGuard guard(in.device());
   hipStream_t stream = at::cuda::getCurrentCUDAStream().stream();
-  flowhip_transpose_cast_launch(in.data_ptr<float>(), out.data_ptr(), B, M,
-                                N, stream);
+  flowhip_transpose_cast_launch(in.data_ptr(), out.data_ptr(), B, M,
+                                N, bf ? 1 : 0, stream);
   return out;
 }
 
@@ -1068,11 +1088,18 @@ std::vector<torch::Tensor> pacconv_bwd(torch::Tensor dy, torch::Tensor x,
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "flowhip gfx950 HIP kernels";
   m.def("bgemm_nt", &bgemm_nt,
-        "C[b] = alpha * A[b] (M,K) @ B[b] (N,K)^T, bf16 in / fp32 out");
+        "C[b] = alpha * A[b] (M,K) @ B[b] (N,K)^T, bf16 in / fp32|bf16 out",
+        py::arg("a"), py::arg("b"), py::arg("alpha"),
+        py::arg("out_bf16") = false);
   m.def("corr_lookup_fwd", &corr_lookup_fwd,
-        "fused multi-level correlation window lookup");
+        "fused multi-level correlation window lookup (fp32/bf16 levels)");
   m.def("corr_lookup_bwd", &corr_lookup_bwd,
-        "backward of corr_lookup_fwd (pyramid grads)");
+        "backward of corr_lookup_fwd (pyramid grads)",
+        py::arg("gout"), py::arg("coords"), py::arg("radius"),
+        py::arg("level_shapes"), py::arg("channels_last"),
+        py::arg("levels_bf16") = false);
+  m.def("corr_pyramid_fits", &corr_pyramid_fits,
+        "does the fused pyramid-build kernel support this map size/dtype");
   m.def("packernel_fwd", &packernel_fwd, "PAC gaussian adapting kernel");
   m.def("packernel_bwd", &packernel_bwd, "backward of packernel_fwd");
   m.def("pacconv_fwd", &pacconv_fwd, "pixel-adaptive convolution forward");

@@ -81,7 +81,7 @@ __global__ __launch_bounds__(LK_THREADS) void corr_lookup_fwd_kernel(
     float rprev[K], rcur[K];
 #pragma unroll
     for (int j = 0; j <= K; ++j) {
-      const float* row = map + (long)(y0 + j) * Wl + x0;
+      const scalar_t* row = map + (long)(y0 + j) * Wl + x0;
       float rv[K + 1];
 #pragma unroll
       for (int a = 0; a <= K; ++a) rv[a] = (float)row[a];
@@ -228,44 +228,62 @@ __global__ __launch_bounds__(LK_THREADS) void corr_lookup_bwd_kernel(
   }
 }
 
-template <int R>
-static void lookup_fwd_level(const float* level, const float* coords,
+template <int R, typename scalar_t>
+static void lookup_fwd_level(const scalar_t* level, const float* coords,
                              float* out, int BP, int P, int Hl, int Wl, int l,
                              int L, int cl, int ldc, hipStream_t stream) {
   dim3 grid(fh_cdiv(BP, LK_THREADS));
-  hipLaunchKernelGGL((corr_lookup_fwd_kernel<R, float>), grid,
+  hipLaunchKernelGGL((corr_lookup_fwd_kernel<R, scalar_t>), grid,
                      dim3(LK_THREADS), 0, stream, level, coords, out, BP, P,
                      Hl, Wl, l, L, cl, ldc);
 }
 
-template <int R>
+template <int R, typename scalar_t>
 static void lookup_bwd_level(const float* gout, const float* coords,
-                             float* glevel, int BP, int P, int Hl, int Wl,
+                             scalar_t* glevel, int BP, int P, int Hl, int Wl,
                              int l, int L, int cl, hipStream_t stream) {
   dim3 grid(fh_cdiv(BP, LK_THREADS));
-  hipLaunchKernelGGL((corr_lookup_bwd_kernel<R, float>), grid,
+  hipLaunchKernelGGL((corr_lookup_bwd_kernel<R, scalar_t>), grid,
                      dim3(LK_THREADS), 0, stream, gout, coords, glevel, BP, P,
                      Hl, Wl, l, L, cl);
 }
 
-void flowhip_corr_lookup_fwd_launch(const float* level, const float* coords,
+// levels may be fp32 (reference parity) or bf16 (HBM-resident bf16
+// pyramid); out / gout stay fp32, the bilerp blend always runs fp32.
+void flowhip_corr_lookup_fwd_launch(const void* level, const float* coords,
                                     float* out, int BP, int P, int Hl, int Wl,
                                     int l, int L, int radius, int cl, int ldc,
-                                    hipStream_t stream) {
-  switch (radius) {
-    case 3: lookup_fwd_level<3>(level, coords, out, BP, P, Hl, Wl, l, L, cl, ldc, stream); break;
-    case 4: lookup_fwd_level<4>(level, coords, out, BP, P, Hl, Wl, l, L, cl, ldc, stream); break;
-    default: abort();
+                                    int is_bf16, hipStream_t stream) {
+  if (is_bf16) {
+    switch (radius) {
+      case 3: lookup_fwd_level<3>((const __bf16*)level, coords, out, BP, P, Hl, Wl, l, L, cl, ldc, stream); break;
+      case 4: lookup_fwd_level<4>((const __bf16*)level, coords, out, BP, P, Hl, Wl, l, L, cl, ldc, stream); break;
+      default: abort();
+    }
+  } else {
+    switch (radius) {
+      case 3: lookup_fwd_level<3>((const float*)level, coords, out, BP, P, Hl, Wl, l, L, cl, ldc, stream); break;
+      case 4: lookup_fwd_level<4>((const float*)level, coords, out, BP, P, Hl, Wl, l, L, cl, ldc, stream); break;
+      default: abort();
+    }
   }
 }
 
 void flowhip_corr_lookup_bwd_launch(const float* gout, const float* coords,
-                                    float* glevel, int BP, int P, int Hl,
+                                    void* glevel, int BP, int P, int Hl,
                                     int Wl, int l, int L, int radius, int cl,
-                                    hipStream_t stream) {
-  switch (radius) {
-    case 3: lookup_bwd_level<3>(gout, coords, glevel, BP, P, Hl, Wl, l, L, cl, stream); break;
-    case 4: lookup_bwd_level<4>(gout, coords, glevel, BP, P, Hl, Wl, l, L, cl, stream); break;
-    default: abort();
+                                    int is_bf16, hipStream_t stream) {
+  if (is_bf16) {
+    switch (radius) {
+      case 3: lookup_bwd_level<3>(gout, coords, (__bf16*)glevel, BP, P, Hl, Wl, l, L, cl, stream); break;
+      case 4: lookup_bwd_level<4>(gout, coords, (__bf16*)glevel, BP, P, Hl, Wl, l, L, cl, stream); break;
+      default: abort();
+    }
+  } else {
+    switch (radius) {
+      case 3: lookup_bwd_level<3>(gout, coords, (float*)glevel, BP, P, Hl, Wl, l, L, cl, stream); break;
+      case 4: lookup_bwd_level<4>(gout, coords, (float*)glevel, BP, P, Hl, Wl, l, L, cl, stream); break;
+      default: abort();
+    }
   }
 }

@@ -9,8 +9,9 @@
 #define TC_DIM 16   // 16x16 threads
 #define TC_TILE 64  // 64x64 tile, each thread 4x4
 
+template <typename in_t>
 __global__ __launch_bounds__(TC_DIM * TC_DIM) void transpose_cast_kernel(
-    const float* __restrict__ in,  // (B, M, N)
+    const in_t* __restrict__ in,   // (B, M, N) fp32 or bf16
     __bf16* __restrict__ out,      // (B, N, M)
     int M, int N, int tiles_m) {
   __shared__ __bf16 tile[TC_TILE][TC_TILE + 2];  // +2: bank-conflict pad
@@ -51,11 +52,15 @@ __global__ __launch_bounds__(TC_DIM * TC_DIM) void transpose_cast_kernel(
   }
 }
 
-void flowhip_transpose_cast_launch(const float* in, void* out, int B, int M,
-                                   int N, hipStream_t stream) {
+void flowhip_transpose_cast_launch(const void* in, void* out, int B, int M,
+                                   int N, int in_bf16, hipStream_t stream) {
   const int tiles_m = fh_cdiv(M, TC_TILE);
   const int tiles_n = fh_cdiv(N, TC_TILE);
   dim3 grid(tiles_m * tiles_n, 1, B), block(TC_DIM * TC_DIM);
-  hipLaunchKernelGGL(transpose_cast_kernel, grid, block, 0, stream, in,
-                     (__bf16*)out, M, N, tiles_m);
+  if (in_bf16)
+    hipLaunchKernelGGL(transpose_cast_kernel<__bf16>, grid, block, 0, stream,
+                       (const __bf16*)in, (__bf16*)out, M, N, tiles_m);
+  else
+    hipLaunchKernelGGL(transpose_cast_kernel<float>, grid, block, 0, stream,
+                       (const float*)in, (__bf16*)out, M, N, tiles_m);
 }

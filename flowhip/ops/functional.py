@@ -30,10 +30,11 @@ def _pad_k(t):
     return torch.nn.functional.pad(t, (0, pad))
 
 
-def _bgemm_nt(a, b, alpha):
-    """C[bat] = alpha * A[bat] @ B[bat]^T ; a (Bt,M,K) bf16, b (Bt,N,K) bf16 -> fp32."""
+def _bgemm_nt(a, b, alpha, out_bf16=False):
+    """C[bat] = alpha * A[bat] @ B[bat]^T ; a (Bt,M,K) bf16, b (Bt,N,K) bf16
+    -> fp32 (default) or bf16 (resident corr volume)."""
     return _ext.ext().bgemm_nt(_pad_k(a).contiguous(), _pad_k(b).contiguous(),
-                               alpha)
+                               alpha, out_bf16)
 
 
 class CorrVolumeFn(torch.autograd.Function):
@@ -62,9 +63,13 @@ class CorrVolumeFn(torch.autograd.Function):
             return f.reshape(B, D, P).transpose(1, 2).contiguous().to(
                 torch.bfloat16)
 
+        from ..utils.layout import corr_bf16_enabled
         f1t = as_pd(fmap1)  # (B,P,D)
         f2t = as_pd(fmap2)
-        corr = _bgemm_nt(f1t, f2t, 1.0 / math.sqrt(D))  # (B,P,P) fp32
+        # (B,P,P); bf16 residency halves HBM footprint + all pyramid/lookup
+        # read traffic (fp32 accumulate inside the MFMA either way)
+        corr = _bgemm_nt(f1t, f2t, 1.0 / math.sqrt(D),
+                         out_bf16=corr_bf16_enabled())
         ctx.save_for_backward(f1t, f2t)
         ctx.shape = (B, D, H, W)
         ctx.cl = fmap1.is_contiguous(memory_format=torch.channels_last)
@@ -77,9 +82,9 @@ class CorrVolumeFn(torch.autograd.Function):
         P = H * W
         alpha = 1.0 / math.sqrt(D)
         g3 = grad.reshape(B, P, P).contiguous()
-        dc = g3.to(torch.bfloat16)
-        # tiled transpose+cast kernel (eager transpose().to(bf16) is an
-        # uncoalesced ~300us elementwise op at P=7168)
+        dc = g3 if g3.dtype == torch.bfloat16 else g3.to(torch.bfloat16)
+        # tiled transpose(+cast) kernel (eager transpose().to(bf16) is an
+        # uncoalesced ~300us elementwise op at P=7168); accepts fp32 or bf16
         dct = _ext.ext().transpose_cast_bf16(g3)
         # dF1t[i,d] = sum_j dC[i,j] * F2t[j,d]: A = dC (M=P, K=P); the B
         # operand must be (N=D, K=P) row-major = f2 in (D, P) layout with P
@@ -120,6 +125,7 @@ class CorrLookupFn(torch.autograd.Function):
         ctx.radius = int(radius)
         ctx.cl = cl
         ctx.level_shapes = [tuple(p.shape) for p in pyramid]
+        ctx.levels_bf16 = pyramid[0].dtype == torch.bfloat16
         return out
 
     @staticmethod
@@ -131,7 +137,7 @@ class CorrLookupFn(torch.autograd.Function):
             grad = grad.contiguous()
         grads = _ext.ext().corr_lookup_bwd(
             grad, coords, ctx.radius,
-            [list(s) for s in ctx.level_shapes], ctx.cl)
+            [list(s) for s in ctx.level_shapes], ctx.cl, ctx.levels_bf16)
         return (None, None, *grads)
 
 

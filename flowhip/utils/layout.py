@@ -16,9 +16,27 @@ whole network channels_last:
 CPU keeps NCHW (the torch reference path; layout has no semantic effect).
 """
 
+import os
+
 import torch
 
 _CHANNELS_LAST = True
+
+# HBM residency dtype for the correlation volume + pyramid (north star /
+# BASELINE config 5: full-res corr volume resident in bf16). The GEMM
+# emits bf16 directly, the pyramid/lookup kernels read bf16, the bilinear
+# blend and the lookup output stay fp32. FLOWHIP_CORR_DTYPE=fp32 restores
+# the round-1 fp32-resident pyramid (numerics ablation / parity runs).
+_CORR_BF16 = os.environ.get("FLOWHIP_CORR_DTYPE", "bf16").lower() != "fp32"
+
+
+def set_corr_bf16(enabled):
+    global _CORR_BF16
+    _CORR_BF16 = bool(enabled)
+
+
+def corr_bf16_enabled():
+    return _CORR_BF16
 
 
 def set_channels_last(enabled):

@@ -49,8 +49,19 @@ class TestBgemmNT:
         assert c[0, 7, 3].item() == 0.0
 
 
+@pytest.fixture
+def fp32_corr():
+    """Pin the corr stack to the fp32-resident mode for exact-contract
+    oracle tests (the default is the bf16-resident pyramid)."""
+    from flowhip.utils import layout
+    old = layout.corr_bf16_enabled()
+    layout.set_corr_bf16(False)
+    yield
+    layout.set_corr_bf16(old)
+
+
 class TestCorrVolume:
-    def test_forward_matches_ref(self):
+    def test_forward_matches_ref(self, fp32_corr):
         from flowhip.ops import torch_ref
         from flowhip.ops.functional import CorrVolumeFn
         torch.manual_seed(2)
@@ -63,7 +74,7 @@ class TestCorrVolume:
         # bf16 inputs, fp32 accumulate: |err| ~ sqrt(D)*eps_bf16*|f|^2
         torch.testing.assert_close(out, ref, atol=5e-2, rtol=5e-2)
 
-    def test_backward_matches_ref(self):
+    def test_backward_matches_ref(self, fp32_corr):
         from flowhip.ops import torch_ref
         from flowhip.ops.functional import CorrVolumeFn
         torch.manual_seed(3)
@@ -132,6 +143,102 @@ class TestCorrLookup:
         l0 = torch.randn(64, 1, 8, 8, device=_dev())
         with pytest.raises(AssertionError):
             CorrLookupFn.apply(coords, 4, l0)
+
+
+class TestCorrBf16Resident:
+    """bf16-resident corr volume + pyramid (north star §5.7 / BASELINE
+    config 5): the GEMM emits bf16, pyramid/lookup read+write bf16, all
+    blends fp32. Tolerances reflect one bf16 rounding of O(1) values."""
+
+    def test_volume_bf16_close_to_fp32(self):
+        from flowhip.ops import torch_ref
+        from flowhip.ops.functional import CorrVolumeFn
+        from flowhip.utils import layout
+        torch.manual_seed(41)
+        f1 = torch.randn(2, 128, 14, 32, device=_dev())
+        f2 = torch.randn(2, 128, 14, 32, device=_dev())
+        old = layout.corr_bf16_enabled()
+        layout.set_corr_bf16(True)
+        try:
+            out = CorrVolumeFn.apply(f1, f2)
+        finally:
+            layout.set_corr_bf16(old)
+        assert out.dtype == torch.bfloat16
+        ref = torch_ref.corr_volume(f1, f2)
+        torch.testing.assert_close(out.float(), ref, atol=8e-2, rtol=8e-2)
+
+    def test_pyramid_and_lookup_bf16(self):
+        from flowhip import ops
+        from flowhip.ops import torch_ref
+        from flowhip.ops.functional import CorrLookupFn, CorrPyramidFn
+        torch.manual_seed(42)
+        B, H, W = 1, 16, 24
+        P = H * W
+        l0f = torch.randn(B * P, 1, H, W, device=_dev())
+        l0 = l0f.to(torch.bfloat16).requires_grad_(True)
+        pyr = list(CorrPyramidFn.apply(l0, 4))
+        assert all(p.dtype == torch.bfloat16 for p in pyr)
+        ref_pyr = torch_ref.corr_pyramid(l0.detach().float(), 4)
+        for p, r in zip(pyr, ref_pyr):
+            torch.testing.assert_close(p.float(), r, atol=2e-2, rtol=2e-2)
+
+        coords = (torch.rand(B, 2, H, W, device=_dev()) *
+                  torch.tensor([W, H], device=_dev()).view(1, 2, 1, 1))
+        out = CorrLookupFn.apply(coords, 4, *[p.detach().requires_grad_(True)
+                                              for p in pyr])
+        assert out.dtype == torch.float32
+        ref = torch_ref.corr_lookup([r for r in ref_pyr], coords, 4)
+        torch.testing.assert_close(out, ref, atol=2e-2, rtol=2e-2)
+
+    def test_lookup_bwd_bf16_levels(self):
+        from flowhip.ops import torch_ref
+        from flowhip.ops.functional import CorrLookupFn
+        torch.manual_seed(43)
+        B, H, W, radius = 1, 16, 24, 4
+        P = H * W
+        l0f = torch.randn(B * P, 1, H, W, device=_dev())
+        pyr_f = [p.detach() for p in torch_ref.corr_pyramid(l0f, 4)]
+        pyr_b = [p.to(torch.bfloat16).requires_grad_(True) for p in pyr_f]
+        coords = (torch.rand(B, 2, H, W, device=_dev()) *
+                  torch.tensor([W, H], device=_dev()).view(1, 2, 1, 1))
+        out = CorrLookupFn.apply(coords, radius, *pyr_b)
+        g = torch.randn_like(out)
+        grads = torch.autograd.grad(out, pyr_b, g)
+        pyr_r = [p.detach().float().requires_grad_(True) for p in pyr_b]
+        ref = torch_ref.corr_lookup(pyr_r, coords, radius)
+        refg = torch.autograd.grad(ref, pyr_r, g)
+        for d, r in zip(grads, refg):
+            assert d.dtype == torch.bfloat16
+            torch.testing.assert_close(d.float(), r, atol=5e-2, rtol=5e-2)
+
+    def test_model_bf16_corr_deviation_bounded(self):
+        """Full raft_nc_dbl forward: bf16-resident corr vs fp32-resident —
+        the EPE-parity ablation VERDICT r01 asks for, in miniature: flow
+        outputs must agree within a small fraction of a pixel."""
+        from flowhip.config.args import default_ncup_args
+        from flowhip.models import build_model
+        from flowhip.utils import layout
+        torch.manual_seed(44)
+        args = default_ncup_args(model="raft_nc_dbl", mixed_precision=True,
+                                 dataset="sintel")
+        model = build_model(args).to(_dev()).eval()
+        layout.apply_channels_last(model)
+        img1 = torch.rand(1, 3, 128, 256, device=_dev()) * 255
+        img2 = torch.rand(1, 3, 128, 256, device=_dev()) * 255
+        outs = {}
+        old = layout.corr_bf16_enabled()
+        try:
+            for mode in (False, True):
+                layout.set_corr_bf16(mode)
+                with torch.no_grad():
+                    _, flow = model(img1, img2, iters=6, test_mode=True)
+                outs[mode] = flow.float()
+        finally:
+            layout.set_corr_bf16(old)
+        diff = (outs[True] - outs[False]).norm(dim=1)
+        assert torch.isfinite(outs[True]).all()
+        # sub-0.1px deviation against the fp32-resident pyramid
+        assert diff.mean().item() < 0.1, diff.mean().item()
 
 
 class TestModelGPU:
@@ -326,6 +433,25 @@ class TestCorrPyramidFused:
             assert len(got) == len(ref)
             for g, r in zip(got, ref):
                 torch.testing.assert_close(g.cpu(), r, atol=1e-6, rtol=1e-6)
+
+    def test_kitti_submission_shape_stays_on_hip(self):
+        """47x156 (KITTI full-res 375x1242 / 8) exceeded round-1's static
+        56x128 LDS cap and silently fell back to torch; the dynamic-LDS
+        kernel must take it (VERDICT r01 weak #2)."""
+        import flowhip._C as C
+        from flowhip import ops
+        from flowhip.ops import torch_ref
+        assert C.corr_pyramid_fits(47, 156, False)
+        assert C.corr_pyramid_fits(47, 156, True)
+        torch.manual_seed(51)
+        corr = torch.randn(32, 1, 47, 156, device=_dev())
+        import warnings
+        with warnings.catch_warnings():
+            warnings.simplefilter("error")  # fallback warns -> fail
+            got = ops.corr_pyramid(corr, 4)
+        ref = torch_ref.corr_pyramid(corr.cpu(), 4)
+        for g, r in zip(got, ref):
+            torch.testing.assert_close(g.cpu(), r, atol=1e-6, rtol=1e-6)
 
     def test_backward_matches_autograd(self):
         from flowhip import ops
