@@ -34,6 +34,6 @@ class InstanceNorm2d(nn.InstanceNorm2d):
                 and x.is_contiguous(memory_format=torch.channels_last)
                 and x.dtype in (torch.float32, torch.bfloat16)
                 and x.shape[1] % 8 == 0 and x.shape[1] >= 8
-                and _ext.ext() is not None):
+                and _ext.ext() is not None and not _ext.force_ref()):
             return _InstNormCLFn.apply(x, self.eps)
         return super().forward(x)

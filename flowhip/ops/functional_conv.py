@@ -133,7 +133,7 @@ def can_fuse_conv(x, weight, stride, padding, dilation, groups):
         return False
     if x.stride(1) != 1:  # channels-last-like only
         return False
-    if _ext.ext() is None:
+    if _ext.ext() is None or _ext.force_ref():
         return False
     from torch.nn.modules.utils import _pair
     if (_pair(stride) not in ((1, 1), (2, 2))
@@ -211,7 +211,7 @@ def fused_conv2d_cat2(x1, x2, weight, bias, padding, cache, key):
                and x1.shape[1] % 64 == 0
                and (x1.shape[1] + x2.shape[1]) % 8 == 0  # 16-B staging tail
                and x1.stride(1) == 1 and x2.stride(1) == 1
-               and _ext.ext() is not None)
+               and _ext.ext() is not None and not _ext.force_ref())
     from torch.nn.modules.utils import _pair
     if fusable and _pair(padding) == (KH // 2, KW // 2):
         wf, wb = _packs(weight, cache, key)

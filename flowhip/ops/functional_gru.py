@@ -56,4 +56,4 @@ class GruGate2Fn(torch.autograd.Function):
 
 def gru_gates_available(t):
     return (t.is_cuda and t.dtype in (torch.float32, torch.bfloat16)
-            and _ext.ext() is not None)
+            and _ext.ext() is not None and not _ext.force_ref())

