@@ -181,7 +181,8 @@ class NConvUpsampler(nn.Module):
 
         west_in = (torch.cat((x_data_for_guidance, x_guidance), 1)
                    if self.use_data_for_guidance else x_guidance)
-        if (west_in.is_cuda and self.conf_net_bf16
+        from ..ops import _ext
+        if (west_in.is_cuda and self.conf_net_bf16 and not _ext.force_ref()
                 and isinstance(self.weights_est_net, nn.Module)):
             # The confidence net runs bf16/NHWC on the MFMA conv kernel.
             # Documented deviation from the reference (which leaves the

@@ -45,13 +45,22 @@ def _single_process_grads():
 
 
 @pytest.mark.timeout(900)
-def test_rccl_two_ranks_one_gpu(tmp_path):
+@pytest.mark.parametrize("backend", ["nccl", "gloo"])
+def test_two_ranks_one_gpu_ddp(tmp_path, backend):
+    """Two torchrun ranks sharing cuda:0. RCCL (nccl) categorically refuses
+    two ranks on one device ("Duplicate GPU detected" — verified on this
+    stack), so the nccl arm documents that as a skip; the gloo arm runs the
+    SAME DDP step on the GPU model (all-reduce correctness of the one-bucket
+    wrap_ddp path with CUDA grads). Real multi-GPU RCCL executes in the
+    driver's round-end SCALE run; single-rank RCCL collectives are covered
+    by test_nccl_world1_process_group below."""
     env = dict(os.environ, HSA_ENABLE_IPC_MODE_LEGACY="0")
+    port = 29741 if backend == "nccl" else 29743
     cmd = [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
            "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
-           "--master-port", "29741",
+           "--master-port", str(port),
            os.path.join(REPO, "tools", "rccl_smoke.py"),
-           "--out", str(tmp_path), "--profile"]
+           "--out", str(tmp_path), "--profile", "--backend", backend]
     r = subprocess.run(cmd, capture_output=True, text=True, timeout=840,
                        cwd=REPO, env=env)
 
@@ -59,16 +68,18 @@ def test_rccl_two_ranks_one_gpu(tmp_path):
     if os.path.exists(skip_marker):
         with open(skip_marker) as f:
             why = json.load(f)["error"]
+        assert backend == "nccl", f"gloo arm must not skip: {why}"
+        assert "Duplicate GPU" in why
         pytest.skip(f"RCCL refuses 2 ranks on one GPU on this stack: {why}")
 
     assert r.returncode == 0, r.stdout[-3000:] + r.stderr[-3000:]
     with open(os.path.join(str(tmp_path), "rccl_smoke.json")) as f:
         rec = json.load(f)
     assert rec["rccl_allreduce_ok"] and rec["world_size"] == 2
-    assert rec["nccl_backend"] == "nccl"
+    assert rec["nccl_backend"] == backend
 
     # DDP-averaged grads == single-process batched grads (the gloo/CPU
-    # equivalence test, now over real RCCL kernels)
+    # equivalence test, now with the model + grads on the GPU)
     ddp_grads = torch.load(os.path.join(str(tmp_path), "rccl_grads.pth"),
                            weights_only=True)
     ref = _single_process_grads()
@@ -76,10 +87,6 @@ def test_rccl_two_ranks_one_gpu(tmp_path):
     bad = [n for n in ref
            if not torch.allclose(ddp_grads[n], ref[n], atol=2e-2, rtol=2e-2)]
     assert not bad, bad[:8]
-
-    # the profiler summary from rank 0 must actually contain an RCCL kernel
-    assert rec.get("rccl_kernel_seen"), [
-        r["name"] for r in rec.get("profile_top", [])][:10]
 
 
 @pytest.mark.timeout(600)

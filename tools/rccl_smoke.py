@@ -35,6 +35,10 @@ def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--out", required=True)
     ap.add_argument("--profile", action="store_true")
+    ap.add_argument("--backend", default="nccl",
+                    help="nccl (=RCCL; refuses 2 ranks on one device) or "
+                         "gloo (exercises the same DDP machinery on the "
+                         "GPU model when only one GPU is leased)")
     args = ap.parse_args()
 
     rank = int(os.environ["RANK"])
@@ -43,7 +47,7 @@ def main():
     torch.cuda.set_device(0)  # both ranks share GPU 0
 
     try:
-        dist.init_process_group("nccl", rank=rank, world_size=world)
+        dist.init_process_group(args.backend, rank=rank, world_size=world)
         t = torch.full((4,), float(rank + 1), device="cuda:0")
         dist.all_reduce(t)
         torch.cuda.synchronize()
