@@ -264,13 +264,44 @@ void conv_gemm_fwd_kernel(
 
 // ---------------------------------------------------------------------------
 // Weight gradient: dWp[o, kyx, c] partial over an M chunk.
-// grid: (tiles_o * tiles_c, KYX, CHUNKS). LDS tiles stay m-major (glds
-// lane-linear); fragments are read transposed element-wise (u16 loads) —
-// correctness-first; the m-contraction MFMA dominates.
+// grid: (tiles_o * tiles_c, KYX, CHUNKS); the GEMM contracts over m, so
+// both operands need TRANSPOSED (k=m-major) fragments from tiles whose
+// global layout is m-major. gfx950's ds_read_b64_tr_b16 does that
+// transpose in hardware: each 16-lane group reads one [4 m][16 col] block
+// lane-linearly and receives per lane the 4 m-values of its own column
+// (guide T10). The LDS image is therefore laid out as 64 subblocks of
+// [4 m][16 c] halfwords (cb-major, then mq), with the tile's m rows
+// PERMUTED (wrw_row_to_m) so that the two tr reads of a fragment deliver
+// exactly the MFMA a/b-operand k-order (k = (lane>>4)*8 + e). The glds
+// staging writes the permuted image directly — the per-piece SOURCE
+// address does the permutation, the LDS destination stays lane-linear.
+// Replaces the round-1 per-element u16 fragment reads (~13% MFMA util,
+// profiles headroom #3).
 // partials layout: (CHUNKS, KYX, tiles_o*64, cpad) fp32.
 // Strided conv support: m walks dy's (OH, OW) grid, x is gathered at
 // sy = yy*sH + tap (x dims srcH, srcW).
 // ---------------------------------------------------------------------------
+
+typedef __attribute__((ext_vector_type(4))) __bf16 bf16x4;
+
+// image row (0..63) -> tile m row: per 32-row half, rows [q*4..q*4+4) hold
+// m = (q&3)*8 + (q>>2)*4 + e so a 16-row tr window yields k-consecutive
+// 8-element fragments
+__device__ __forceinline__ int wrw_row_to_m(int row) {
+  const int half = row & 32;
+  const int q = (row & 31) >> 2;
+  return half + ((q & 3) << 3) + ((q >> 2) << 2) + (row & 3);
+}
+
+__device__ __forceinline__ bf16x4 wrw_tr_read(const char* buf,
+                                              unsigned byte_off) {
+  bf16x4 v;
+  const __attribute__((address_space(3))) char* p =
+      (const __attribute__((address_space(3))) char*)(buf + byte_off);
+  asm volatile("ds_read_b64_tr_b16 %0, %1" : "=v"(v) : "v"(p));
+  return v;
+}
+
 __global__ __launch_bounds__(CG_THREADS, 2) void conv_gemm_wrw_kernel(
     const __bf16* __restrict__ dy,  // (Mtot, Cout)
     const __bf16* __restrict__ x,   // (N, srcH, srcW, ld_x)
@@ -310,17 +341,21 @@ __global__ __launch_bounds__(CG_THREADS, 2) void conv_gemm_wrw_kernel(
     for (int j = 0; j < 2; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
   const unsigned TS = CG_BM * CG_BK * 2;
+  // piece -> permuted-image coordinates: subblock sb = cb*16 + mq holds
+  // [4 m][16 col] halfwords at sb*128 B; piece = sb*8 + pr*2 + ch covers
+  // row pr, columns cb*16 + ch*8 .. +8 of image row mq*4 + pr.
   auto stage_dy = [&](long mt, char* buf) {
 #pragma unroll
     for (int j = 0; j < 2; ++j) {
       const int piece0 = wave * 64 + CG_THREADS * j;
       const int piece = piece0 + lane;
-      const int row = piece >> 3;
-      const int slot = piece & 7;
-      const int sslot = cg_swz(row, slot);
-      long m = mt * CG_BM + row;
+      const int sb = piece >> 3;
+      const int pr = (piece >> 1) & 3;
+      const int ch = piece & 1;
+      const int irow = (sb & 15) * 4 + pr;
+      long m = mt * CG_BM + wrw_row_to_m(irow);
       const __bf16* src;
-      const int o = o0 + sslot * 8;
+      const int o = o0 + (sb >> 4) * 16 + ch * 8;
       if (m < Mtot && o < Cout)
         src = dy + m * Cout + o;
       else
@@ -336,17 +371,18 @@ __global__ __launch_bounds__(CG_THREADS, 2) void conv_gemm_wrw_kernel(
     for (int j = 0; j < 2; ++j) {
       const int piece0 = wave * 64 + CG_THREADS * j;
       const int piece = piece0 + lane;
-      const int row = piece >> 3;
-      const int slot = piece & 7;
-      const int sslot = cg_swz(row, slot);
-      const long m = mt * CG_BM + row;
+      const int sb = piece >> 3;
+      const int pr = (piece >> 1) & 3;
+      const int ch = piece & 1;
+      const int irow = (sb & 15) * 4 + pr;
+      const long m = mt * CG_BM + wrw_row_to_m(irow);
       const __bf16* src = zpage;
       if (m < Mtot) {
         const int xx = (int)(m % WW);
         const int yy = (int)((m / WW) % HH);
         const long n = m / ((long)WW * HH);
         const int sy = yy * sH + dyo, sx = xx * sW + dxo;
-        const int c = c0 + sslot * 8;
+        const int c = c0 + (sb >> 4) * 16 + ch * 8;
         if (sy >= 0 && sy < srcH && sx >= 0 && sx < srcW && c < Cin) {
           if (x2 == nullptr || c < C1)
             src = x + (((long)n * srcH + sy) * srcW + sx) * ld_x + c;
@@ -375,51 +411,52 @@ __global__ __launch_bounds__(CG_THREADS, 2) void conv_gemm_wrw_kernel(
       stage_x(t + 1, lds + (cur ^ 1) * 2 * TS + TS);
     }
 
-    // transposed fragment reads from the m-major images:
-    //   A-frag (o rows, k=m): a[i] lane reads 8 m-consecutive u16 at
-    //   column (wr + i*16 + frow); image element (m, col) at m*128B + col*2
+    // hardware-transposed fragment reads (ds_read_b64_tr_b16): fragment
+    // (i, kk) = two tr windows of the permuted image — lane receives its
+    // own column's m-run (k = (lane>>4)*8 + e) directly in MFMA order.
+    // Window (cb, kk, w) sits at byte offset (cb*16 + kk*8 + w*4)*128;
+    // lane-linear +lane*8 addresses are conflict-free (banks 2l, 2l+1).
     const char* dbuf = lds + cur * 2 * TS;
     const char* xbuf = dbuf + TS;
-    const int frow = lane & 15;
-    const int fk = lane >> 4;  // which 8-m chunk
+    const unsigned lb = (unsigned)lane * 8;
 
 #pragma unroll
     for (int kk = 0; kk < 2; ++kk) {
-      bf16x8 af[2], bf[2];
-      const int mbase = kk * 32 + fk * 8;
+      bf16x4 a0[2], a1[2], b0[2], b1[2];
 #pragma unroll
       for (int i = 0; i < 2; ++i) {
-        const int col = wr + i * 16 + frow;  // o column
-        bf16x8 v;
-#pragma unroll
-        for (int e = 0; e < 8; ++e) {
-          const int m = mbase + e;
-          const int slot = col >> 3;
-          v[e] = *(const __bf16*)(dbuf + m * (CG_BK * 2) +
-                                  cg_swz(m, slot) * 16 + (col & 7) * 2);
-        }
-        af[i] = v;
+        const unsigned cb = (unsigned)(wr >> 4) + i;  // o column block
+        a0[i] = wrw_tr_read(dbuf, (cb * 16 + kk * 8 + 0) * 128 + lb);
+        a1[i] = wrw_tr_read(dbuf, (cb * 16 + kk * 8 + 4) * 128 + lb);
       }
 #pragma unroll
       for (int j = 0; j < 2; ++j) {
-        const int col = wc + j * 16 + frow;  // c column
-        bf16x8 v;
-#pragma unroll
-        for (int e = 0; e < 8; ++e) {
-          const int m = mbase + e;
-          const int slot = col >> 3;
-          v[e] = *(const __bf16*)(xbuf + m * (CG_BK * 2) +
-                                  cg_swz(m, slot) * 16 + (col & 7) * 2);
-        }
-        bf[j] = v;
+        const unsigned cb = (unsigned)(wc >> 4) + j;  // c column block
+        b0[j] = wrw_tr_read(xbuf, (cb * 16 + kk * 8 + 0) * 128 + lb);
+        b1[j] = wrw_tr_read(xbuf, (cb * 16 + kk * 8 + 4) * 128 + lb);
       }
+      // hipcc does not track asm loads: drain lgkm with every destination
+      // tied, then fence the (register-only) MFMAs below the wait (§5.4
+      // rule 18)
+      asm volatile("s_waitcnt lgkmcnt(0)"
+                   : "+v"(a0[0]), "+v"(a0[1]), "+v"(a1[0]), "+v"(a1[1]),
+                     "+v"(b0[0]), "+v"(b0[1]), "+v"(b1[0]), "+v"(b1[1])
+                   :: "memory");
+      __builtin_amdgcn_sched_barrier(0);
+
 #pragma unroll
-      for (int i = 0; i < 2; ++i)
+      for (int i = 0; i < 2; ++i) {
+        const bf16x8 af = __builtin_shufflevector(a0[i], a1[i], 0, 1, 2, 3,
+                                                  4, 5, 6, 7);
 #pragma unroll
-        for (int j = 0; j < 2; ++j)
-          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af[i], bf[j],
+        for (int j = 0; j < 2; ++j) {
+          const bf16x8 bf = __builtin_shufflevector(b0[j], b1[j], 0, 1, 2,
+                                                    3, 4, 5, 6, 7);
+          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bf,
                                                               acc[i][j], 0,
                                                               0, 0);
+        }
+      }
     }
 
     asm volatile("s_waitcnt vmcnt(0)" ::: "memory");

@@ -166,6 +166,11 @@ def fused_conv2d(x, weight, bias, stride, padding, dilation, groups, cache,
         from torch.nn.modules.utils import _pair
         sH, sW = _pair(stride)
         return ConvGemmFn.apply(x, weight, bias, wf, wb, sH, sW)
+    import os
+    if x.is_cuda and os.environ.get("FLOWHIP_LOG_FALLBACK", "0") == "1":
+        print(f"[conv fallback] w={tuple(weight.shape)} stride={stride} "
+              f"pad={padding} dil={dilation} g={groups} x={tuple(x.shape)} "
+              f"xs={x.stride()} dt={x.dtype}", flush=True)
     return F.conv2d(x, weight, bias, stride, padding, dilation, groups)
 
 
