@@ -157,6 +157,18 @@ def can_fuse_conv(x, weight, stride, padding, dilation, groups):
 def fused_conv2d(x, weight, bias, stride, padding, dilation, groups, cache,
                  key=None):
     """F.conv2d drop-in that routes supported shapes to the MFMA kernel."""
+    # autocast does NOT auto-cast custom autograd Functions: an fp32 input
+    # under an active bf16 autocast region (convc1's corr features, convf1's
+    # flow) would silently fall back to MIOpen — do autocast's cast here
+    if (x.is_cuda and x.dtype == torch.float32
+            and torch.is_autocast_enabled("cuda")
+            and torch.get_autocast_dtype("cuda") == torch.bfloat16):
+        xc = x.to(torch.bfloat16)
+        if xc.stride(1) != 1 or not xc.is_contiguous(
+                memory_format=torch.channels_last):
+            xc = xc.contiguous(memory_format=torch.channels_last)
+        if can_fuse_conv(xc, weight, stride, padding, dilation, groups):
+            x = xc
     if can_fuse_conv(x, weight, stride, padding, dilation, groups):
         if key is None:
             key = (weight._version, weight.data_ptr())

@@ -784,13 +784,15 @@ class TestConvGemm:
         with torch.autocast("cuda", dtype=torch.bfloat16):
             yf = enc_f(x.contiguous(memory_format=torch.channels_last))
             ym = enc_m(x.contiguous(memory_format=torch.channels_last))
-        # 8 bf16 convs + instance norms chained: MFMA-vs-MIOpen rounding
-        # drifts a handful of near-zero post-norm values past elementwise
-        # tolerances — bound the distribution, not the worst element
-        diff = (yf.float() - ym.float()).abs()
-        assert diff.mean().item() < 2e-2, diff.mean().item()
-        assert diff.max().item() < 0.5, diff.max().item()
-        assert (diff > 0.1).float().mean().item() < 5e-3
+        # 8 bf16 convs + instance norms chained amplify per-layer rounding
+        # (different k-orders) multiplicatively; anchor both bf16 paths
+        # against the fp32 oracle instead of against each other: the MFMA
+        # path may not sit materially farther from fp32 than MIOpen does.
+        with torch.no_grad():
+            y32 = enc_m(x)  # fp32, no autocast
+        d_f = (yf.float() - y32).abs().mean().item()
+        d_m = (ym.float() - y32).abs().mean().item()
+        assert d_f < max(2.0 * d_m, 0.02), (d_f, d_m)
 
     def test_narrowed_input_view(self):
         # a channel-narrowed channels-last view (the 324-of-328 corr case)
