@@ -13,10 +13,10 @@ void flowhip_bgemm_nt_launch(const void* A, const void* B, void* C,
                              float alpha, int batch, int M, int N, int K,
                              int out_bf16, hipStream_t stream);
 void flowhip_corr_lookup_fwd_launch(const void* level, const float* coords,
-                                    float* out, int BP, int P, int Hl, int Wl,
+                                    void* out, int BP, int P, int Hl, int Wl,
                                     int l, int L, int radius, int cl, int ldc,
                                     int is_bf16, hipStream_t stream);
-void flowhip_corr_lookup_bwd_launch(const float* gout, const float* coords,
+void flowhip_corr_lookup_bwd_launch(const void* gout, const float* coords,
                                     void* glevel, int BP, int P, int Hl,
                                     int Wl, int l, int L, int radius, int cl,
                                     int is_bf16, hipStream_t stream);
@@ -133,6 +133,9 @@ void flowhip_nconv_bwd_prep_launch(const float* gout, const float* gcout,
                                    float* dnomin, float* ddenom, long total,
                                    long plane, int Co, float eps,
                                    hipStream_t stream);
+bool flowhip_col_sum2_launch(const void* g, const void* x, float* partials,
+                             float* out, long M, int C, int nchunk,
+                             hipStream_t stream);
 bool flowhip_col_sum_launch(const void* dy, float* partials, float* out,
                             long M, int C, int nchunk, hipStream_t stream);
 void flowhip_up2x_cat_fwd_launch(const float* low, const float* skip,
@@ -208,31 +211,34 @@ torch::Tensor corr_lookup_fwd(std::vector<torch::Tensor> pyramid,
   const int K = 2 * (int)radius + 1;
 
   // channels-last: allocate with the channel count rounded up to 8 and
-  // return a narrow view — the NHWC conv consumer reads whole 16-B pieces
+  // return a narrow view — the NHWC conv consumer reads whole 16-B pieces.
+  // Output dtype follows the pyramid dtype (bf16-resident pyramid feeds
+  // the bf16 motion-encoder conv with no cast pass).
+  const bool bf = pyramid[0].dtype() == torch::kBFloat16;
+  const auto odt = bf ? torch::kBFloat16 : torch::kFloat32;
   const long C = (long)L * K * K;
   const long C8 = channels_last ? (C + 7) / 8 * 8 : C;
   auto full = channels_last
                   ? torch::empty({B, C8, H, W},
-                                 coords.options().dtype(torch::kFloat32)
+                                 coords.options().dtype(odt)
                                      .memory_format(torch::MemoryFormat::ChannelsLast))
                   : torch::empty({B, C, H, W},
-                                 coords.options().dtype(torch::kFloat32));
+                                 coords.options().dtype(odt));
   auto out = (C8 != C) ? full.narrow(1, 0, C) : full;
   const c10::cuda::CUDAGuard guard(coords.device());
   hipStream_t stream = at::cuda::getCurrentCUDAStream().stream();
 
   for (int l = 0; l < L; ++l) {
     auto& lvl = pyramid[l];
-    const bool lb = lvl.dtype() == torch::kBFloat16;
     TORCH_CHECK(lvl.is_cuda() && lvl.is_contiguous() &&
-                (lb || lvl.dtype() == torch::kFloat32),
-                "corr_lookup: fp32/bf16 contiguous pyramid levels required");
+                lvl.dtype() == (bf ? torch::kBFloat16 : torch::kFloat32),
+                "corr_lookup: same-dtype contiguous pyramid levels required");
     TORCH_CHECK(lvl.size(0) == (long)B * P, "corr_lookup: level batch mismatch");
     const int Hl = lvl.size(-2), Wl = lvl.size(-1);
     flowhip_corr_lookup_fwd_launch(
         lvl.data_ptr(), coords.data_ptr<float>(),
-        full.data_ptr<float>(), B * P, P, Hl, Wl, l, L, (int)radius,
-        channels_last ? 1 : 0, (int)C8, lb ? 1 : 0, stream);
+        full.data_ptr(), B * P, P, Hl, Wl, l, L, (int)radius,
+        channels_last ? 1 : 0, (int)C8, bf ? 1 : 0, stream);
   }
   return out;
 }
@@ -244,7 +250,10 @@ std::vector<torch::Tensor> corr_lookup_bwd(torch::Tensor gout,
                                                level_shapes,
                                            bool channels_last,
                                            bool levels_bf16) {
-  TORCH_CHECK(gout.is_cuda() && gout.dtype() == torch::kFloat32);
+  TORCH_CHECK(gout.is_cuda() &&
+              gout.dtype() == (levels_bf16 ? torch::kBFloat16
+                                           : torch::kFloat32),
+              "corr_lookup_bwd: gout dtype must match the pyramid dtype");
   TORCH_CHECK(channels_last
                   ? gout.is_contiguous(torch::MemoryFormat::ChannelsLast)
                   : gout.is_contiguous());
@@ -277,7 +286,7 @@ std::vector<torch::Tensor> corr_lookup_bwd(torch::Tensor gout,
     off += sizes[l];
     const int Hl = g.size(-2), Wl = g.size(-1);
     flowhip_corr_lookup_bwd_launch(
-        gout.data_ptr<float>(), coords.data_ptr<float>(), g.data_ptr(),
+        gout.data_ptr(), coords.data_ptr<float>(), g.data_ptr(),
         B * P, P, Hl, Wl, l, L, (int)radius, channels_last ? 1 : 0,
         levels_bf16 ? 1 : 0, stream);
     grads.push_back(g);
@@ -956,6 +965,29 @@ torch::Tensor col_sum_bf16(torch::Tensor dy) {
   return out;
 }
 
+torch::Tensor col_sum2_bf16(torch::Tensor g, torch::Tensor x) {
+  TORCH_CHECK(g.is_cuda() && g.dim() == 4 &&
+              g.scalar_type() == torch::kBFloat16 && g.stride(1) == 1 &&
+              g.stride(3) == g.size(1));
+  TORCH_CHECK(x.sizes() == g.sizes() && x.strides() == g.strides() &&
+              x.scalar_type() == torch::kBFloat16);
+  const int C = g.size(1);
+  const long M = g.numel() / C;
+  const int nchunk = (int)((M + 255) / 256);
+  const int ncb = (C + 63) / 64;
+  auto partials = torch::empty({(long)nchunk * 2 * ncb * 64},
+                               g.options().dtype(torch::kFloat32));
+  auto out = torch::empty({2, (long)C}, g.options().dtype(torch::kFloat32));
+  const c10::cuda::CUDAGuard guard(g.device());
+  hipStream_t stream = at::cuda::getCurrentCUDAStream().stream();
+  bool ok = flowhip_col_sum2_launch(g.data_ptr(), x.data_ptr(),
+                                    partials.data_ptr<float>(),
+                                    out.data_ptr<float>(), M, C, nchunk,
+                                    stream);
+  TORCH_CHECK(ok, "col_sum2_bf16: C must be a multiple of 8");
+  return out;
+}
+
 torch::Tensor up2x_cat_fwd(torch::Tensor low, torch::Tensor skip) {
   TORCH_CHECK(low.is_cuda() && low.is_contiguous() &&
               low.dtype() == torch::kFloat32);
@@ -1112,6 +1144,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("convex_up_bwd", &convex_up_bwd, "backward of convex_up_fwd");
   m.def("nconv_fwd", &nconv_fwd,
         "fused normalized convolution forward (out, cout)");
+  m.def("col_sum2_bf16", &col_sum2_bf16,
+        "(sum_m g, sum_m g*x) per channel in one pass (frozen-BN backward)");
   m.def("col_sum_bf16", &col_sum_bf16,
         "(N,C,H,W) channels-last bf16 -> (C) fp32 bias-grad column sum");
   m.def("up2x_cat_fwd", &up2x_cat_fwd,

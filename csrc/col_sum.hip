@@ -78,6 +78,112 @@ __global__ __launch_bounds__(512) void col_sum_finalize_kernel(
   }
 }
 
+// Dual column sum for the frozen-BatchNorm backward: sum_m g[m,c] AND
+// sum_m g[m,c]*x[m,c] in one pass over both tensors (torch's
+// native_batch_norm_backward spends ~105 us/call on the same reduction
+// pair at encoder shapes). partials layout: (chunk, 2, ncb*64).
+__global__ __launch_bounds__(CS_THREADS) void col_sum2_partial_kernel(
+    const __bf16* __restrict__ g, const __bf16* __restrict__ x,
+    float* __restrict__ partials, long M, int C, int nchunk) {
+  const int ncb = (C + 63) / 64;
+  int b = blockIdx.x;
+  const int chunk = b % nchunk; b /= nchunk;
+  const int cb = b;
+  const int CBW = min(64, C - cb * 64);
+  const int CG = CBW / 8;
+  const int gi = threadIdx.x % CG;
+  const int s = threadIdx.x / CG;
+  const int S = CS_THREADS / CG;
+  const int c0 = cb * 64 + gi * 8;
+
+  const long m0 = (long)chunk * CS_MCHUNK;
+  const long m1 = min(m0 + CS_MCHUNK, M);
+
+  float ag[8], agx[8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) ag[j] = agx[j] = 0.f;
+  if (s < S) {
+    for (long m = m0 + s; m < m1; m += S) {
+      const bf16x8 vg = *(const bf16x8*)(g + m * C + c0);
+      const bf16x8 vx = *(const bf16x8*)(x + m * C + c0);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const float fg = (float)vg[j];
+        ag[j] += fg;
+        agx[j] += fg * (float)vx[j];
+      }
+    }
+  }
+
+  __shared__ float red[2][64 * 33];
+  for (int j = 0; j < 8; ++j)
+    if (s < S && s < 32) {
+      red[0][(gi * 8 + j) * 33 + s] = ag[j];
+      red[1][(gi * 8 + j) * 33 + s] = agx[j];
+    }
+  __syncthreads();
+  if (s >= 32 && s < S)
+    for (int j = 0; j < 8; ++j) {
+      atomicAdd(&red[0][(gi * 8 + j) * 33 + (s & 31)], ag[j]);
+      atomicAdd(&red[1][(gi * 8 + j) * 33 + (s & 31)], agx[j]);
+    }
+  __syncthreads();
+  if (threadIdx.x < (unsigned)CBW) {
+    const int smax = S < 32 ? S : 32;
+    float t0 = 0.f, t1 = 0.f;
+    for (int ss = 0; ss < smax; ++ss) {
+      t0 += red[0][threadIdx.x * 33 + ss];
+      t1 += red[1][threadIdx.x * 33 + ss];
+    }
+    float* pc = partials + ((long)chunk * 2) * (ncb * 64);
+    pc[cb * 64 + threadIdx.x] = t0;
+    pc[ncb * 64 + cb * 64 + threadIdx.x] = t1;
+  }
+}
+
+__global__ __launch_bounds__(512) void col_sum2_finalize_kernel(
+    const float* __restrict__ partials, float* __restrict__ out, int C,
+    int nchunk) {
+  const int ncb = (C + 63) / 64;
+  const int cb = blockIdx.x;
+  const int lane = threadIdx.x & 63;
+  const int slice = threadIdx.x >> 6;
+  float t0 = 0.f, t1 = 0.f;
+  for (int ch = slice; ch < nchunk; ch += 8) {
+    const float* pc = partials + ((long)ch * 2) * (ncb * 64);
+    t0 += pc[cb * 64 + lane];
+    t1 += pc[ncb * 64 + cb * 64 + lane];
+  }
+  __shared__ float red[2][8][64];
+  red[0][slice][lane] = t0;
+  red[1][slice][lane] = t1;
+  __syncthreads();
+  const int c = cb * 64 + threadIdx.x;
+  if (threadIdx.x < 64 && c < C) {
+    float s0 = 0.f, s1 = 0.f;
+#pragma unroll
+    for (int k = 0; k < 8; ++k) {
+      s0 += red[0][k][threadIdx.x];
+      s1 += red[1][k][threadIdx.x];
+    }
+    out[c] = s0;
+    out[C + c] = s1;
+  }
+}
+
+bool flowhip_col_sum2_launch(const void* g, const void* x, float* partials,
+                             float* out, long M, int C, int nchunk,
+                             hipStream_t stream) {
+  if (C % 8 != 0 || C < 8) return false;
+  const int ncb = (C + 63) / 64;
+  hipLaunchKernelGGL(col_sum2_partial_kernel, dim3(ncb * nchunk),
+                     dim3(CS_THREADS), 0, stream, (const __bf16*)g,
+                     (const __bf16*)x, partials, M, C, nchunk);
+  hipLaunchKernelGGL(col_sum2_finalize_kernel, dim3(ncb), dim3(512), 0,
+                     stream, partials, out, C, nchunk);
+  return true;
+}
+
 bool flowhip_col_sum_launch(const void* dy, float* partials, float* out,
                             long M, int C, int nchunk, hipStream_t stream) {
   if (C % 8 != 0 || C < 8) return false;

@@ -40,7 +40,7 @@ template <int R, typename scalar_t>
 __global__ __launch_bounds__(LK_THREADS) void corr_lookup_fwd_kernel(
     const scalar_t* __restrict__ level,  // (B*P, Hl, Wl)
     const float* __restrict__ coords,    // (B, 2, H, W)
-    float* __restrict__ out,             // (B, L*K2, H, W) NCHW or NHWC
+    scalar_t* __restrict__ out,          // (B, L*K2, H, W) NCHW or NHWC
     int BP, int P, int Hl, int Wl, int l, int L, int cl, int ldc) {
   constexpr int K = 2 * R + 1;
   constexpr int K2 = K * K;
@@ -62,14 +62,14 @@ __global__ __launch_bounds__(LK_THREADS) void corr_lookup_fwd_kernel(
 
   const scalar_t* map = level + (long)pix * Hl * Wl;
   const long tap_stride = cl ? 1 : (long)P;
-  float* outb = cl ? out + ((long)b * P + i) * ldc + (long)l * K2
-                   : out + ((long)b * L * K2 + (long)l * K2) * P + i;
+  scalar_t* outb = cl ? out + ((long)b * P + i) * ldc + (long)l * K2
+                      : out + ((long)b * L * K2 + (long)l * K2) * P + i;
   if (cl && l == 0) {
     // zero the channel-pad tail once (ldc > L*K2: the emitted tensor is a
     // narrow view of an 8-channel-aligned allocation so the NHWC conv
     // consumer's 16-B staging reads stay in-row)
-    float* rowb = out + ((long)b * P + i) * ldc;
-    for (int c = L * K2; c < ldc; ++c) rowb[c] = 0.0f;
+    scalar_t* rowb = out + ((long)b * P + i) * ldc;
+    for (int c = L * K2; c < ldc; ++c) rowb[c] = (scalar_t)0.0f;
   }
 
   const bool interior = (x0 >= 0) & (y0 >= 0) & (x0 + K < Wl) &
@@ -91,7 +91,7 @@ __global__ __launch_bounds__(LK_THREADS) void corr_lookup_fwd_kernel(
 #pragma unroll
         for (int a = 0; a < K; ++a)
           outb[(long)(a * K + (j - 1)) * tap_stride] =
-              wy0 * rprev[a] + wy1 * rcur[a];
+              (scalar_t)(wy0 * rprev[a] + wy1 * rcur[a]);
       }
 #pragma unroll
       for (int a = 0; a < K; ++a) rprev[a] = rcur[a];
@@ -119,7 +119,7 @@ __global__ __launch_bounds__(LK_THREADS) void corr_lookup_fwd_kernel(
 #pragma unroll
       for (int c = 0; c < K; ++c)
         outb[(long)((a - 1) * K + c) * tap_stride] =
-            wx0 * tprev[c] + wx1 * tcur[c];
+            (scalar_t)(wx0 * tprev[c] + wx1 * tcur[c]);
     }
 #pragma unroll
     for (int c = 0; c < K; ++c) tprev[c] = tcur[c];
@@ -128,7 +128,7 @@ __global__ __launch_bounds__(LK_THREADS) void corr_lookup_fwd_kernel(
 
 template <int R, typename scalar_t>
 __global__ __launch_bounds__(LK_THREADS) void corr_lookup_bwd_kernel(
-    const float* __restrict__ gout,    // (B, L*K2, H, W)
+    const scalar_t* __restrict__ gout,  // (B, L*K2, H, W)
     const float* __restrict__ coords,  // (B, 2, H, W)
     scalar_t* __restrict__ glevel,     // (B*P, Hl, Wl), zero-initialized
     int BP, int P, int Hl, int Wl, int l, int L, int cl) {
@@ -152,8 +152,9 @@ __global__ __launch_bounds__(LK_THREADS) void corr_lookup_bwd_kernel(
 
   scalar_t* gmap = glevel + (long)pix * Hl * Wl;
   const long tap_stride = cl ? 1 : (long)P;
-  const float* gin = cl ? gout + ((long)b * P + i) * (L * K2) + (long)l * K2
-                        : gout + ((long)b * L * K2 + (long)l * K2) * P + i;
+  const scalar_t* gin = cl
+      ? gout + ((long)b * P + i) * (L * K2) + (long)l * K2
+      : gout + ((long)b * L * K2 + (long)l * K2) * P + i;
 
   const bool interior = (x0 >= 0) & (y0 >= 0) & (x0 + K < Wl) &
                         (y0 + K < Hl);
@@ -168,7 +169,7 @@ __global__ __launch_bounds__(LK_THREADS) void corr_lookup_bwd_kernel(
         float grow[K];
 #pragma unroll
         for (int a = 0; a < K; ++a)
-          grow[a] = gin[(long)(a * K + j) * tap_stride];
+          grow[a] = (float)gin[(long)(a * K + j) * tap_stride];
 #pragma unroll
         for (int u = 0; u <= K; ++u) {
           float v = 0.0f;
@@ -202,7 +203,7 @@ __global__ __launch_bounds__(LK_THREADS) void corr_lookup_bwd_kernel(
   for (int a = 0; a <= K; ++a) {
 #pragma unroll
     for (int c = 0; c < K; ++c)
-      gcur[c] = (a < K) ? gin[(long)(a * K + c) * tap_stride] : 0.0f;
+      gcur[c] = (a < K) ? (float)gin[(long)(a * K + c) * tap_stride] : 0.0f;
 
     const int xx = x0 + a;
     if ((xx >= 0) & (xx < Wl)) {
@@ -230,8 +231,9 @@ __global__ __launch_bounds__(LK_THREADS) void corr_lookup_bwd_kernel(
 
 template <int R, typename scalar_t>
 static void lookup_fwd_level(const scalar_t* level, const float* coords,
-                             float* out, int BP, int P, int Hl, int Wl, int l,
-                             int L, int cl, int ldc, hipStream_t stream) {
+                             scalar_t* out, int BP, int P, int Hl, int Wl,
+                             int l, int L, int cl, int ldc,
+                             hipStream_t stream) {
   dim3 grid(fh_cdiv(BP, LK_THREADS));
   hipLaunchKernelGGL((corr_lookup_fwd_kernel<R, scalar_t>), grid,
                      dim3(LK_THREADS), 0, stream, level, coords, out, BP, P,
@@ -239,7 +241,7 @@ static void lookup_fwd_level(const scalar_t* level, const float* coords,
 }
 
 template <int R, typename scalar_t>
-static void lookup_bwd_level(const float* gout, const float* coords,
+static void lookup_bwd_level(const scalar_t* gout, const float* coords,
                              scalar_t* glevel, int BP, int P, int Hl, int Wl,
                              int l, int L, int cl, hipStream_t stream) {
   dim3 grid(fh_cdiv(BP, LK_THREADS));
@@ -250,39 +252,41 @@ static void lookup_bwd_level(const float* gout, const float* coords,
 
 // levels may be fp32 (reference parity) or bf16 (HBM-resident bf16
 // pyramid); out / gout stay fp32, the bilerp blend always runs fp32.
+// out / gout dtype == level dtype (bf16-resident pyramid emits bf16 taps
+// straight into the bf16 NHWC motion-encoder conv; fp32 keeps parity).
 void flowhip_corr_lookup_fwd_launch(const void* level, const float* coords,
-                                    float* out, int BP, int P, int Hl, int Wl,
+                                    void* out, int BP, int P, int Hl, int Wl,
                                     int l, int L, int radius, int cl, int ldc,
                                     int is_bf16, hipStream_t stream) {
   if (is_bf16) {
     switch (radius) {
-      case 3: lookup_fwd_level<3>((const __bf16*)level, coords, out, BP, P, Hl, Wl, l, L, cl, ldc, stream); break;
-      case 4: lookup_fwd_level<4>((const __bf16*)level, coords, out, BP, P, Hl, Wl, l, L, cl, ldc, stream); break;
+      case 3: lookup_fwd_level<3>((const __bf16*)level, coords, (__bf16*)out, BP, P, Hl, Wl, l, L, cl, ldc, stream); break;
+      case 4: lookup_fwd_level<4>((const __bf16*)level, coords, (__bf16*)out, BP, P, Hl, Wl, l, L, cl, ldc, stream); break;
       default: abort();
     }
   } else {
     switch (radius) {
-      case 3: lookup_fwd_level<3>((const float*)level, coords, out, BP, P, Hl, Wl, l, L, cl, ldc, stream); break;
-      case 4: lookup_fwd_level<4>((const float*)level, coords, out, BP, P, Hl, Wl, l, L, cl, ldc, stream); break;
+      case 3: lookup_fwd_level<3>((const float*)level, coords, (float*)out, BP, P, Hl, Wl, l, L, cl, ldc, stream); break;
+      case 4: lookup_fwd_level<4>((const float*)level, coords, (float*)out, BP, P, Hl, Wl, l, L, cl, ldc, stream); break;
       default: abort();
     }
   }
 }
 
-void flowhip_corr_lookup_bwd_launch(const float* gout, const float* coords,
+void flowhip_corr_lookup_bwd_launch(const void* gout, const float* coords,
                                     void* glevel, int BP, int P, int Hl,
                                     int Wl, int l, int L, int radius, int cl,
                                     int is_bf16, hipStream_t stream) {
   if (is_bf16) {
     switch (radius) {
-      case 3: lookup_bwd_level<3>(gout, coords, (__bf16*)glevel, BP, P, Hl, Wl, l, L, cl, stream); break;
-      case 4: lookup_bwd_level<4>(gout, coords, (__bf16*)glevel, BP, P, Hl, Wl, l, L, cl, stream); break;
+      case 3: lookup_bwd_level<3>((const __bf16*)gout, coords, (__bf16*)glevel, BP, P, Hl, Wl, l, L, cl, stream); break;
+      case 4: lookup_bwd_level<4>((const __bf16*)gout, coords, (__bf16*)glevel, BP, P, Hl, Wl, l, L, cl, stream); break;
       default: abort();
     }
   } else {
     switch (radius) {
-      case 3: lookup_bwd_level<3>(gout, coords, (float*)glevel, BP, P, Hl, Wl, l, L, cl, stream); break;
-      case 4: lookup_bwd_level<4>(gout, coords, (float*)glevel, BP, P, Hl, Wl, l, L, cl, stream); break;
+      case 3: lookup_bwd_level<3>((const float*)gout, coords, (float*)glevel, BP, P, Hl, Wl, l, L, cl, stream); break;
+      case 4: lookup_bwd_level<4>((const float*)gout, coords, (float*)glevel, BP, P, Hl, Wl, l, L, cl, stream); break;
       default: abort();
     }
   }

@@ -37,7 +37,8 @@ def _norm(norm_fn, planes, groups_planes=None):
         ng = groups_planes if groups_planes is not None else planes // 8
         return nn.GroupNorm(num_groups=ng, num_channels=planes)
     if norm_fn == "batch":
-        return nn.BatchNorm2d(planes)
+        from .norm import BatchNorm2d
+        return BatchNorm2d(planes)
     if norm_fn == "instance":
         from .norm import InstanceNorm2d
         return InstanceNorm2d(planes)

@@ -46,7 +46,8 @@ class Simple(nn.Module):
                                   padding=pad_for(filter_sz[i], dilation[i]),
                                   dilation=dilation[i], stride=1)]
             if use_bn:
-                layers.append(nn.BatchNorm2d(num_ch[i + 1]))
+                from .norm import BatchNorm2d
+                layers.append(BatchNorm2d(num_ch[i + 1]))
             layers.append(nn.ReLU(inplace=True))
             self.conv.append(nn.Sequential(*layers))
 

@@ -37,3 +37,58 @@ class InstanceNorm2d(nn.InstanceNorm2d):
                 and _ext.ext() is not None and not _ext.force_ref()):
             return _InstNormCLFn.apply(x, self.eps)
         return super().forward(x)
+
+
+class _FrozenBNFn(torch.autograd.Function):
+    """Eval-mode BatchNorm y = (x - rm) * w/sqrt(rv+eps) + b with a fused
+    backward: torch's native_batch_norm_backward spends ~105 us/call on the
+    (sum g, sum g*xhat) reductions at encoder shapes; one col_sum2 pass +
+    an elementwise scale replaces it (~3x). Frozen stats (freeze_bn after
+    the chairs stage — reference train.py:185) make stats constants, so
+        dx = g * s,  db = sum(g),  dw = (sum(g*x) - rm*sum(g)) * invstd.
+    """
+
+    @staticmethod
+    def forward(ctx, x, weight, bias, rm, rv, eps):
+        invstd = (rv.float() + eps).rsqrt()
+        s = weight.float() * invstd
+        t = bias.float() - rm.float() * s
+        # one elementwise pass in the input dtype (stock autocast BN runs
+        # fp32 and pays casts both ways; bf16 affine deviation is bounded
+        # by the encoder/model parity tests)
+        y = torch.addcmul(t.to(x.dtype).view(1, -1, 1, 1), x,
+                          s.to(x.dtype).view(1, -1, 1, 1))
+        ctx.save_for_backward(x, s, invstd, rm)
+        return y
+
+    @staticmethod
+    def backward(ctx, g):
+        x, s, invstd, rm = ctx.saved_tensors
+        g = g.contiguous(memory_format=torch.channels_last)
+        if g.dtype != torch.bfloat16:
+            g = g.to(torch.bfloat16)
+        sums = _ext.ext().col_sum2_bf16(
+            g, x if x.dtype == torch.bfloat16 else x.to(torch.bfloat16))
+        db = sums[0]
+        dw = (sums[1] - rm.float() * db) * invstd
+        dx = g * s.view(1, -1, 1, 1).to(g.dtype)
+        return dx, dw, db, None, None, None
+
+
+class BatchNorm2d(nn.BatchNorm2d):
+    """nn.BatchNorm2d whose FROZEN (eval-mode) GPU path runs the fused
+    backward above. Training mode (chairs stage) and CPU use the stock
+    implementation; state dict unchanged."""
+
+    def forward(self, x):
+        use = (not self.training and x.is_cuda and self.affine
+               and self.track_running_stats
+               and x.is_contiguous(memory_format=torch.channels_last)
+               and x.dtype in (torch.float32, torch.bfloat16)
+               and x.shape[1] % 8 == 0
+               and _ext.ext() is not None and not _ext.force_ref())
+        if use:
+            return _FrozenBNFn.apply(x, self.weight, self.bias,
+                                     self.running_mean, self.running_var,
+                                     self.eps)
+        return super().forward(x)

@@ -160,7 +160,18 @@ class BasicMotionEncoder(nn.Module):
             flow = flow.contiguous(memory_format=torch.channels_last)
         cor = F.relu(self.convc2(F.relu(self.convc1(corr))))
         flo = F.relu(self.convf2(F.relu(self.convf1(flow))))
-        out = F.relu(self.conv(torch.cat([cor, flo], dim=1)))
+        if cor.is_cuda and cor.dtype == torch.bfloat16:
+            # virtually-concatenated conv input (192 | 64): no cat
+            # materialization, no backward narrows
+            if not hasattr(self, "_cv_cache"):
+                self._cv_cache = {}
+            out = F.relu(fused_conv2d_cat2(
+                cor, flo, self.conv.weight, self.conv.bias, (1, 1),
+                self._cv_cache,
+                key=(self.conv.weight.data_ptr(),
+                     self.conv.weight._version)))
+        else:
+            out = F.relu(self.conv(torch.cat([cor, flo], dim=1)))
         return torch.cat([out, flow], dim=1)
 
 

@@ -131,6 +131,9 @@ class CorrLookupFn(torch.autograd.Function):
     @staticmethod
     def backward(ctx, grad):
         (coords,) = ctx.saved_tensors
+        dt = torch.bfloat16 if ctx.levels_bf16 else torch.float32
+        if grad.dtype != dt:
+            grad = grad.to(dt)
         if ctx.cl:
             grad = grad.contiguous(memory_format=torch.channels_last)
         else:

@@ -172,7 +172,12 @@ def fused_conv2d(x, weight, bias, stride, padding, dilation, groups, cache,
     if can_fuse_conv(x, weight, stride, padding, dilation, groups):
         if key is None:
             key = (weight._version, weight.data_ptr())
-        if weight.shape[1] % 8 != 0:
+        if weight.shape[1] % 8 != 0 and x.stride(3) == x.size(1):
+            # ragged Cin on a tight row: pad both (16-B staging tail).
+            # A channel-narrowed view with padded ld (the corr-lookup
+            # output) needs NO pad — its tail bytes are in-row and the
+            # packed weight's zero columns null them (saves a full-tensor
+            # cat copy per GRU iteration).
             x, weight = _pad_c8(x, weight)
         wf, wb = _packs(weight, cache, key)
         from torch.nn.modules.utils import _pair
