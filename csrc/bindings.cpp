@@ -136,6 +136,9 @@ void flowhip_nconv_bwd_prep_launch(const float* gout, const float* gcout,
 bool flowhip_col_sum2_launch(const void* g, const void* x, float* partials,
                              float* out, long M, int C, int nchunk,
                              hipStream_t stream);
+void flowhip_frozen_bn_apply_launch(const void* x, void* y, const float* s,
+                                    const float* t, long M, int C,
+                                    hipStream_t stream);
 bool flowhip_col_sum_launch(const void* dy, float* partials, float* out,
                             long M, int C, int nchunk, hipStream_t stream);
 void flowhip_up2x_cat_fwd_launch(const float* low, const float* skip,
@@ -965,6 +968,29 @@ torch::Tensor col_sum_bf16(torch::Tensor dy) {
   return out;
 }
 
+torch::Tensor frozen_bn_apply(torch::Tensor x, torch::Tensor s,
+                              c10::optional<torch::Tensor> t) {
+  TORCH_CHECK(x.is_cuda() && x.dim() == 4 &&
+              x.scalar_type() == torch::kBFloat16 && x.stride(1) == 1 &&
+              x.stride(3) == x.size(1) && x.size(1) % 8 == 0);
+  TORCH_CHECK(s.is_contiguous() && s.scalar_type() == torch::kFloat32 &&
+              s.numel() == x.size(1));
+  const float* tp = nullptr;
+  if (t.has_value()) {
+    TORCH_CHECK(t->is_contiguous() && t->scalar_type() == torch::kFloat32 &&
+                t->numel() == x.size(1));
+    tp = t->data_ptr<float>();
+  }
+  auto y = torch::empty_like(x);
+  const c10::cuda::CUDAGuard guard(x.device());
+  hipStream_t stream = at::cuda::getCurrentCUDAStream().stream();
+  flowhip_frozen_bn_apply_launch(x.data_ptr(), y.data_ptr(),
+                                 s.data_ptr<float>(), tp,
+                                 x.numel() / x.size(1), (int)x.size(1),
+                                 stream);
+  return y;
+}
+
 torch::Tensor col_sum2_bf16(torch::Tensor g, torch::Tensor x) {
   TORCH_CHECK(g.is_cuda() && g.dim() == 4 &&
               g.scalar_type() == torch::kBFloat16 && g.stride(1) == 1 &&
@@ -1144,6 +1170,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("convex_up_bwd", &convex_up_bwd, "backward of convex_up_fwd");
   m.def("nconv_fwd", &nconv_fwd,
         "fused normalized convolution forward (out, cout)");
+  m.def("frozen_bn_apply", &frozen_bn_apply,
+        "y = bf16(fp32(x)*s + t) per channel, channels-last");
   m.def("col_sum2_bf16", &col_sum2_bf16,
         "(sum_m g, sum_m g*x) per channel in one pass (frozen-BN backward)");
   m.def("col_sum_bf16", &col_sum_bf16,

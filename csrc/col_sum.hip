@@ -171,6 +171,40 @@ __global__ __launch_bounds__(512) void col_sum2_finalize_kernel(
   }
 }
 
+// Frozen-BN affine: y[m,c] = bf16(fp32(x[m,c]) * s[c] + t[c]) on
+// channels-last rows. fp32 math + bf16 store puts the quantization point
+// exactly where the stock autocast path (fp32 BN output, bf16 cast at the
+// next conv) puts it. t == nullptr -> t = 0 (the backward dx = g*s pass).
+__global__ __launch_bounds__(256) void frozen_bn_apply_kernel(
+    const __bf16* __restrict__ x, __bf16* __restrict__ y,
+    const float* __restrict__ s, const float* __restrict__ t, long npieces,
+    int cpieces) {
+  for (long p = (long)blockIdx.x * 256 + threadIdx.x; p < npieces;
+       p += (long)gridDim.x * 256) {
+    const int c0 = (int)(p % cpieces) * 8;
+    const bf16x8 v = *(const bf16x8*)(x + p * 8);
+    bf16x8 o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float r = (float)v[j] * s[c0 + j];
+      if (t) r += t[c0 + j];
+      o[j] = (__bf16)r;
+    }
+    *(bf16x8*)(y + p * 8) = o;
+  }
+}
+
+void flowhip_frozen_bn_apply_launch(const void* x, void* y, const float* s,
+                                    const float* t, long M, int C,
+                                    hipStream_t stream) {
+  const long npieces = M * (C / 8);
+  long blocks = (npieces + 255) / 256;
+  if (blocks > 8192) blocks = 8192;
+  hipLaunchKernelGGL(frozen_bn_apply_kernel, dim3((int)blocks), dim3(256), 0,
+                     stream, (const __bf16*)x, (__bf16*)y, s, t, npieces,
+                     C / 8);
+}
+
 bool flowhip_col_sum2_launch(const void* g, const void* x, float* partials,
                              float* out, long M, int C, int nchunk,
                              hipStream_t stream) {

@@ -15,7 +15,8 @@ import torch
 import torch.nn as nn
 import torch.nn.functional as F
 
-from ..ops.functional_conv import fused_conv2d, fused_conv2d_cat2
+from ..ops.functional_conv import (fused_conv2d, fused_conv2d_cat2,
+                                   fused_gru_zr_conv)
 
 
 class FusedConv2d(nn.Conv2d):
@@ -87,15 +88,10 @@ class SepConvGRU(nn.Module):
         self.convq2 = FusedConv2d(hidden_dim + input_dim, hidden_dim, (5, 1), padding=(2, 0))
 
     def _pass(self, h, x, convz, convr, convq, padding, zr_cache, q_cache):
-        zr_w = torch.cat([convz.weight, convr.weight])
-        zr_b = torch.cat([convz.bias, convr.bias])
-        # virtually-concatenated input (no cat materialization on GPU); the
-        # packed-weight cache is keyed by the source params' versions
-        zr = fused_conv2d_cat2(h, x, zr_w, zr_b, padding, zr_cache,
-                               key=(convz.weight.data_ptr(),
-                                    convz.weight._version,
-                                    convr.weight.data_ptr(),
-                                    convr.weight._version))
+        # packed z+r conv over the virtually-concatenated input: weights,
+        # bias and their packings cached per parameter version (no per-call
+        # torch.cat of weights)
+        zr = fused_gru_zr_conv(h, x, convz, convr, padding, zr_cache)
         if h.is_cuda:
             # fused gate kernels (ops/functional_gru): one kernel for
             # sigmoid/chunk/r*h, one for tanh + lerp, fused backwards

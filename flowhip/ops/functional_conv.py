@@ -224,6 +224,73 @@ class ConvGemmCat2Fn(torch.autograd.Function):
         return dx1, dx2, dw, dbias, None, None
 
 
+class ConvGemmCat2ZrFn(torch.autograd.Function):
+    """Packed z+r gate convolution over the virtually-concatenated GRU
+    input: ONE kernel computes conv(cat([h, x]), cat([wz, wr])) without
+    materializing either cat. The packed weights/bias are cached per
+    parameter version (rebuilt after optimizer steps only); gradients are
+    routed back to the four separate parameters by slicing — replaces the
+    per-call torch.cat of weights AND its CatBackward (4 cat launches per
+    GRU pass)."""
+
+    @staticmethod
+    def forward(ctx, x1, x2, wz, wr, bz, br, bias_f, wpk_fwd, wpk_bwd):
+        Oz, I, KH, KW = wz.shape
+        O = Oz + wr.shape[0]
+        out = _ext.ext().conv_gemm_fwd2(x1, x2, wpk_fwd, bias_f, O, KH, KW,
+                                        0, 0)[0]
+        ctx.save_for_backward(x1, x2, wpk_bwd)
+        ctx.meta = (O, Oz, I, KH, KW, x1.shape[1])
+        return out
+
+    @staticmethod
+    def backward(ctx, dy):
+        x1, x2, wpk_bwd = ctx.saved_tensors
+        O, Oz, I, KH, KW, C1 = ctx.meta
+        dy = dy.contiguous(memory_format=torch.channels_last)
+        if dy.dtype != torch.bfloat16:
+            dy = dy.to(torch.bfloat16)
+        dx1, dx2 = _ext.ext().conv_gemm_fwd2(dy, None, wpk_bwd, None, I, KH,
+                                             KW, C1, 0)
+        dw = _ext.ext().conv_gemm_wrw(dy, x1, x2, KH, KW, 1, 1)
+        db = _col_sum(dy)
+        return (dx1, dx2, dw[:Oz], dw[Oz:].contiguous(), db[:Oz],
+                db[Oz:].contiguous(), None, None, None)
+
+
+def fused_gru_zr_conv(h, x, convz, convr, padding, cache):
+    """conv2d(cat([h, x]), cat([wz, wr])) + cat([bz, br]) in one fused
+    kernel with version-cached packing. Falls back to the eager
+    composition off-GPU."""
+    key = (convz.weight.data_ptr(), convz.weight._version,
+           convr.weight.data_ptr(), convr.weight._version,
+           convz.bias._version, convr.bias._version)
+    O1, I, KH, KW = convz.weight.shape
+    fusable = (h.is_cuda and h.dtype == torch.bfloat16
+               and x.dtype == torch.bfloat16
+               and h.shape[1] % 64 == 0
+               and (h.shape[1] + x.shape[1]) % 8 == 0
+               and h.stride(1) == 1 and x.stride(1) == 1
+               and _ext.ext() is not None and not _ext.force_ref())
+    from torch.nn.modules.utils import _pair
+    if fusable and _pair(padding) == (KH // 2, KW // 2):
+        if cache.get("k") != key:
+            with torch.no_grad():
+                w = torch.cat([convz.weight, convr.weight])
+                cache["k"] = key
+                cache["fwd"] = _pack_fwd(w)
+                cache["bwd"] = _pack_bwd(w)
+                cache["bias"] = torch.cat([convz.bias, convr.bias]) \
+                    .float().contiguous()
+        return ConvGemmCat2ZrFn.apply(h, x, convz.weight, convr.weight,
+                                      convz.bias, convr.bias, cache["bias"],
+                                      cache["fwd"], cache["bwd"])
+    zr_w = torch.cat([convz.weight, convr.weight])
+    zr_b = torch.cat([convz.bias, convr.bias])
+    hx = torch.cat([h, x], dim=1)
+    return F.conv2d(hx, zr_w, zr_b, padding=padding)
+
+
 def fused_conv2d_cat2(x1, x2, weight, bias, padding, cache, key):
     """conv2d(cat([x1, x2], 1), weight) without materializing the cat.
     Falls back to the eager cat + F.conv2d outside the fused envelope."""
