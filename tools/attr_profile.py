@@ -96,6 +96,19 @@ def main():
         for e in sorted(keys, key=lambda e: -e.self_device_time_total)[:50]:
             f.write(f"{e.self_device_time_total/1e3/args.steps:9.3f} ms/step "
                     f"x{e.count//args.steps:5d}  {e.key[:120]}\n")
+
+    # identify any convs still on the torch/MIOpen path by input shape
+    with profile(activities=[ProfilerActivity.CPU, ProfilerActivity.CUDA],
+                 record_shapes=True) as prof2:
+        step()
+        torch.cuda.synchronize()
+    with open(args.out, "a") as f:
+        f.write("\n\n==== library conv calls by input shape ====\n")
+        for e in prof2.key_averages(group_by_input_shape=True):
+            if ("miopen_convolution" in e.key or "convolution_backward"
+                    in e.key or "batch_norm" in e.key):
+                f.write(f"{e.device_time_total/1e3:9.3f} ms x{e.count:4d}  "
+                        f"{e.key[:60]}  shapes={e.input_shapes}\n")
     print("wrote", args.out)
 
 
