@@ -133,7 +133,7 @@ class SmallMotionEncoder(nn.Module):
         cor = F.relu(self.convc1(corr))
         flo = F.relu(self.convf2(F.relu(self.convf1(flow))))
         out = F.relu(self.conv(torch.cat([cor, flo], dim=1)))
-        return torch.cat([out, flow], dim=1)
+        return torch.cat([out, flow.to(out.dtype)], dim=1)
 
 
 class BasicMotionEncoder(nn.Module):
@@ -168,7 +168,11 @@ class BasicMotionEncoder(nn.Module):
                      self.conv.weight._version)))
         else:
             out = F.relu(self.conv(torch.cat([cor, flo], dim=1)))
-        return torch.cat([out, flow], dim=1)
+        # keep the motion features in ONE dtype: cat([bf16, fp32]) promotes
+        # the whole 128-ch feature map to fp32 and silently pushes all four
+        # 384-ch GRU convs off the MFMA path (autocast would cast flow at
+        # the conv anyway — same math, discovered via attr_profile r02)
+        return torch.cat([out, flow.to(out.dtype)], dim=1)
 
 
 class SmallUpdateBlock(nn.Module):
