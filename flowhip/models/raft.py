@@ -78,8 +78,12 @@ class RAFT(nn.Module):
 
         cnet = self.cnet(image1)
         net, inp = torch.split(cnet, [self.hidden_dim, self.context_dim], dim=1)
-        net = torch.tanh(net)
-        inp = torch.relu(inp)
+        # tanh/relu on the narrowed channels-last views materialize plain
+        # NCHW tensors; pin both back to the model layout ONCE here or the
+        # GRU input cat goes NCHW and every 384-ch GRU conv leaves the
+        # MFMA path (attr_profile r02)
+        net = to_model_layout(torch.tanh(net))
+        inp = to_model_layout(torch.relu(inp))
         return image1, corr_fn, net, inp
 
     def forward(self, image1, image2, iters=12, flow_init=None, upsample=True,

@@ -285,6 +285,11 @@ def fused_gru_zr_conv(h, x, convz, convr, padding, cache):
         return ConvGemmCat2ZrFn.apply(h, x, convz.weight, convr.weight,
                                       convz.bias, convr.bias, cache["bias"],
                                       cache["fwd"], cache["bwd"])
+    import os
+    if h.is_cuda and os.environ.get("FLOWHIP_LOG_FALLBACK", "0") == "1":
+        print(f"[zr fallback] h={tuple(h.shape)} hd={h.dtype} "
+              f"hs={h.stride()} x={tuple(x.shape)} xd={x.dtype} "
+              f"xs={x.stride()} pad={padding} K=({KH},{KW})", flush=True)
     zr_w = torch.cat([convz.weight, convr.weight])
     zr_b = torch.cat([convz.bias, convr.bias])
     hx = torch.cat([h, x], dim=1)
@@ -305,5 +310,11 @@ def fused_conv2d_cat2(x1, x2, weight, bias, padding, cache, key):
     if fusable and _pair(padding) == (KH // 2, KW // 2):
         wf, wb = _packs(weight, cache, key)
         return ConvGemmCat2Fn.apply(x1, x2, weight, bias, wf, wb)
+    import os
+    if x1.is_cuda and os.environ.get("FLOWHIP_LOG_FALLBACK", "0") == "1":
+        print(f"[cat2 fallback] x1={tuple(x1.shape)} d={x1.dtype} "
+              f"s={x1.stride()} x2={tuple(x2.shape)} d2={x2.dtype} "
+              f"s2={x2.stride()} w={tuple(weight.shape)} pad={padding}",
+              flush=True)
     hx = torch.cat([x1, x2], dim=1)
     return F.conv2d(hx, weight, bias, padding=padding)
