@@ -128,7 +128,7 @@ class SmallMotionEncoder(nn.Module):
         self.conv = FusedConv2d(128, 80, 3, padding=1)
 
     def forward(self, flow, corr):
-        if corr.is_contiguous(memory_format=torch.channels_last):
+        if corr.is_cuda and corr.stride(1) == 1:
             flow = flow.contiguous(memory_format=torch.channels_last)
         cor = F.relu(self.convc1(corr))
         flo = F.relu(self.convf2(F.relu(self.convf1(flow))))
@@ -149,10 +149,12 @@ class BasicMotionEncoder(nn.Module):
         self.conv = FusedConv2d(64 + 192, 128 - 2, 3, padding=1)
 
     def forward(self, flow, corr):
-        if corr.is_contiguous(memory_format=torch.channels_last):
+        if corr.is_cuda and corr.stride(1) == 1:
             # keep the whole motion-feature chain in one layout: a mixed
             # cat([cl, nchw]) falls back to NCHW and costs an uncoalesced
-            # layout copy on every GRU conv input (profiles/, tprof6)
+            # layout copy on every GRU conv input (profiles/, tprof6).
+            # stride(1)==1, not is_contiguous(channels_last): the corr
+            # lookup output is a channel-NARROWED channels-last view
             flow = flow.contiguous(memory_format=torch.channels_last)
         cor = F.relu(self.convc2(F.relu(self.convc1(corr))))
         flo = F.relu(self.convf2(F.relu(self.convf1(flow))))
