@@ -19,7 +19,8 @@ void flowhip_corr_lookup_fwd_launch(const void* level, const float* coords,
 void flowhip_corr_lookup_bwd_launch(const void* gout, const float* coords,
                                     void* glevel, int BP, int P, int Hl,
                                     int Wl, int l, int L, int radius, int cl,
-                                    int is_bf16, hipStream_t stream);
+                                    int is_bf16, int acc,
+                                    hipStream_t stream);
 void flowhip_convex_up_fwd_launch(const float* flow, const float* mask,
                                   float* out, int N, int H, int W, int factor,
                                   hipStream_t stream);
@@ -252,7 +253,9 @@ std::vector<torch::Tensor> corr_lookup_bwd(torch::Tensor gout,
                                            std::vector<std::vector<int64_t>>
                                                level_shapes,
                                            bool channels_last,
-                                           bool levels_bf16) {
+                                           bool levels_bf16,
+                                           c10::optional<std::vector<torch::Tensor>>
+                                               prev) {
   TORCH_CHECK(gout.is_cuda() &&
               gout.dtype() == (levels_bf16 ? torch::kBFloat16
                                            : torch::kFloat32),
@@ -267,32 +270,47 @@ std::vector<torch::Tensor> corr_lookup_bwd(torch::Tensor gout,
   const c10::cuda::CUDAGuard guard(coords.device());
   hipStream_t stream = at::cuda::getCurrentCUDAStream().stream();
 
-  // one flat zero-fill for all levels (4x fewer fill kernels), sliced into
-  // per-level views; level-grad dtype matches the resident pyramid dtype
-  int64_t total = 0;
-  std::vector<int64_t> sizes(L);
-  for (int l = 0; l < L; ++l) {
-    int64_t n = 1;
-    for (auto d : level_shapes[l]) n *= d;
-    sizes[l] = n;
-    total += n;
-  }
-  auto flat = torch::zeros(
-      {total}, gout.options().dtype(levels_bf16 ? torch::kBFloat16
-                                                : torch::kFloat32));
-
+  // `prev` = the NEXT iteration's level grads (the pyramid is threaded
+  // through the iteration loop as a chain): accumulate this lookup's
+  // contribution into them IN KERNEL — no fresh zero-fill, no autograd
+  // fan-in add_. Without prev: one flat zero-fill for all levels, plain
+  // stores. Level-grad dtype matches the resident pyramid dtype.
   std::vector<torch::Tensor> grads;
   grads.reserve(L);
-  int64_t off = 0;
+  const bool acc = prev.has_value();
+  if (acc) {
+    TORCH_CHECK((int)prev->size() == L);
+    for (int l = 0; l < L; ++l) {
+      auto& g = (*prev)[l];
+      TORCH_CHECK(g.is_cuda() && g.is_contiguous() &&
+                  g.dtype() == gout.dtype());
+      grads.push_back(g);
+    }
+  } else {
+    int64_t total = 0;
+    std::vector<int64_t> sizes(L);
+    for (int l = 0; l < L; ++l) {
+      int64_t n = 1;
+      for (auto d : level_shapes[l]) n *= d;
+      sizes[l] = n;
+      total += n;
+    }
+    auto flat = torch::zeros(
+        {total}, gout.options().dtype(levels_bf16 ? torch::kBFloat16
+                                                  : torch::kFloat32));
+    int64_t off = 0;
+    for (int l = 0; l < L; ++l) {
+      grads.push_back(flat.narrow(0, off, sizes[l]).view(level_shapes[l]));
+      off += sizes[l];
+    }
+  }
   for (int l = 0; l < L; ++l) {
-    auto g = flat.narrow(0, off, sizes[l]).view(level_shapes[l]);
-    off += sizes[l];
+    auto& g = grads[l];
     const int Hl = g.size(-2), Wl = g.size(-1);
     flowhip_corr_lookup_bwd_launch(
         gout.data_ptr(), coords.data_ptr<float>(), g.data_ptr(),
         B * P, P, Hl, Wl, l, L, (int)radius, channels_last ? 1 : 0,
-        levels_bf16 ? 1 : 0, stream);
-    grads.push_back(g);
+        levels_bf16 ? 1 : 0, acc ? 1 : 0, stream);
   }
   return grads;
 }
@@ -1152,10 +1170,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("corr_lookup_fwd", &corr_lookup_fwd,
         "fused multi-level correlation window lookup (fp32/bf16 levels)");
   m.def("corr_lookup_bwd", &corr_lookup_bwd,
-        "backward of corr_lookup_fwd (pyramid grads)",
+        "backward of corr_lookup_fwd (pyramid grads; accumulates into "
+        "`prev` when the pyramid is iteration-chained)",
         py::arg("gout"), py::arg("coords"), py::arg("radius"),
         py::arg("level_shapes"), py::arg("channels_last"),
-        py::arg("levels_bf16") = false);
+        py::arg("levels_bf16") = false,
+        py::arg("prev") = py::none());
   m.def("corr_pyramid_fits", &corr_pyramid_fits,
         "does the fused pyramid-build kernel support this map size/dtype");
   m.def("packernel_fwd", &packernel_fwd, "PAC gaussian adapting kernel");

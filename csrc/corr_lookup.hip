@@ -130,8 +130,10 @@ template <int R, typename scalar_t>
 __global__ __launch_bounds__(LK_THREADS) void corr_lookup_bwd_kernel(
     const scalar_t* __restrict__ gout,  // (B, L*K2, H, W)
     const float* __restrict__ coords,  // (B, 2, H, W)
-    scalar_t* __restrict__ glevel,     // (B*P, Hl, Wl), zero-initialized
-    int BP, int P, int Hl, int Wl, int l, int L, int cl) {
+    scalar_t* __restrict__ glevel,     // (B*P, Hl, Wl); zero-initialized
+                                       // (acc=0) or accumulated into (acc=1:
+                                       // the iteration-chained grad buffer)
+    int BP, int P, int Hl, int Wl, int l, int L, int cl, int acc) {
   constexpr int K = 2 * R + 1;
   constexpr int K2 = K * K;
 
@@ -186,7 +188,7 @@ __global__ __launch_bounds__(LK_THREADS) void corr_lookup_bwd_kernel(
       for (int u = 0; u <= K; ++u) {
         float v = wy0 * acur[u];
         if (j > 0) v += wy1 * aprev[u];
-        dst[u] = (scalar_t)v;
+        dst[u] = (scalar_t)(acc ? (float)dst[u] + v : v);
       }
 #pragma unroll
       for (int u = 0; u <= K; ++u) aprev[u] = acur[u];
@@ -217,10 +219,11 @@ __global__ __launch_bounds__(LK_THREADS) void corr_lookup_bwd_kernel(
           float v = 0.0f;
           if (j < K) v += wy0 * tx[j];
           if (j > 0) v += wy1 * tx[j - 1];
-          // each patch position of this thread's PRIVATE map is written
-          // exactly once per call (buffer is fresh-zeroed): plain store,
-          // no read-modify-write
-          gmap[(long)yy * Wl + xx] = (scalar_t)v;
+          // each patch position of this thread's PRIVATE map is touched
+          // exactly once per call: plain store on a fresh buffer, one RMW
+          // on the iteration-chained buffer
+          scalar_t* d = gmap + (long)yy * Wl + xx;
+          *d = (scalar_t)(acc ? (float)*d + v : v);
         }
       }
     }
@@ -243,11 +246,12 @@ static void lookup_fwd_level(const scalar_t* level, const float* coords,
 template <int R, typename scalar_t>
 static void lookup_bwd_level(const scalar_t* gout, const float* coords,
                              scalar_t* glevel, int BP, int P, int Hl, int Wl,
-                             int l, int L, int cl, hipStream_t stream) {
+                             int l, int L, int cl, int acc,
+                             hipStream_t stream) {
   dim3 grid(fh_cdiv(BP, LK_THREADS));
   hipLaunchKernelGGL((corr_lookup_bwd_kernel<R, scalar_t>), grid,
                      dim3(LK_THREADS), 0, stream, gout, coords, glevel, BP, P,
-                     Hl, Wl, l, L, cl);
+                     Hl, Wl, l, L, cl, acc);
 }
 
 // levels may be fp32 (reference parity) or bf16 (HBM-resident bf16
@@ -276,17 +280,18 @@ void flowhip_corr_lookup_fwd_launch(const void* level, const float* coords,
 void flowhip_corr_lookup_bwd_launch(const void* gout, const float* coords,
                                     void* glevel, int BP, int P, int Hl,
                                     int Wl, int l, int L, int radius, int cl,
-                                    int is_bf16, hipStream_t stream) {
+                                    int is_bf16, int acc,
+                                    hipStream_t stream) {
   if (is_bf16) {
     switch (radius) {
-      case 3: lookup_bwd_level<3>((const __bf16*)gout, coords, (__bf16*)glevel, BP, P, Hl, Wl, l, L, cl, stream); break;
-      case 4: lookup_bwd_level<4>((const __bf16*)gout, coords, (__bf16*)glevel, BP, P, Hl, Wl, l, L, cl, stream); break;
+      case 3: lookup_bwd_level<3>((const __bf16*)gout, coords, (__bf16*)glevel, BP, P, Hl, Wl, l, L, cl, acc, stream); break;
+      case 4: lookup_bwd_level<4>((const __bf16*)gout, coords, (__bf16*)glevel, BP, P, Hl, Wl, l, L, cl, acc, stream); break;
       default: abort();
     }
   } else {
     switch (radius) {
-      case 3: lookup_bwd_level<3>((const float*)gout, coords, (float*)glevel, BP, P, Hl, Wl, l, L, cl, stream); break;
-      case 4: lookup_bwd_level<4>((const float*)gout, coords, (float*)glevel, BP, P, Hl, Wl, l, L, cl, stream); break;
+      case 3: lookup_bwd_level<3>((const float*)gout, coords, (float*)glevel, BP, P, Hl, Wl, l, L, cl, acc, stream); break;
+      case 4: lookup_bwd_level<4>((const float*)gout, coords, (float*)glevel, BP, P, Hl, Wl, l, L, cl, acc, stream); break;
       default: abort();
     }
   }

@@ -19,4 +19,6 @@ class CorrBlock:
         self.corr_pyramid = ops.corr_pyramid(corr, num_levels)
 
     def __call__(self, coords):
-        return ops.corr_lookup(self.corr_pyramid, coords, self.radius)
+        out, self.corr_pyramid = ops.corr_lookup_chained(
+            self.corr_pyramid, coords, self.radius)
+        return out

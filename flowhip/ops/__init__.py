@@ -69,8 +69,20 @@ def corr_lookup(pyramid, coords, radius):
     # HIP kernel instantiates the RAFT radii (3 small / 4 basic)
     if radius in (3, 4) and _ext.use_hip(coords):
         from .functional import CorrLookupFn
-        return CorrLookupFn.apply(coords, radius, *pyramid)
+        return CorrLookupFn.apply(coords, radius, *pyramid)[0]
     return torch_ref.corr_lookup(pyramid, coords, radius)
+
+
+def corr_lookup_chained(pyramid, coords, radius):
+    """Lookup + pyramid passthrough for the iteration loop (see
+    CorrLookupFn): returns (out, new_pyramid) where new_pyramid must be
+    fed to the NEXT iteration's lookup so level grads accumulate along
+    the chain."""
+    if radius in (3, 4) and _ext.use_hip(coords):
+        from .functional import CorrLookupFn
+        res = CorrLookupFn.apply(coords, radius, *pyramid)
+        return res[0], list(res[1:])
+    return torch_ref.corr_lookup(pyramid, coords, radius), pyramid
 
 
 def nconv2d(data, conf, weight, bias=None, stride=1, padding=0, dilation=1,

@@ -109,6 +109,15 @@ class CorrLookupFn(torch.autograd.Function):
     coords never require grad in the RAFT iteration loop (coords1 is detached
     right before every lookup — raft.py:122, raft_nc_dbl.py:149), so backward
     produces gradients for the pyramid levels only and asserts that contract.
+
+    Returns (out, *levels): the pyramid is passed THROUGH so CorrBlock can
+    thread it along the iteration loop — the autograd graph then forms a
+    CHAIN over the 12 lookups instead of a 12-way fan-out, and each
+    backward ACCUMULATES its patch contribution into the incoming level
+    grads in-kernel (no per-iteration zero-fill, no autograd add_ fan-in).
+    Callers that drop the passthrough (tests, single lookups) still get
+    correct fan-out semantics: missing chain grads select the fresh-buffer
+    path.
     """
 
     @staticmethod
@@ -126,10 +135,10 @@ class CorrLookupFn(torch.autograd.Function):
         ctx.cl = cl
         ctx.level_shapes = [tuple(p.shape) for p in pyramid]
         ctx.levels_bf16 = pyramid[0].dtype == torch.bfloat16
-        return out
+        return (out, *pyramid)
 
     @staticmethod
-    def backward(ctx, grad):
+    def backward(ctx, grad, *glevels):
         (coords,) = ctx.saved_tensors
         dt = torch.bfloat16 if ctx.levels_bf16 else torch.float32
         if grad.dtype != dt:
@@ -138,9 +147,16 @@ class CorrLookupFn(torch.autograd.Function):
             grad = grad.contiguous(memory_format=torch.channels_last)
         else:
             grad = grad.contiguous()
+        have = [g is not None for g in glevels]
+        prev = None
+        if all(have):
+            prev = [g.contiguous() for g in glevels]
+        else:
+            assert not any(have), "corr_lookup chain: partial level grads"
         grads = _ext.ext().corr_lookup_bwd(
             grad, coords, ctx.radius,
-            [list(s) for s in ctx.level_shapes], ctx.cl, ctx.levels_bf16)
+            [list(s) for s in ctx.level_shapes], ctx.cl, ctx.levels_bf16,
+            prev)
         return (None, None, *grads)
 
 

@@ -109,7 +109,7 @@ class TestCorrLookup:
         coords = (torch.rand(B, 2, H, W, device=_dev()) *
                   torch.tensor([W, H], device=_dev()).view(1, 2, 1, 1))
 
-        out = CorrLookupFn.apply(coords, radius, *pyramid)
+        out = CorrLookupFn.apply(coords, radius, *pyramid)[0]
         ref = torch_ref.corr_lookup(pyramid, coords, radius)
         assert out.shape == ref.shape
         torch.testing.assert_close(out, ref, atol=1e-4, rtol=1e-4)
@@ -126,7 +126,7 @@ class TestCorrLookup:
         coords = (torch.rand(B, 2, H, W, device=_dev()) *
                   torch.tensor([W, H], device=_dev()).view(1, 2, 1, 1))
 
-        out = CorrLookupFn.apply(coords, radius, *pyramid)
+        out = CorrLookupFn.apply(coords, radius, *pyramid)[0]
         g = torch.randn_like(out)
         (dl0,) = torch.autograd.grad(out, l0, g)
 
@@ -185,7 +185,7 @@ class TestCorrBf16Resident:
         coords = (torch.rand(B, 2, H, W, device=_dev()) *
                   torch.tensor([W, H], device=_dev()).view(1, 2, 1, 1))
         out = CorrLookupFn.apply(coords, 4, *[p.detach().requires_grad_(True)
-                                              for p in pyr])
+                                              for p in pyr])[0]
         # lookup output dtype follows the pyramid residency dtype
         assert out.dtype == torch.bfloat16
         ref = torch_ref.corr_lookup([r for r in ref_pyr], coords, 4)
@@ -202,7 +202,7 @@ class TestCorrBf16Resident:
         pyr_b = [p.to(torch.bfloat16).requires_grad_(True) for p in pyr_f]
         coords = (torch.rand(B, 2, H, W, device=_dev()) *
                   torch.tensor([W, H], device=_dev()).view(1, 2, 1, 1))
-        out = CorrLookupFn.apply(coords, radius, *pyr_b)
+        out = CorrLookupFn.apply(coords, radius, *pyr_b)[0]
         g = torch.randn_like(out)
         grads = torch.autograd.grad(out, pyr_b, g)
         pyr_r = [p.detach().float().requires_grad_(True) for p in pyr_b]
