@@ -140,6 +140,8 @@ bool flowhip_col_sum2_launch(const void* g, const void* x, float* partials,
 void flowhip_frozen_bn_apply_launch(const void* x, void* y, const float* s,
                                     const float* t, long M, int C,
                                     hipStream_t stream);
+void flowhip_plane_dot_sum_launch(const float* x, const float* y, float* out,
+                                  long P, int B, int C, hipStream_t stream);
 bool flowhip_col_sum_launch(const void* dy, float* partials, float* out,
                             long M, int C, int nchunk, hipStream_t stream);
 void flowhip_up2x_cat_fwd_launch(const float* low, const float* skip,
@@ -986,6 +988,25 @@ torch::Tensor col_sum_bf16(torch::Tensor dy) {
   return out;
 }
 
+torch::Tensor plane_sum_nchw(torch::Tensor x, c10::optional<torch::Tensor> y) {
+  TORCH_CHECK(x.is_cuda() && x.dim() == 4 && x.is_contiguous() &&
+              x.scalar_type() == torch::kFloat32);
+  const float* yp = nullptr;
+  if (y.has_value()) {
+    TORCH_CHECK(y->sizes() == x.sizes() && y->is_contiguous() &&
+                y->scalar_type() == torch::kFloat32);
+    yp = y->data_ptr<float>();
+  }
+  const int B = x.size(0), C = x.size(1);
+  const long P = (long)x.size(2) * x.size(3);
+  auto out = torch::zeros({(long)C}, x.options());
+  const c10::cuda::CUDAGuard guard(x.device());
+  hipStream_t stream = at::cuda::getCurrentCUDAStream().stream();
+  flowhip_plane_dot_sum_launch(x.data_ptr<float>(), yp,
+                               out.data_ptr<float>(), P, B, C, stream);
+  return out;
+}
+
 torch::Tensor frozen_bn_apply(torch::Tensor x, torch::Tensor s,
                               c10::optional<torch::Tensor> t) {
   TORCH_CHECK(x.is_cuda() && x.dim() == 4 &&
@@ -1190,6 +1211,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("convex_up_bwd", &convex_up_bwd, "backward of convex_up_fwd");
   m.def("nconv_fwd", &nconv_fwd,
         "fused normalized convolution forward (out, cout)");
+  m.def("plane_sum_nchw", &plane_sum_nchw,
+        "out[c] = sum_{b,h,w} x (or x*y), NCHW fp32",
+        py::arg("x"), py::arg("y") = py::none());
   m.def("frozen_bn_apply", &frozen_bn_apply,
         "y = bf16(fp32(x)*s + t) per channel, channels-last");
   m.def("col_sum2_bf16", &col_sum2_bf16,

@@ -171,6 +171,63 @@ __global__ __launch_bounds__(512) void col_sum2_finalize_kernel(
   }
 }
 
+// Per-channel plane reduction on NCHW fp32: out[c] = sum_{b,h,w} x (or
+// x*y). Replaces the eager mul+sum pair in the nconv backward's
+// ds-through-sum(w) term (full-res (B,2,H,W) tensors, ~48 pairs/step).
+__global__ __launch_bounds__(256) void plane_dot_sum_kernel(
+    const float* __restrict__ x, const float* __restrict__ y,
+    float* __restrict__ out, long P, int C, int pchunks) {
+  const int bc = blockIdx.x;
+  const int c = bc % C;
+  const int chunk = blockIdx.y;
+  // chunk bounds rounded to float4 granularity (tail in the last chunk)
+  long p0 = ((P * (long)chunk) / pchunks) & ~3L;
+  long p1 = (chunk + 1 == pchunks) ? P
+                                   : ((P * (long)(chunk + 1)) / pchunks) & ~3L;
+  const float* xp = x + (long)bc * P;
+  const float* yp = y ? y + (long)bc * P : nullptr;
+  float acc = 0.f;
+  if ((P & 3) == 0) {  // 16-B aligned planes: float4 stream
+    const long p1v = p1 & ~3L;
+    for (long p = p0 + threadIdx.x * 4; p < p1v; p += 256 * 4) {
+      const float4 vx = *(const float4*)(xp + p);
+      if (yp) {
+        const float4 vy = *(const float4*)(yp + p);
+        acc += vx.x * vy.x + vx.y * vy.y + vx.z * vy.z + vx.w * vy.w;
+      } else {
+        acc += vx.x + vx.y + vx.z + vx.w;
+      }
+    }
+    const long t = p1v + threadIdx.x;
+    if (t < p1) acc += yp ? xp[t] * yp[t] : xp[t];
+  } else {
+    for (long p = p0 + threadIdx.x; p < p1; p += 256)
+      acc += yp ? xp[p] * yp[p] : xp[p];
+  }
+
+  __shared__ float red[256];
+  red[threadIdx.x] = acc;
+  __syncthreads();
+#pragma unroll
+  for (int s = 128; s > 0; s >>= 1) {
+    if ((int)threadIdx.x < s) red[threadIdx.x] += red[threadIdx.x + s];
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) atomicAdd(&out[c], red[0]);
+}
+
+void flowhip_plane_dot_sum_launch(const float* x, const float* y, float* out,
+                                  long P, int B, int C, hipStream_t stream) {
+  int pchunks = (int)((512 + (long)B * C - 1) / ((long)B * C));
+  if (pchunks < 1) pchunks = 1;
+  if (pchunks > 1024) pchunks = 1024;
+  if ((long)pchunks > (P + 1023) / 1024) pchunks = (int)((P + 1023) / 1024);
+  if (pchunks < 1) pchunks = 1;
+  dim3 grid(B * C, pchunks), block(256);
+  hipLaunchKernelGGL(plane_dot_sum_kernel, grid, block, 0, stream, x, y, out,
+                     P, C, pchunks);
+}
+
 // Frozen-BN affine: y[m,c] = bf16(fp32(x[m,c]) * s[c] + t[c]) on
 // channels-last rows. fp32 math + bf16 store puts the quantization point
 // exactly where the stock autocast path (fp32 BN output, bf16 cast at the

@@ -20,11 +20,12 @@ from .update import FusedConv2d
 
 
 def _enc_conv(*a, **k):
-    """Encoder convs run on MIOpen by default (measured: 34.3 vs 32.4
-    pairs/s with the 64x64-tile fused kernel — at the encoder's huge-M
-    shapes the library's 128x128/256x128 tiles amortize the B-operand far
-    better). FLOWHIP_FUSED_ENCODER=1 routes them through the MFMA kernel."""
-    if os.environ.get("FLOWHIP_FUSED_ENCODER", "0") == "1":
+    """Encoder convs run on the hand-written MFMA kernel by DEFAULT since
+    round 2: the BM=128 8-wave tile + tr_b16 wrw beat MIOpen in the
+    same-box A/B (34.69 vs 34.38 pairs/s at flip time, wider after the
+    later eager-tax work). FLOWHIP_FUSED_ENCODER=0 restores the library
+    path (A/B runs)."""
+    if os.environ.get("FLOWHIP_FUSED_ENCODER", "1") == "1":
         return FusedConv2d(*a, **k)
     return nn.Conv2d(*a, **k)
 

@@ -64,11 +64,16 @@ class NConv2dFn(torch.autograd.Function):
         if not ctx.needs_input_grad[2]:
             dweight = None
         elif gcout is not None:
-            # cout = denom/s depends on s = sum(w) per out channel
-            ds = -(cout * gcout).sum(dim=(0, 2, 3)) / s
+            # cout = denom/s depends on s = sum(w) per out channel; one
+            # fused per-channel plane reduction instead of the full-res
+            # mul + sum pair (~48 pairs/step at (B*2, Co, H, W))
+            ds = -_ext.ext().plane_sum_nchw(cout, gcout) / s
             dweight = dweight + ds.view(-1, 1, 1, 1)
 
-        dbias = gout.sum(dim=(0, 2, 3)) if ctx.has_bias else None
+        if ctx.has_bias:
+            dbias = _ext.ext().plane_sum_nchw(gout.contiguous())
+        else:
+            dbias = None
         return ddata, dconf, dweight, dbias, None, None, None
 
 

@@ -558,6 +558,19 @@ class TestFrozenBatchNorm:
             torch.testing.assert_close(out[0], ref0, atol=2e-1, rtol=1e-2)
             torch.testing.assert_close(out[1], ref1, atol=5e-1, rtol=2e-2)
 
+    def test_plane_sum_nchw(self):
+        import flowhip._C as C
+        torch.manual_seed(63)
+        for (b, c, h, w) in [(4, 2, 30, 44), (2, 3, 23, 31), (1, 8, 64, 64)]:
+            x = torch.randn(b, c, h, w, device=_dev())
+            y = torch.randn(b, c, h, w, device=_dev())
+            torch.testing.assert_close(C.plane_sum_nchw(x),
+                                       x.sum(dim=(0, 2, 3)),
+                                       atol=1e-3, rtol=1e-4)
+            torch.testing.assert_close(C.plane_sum_nchw(x, y),
+                                       (x * y).sum(dim=(0, 2, 3)),
+                                       atol=1e-3, rtol=1e-4)
+
     def test_frozen_bn_matches_stock_eval_bn(self):
         from flowhip.nn.norm import BatchNorm2d
         torch.manual_seed(62)
