@@ -1,56 +1,80 @@
-// LDS-tiled normalized-convolution kernels, v2 (kernels #6/#7 of
+// LDS-tiled normalized-convolution kernels, v3 (kernels #6/#7 of
 // SURVEY.md §2.2; math contract = reference nconv_modules.py:164-199).
 //
-// Replaces the v1 global-gather kernels in nconv.hip for K in {3,5}: the
-// v1 kernels re-read every input element K*K times through L1 and (wrw)
-// funnel 200k+ atomics onto ~100 words; rocprofv3 showed nconv_wrw alone
-// at 31.7% of a training step (profiles/r01_trace_step.md). v2 stages the
-// (conf, data*conf) halo tile through LDS once per workgroup and reduces
-// the weight gradient hierarchically: registers -> wave shuffle -> LDS ->
-// one partial row per workgroup -> tiny second-stage reduce kernel. No
-// atomics anywhere, fully deterministic.
+// v2 (round 1) staged the (conf, data*conf) halo tile through LDS once per
+// workgroup with an atomic-free hierarchical weight-grad reduction, but
+// moved every global byte as a SCALAR 4-B-per-lane access and read K*K LDS
+// taps per output pixel — profiles put the forward ~2.5x above its HBM
+// bound (headroom #5). v3 keeps the structure and:
+//   - stages through ALIGNED float4 chunks (16 B/lane; covers the halo by
+//     loading [x0-4, x0+TW+4) and scatter-clipping into the LDS tile);
+//   - each thread owns 4 ADJACENT output pixels: the K-tap windows of the
+//     4 pixels overlap, so one (K+3)-element LDS row read feeds all four
+//     (2.5x fewer LDS cycles), and every output store is one float4.
+// Requires W % 4 == 0 (all BASELINE/NCUP shapes; others fall back to the
+// v1 global-gather kernels in nconv.hip).
 //
-// Tile geometry: TW=64 (one wave covers a row), TH=16 output rows per
-// workgroup of 256 threads (each thread owns 4 pixels, stride-4 rows).
-// LDS per plane pair = (TH+K-1)*(TW+K-1)*2 floats; CI<=2 K=5 ~22 KB,
-// CI=4 K=3 ~38 KB -> 4+ workgroups/CU.
+// Tile geometry: TW=64, TH=16 output rows per 256-thread workgroup
+// (fwd/bwd; wrw uses TH=32 in two passes). 16 lanes x 4 px cover a row.
 
 #include "common.h"
 
 #define NCT_THREADS 256
 #define NCT_TW 64
 #define NCT_TH 16
+#define NCT_TH_WRW 32
 
 // ---------------------------------------------------------------------------
-// Shared tile loader: stages conf and data*conf (with K/2 halo, zero-padded)
-// for all CI channels into LDS.  LW = TW+K-1 row stride.
+// Vectorized halo stage: loads aligned float4 chunks of planes (a, b) and
+// stores (a, MUL ? b*a : b) into the LDS tile pair. LW = TW+K-1 row
+// stride; chunks cover [x0-4, x0+TW+4) so every halo column lands.
 // ---------------------------------------------------------------------------
-template <int K, int CI, int TH = NCT_TH>
-__device__ inline void nct_stage_tile(const float* __restrict__ data,
-                                      const float* __restrict__ conf,
-                                      float* __restrict__ lds_c,
-                                      float* __restrict__ lds_dc,
-                                      int n, int x0, int y0, int Ci_stride_n,
-                                      int H, int W) {
+template <int K, int CH, int TH, bool MUL>
+__device__ inline void nct_stage_v(const float* __restrict__ a,
+                                   const float* __restrict__ b,
+                                   float* __restrict__ lds_a,
+                                   float* __restrict__ lds_b,
+                                   int n, int x0, int y0, int H, int W) {
   constexpr int LW = NCT_TW + K - 1;
   constexpr int LH = TH + K - 1;
+  constexpr int CHUNKS = (NCT_TW + 8) / 4;  // 18 aligned float4 per row
   const long plane = (long)H * W;
-  for (int idx = threadIdx.x; idx < LH * LW; idx += NCT_THREADS) {
-    const int row = idx / LW, col = idx - row * LW;
+  for (int u = threadIdx.x; u < LH * CHUNKS; u += NCT_THREADS) {
+    const int row = u / CHUNKS, k = u - row * CHUNKS;
     const int gy = y0 - K / 2 + row;
-    const int gx = x0 - K / 2 + col;
-    const bool in = (gy >= 0 && gy < H && gx >= 0 && gx < W);
-    const long goff = (long)gy * W + gx;
+    const int gx = x0 - 4 + k * 4;
+    const bool rin = (gy >= 0 && gy < H);
+    const int lc0 = gx - (x0 - K / 2);
 #pragma unroll
-    for (int ci = 0; ci < CI; ++ci) {
-      const long base = ((long)n * Ci_stride_n + ci) * plane;
-      float c = 0.f, d = 0.f;
-      if (in) {
-        c = conf[base + goff];
-        d = data[base + goff];
+    for (int ch = 0; ch < CH; ++ch) {
+      const long base = ((long)n * CH + ch) * plane + (long)gy * W;
+      float4 va = {0.f, 0.f, 0.f, 0.f}, vb = {0.f, 0.f, 0.f, 0.f};
+      if (rin) {
+        if (gx >= 0 && gx + 3 < W) {
+          va = *(const float4*)(a + base + gx);
+          vb = *(const float4*)(b + base + gx);
+        } else {
+#pragma unroll
+          for (int e = 0; e < 4; ++e) {
+            const int xx = gx + e;
+            if (xx >= 0 && xx < W) {
+              (&va.x)[e] = a[base + xx];
+              (&vb.x)[e] = b[base + xx];
+            }
+          }
+        }
       }
-      lds_c[ci * LH * LW + idx] = c;
-      lds_dc[ci * LH * LW + idx] = d * c;
+      float* la = lds_a + ch * LH * LW + row * LW;
+      float* lb = lds_b + ch * LH * LW + row * LW;
+#pragma unroll
+      for (int e = 0; e < 4; ++e) {
+        const int lcol = lc0 + e;
+        if (lcol >= 0 && lcol < LW) {
+          const float av = (&va.x)[e];
+          la[lcol] = av;
+          lb[lcol] = MUL ? (&vb.x)[e] * av : (&vb.x)[e];
+        }
+      }
     }
   }
 }
@@ -58,7 +82,7 @@ __device__ inline void nct_stage_tile(const float* __restrict__ data,
 // ---------------------------------------------------------------------------
 // Forward:  out = conv(dc, w) / (conv(c, w) + 1e-20) [+bias]
 //           cout = conv(c, w) / sum_w   (confidence propagation)
-// Grid: (ntx*nty*N); each thread computes NCT_TH/4 pixels for ALL Co.
+// Grid: (ntx*nty*N); thread = 4 adjacent pixels of one row, all Co.
 // ---------------------------------------------------------------------------
 template <int K, int CI, int CO>
 __global__ __launch_bounds__(NCT_THREADS) void nconv_fwd_tiled_kernel(
@@ -87,51 +111,64 @@ __global__ __launch_bounds__(NCT_THREADS) void nconv_fwd_tiled_kernel(
     for (int i = 0; i < CI * K * K; ++i) s += wsh[threadIdx.x * CI * K * K + i];
     winv[threadIdx.x] = 1.0f / s;
   }
-  nct_stage_tile<K, CI>(data, conf, lds_c, lds_dc, n, x0, y0, CI, H, W);
+  nct_stage_v<K, CI, NCT_TH, true>(conf, data, lds_c, lds_dc, n, x0, y0, H,
+                                   W);
   __syncthreads();
 
   const long plane = (long)H * W;
-  const int lx = threadIdx.x & 63;          // 0..63 within tile row
-  const int ly0 = threadIdx.x >> 6;         // 0..3
-  const int x = x0 + lx;
+  const int lxb = (threadIdx.x & 15) * 4;   // tile-local x of pixel 0
+  const int row = threadIdx.x >> 4;         // 0..15
+  const int x = x0 + lxb;
+  const int y = y0 + row;
+  if (x >= W || y >= H) return;  // W%4==0: the 4-chunk is all-in or all-out
 
+  float denom[CO][4], nomin[CO][4];
+#pragma unroll
+  for (int co = 0; co < CO; ++co)
+#pragma unroll
+    for (int e = 0; e < 4; ++e) { denom[co][e] = 0.f; nomin[co][e] = 0.f; }
+
+#pragma unroll
+  for (int ci = 0; ci < CI; ++ci) {
+    const float* lc = lds_c + ci * LH * LW;
+    const float* ldc = lds_dc + ci * LH * LW;
 #pragma unroll 1
-  for (int j = 0; j < NCT_TH / 4; ++j) {
-    const int lyy = ly0 + 4 * j;
-    const int y = y0 + lyy;
-    if (x >= W || y >= H) continue;
-    float denom[CO], nomin[CO];
+    for (int ky = 0; ky < K; ++ky) {
+      // one (K+3)-wide row read feeds all 4 pixels' kx windows
+      float cbuf[K + 3], dcbuf[K + 3];
+      const float* rc = lc + (row + ky) * LW + lxb;
+      const float* rdc = ldc + (row + ky) * LW + lxb;
 #pragma unroll
-    for (int co = 0; co < CO; ++co) { denom[co] = 0.f; nomin[co] = 0.f; }
+      for (int tap = 0; tap < K + 3; ++tap) {
+        cbuf[tap] = rc[tap];
+        dcbuf[tap] = rdc[tap];
+      }
 #pragma unroll
-    for (int ci = 0; ci < CI; ++ci) {
-      const float* lc = lds_c + ci * LH * LW;
-      const float* ldc = lds_dc + ci * LH * LW;
-      // bounded unroll (K*CO*2 fma + K*2 LDS loads in flight per row):
-      // full K*K unrolling costs 256 VGPRs -> 1 wave/SIMD
-#pragma unroll 1
-      for (int ky = 0; ky < K; ++ky) {
+      for (int kx = 0; kx < K; ++kx) {
 #pragma unroll
-        for (int kx = 0; kx < K; ++kx) {
-          const float c = lc[(lyy + ky) * LW + lx + kx];
-          const float dc = ldc[(lyy + ky) * LW + lx + kx];
+        for (int co = 0; co < CO; ++co) {
+          const float w = wsh[((co * CI + ci) * K + ky) * K + kx];
 #pragma unroll
-          for (int co = 0; co < CO; ++co) {
-            const float w = wsh[((co * CI + ci) * K + ky) * K + kx];
-            denom[co] += w * c;
-            nomin[co] += w * dc;
+          for (int e = 0; e < 4; ++e) {
+            denom[co][e] += w * cbuf[kx + e];
+            nomin[co][e] += w * dcbuf[kx + e];
           }
         }
       }
     }
+  }
 #pragma unroll
-    for (int co = 0; co < CO; ++co) {
-      float v = nomin[co] / (denom[co] + 1e-20f);
-      if (bias != nullptr) v += bias[co];
-      const long o = ((long)n * CO + co) * plane + (long)y * W + x;
-      out[o] = v;
-      cout[o] = denom[co] * winv[co];
+  for (int co = 0; co < CO; ++co) {
+    const float b = bias != nullptr ? bias[co] : 0.f;
+    float4 vo, vc;
+#pragma unroll
+    for (int e = 0; e < 4; ++e) {
+      (&vo.x)[e] = nomin[co][e] / (denom[co][e] + 1e-20f) + b;
+      (&vc.x)[e] = denom[co][e] * winv[co];
     }
+    const long o = ((long)n * CO + co) * plane + (long)y * W + x;
+    *(float4*)(out + o) = vo;
+    *(float4*)(cout + o) = vc;
   }
 }
 
@@ -139,7 +176,6 @@ __global__ __launch_bounds__(NCT_THREADS) void nconv_fwd_tiled_kernel(
 // Backward data: transposed-conv gather of (dnomin, ddenom) staged in LDS.
 //   g  = convT(dnomin, w[.,ci]);  gd = convT(ddenom, w[.,ci])
 //   ddata = conf * g ;  dconf = data * g + gd
-// LDS planes here are the Co gradient channels (gn, gd).
 // ---------------------------------------------------------------------------
 template <int K, int CO>
 __global__ __launch_bounds__(NCT_THREADS) void nconv_bwd_data_tiled_kernel(
@@ -152,7 +188,7 @@ __global__ __launch_bounds__(NCT_THREADS) void nconv_bwd_data_tiled_kernel(
   constexpr int LH = NCT_TH + K - 1;
   __shared__ float lds_gn[CO * LH * LW];
   __shared__ float lds_gd[CO * LH * LW];
-  __shared__ float wsh[8 * CO * K * K];  // [co][ci][ky][kx] layout below
+  __shared__ float wsh[8 * CO * K * K];  // [co][ci][ky][kx]
 
   int t = blockIdx.x;
   const int tx = t % ntx; t /= ntx;
@@ -163,77 +199,68 @@ __global__ __launch_bounds__(NCT_THREADS) void nconv_bwd_data_tiled_kernel(
   const int nw = CO * Ci * K * K;
   for (int i = threadIdx.x; i < nw; i += NCT_THREADS) wsh[i] = weight[i];
   __syncthreads();
-
-  // stage gn, gd (note: "data=dnomin, conf=ddenom" pairing abuse of the
-  // generic loader would multiply them; stage directly instead)
-  {
-    const long plane = (long)H * W;
-    for (int idx = threadIdx.x; idx < LH * LW; idx += NCT_THREADS) {
-      const int row = idx / LW, col = idx - row * LW;
-      const int gy = y0 - K / 2 + row;
-      const int gx = x0 - K / 2 + col;
-      const bool in = (gy >= 0 && gy < H && gx >= 0 && gx < W);
-      const long goff = (long)gy * W + gx;
-#pragma unroll
-      for (int co = 0; co < CO; ++co) {
-        const long base = ((long)n * CO + co) * plane;
-        lds_gn[co * LH * LW + idx] = in ? dnomin[base + goff] : 0.f;
-        lds_gd[co * LH * LW + idx] = in ? ddenom[base + goff] : 0.f;
-      }
-    }
-  }
+  nct_stage_v<K, CO, NCT_TH, false>(dnomin, ddenom, lds_gn, lds_gd, n, x0,
+                                    y0, H, W);
   __syncthreads();
 
   const long plane = (long)H * W;
-  const int lx = threadIdx.x & 63;
-  const int ly0 = threadIdx.x >> 6;
-  const int x = x0 + lx;
+  const int lxb = (threadIdx.x & 15) * 4;
+  const int row = threadIdx.x >> 4;
+  const int x = x0 + lxb;
+  const int y = y0 + row;
+  if (x >= W || y >= H) return;
 
+  for (int ci = 0; ci < Ci; ++ci) {
+    float g[4] = {0.f, 0.f, 0.f, 0.f}, gd[4] = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
-  for (int j = 0; j < NCT_TH / 4; ++j) {
-    const int lyy = ly0 + 4 * j;
-    const int y = y0 + lyy;
-    if (x >= W || y >= H) continue;
-    for (int ci = 0; ci < Ci; ++ci) {
-      float g = 0.f, gd = 0.f;
-#pragma unroll
-      for (int co = 0; co < CO; ++co) {
-        const float* lgn = lds_gn + co * LH * LW;
-        const float* lgd = lds_gd + co * LH * LW;
-        // bounded unroll: full K*K*CO unrolling put 50+ LDS loads in
-        // flight and cost 256 VGPRs (1 wave/SIMD); K kx-taps of ILP is
-        // plenty for ~50-cycle LDS latency at 4+ waves
+    for (int co = 0; co < CO; ++co) {
+      const float* lgn = lds_gn + co * LH * LW;
+      const float* lgd = lds_gd + co * LH * LW;
 #pragma unroll 1
-        for (int ky = 0; ky < K; ++ky) {
+      for (int ky = 0; ky < K; ++ky) {
+        // transposed conv: out[x+e] gathers in[x+e + K-1-kx - (K-1)/2...]
+        // in LDS coords: row = row + K-1-ky, cols = lxb + e + K-1-kx
+        float nbuf[K + 3], dbuf[K + 3];
+        const float* rn = lgn + (row + K - 1 - ky) * LW + lxb;
+        const float* rd = lgd + (row + K - 1 - ky) * LW + lxb;
 #pragma unroll
-          for (int kx = 0; kx < K; ++kx) {
-            // transposed conv: out[y] gathers in[y - ky + K/2]; in LDS
-            // coords (center at +K/2): row = lyy + K-1-ky, col = lx + K-1-kx
-            const float w = wsh[((co * Ci + ci) * K + ky) * K + kx];
-            g += w * lgn[(lyy + K - 1 - ky) * LW + lx + K - 1 - kx];
-            gd += w * lgd[(lyy + K - 1 - ky) * LW + lx + K - 1 - kx];
+        for (int tap = 0; tap < K + 3; ++tap) {
+          nbuf[tap] = rn[tap];
+          dbuf[tap] = rd[tap];
+        }
+#pragma unroll
+        for (int kx = 0; kx < K; ++kx) {
+          const float w = wsh[((co * Ci + ci) * K + ky) * K + kx];
+#pragma unroll
+          for (int e = 0; e < 4; ++e) {
+            g[e] += w * nbuf[K - 1 - kx + e];
+            gd[e] += w * dbuf[K - 1 - kx + e];
           }
         }
       }
-      const long p = ((long)n * Ci + ci) * plane + (long)y * W + x;
-      const float c = conf[p];
-      const float d = data[p];
-      ddata[p] = c * g;
-      dconf[p] = d * g + gd;
     }
+    const long p = ((long)n * Ci + ci) * plane + (long)y * W + x;
+    const float4 c4 = *(const float4*)(conf + p);
+    const float4 d4 = *(const float4*)(data + p);
+    float4 vd, vc;
+#pragma unroll
+    for (int e = 0; e < 4; ++e) {
+      (&vd.x)[e] = (&c4.x)[e] * g[e];
+      (&vc.x)[e] = (&d4.x)[e] * g[e] + gd[e];
+    }
+    *(float4*)(ddata + p) = vd;
+    *(float4*)(dconf + p) = vc;
   }
 }
 
 // ---------------------------------------------------------------------------
 // Weight gradient, stage 1: per-workgroup partial sums.
 //   dw[co,ci,ky,kx] = sum_p dnomin[co,p] * dc[ci,p+d] + ddenom[co,p] * c[ci,p+d]
-// Each thread accumulates the full (CI*K*K) slice for one co at a time in
-// registers over its 4 pixels (taps from LDS), then the workgroup reduces
-// (wave shuffle -> LDS across waves) and writes one row of partials.
+// Thread accumulates the (CI*K*K) slice for one co at a time over its 4
+// adjacent pixels x 2 row passes; hierarchical reduction (wave shuffle ->
+// LDS across waves) emits one partial row per workgroup.
 // partials layout: (nblocks, Co*CI*K*K).
 // ---------------------------------------------------------------------------
-#define NCT_TH_WRW 32
-
 template <int K, int CI>
 __global__ __launch_bounds__(NCT_THREADS) void nconv_wrw_tiled_kernel(
     const float* __restrict__ dnomin, const float* __restrict__ ddenom,
@@ -245,7 +272,7 @@ __global__ __launch_bounds__(NCT_THREADS) void nconv_wrw_tiled_kernel(
   constexpr int NW = CI * K * K;  // weights per co
   __shared__ float lds_c[CI * LH * LW];
   __shared__ float lds_dc[CI * LH * LW];
-  __shared__ float red[4 * NW];  // cross-wave reduction buffer
+  __shared__ float red[4 * NW];
 
   int t = blockIdx.x;
   const int tx = t % ntx; t /= ntx;
@@ -253,14 +280,14 @@ __global__ __launch_bounds__(NCT_THREADS) void nconv_wrw_tiled_kernel(
   const int n = t;
   const int x0 = tx * NCT_TW, y0 = ty * NCT_TH_WRW;
 
-  nct_stage_tile<K, CI, NCT_TH_WRW>(data, conf, lds_c, lds_dc, n, x0, y0,
-                                    CI, H, W);
+  nct_stage_v<K, CI, NCT_TH_WRW, true>(conf, data, lds_c, lds_dc, n, x0, y0,
+                                       H, W);
   __syncthreads();
 
   const long plane = (long)H * W;
-  const int lx = threadIdx.x & 63;
-  const int ly0 = threadIdx.x >> 6;
-  const int x = x0 + lx;
+  const int lxb = (threadIdx.x & 15) * 4;
+  const int row0 = threadIdx.x >> 4;  // 0..15; two row passes cover TH=32
+  const int x = x0 + lxb;
   const int wave = threadIdx.x >> 6;
   const int lane = threadIdx.x & 63;
 
@@ -270,24 +297,34 @@ __global__ __launch_bounds__(NCT_THREADS) void nconv_wrw_tiled_kernel(
     for (int i = 0; i < NW; ++i) acc[i] = 0.f;
 
 #pragma unroll
-    for (int j = 0; j < NCT_TH_WRW / 4; ++j) {
-      const int lyy = ly0 + 4 * j;
+    for (int j = 0; j < NCT_TH_WRW / 16; ++j) {
+      const int lyy = row0 + 16 * j;
       const int y = y0 + lyy;
       if (x >= W || y >= H) continue;
       const long go = ((long)n * Co + co) * plane + (long)y * W + x;
-      const float gn = dnomin[go];
-      const float gd = ddenom[go];
+      const float4 gn4 = *(const float4*)(dnomin + go);
+      const float4 gd4 = *(const float4*)(ddenom + go);
 #pragma unroll
       for (int ci = 0; ci < CI; ++ci) {
         const float* lc = lds_c + ci * LH * LW;
         const float* ldc = lds_dc + ci * LH * LW;
-#pragma unroll
+#pragma unroll 1
         for (int ky = 0; ky < K; ++ky) {
+          float cbuf[K + 3], dcbuf[K + 3];
+          const float* rc = lc + (lyy + ky) * LW + lxb;
+          const float* rdc = ldc + (lyy + ky) * LW + lxb;
+#pragma unroll
+          for (int tap = 0; tap < K + 3; ++tap) {
+            cbuf[tap] = rc[tap];
+            dcbuf[tap] = rdc[tap];
+          }
 #pragma unroll
           for (int kx = 0; kx < K; ++kx) {
-            const float c = lc[(lyy + ky) * LW + lx + kx];
-            const float dc = ldc[(lyy + ky) * LW + lx + kx];
-            acc[(ci * K + ky) * K + kx] += gn * dc + gd * c;
+            float v = 0.f;
+#pragma unroll
+            for (int e = 0; e < 4; ++e)
+              v += (&gn4.x)[e] * dcbuf[kx + e] + (&gd4.x)[e] * cbuf[kx + e];
+            acc[(ci * K + ky) * K + kx] += v;
           }
         }
       }
@@ -302,7 +339,6 @@ __global__ __launch_bounds__(NCT_THREADS) void nconv_wrw_tiled_kernel(
       if (lane == 0) red[wave * NW + i] = v;
     }
     __syncthreads();
-    // first NW threads fold the 4 wave rows and emit the partial row
     if (threadIdx.x < NW) {
       const float v = red[threadIdx.x] + red[NW + threadIdx.x] +
                       red[2 * NW + threadIdx.x] + red[3 * NW + threadIdx.x];
@@ -332,9 +368,10 @@ __global__ __launch_bounds__(NCT_THREADS) void nconv_wrw_reduce_kernel(
 }
 
 // ---------------------------------------------------------------------------
-// Launchers.  K in {3,5}; CI (fwd/wrw) and Co (bwd-data) dispatched over
-// {1,2,4,8}.  Returns false if the shape is outside the tiled space (caller
-// falls back to the v1 kernels).
+// Launchers.  K in {1,3,5}; CI (fwd/wrw) and Co (bwd-data) dispatched over
+// {1,2,4,8}.  Returns false if the shape is outside the tiled space — W
+// not a multiple of 4 (float4 rows) included — and the caller falls back
+// to the v1 kernels.
 // ---------------------------------------------------------------------------
 
 #define NCT_CASE_FWD(KK, CIV, COV)                                             \
@@ -350,6 +387,7 @@ bool flowhip_nconv_fwd_tiled_launch(const float* data, const float* conf,
                                     float* out, float* cout, int N, int Ci,
                                     int Co, int H, int W, int K,
                                     hipStream_t stream) {
+  if (W % 4 != 0) return false;
   const int ntx = fh_cdiv(W, NCT_TW), nty = fh_cdiv(H, NCT_TH);
   dim3 grid(ntx * nty * N), block(NCT_THREADS);
   NCT_CASE_FWD(5, 1, 1) NCT_CASE_FWD(5, 1, 2) NCT_CASE_FWD(5, 1, 4)
@@ -375,6 +413,7 @@ bool flowhip_nconv_bwd_data_tiled_launch(
     const float* dnomin, const float* ddenom, const float* data,
     const float* conf, const float* weight, float* ddata, float* dconf, int N,
     int Ci, int Co, int H, int W, int K, hipStream_t stream) {
+  if (W % 4 != 0) return false;
   const int ntx = fh_cdiv(W, NCT_TW), nty = fh_cdiv(H, NCT_TH);
   dim3 grid(ntx * nty * N), block(NCT_THREADS);
   NCT_CASE_BWD(5, 1) NCT_CASE_BWD(5, 2) NCT_CASE_BWD(5, 4)
@@ -400,6 +439,7 @@ bool flowhip_nconv_wrw_tiled_launch(const float* dnomin, const float* ddenom,
                                     float* partials, float* dweight, int N,
                                     int Ci, int Co, int H, int W, int K,
                                     hipStream_t stream) {
+  if (W % 4 != 0) return false;
   const int ntx = fh_cdiv(W, NCT_TW), nty = fh_cdiv(H, NCT_TH_WRW);
   const int nblocks = ntx * nty * N;
   const int nw = Co * Ci * K * K;
