@@ -41,7 +41,7 @@ void flowhip_nconv_wrw_launch(const float* dnomin, const float* ddenom,
                               const float* data, const float* conf,
                               float* dweight, int N, int Ci, int Co, int H,
                               int W, int K, hipStream_t stream);
-int flowhip_nconv_tiled_nblocks(int N, int H, int W);
+int flowhip_nconv_tiled_nblocks(int N, int H, int W, int Ci);
 bool flowhip_nconv_fwd_tiled_launch(const float* data, const float* conf,
                                     const float* weight, const float* bias,
                                     float* out, float* cout, int N, int Ci,
@@ -485,7 +485,7 @@ std::vector<torch::Tensor> nconv_bwd(torch::Tensor dnomin,
   }
   bool wrw_done = false;
   if (K == 1 || K == 3 || K == 5) {
-    const int nblocks = flowhip_nconv_tiled_nblocks(N, H, W);
+    const int nblocks = flowhip_nconv_tiled_nblocks(N, H, W, Ci);
     auto partials = torch::empty({nblocks, (long)Co * Ci * K * K},
                                  data.options());
     wrw_done = flowhip_nconv_wrw_tiled_launch(

@@ -261,14 +261,23 @@ __global__ __launch_bounds__(NCT_THREADS) void nconv_bwd_data_tiled_kernel(
 // LDS across waves) emits one partial row per workgroup.
 // partials layout: (nblocks, Co*CI*K*K).
 // ---------------------------------------------------------------------------
+// wrw tile height by channel count: as tall as the LDS budget allows —
+// the per-co wave-shuffle reduction (6 shfl per accumulator) amortizes
+// over TH*TW pixels, so taller tiles cut its share 2-4x.
+template <int CI, int K>
+constexpr int nct_th_wrw() {
+  return CI == 1 ? 64 : CI == 2 ? 48 : CI == 4 ? 32 : 16;
+}
+
 template <int K, int CI>
 __global__ __launch_bounds__(NCT_THREADS) void nconv_wrw_tiled_kernel(
     const float* __restrict__ dnomin, const float* __restrict__ ddenom,
     const float* __restrict__ data, const float* __restrict__ conf,
     float* __restrict__ partials,
     int N, int Co, int H, int W, int ntx, int nty) {
+  constexpr int TH = nct_th_wrw<CI, K>();
   constexpr int LW = NCT_TW + K - 1;
-  constexpr int LH = NCT_TH_WRW + K - 1;
+  constexpr int LH = TH + K - 1;
   constexpr int NW = CI * K * K;  // weights per co
   __shared__ float lds_c[CI * LH * LW];
   __shared__ float lds_dc[CI * LH * LW];
@@ -278,10 +287,10 @@ __global__ __launch_bounds__(NCT_THREADS) void nconv_wrw_tiled_kernel(
   const int tx = t % ntx; t /= ntx;
   const int ty = t % nty; t /= nty;
   const int n = t;
-  const int x0 = tx * NCT_TW, y0 = ty * NCT_TH_WRW;
+  const int x0 = tx * NCT_TW, y0 = ty * TH;
 
-  nct_stage_v<K, CI, NCT_TH_WRW, true>(conf, data, lds_c, lds_dc, n, x0, y0,
-                                       H, W);
+  nct_stage_v<K, CI, TH, true>(conf, data, lds_c, lds_dc, n, x0, y0,
+                               H, W);
   __syncthreads();
 
   const long plane = (long)H * W;
@@ -297,7 +306,7 @@ __global__ __launch_bounds__(NCT_THREADS) void nconv_wrw_tiled_kernel(
     for (int i = 0; i < NW; ++i) acc[i] = 0.f;
 
 #pragma unroll
-    for (int j = 0; j < NCT_TH_WRW / 16; ++j) {
+    for (int j = 0; j < TH / 16; ++j) {
       const int lyy = row0 + 16 * j;
       const int y = y0 + lyy;
       if (x >= W || y >= H) continue;
@@ -430,8 +439,12 @@ bool flowhip_nconv_bwd_data_tiled_launch(
     launched = true;                                                           \
   }
 
-int flowhip_nconv_tiled_nblocks(int N, int H, int W) {
-  return fh_cdiv(W, NCT_TW) * fh_cdiv(H, NCT_TH_WRW) * N;
+static int nct_th_wrw_rt(int Ci) {
+  return Ci == 1 ? 64 : Ci == 2 ? 48 : Ci == 4 ? 32 : 16;
+}
+
+int flowhip_nconv_tiled_nblocks(int N, int H, int W, int Ci) {
+  return fh_cdiv(W, NCT_TW) * fh_cdiv(H, nct_th_wrw_rt(Ci)) * N;
 }
 
 bool flowhip_nconv_wrw_tiled_launch(const float* dnomin, const float* ddenom,
@@ -440,7 +453,7 @@ bool flowhip_nconv_wrw_tiled_launch(const float* dnomin, const float* ddenom,
                                     int Ci, int Co, int H, int W, int K,
                                     hipStream_t stream) {
   if (W % 4 != 0) return false;
-  const int ntx = fh_cdiv(W, NCT_TW), nty = fh_cdiv(H, NCT_TH_WRW);
+  const int ntx = fh_cdiv(W, NCT_TW), nty = fh_cdiv(H, nct_th_wrw_rt(Ci));
   const int nblocks = ntx * nty * N;
   const int nw = Co * Ci * K * K;
   dim3 grid(nblocks), block(NCT_THREADS);
