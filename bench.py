@@ -171,6 +171,7 @@ def main():
     global_batch = batch * world_size
     ms_per_step = elapsed / cli.steps * 1000.0
     pairs_per_sec = global_batch * cli.steps / elapsed
+    peak_gb = (torch.cuda.max_memory_allocated() / 2**30) if on_gpu else None
 
     if rank == 0:
         print(json.dumps({
@@ -192,6 +193,7 @@ def main():
                 "image_size": [h, w],
                 "refinement_iters": iters,
                 "parallelism": f"dp{world_size}" if on_gpu else "cpu",
+                "peak_mem_gb": round(peak_gb, 2) if peak_gb else None,
             },
         }))
 
