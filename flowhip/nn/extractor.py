@@ -132,7 +132,10 @@ class _Encoder(nn.Module):
 
         self.norm1 = _norm(norm_fn, self.stem_planes,
                            groups_planes=8 if norm_fn == "group" else None)
-        self.conv1 = nn.Conv2d(3, self.stem_planes, kernel_size=7, stride=2, padding=3)
+        # the 7x7 s2 stem runs on the MFMA kernel too (3->8 channel pad,
+        # zero-page K tail): the last library conv out of the hot path
+        self.conv1 = _enc_conv(3, self.stem_planes, kernel_size=7, stride=2,
+                               padding=3)
         self.relu1 = nn.ReLU(inplace=True)
 
         self.in_planes = self.stem_planes

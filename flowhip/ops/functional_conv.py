@@ -115,12 +115,15 @@ class ConvGemmFn(torch.autograd.Function):
         if dy.dtype != torch.bfloat16:
             dy = dy.to(torch.bfloat16)
         dyp, O_real = _pad_dy8(dy)
-        if sH == 1 and sW == 1:
-            dx = _ext.ext().conv_gemm_fwd(dyp, wpk_bwd, None, I, KH, KW, 0)
-        else:
-            dx = _ext.ext().conv_gemm_fwd2(dyp, None, wpk_bwd, None, I, KH,
-                                           KW, 0, 0, sH, sW, 2,
-                                           x.shape[2], x.shape[3])[0]
+        dx = None
+        if ctx.needs_input_grad[0]:  # leaf inputs (the stem images) skip dx
+            if sH == 1 and sW == 1:
+                dx = _ext.ext().conv_gemm_fwd(dyp, wpk_bwd, None, I, KH, KW,
+                                              0)
+            else:
+                dx = _ext.ext().conv_gemm_fwd2(dyp, None, wpk_bwd, None, I,
+                                               KH, KW, 0, 0, sH, sW, 2,
+                                               x.shape[2], x.shape[3])[0]
         dw = _ext.ext().conv_gemm_wrw(dyp, x, None, KH, KW, sH, sW)
         if dw.shape[0] != O_real:
             dw = dw[:O_real].contiguous()

@@ -775,7 +775,9 @@ class TestConvGemm:
 
     STRIDED = [
         # (Cin, Cout, K, stride, H, W) — encoder inventory (extractor.py):
-        # stage-transition 3x3 s2, downsample 1x1 s2; odd/hostile dims too
+        # 7x7 s2 stem (3ch zero-pad path), stage-transition 3x3 s2,
+        # downsample 1x1 s2; odd/hostile dims too
+        (3, 64, 7, 2, 30, 48),
         (64, 96, 3, 2, 24, 40),
         (96, 128, 3, 2, 23, 37),
         (64, 96, 1, 2, 24, 40),
