@@ -12,13 +12,15 @@
 void flowhip_bgemm_nt_launch(const void* A, const void* B, void* C,
                              float alpha, int batch, int M, int N, int K,
                              int out_bf16, hipStream_t stream);
-void flowhip_corr_lookup_fwd_launch(const void* level, const float* coords,
-                                    void* out, int BP, int P, int Hl, int Wl,
-                                    int l, int L, int radius, int cl, int ldc,
+void flowhip_corr_lookup_fwd_launch(const void* const* levels, const int* Hs,
+                                    const int* Ws, const float* coords,
+                                    void* out, int BP, int P,
+                                    int L, int radius, int cl, int ldc,
                                     int is_bf16, hipStream_t stream);
 void flowhip_corr_lookup_bwd_launch(const void* gout, const float* coords,
-                                    void* glevel, int BP, int P, int Hl,
-                                    int Wl, int l, int L, int radius, int cl,
+                                    void* const* glevels, const int* Hs,
+                                    const int* Ws, int BP, int P,
+                                    int L, int radius, int cl,
                                     int is_bf16, int acc,
                                     hipStream_t stream);
 void flowhip_convex_up_fwd_launch(const float* flow, const float* mask,
@@ -234,18 +236,22 @@ torch::Tensor corr_lookup_fwd(std::vector<torch::Tensor> pyramid,
   const c10::cuda::CUDAGuard guard(coords.device());
   hipStream_t stream = at::cuda::getCurrentCUDAStream().stream();
 
+  const void* ptrs[4] = {nullptr, nullptr, nullptr, nullptr};
+  int Hs[4] = {0, 0, 0, 0}, Ws[4] = {0, 0, 0, 0};
+  TORCH_CHECK(L <= 4);
   for (int l = 0; l < L; ++l) {
     auto& lvl = pyramid[l];
     TORCH_CHECK(lvl.is_cuda() && lvl.is_contiguous() &&
                 lvl.dtype() == (bf ? torch::kBFloat16 : torch::kFloat32),
                 "corr_lookup: same-dtype contiguous pyramid levels required");
     TORCH_CHECK(lvl.size(0) == (long)B * P, "corr_lookup: level batch mismatch");
-    const int Hl = lvl.size(-2), Wl = lvl.size(-1);
-    flowhip_corr_lookup_fwd_launch(
-        lvl.data_ptr(), coords.data_ptr<float>(),
-        full.data_ptr(), B * P, P, Hl, Wl, l, L, (int)radius,
-        channels_last ? 1 : 0, (int)C8, bf ? 1 : 0, stream);
+    ptrs[l] = lvl.data_ptr();
+    Hs[l] = lvl.size(-2);
+    Ws[l] = lvl.size(-1);
   }
+  flowhip_corr_lookup_fwd_launch(
+      ptrs, Hs, Ws, coords.data_ptr<float>(), full.data_ptr(), B * P, P, L,
+      (int)radius, channels_last ? 1 : 0, (int)C8, bf ? 1 : 0, stream);
   return out;
 }
 
@@ -306,14 +312,18 @@ std::vector<torch::Tensor> corr_lookup_bwd(torch::Tensor gout,
       off += sizes[l];
     }
   }
+  void* gptrs[4] = {nullptr, nullptr, nullptr, nullptr};
+  int Hs[4] = {0, 0, 0, 0}, Ws[4] = {0, 0, 0, 0};
+  TORCH_CHECK(L <= 4);
   for (int l = 0; l < L; ++l) {
-    auto& g = grads[l];
-    const int Hl = g.size(-2), Wl = g.size(-1);
-    flowhip_corr_lookup_bwd_launch(
-        gout.data_ptr(), coords.data_ptr<float>(), g.data_ptr(),
-        B * P, P, Hl, Wl, l, L, (int)radius, channels_last ? 1 : 0,
-        levels_bf16 ? 1 : 0, acc ? 1 : 0, stream);
+    gptrs[l] = grads[l].data_ptr();
+    Hs[l] = grads[l].size(-2);
+    Ws[l] = grads[l].size(-1);
   }
+  flowhip_corr_lookup_bwd_launch(
+      gout.data_ptr(), coords.data_ptr<float>(), gptrs, Hs, Ws,
+      B * P, P, L, (int)radius, channels_last ? 1 : 0,
+      levels_bf16 ? 1 : 0, acc ? 1 : 0, stream);
   return grads;
 }
 

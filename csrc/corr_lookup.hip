@@ -36,15 +36,29 @@
 
 #define LK_THREADS 256
 
+// all pyramid levels in ONE launch (blockIdx.y = level): a per-level
+// launch at level 0 is only BP/256 workgroups (84 at the flagship shape)
+// -- a 3x underfill of the 256-CU chip; batching the levels quadruples
+// the resident blocks and cuts 8 launches per iteration to 2.
+struct LkLevels {
+  const void* p[4];
+  void* g[4];
+  int H[4];
+  int W[4];
+};
+
 template <int R, typename scalar_t>
 __global__ __launch_bounds__(LK_THREADS) void corr_lookup_fwd_kernel(
-    const scalar_t* __restrict__ level,  // (B*P, Hl, Wl)
+    LkLevels lv,                         // levels (B*P, Hl, Wl)
     const float* __restrict__ coords,    // (B, 2, H, W)
     scalar_t* __restrict__ out,          // (B, L*K2, H, W) NCHW or NHWC
-    int BP, int P, int Hl, int Wl, int l, int L, int cl, int ldc) {
+    int BP, int P, int L, int cl, int ldc) {
   constexpr int K = 2 * R + 1;
   constexpr int K2 = K * K;
 
+  const int l = blockIdx.y;
+  const scalar_t* __restrict__ level = (const scalar_t*)lv.p[l];
+  const int Hl = lv.H[l], Wl = lv.W[l];
   const int pix = blockIdx.x * LK_THREADS + threadIdx.x;
   if (pix >= BP) return;
   const int b = pix / P;
@@ -130,13 +144,16 @@ template <int R, typename scalar_t>
 __global__ __launch_bounds__(LK_THREADS) void corr_lookup_bwd_kernel(
     const scalar_t* __restrict__ gout,  // (B, L*K2, H, W)
     const float* __restrict__ coords,  // (B, 2, H, W)
-    scalar_t* __restrict__ glevel,     // (B*P, Hl, Wl); zero-initialized
+    LkLevels lv,                       // glevels (B*P, Hl, Wl); zeroed
                                        // (acc=0) or accumulated into (acc=1:
                                        // the iteration-chained grad buffer)
-    int BP, int P, int Hl, int Wl, int l, int L, int cl, int acc) {
+    int BP, int P, int L, int cl, int acc) {
   constexpr int K = 2 * R + 1;
   constexpr int K2 = K * K;
 
+  const int l = blockIdx.y;
+  scalar_t* __restrict__ glevel = (scalar_t*)lv.g[l];
+  const int Hl = lv.H[l], Wl = lv.W[l];
   const int pix = blockIdx.x * LK_THREADS + threadIdx.x;
   if (pix >= BP) return;
   const int b = pix / P;
@@ -233,65 +250,70 @@ __global__ __launch_bounds__(LK_THREADS) void corr_lookup_bwd_kernel(
 }
 
 template <int R, typename scalar_t>
-static void lookup_fwd_level(const scalar_t* level, const float* coords,
-                             scalar_t* out, int BP, int P, int Hl, int Wl,
-                             int l, int L, int cl, int ldc,
-                             hipStream_t stream) {
-  dim3 grid(fh_cdiv(BP, LK_THREADS));
+static void lookup_fwd_all(const LkLevels& lv, const float* coords,
+                           scalar_t* out, int BP, int P, int L, int cl,
+                           int ldc, hipStream_t stream) {
+  dim3 grid(fh_cdiv(BP, LK_THREADS), L);
   hipLaunchKernelGGL((corr_lookup_fwd_kernel<R, scalar_t>), grid,
-                     dim3(LK_THREADS), 0, stream, level, coords, out, BP, P,
-                     Hl, Wl, l, L, cl, ldc);
+                     dim3(LK_THREADS), 0, stream, lv, coords, out, BP, P,
+                     L, cl, ldc);
 }
 
 template <int R, typename scalar_t>
-static void lookup_bwd_level(const scalar_t* gout, const float* coords,
-                             scalar_t* glevel, int BP, int P, int Hl, int Wl,
-                             int l, int L, int cl, int acc,
-                             hipStream_t stream) {
-  dim3 grid(fh_cdiv(BP, LK_THREADS));
+static void lookup_bwd_all(const scalar_t* gout, const float* coords,
+                           const LkLevels& lv, int BP, int P, int L, int cl,
+                           int acc, hipStream_t stream) {
+  dim3 grid(fh_cdiv(BP, LK_THREADS), L);
   hipLaunchKernelGGL((corr_lookup_bwd_kernel<R, scalar_t>), grid,
-                     dim3(LK_THREADS), 0, stream, gout, coords, glevel, BP, P,
-                     Hl, Wl, l, L, cl, acc);
+                     dim3(LK_THREADS), 0, stream, gout, coords, lv, BP, P,
+                     L, cl, acc);
 }
 
 // levels may be fp32 (reference parity) or bf16 (HBM-resident bf16
 // pyramid); out / gout stay fp32, the bilerp blend always runs fp32.
 // out / gout dtype == level dtype (bf16-resident pyramid emits bf16 taps
 // straight into the bf16 NHWC motion-encoder conv; fp32 keeps parity).
-void flowhip_corr_lookup_fwd_launch(const void* level, const float* coords,
-                                    void* out, int BP, int P, int Hl, int Wl,
-                                    int l, int L, int radius, int cl, int ldc,
+void flowhip_corr_lookup_fwd_launch(const void* const* levels,
+                                    const int* Hs, const int* Ws,
+                                    const float* coords,
+                                    void* out, int BP, int P,
+                                    int L, int radius, int cl, int ldc,
                                     int is_bf16, hipStream_t stream) {
+  LkLevels lv{};
+  for (int l = 0; l < L; ++l) { lv.p[l] = levels[l]; lv.H[l] = Hs[l]; lv.W[l] = Ws[l]; }
   if (is_bf16) {
     switch (radius) {
-      case 3: lookup_fwd_level<3>((const __bf16*)level, coords, (__bf16*)out, BP, P, Hl, Wl, l, L, cl, ldc, stream); break;
-      case 4: lookup_fwd_level<4>((const __bf16*)level, coords, (__bf16*)out, BP, P, Hl, Wl, l, L, cl, ldc, stream); break;
+      case 3: lookup_fwd_all<3>(lv, coords, (__bf16*)out, BP, P, L, cl, ldc, stream); break;
+      case 4: lookup_fwd_all<4>(lv, coords, (__bf16*)out, BP, P, L, cl, ldc, stream); break;
       default: abort();
     }
   } else {
     switch (radius) {
-      case 3: lookup_fwd_level<3>((const float*)level, coords, (float*)out, BP, P, Hl, Wl, l, L, cl, ldc, stream); break;
-      case 4: lookup_fwd_level<4>((const float*)level, coords, (float*)out, BP, P, Hl, Wl, l, L, cl, ldc, stream); break;
+      case 3: lookup_fwd_all<3>(lv, coords, (float*)out, BP, P, L, cl, ldc, stream); break;
+      case 4: lookup_fwd_all<4>(lv, coords, (float*)out, BP, P, L, cl, ldc, stream); break;
       default: abort();
     }
   }
 }
 
 void flowhip_corr_lookup_bwd_launch(const void* gout, const float* coords,
-                                    void* glevel, int BP, int P, int Hl,
-                                    int Wl, int l, int L, int radius, int cl,
+                                    void* const* glevels, const int* Hs,
+                                    const int* Ws, int BP, int P,
+                                    int L, int radius, int cl,
                                     int is_bf16, int acc,
                                     hipStream_t stream) {
+  LkLevels lv{};
+  for (int l = 0; l < L; ++l) { lv.g[l] = glevels[l]; lv.H[l] = Hs[l]; lv.W[l] = Ws[l]; }
   if (is_bf16) {
     switch (radius) {
-      case 3: lookup_bwd_level<3>((const __bf16*)gout, coords, (__bf16*)glevel, BP, P, Hl, Wl, l, L, cl, acc, stream); break;
-      case 4: lookup_bwd_level<4>((const __bf16*)gout, coords, (__bf16*)glevel, BP, P, Hl, Wl, l, L, cl, acc, stream); break;
+      case 3: lookup_bwd_all<3>((const __bf16*)gout, coords, lv, BP, P, L, cl, acc, stream); break;
+      case 4: lookup_bwd_all<4>((const __bf16*)gout, coords, lv, BP, P, L, cl, acc, stream); break;
       default: abort();
     }
   } else {
     switch (radius) {
-      case 3: lookup_bwd_level<3>((const float*)gout, coords, (float*)glevel, BP, P, Hl, Wl, l, L, cl, acc, stream); break;
-      case 4: lookup_bwd_level<4>((const float*)gout, coords, (float*)glevel, BP, P, Hl, Wl, l, L, cl, acc, stream); break;
+      case 3: lookup_bwd_all<3>((const float*)gout, coords, lv, BP, P, L, cl, acc, stream); break;
+      case 4: lookup_bwd_all<4>((const float*)gout, coords, lv, BP, P, L, cl, acc, stream); break;
       default: abort();
     }
   }
