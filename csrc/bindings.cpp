@@ -172,6 +172,15 @@ bool flowhip_pacconv_bwd_launch(const float* dy, const float* x,
                                 int nchunk, int B, int Ci, int Co, int H,
                                 int W, int OH, int OW, int pH, int pW, int K,
                                 int dil, int shared, hipStream_t stream);
+void flowhip_pacpool_fwd_launch(const float* x, const float* kr, float* out,
+                                int B, int C, int KCH, int H, int W, int OH,
+                                int OW, int K, int sH, int sW, int pH,
+                                int pW, int dil, hipStream_t stream);
+void flowhip_pacpool_bwd_launch(const float* dy, const float* x,
+                                const float* kr, float* dx, float* dk, int B,
+                                int C, int KCH, int H, int W, int OH, int OW,
+                                int K, int sH, int sW, int pH, int pW,
+                                int dil, hipStream_t stream);
 bool flowhip_corr_pyramid_fwd_launch(const void* corr, void* l1, void* l2,
                                      void* l3, int BP, int H0, int W0,
                                      int nlev, int is_bf16,
@@ -1190,6 +1199,46 @@ std::vector<torch::Tensor> pacconv_bwd(torch::Tensor dy, torch::Tensor x,
   return {dx, dk, dw};
 }
 
+// kr: (B, KCH, K2, OH, OW) fp32; x: (B, C, H, W) fp32
+torch::Tensor pacpool_fwd(torch::Tensor x, torch::Tensor kr, int64_t K,
+                          int64_t sH, int64_t sW, int64_t pH, int64_t pW,
+                          int64_t dil) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous() &&
+              x.scalar_type() == torch::kFloat32);
+  TORCH_CHECK(kr.is_cuda() && kr.is_contiguous() && kr.dim() == 5 &&
+              kr.size(2) == K * K);
+  const int B = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
+  const int KCH = kr.size(1), OH = kr.size(3), OW = kr.size(4);
+  TORCH_CHECK(KCH == 1 || KCH == C);
+  auto out = torch::empty({(long)B, (long)C, (long)OH, (long)OW},
+                          x.options());
+  const c10::cuda::CUDAGuard guard(x.device());
+  hipStream_t stream = at::cuda::getCurrentCUDAStream().stream();
+  flowhip_pacpool_fwd_launch(x.data_ptr<float>(), kr.data_ptr<float>(),
+                             out.data_ptr<float>(), B, C, KCH, H, W, OH, OW,
+                             (int)K, (int)sH, (int)sW, (int)pH, (int)pW,
+                             (int)dil, stream);
+  return out;
+}
+
+std::vector<torch::Tensor> pacpool_bwd(torch::Tensor dy, torch::Tensor x,
+                                       torch::Tensor kr, int64_t K,
+                                       int64_t sH, int64_t sW, int64_t pH,
+                                       int64_t pW, int64_t dil) {
+  const int B = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
+  const int KCH = kr.size(1), OH = kr.size(3), OW = kr.size(4);
+  auto dx = torch::zeros_like(x);
+  auto dk = torch::empty_like(kr);
+  const c10::cuda::CUDAGuard guard(x.device());
+  hipStream_t stream = at::cuda::getCurrentCUDAStream().stream();
+  flowhip_pacpool_bwd_launch(dy.contiguous().data_ptr<float>(),
+                             x.data_ptr<float>(), kr.data_ptr<float>(),
+                             dx.data_ptr<float>(), dk.data_ptr<float>(), B,
+                             C, KCH, H, W, OH, OW, (int)K, (int)sH, (int)sW,
+                             (int)pH, (int)pW, (int)dil, stream);
+  return {dx, dk};
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -1213,6 +1262,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("packernel_bwd", &packernel_bwd, "backward of packernel_fwd");
   m.def("pacconv_fwd", &pacconv_fwd, "pixel-adaptive convolution forward");
   m.def("pacconv_bwd", &pacconv_bwd, "pixel-adaptive conv backward (dx,dk,dw)");
+  m.def("pacpool_fwd", &pacpool_fwd, "pixel-adaptive pooling forward");
+  m.def("pacpool_bwd", &pacpool_bwd,
+        "pixel-adaptive pooling backward (dx via atomic scatter, dk)");
   m.def("corr_pyramid_fwd", &corr_pyramid_fwd,
         "fused avg-pool pyramid build (levels 1..n-1)");
   m.def("corr_pyramid_bwd", &corr_pyramid_bwd,

@@ -456,3 +456,130 @@ bool flowhip_pacconv_bwd_launch(const float* dy, const float* x,
   }
   return true;
 }
+
+// ---------------------------------------------------------------------------
+// PAC pooling (kernel #10 tail; reference pac_modules.py:288-329 PacPool2dFn):
+//   out[b, c, oy, ox] = sum_{ky,kx} kernel[b, kk(c), ky*K+kx, oy, ox] *
+//                       x[b, c, oy*s - p + ky*d, ...]
+// kr is passed flattened (B, KCH, K2, OH, OW) with KCH in {1 (shared), C}.
+// Runtime K/stride/dilation (baseline-head op, not on the hot path).
+// Backward: dx by atomicAdd scatter (overlapping windows under stride <
+// K*d), dkernel by per-(b,kch,tap,pixel) gather over channels.
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(PAC_THREADS) void pacpool_fwd_kernel(
+    const float* __restrict__ x, const float* __restrict__ kr,
+    float* __restrict__ out, long total, int B, int C, int KCH, int H, int W,
+    int OH, int OW, int K, int sH, int sW, int pH, int pW, int dil) {
+  for (long idx = (long)blockIdx.x * PAC_THREADS + threadIdx.x; idx < total;
+       idx += (long)gridDim.x * PAC_THREADS) {
+    long t = idx;
+    const int ox = t % OW; t /= OW;
+    const int oy = t % OH; t /= OH;
+    const int c = t % C; t /= C;
+    const int b = (int)t;
+    const int kc = KCH == 1 ? 0 : c;
+    const float* kb = kr + (((long)b * KCH + kc) * K * K) * OH * OW +
+                      (long)oy * OW + ox;
+    const float* xb = x + ((long)b * C + c) * H * W;
+    float acc = 0.f;
+    for (int ky = 0; ky < K; ++ky) {
+      const int iy = oy * sH - pH + ky * dil;
+      if (iy < 0 || iy >= H) continue;
+      for (int kx = 0; kx < K; ++kx) {
+        const int ix = ox * sW - pW + kx * dil;
+        if (ix < 0 || ix >= W) continue;
+        acc += kb[(long)(ky * K + kx) * OH * OW] * xb[(long)iy * W + ix];
+      }
+    }
+    out[idx] = acc;
+  }
+}
+
+__global__ __launch_bounds__(PAC_THREADS) void pacpool_bwd_dx_kernel(
+    const float* __restrict__ dy, const float* __restrict__ kr,
+    float* __restrict__ dx, long total, int B, int C, int KCH, int H, int W,
+    int OH, int OW, int K, int sH, int sW, int pH, int pW, int dil) {
+  for (long idx = (long)blockIdx.x * PAC_THREADS + threadIdx.x; idx < total;
+       idx += (long)gridDim.x * PAC_THREADS) {
+    long t = idx;
+    const int ox = t % OW; t /= OW;
+    const int oy = t % OH; t /= OH;
+    const int c = t % C; t /= C;
+    const int b = (int)t;
+    const int kc = KCH == 1 ? 0 : c;
+    const float g = dy[idx];
+    const float* kb = kr + (((long)b * KCH + kc) * K * K) * OH * OW +
+                      (long)oy * OW + ox;
+    float* xg = dx + ((long)b * C + c) * H * W;
+    for (int ky = 0; ky < K; ++ky) {
+      const int iy = oy * sH - pH + ky * dil;
+      if (iy < 0 || iy >= H) continue;
+      for (int kx = 0; kx < K; ++kx) {
+        const int ix = ox * sW - pW + kx * dil;
+        if (ix < 0 || ix >= W) continue;
+        atomicAdd(&xg[(long)iy * W + ix],
+                  g * kb[(long)(ky * K + kx) * OH * OW]);
+      }
+    }
+  }
+}
+
+__global__ __launch_bounds__(PAC_THREADS) void pacpool_bwd_dk_kernel(
+    const float* __restrict__ dy, const float* __restrict__ x,
+    float* __restrict__ dk, long total, int B, int C, int KCH, int H, int W,
+    int OH, int OW, int K, int sH, int sW, int pH, int pW, int dil) {
+  // total = B * KCH * K2 * OH * OW
+  const int K2 = K * K;
+  for (long idx = (long)blockIdx.x * PAC_THREADS + threadIdx.x; idx < total;
+       idx += (long)gridDim.x * PAC_THREADS) {
+    long t = idx;
+    const int ox = t % OW; t /= OW;
+    const int oy = t % OH; t /= OH;
+    const int kk = t % K2; t /= K2;
+    const int kc = t % KCH; t /= KCH;
+    const int b = (int)t;
+    const int ky = kk / K, kx = kk - ky * K;
+    const int iy = oy * sH - pH + ky * dil;
+    const int ix = ox * sW - pW + kx * dil;
+    float acc = 0.f;
+    if (iy >= 0 && iy < H && ix >= 0 && ix < W) {
+      const int c0 = KCH == 1 ? 0 : kc;
+      const int c1 = KCH == 1 ? C : kc + 1;
+      for (int c = c0; c < c1; ++c)
+        acc += dy[(((long)b * C + c) * OH + oy) * OW + ox] *
+               x[(((long)b * C + c) * H + iy) * W + ix];
+    }
+    dk[idx] = acc;
+  }
+}
+
+void flowhip_pacpool_fwd_launch(const float* x, const float* kr, float* out,
+                                int B, int C, int KCH, int H, int W, int OH,
+                                int OW, int K, int sH, int sW, int pH,
+                                int pW, int dil, hipStream_t stream) {
+  const long total = (long)B * C * OH * OW;
+  dim3 grid(pac_blocks(total)), block(PAC_THREADS);
+  hipLaunchKernelGGL(pacpool_fwd_kernel, grid, block, 0, stream, x, kr, out,
+                     total, B, C, KCH, H, W, OH, OW, K, sH, sW, pH, pW, dil);
+}
+
+void flowhip_pacpool_bwd_launch(const float* dy, const float* x,
+                                const float* kr, float* dx, float* dk, int B,
+                                int C, int KCH, int H, int W, int OH, int OW,
+                                int K, int sH, int sW, int pH, int pW,
+                                int dil, hipStream_t stream) {
+  {
+    const long total = (long)B * C * OH * OW;
+    dim3 grid(pac_blocks(total)), block(PAC_THREADS);
+    hipLaunchKernelGGL(pacpool_bwd_dx_kernel, grid, block, 0, stream, dy, kr,
+                       dx, total, B, C, KCH, H, W, OH, OW, K, sH, sW, pH, pW,
+                       dil);
+  }
+  {
+    const long total = (long)B * KCH * K * K * OH * OW;
+    dim3 grid(pac_blocks(total)), block(PAC_THREADS);
+    hipLaunchKernelGGL(pacpool_bwd_dk_kernel, grid, block, 0, stream, dy, x,
+                       dk, total, B, C, KCH, H, W, OH, OW, K, sH, sW, pH, pW,
+                       dil);
+  }
+}

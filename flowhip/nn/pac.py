@@ -234,6 +234,21 @@ def pacpool2d(input, kernel, kernel_size, stride=1, padding=0, dilation=1):
     """Adaptive pooling: sum over window of (unfold(input) * kernel)."""
     kernel_size = _pair(kernel_size)
     bs, in_ch = input.shape[:2]
+
+    from ..ops import _ext
+    from ..ops.functional_pac import PacPool2dFn
+    s, p, d = _pair(stride), _pair(padding), _pair(dilation)
+    if (input.is_cuda and input.dtype == torch.float32
+            and kernel.dtype == torch.float32
+            and _ext.ext() is not None and not _ext.force_ref()
+            and kernel_size[0] == kernel_size[1] and d[0] == d[1]
+            and kernel.dim() >= 5):
+        K = kernel_size[0]
+        kr = kernel.reshape(bs, -1, K * K, *kernel.shape[-2:])
+        if kr.shape[1] in (1, in_ch):
+            return PacPool2dFn.apply(input, kr, K, s[0], s[1], p[0], p[1],
+                                     d[0])
+
     cols = nd2col(input, kernel_size, stride=stride, padding=padding,
                   dilation=dilation)
     cols = cols * kernel

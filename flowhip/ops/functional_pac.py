@@ -50,6 +50,30 @@ class PacConv2dFn(torch.autograd.Function):
         return dx, dk, dw, dbias, None, None, None, None
 
 
+class PacPool2dFn(torch.autograd.Function):
+    """Adaptive pooling (reference pac_modules.py:288-329): one gather
+    kernel forward; backward = atomic-scatter dx + per-tap channel-gather
+    dk (dx accumulation order is non-deterministic under overlapping
+    windows — baseline-head op, not on the training hot path)."""
+
+    @staticmethod
+    def forward(ctx, x, kr, K, sH, sW, pH, pW, dil):
+        x = x.contiguous()
+        kr = kr.contiguous()
+        out = _ext.ext().pacpool_fwd(x, kr, K, sH, sW, pH, pW, dil)
+        ctx.save_for_backward(x, kr)
+        ctx.meta = (K, sH, sW, pH, pW, dil)
+        return out
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, kr = ctx.saved_tensors
+        K, sH, sW, pH, pW, dil = ctx.meta
+        dx, dk = _ext.ext().pacpool_bwd(dy.contiguous(), x, kr, K, sH, sW,
+                                        pH, pW, dil)
+        return dx, dk, None, None, None, None, None, None
+
+
 def pac_kernel_fusable(input, mask, kernel_type, smooth_kernel_type,
                        channel_wise, kernel_size, dilation, eff_stride,
                        eff_padding):

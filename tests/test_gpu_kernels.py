@@ -542,6 +542,36 @@ class TestCorrPyramidFused:
                                    atol=1e-5, rtol=1e-5)
 
 
+class TestPacPool:
+    @pytest.mark.parametrize("kch,stride", [(1, 1), (1, 2), (0, 1)])
+    def test_matches_unfold_ref(self, kch, stride):
+        """HIP pacpool vs the unfold composition (reference
+        pac_modules.py:288-329); kch 0 means per-channel kernels."""
+        from flowhip.nn.pac import nd2col, pacpool2d
+        torch.manual_seed(71)
+        B, C, H, W, K = 2, 4, 18, 22, 3
+        pad = K // 2
+        OH = (H + 2 * pad - K) // stride + 1
+        OW = (W + 2 * pad - K) // stride + 1
+        x = torch.randn(B, C, H, W, device=_dev(), requires_grad=True)
+        ch = 1 if kch == 1 else C
+        kr = torch.rand(B, ch, K, K, OH, OW, device=_dev(),
+                        requires_grad=True)
+
+        out = pacpool2d(x, kr, K, stride=stride, padding=pad)
+        x2 = x.detach().clone().requires_grad_(True)
+        k2 = kr.detach().clone().requires_grad_(True)
+        cols = nd2col(x2, (K, K), stride=stride, padding=pad)
+        ref = (cols * k2).view(B, C, -1, OH, OW).sum(dim=2)
+        torch.testing.assert_close(out, ref, atol=1e-4, rtol=1e-4)
+
+        g = torch.randn_like(ref)
+        out.backward(g)
+        ref.backward(g)
+        torch.testing.assert_close(x.grad, x2.grad, atol=1e-4, rtol=1e-4)
+        torch.testing.assert_close(kr.grad, k2.grad, atol=1e-4, rtol=1e-4)
+
+
 class TestFrozenBatchNorm:
     def test_col_sum2_matches_torch(self):
         import flowhip._C as C
