@@ -1,0 +1,201 @@
+// Halo-staged 3x3 stride-1 implicit-GEMM convolution (the dominant conv
+// shape: every encoder stage conv + the update block's 3x3 convs).
+//
+// The generic conv_gemm kernel restages its A (im2col) tile from global
+// memory once per TAP — 9x the input traffic for a 3x3 kernel. Here a
+// workgroup stages ONE [TH+2][TW+2][64] channels-last halo tile of the
+// input per 64-channel slab and derives all 9 taps' MFMA A-fragments from
+// it by shifted LDS reads; the 9 taps' MFMAs then run back-to-back from
+// one staged image (one vmcnt/barrier pair per SLAB instead of per tap).
+// B (the packed weight [kyx][o][cpad]) is read as direct global b128
+// fragments — 8 KB per tap, L1/L2-resident and shared by every workgroup.
+//
+//   out[n, y, x, o] = act( sum_{ky,kx,c} x[n, y+ky-1, x+kx-1, c] *
+//                          wpk[kyx][o][c] + bias[o] )
+//
+// Tile: TH=4 rows x TW=32 px = 128 output pixels x 64 output channels per
+// workgroup (256 threads, 4 waves as 2Mx2N of 64px x 32co). Per-slab LDS:
+// 6*34*64 bf16 = 26 KB, double-buffered = 52 KB -> 3 workgroups/CU.
+// The XOR bank swizzle lives on the glds SOURCE channel slot (guide §5
+// rule 21), keyed by the tile-local pixel index.
+//
+// Same-kernel backward-data: stride-1 3x3 bwd-data is this conv with the
+// flipped/transposed pack (the caller swaps wpk). Cin ragged (%8) reads
+// the zero page past Cin like the generic kernel; Cout ragged clamps the
+// B row and masks the store.
+
+#include "common.h"
+
+#define CH_TH 4
+#define CH_TW 32
+#define CH_THREADS 256
+#define CH_LW (CH_TW + 2)        // 34
+#define CH_LH (CH_TH + 2)        // 6
+#define CH_PIECES (CH_LH * CH_LW * 8)  // 16-B pieces per slab tile (1632)
+
+// conflict-free for the b128 fragment read: 16 consecutive pixels' slot
+// indices (pix%2)*8 + slot^((pix>>1)&7) cover all 16 positions of the
+// 256-B bank row
+__device__ __forceinline__ unsigned ch_swz(unsigned pix, unsigned slot) {
+  return slot ^ ((pix >> 1) & 7u);
+}
+
+// stage one 64-channel slab of the halo tile; OOB pixels/channels read the
+// zero page. Optional second source (virtually-concatenated input).
+__device__ __forceinline__ void ch_stage(
+    const __bf16* __restrict__ x, const __bf16* __restrict__ x2,
+    const __bf16* __restrict__ zpage, char* lds_buf, int n, int y0, int x0,
+    int H, int W, int ld_x, int ld_x2, int C1, int Cin, int c0) {
+  for (int piece0 = threadIdx.x; piece0 < CH_PIECES; piece0 += CH_THREADS) {
+    const int pix = piece0 >> 3;        // 0 .. LH*LW-1
+    const int slot = piece0 & 7;
+    const int sslot = ch_swz(pix, slot);
+    const int py = pix / CH_LW, px = pix - py * CH_LW;
+    const int gy = y0 + py - 1;
+    const int gx = x0 + px - 1;
+    const int c = c0 + sslot * 8;
+    const __bf16* src = zpage;
+    if (gy >= 0 && gy < H && gx >= 0 && gx < W && c < Cin) {
+      if (x2 == nullptr || c < C1)
+        src = x + (((long)n * H + gy) * W + gx) * ld_x + c;
+      else
+        src = x2 + (((long)n * H + gy) * W + gx) * ld_x2 + (c - C1);
+    }
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) void*)src,
+        (__attribute__((address_space(3))) void*)(lds_buf + piece0 * 16),
+        16, 0, 0);
+  }
+}
+
+template <int ACT>
+__global__ __launch_bounds__(CH_THREADS, 4) void conv_halo3_fwd_kernel(
+    const __bf16* __restrict__ x, const __bf16* __restrict__ x2,
+    const __bf16* __restrict__ wpk,   // (9, Cout, cpad)
+    const float* __restrict__ bias,
+    __bf16* __restrict__ out,         // (N*H*W, ldo) channels-last rows
+    const __bf16* __restrict__ zpage,
+    int N, int H, int W, int ld_x, int ld_x2, int C1, int Cin, int Cout,
+    int cpad, int ldo, int ntx, int nty, int nco) {
+  __shared__ __attribute__((aligned(16))) char lds[2 * CH_PIECES * 16];
+
+  int t = blockIdx.x;
+  const int tx = t % ntx; t /= ntx;
+  const int ty = t % nty; t /= nty;
+  const int co_blk = t % nco; t /= nco;
+  const int n = t;
+  const int x0 = tx * CH_TW, y0 = ty * CH_TH;
+  const int n0 = co_blk * 64;
+
+  const int wave = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const int wr = (wave >> 1) * 64;   // pixel offset of the wave (0|64)
+  const int wc = (wave & 1) * 32;    // cout offset (0|32)
+
+  f32x4 acc[4][2];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 2; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  const int cslabs = cpad / 64;
+  ch_stage(x, x2, zpage, lds, n, y0, x0, H, W, ld_x, ld_x2, C1, Cin, 0);
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __syncthreads();
+
+  const int frow = lane & 15;  // fragment row (pixel within 16)
+  const int fk = lane >> 4;    // k subchunk
+
+  int cur = 0;
+  for (int cs = 0; cs < cslabs; ++cs) {
+    if (cs + 1 < cslabs)
+      ch_stage(x, x2, zpage, lds + (cur ^ 1) * CH_PIECES * 16, n, y0, x0, H,
+               W, ld_x, ld_x2, C1, Cin, (cs + 1) * 64);
+
+    const char* abuf = lds + cur * CH_PIECES * 16;
+#pragma unroll 1
+    for (int kyx = 0; kyx < 9; ++kyx) {
+      const int ky = kyx / 3, kx = kyx - ky * 3;
+      const __bf16* wsec = wpk + ((long)kyx * Cout + n0) * cpad + cs * 64;
+#pragma unroll
+      for (int kk = 0; kk < 2; ++kk) {
+        // B fragments: direct global b128 (weights are L2-hot)
+        bf16x8 bfr[2];
+#pragma unroll
+        for (int j = 0; j < 2; ++j) {
+          int o = wc + j * 16 + frow;
+          if (n0 + o >= Cout) o = Cout - 1 - n0;  // clamp; store masks
+          bfr[j] = *(const bf16x8*)(wsec + (long)o * cpad + kk * 32 +
+                                    fk * 8);
+        }
+#pragma unroll
+        for (int i = 0; i < 4; ++i) {
+          // A fragment: 16 pixels' tap-shifted 16-B channel chunks
+          const int p = wr + i * 16 + frow;           // tile pixel 0..127
+          const int py = p >> 5, px = p & 31;
+          const unsigned pix = (unsigned)((py + ky) * CH_LW + (px + kx));
+          const unsigned slot = (unsigned)(kk * 4 + fk);
+          const bf16x8 afr = *(const bf16x8*)(
+              abuf + pix * 128 + ch_swz(pix, slot) * 16);
+#pragma unroll
+          for (int j = 0; j < 2; ++j)
+            acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                afr, bfr[j], acc[i][j], 0, 0, 0);
+        }
+      }
+    }
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __syncthreads();
+    cur ^= 1;
+  }
+
+  // epilogue: bias + activation, masked channels-last stores
+  const int fcol = lane & 15;
+  const int frow0 = (lane >> 4) * 4;
+#pragma unroll
+  for (int j = 0; j < 2; ++j) {
+    const int gco = n0 + wc + j * 16 + fcol;
+    if (gco >= Cout) continue;
+    const float b = bias ? bias[gco] : 0.f;
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int p = wr + i * 16 + frow0 + r;
+        const int py = p >> 5, px = p & 31;
+        const int gy = y0 + py, gx = x0 + px;
+        if (gy >= H || gx >= W) continue;
+        float v = acc[i][j][r] + b;
+        if (ACT == 1) v = fmaxf(v, 0.f);
+        out[(((long)n * H + gy) * W + gx) * ldo + gco] = (__bf16)v;
+      }
+    }
+  }
+}
+
+bool flowhip_conv_halo3_fwd_launch(const void* x, const void* x2,
+                                   const void* wpk, const float* bias,
+                                   void* out, const void* zpage, int N,
+                                   int H, int W, int ld_x, int ld_x2, int C1,
+                                   int Cin, int Cout, int cpad, int ldo,
+                                   int act, hipStream_t stream) {
+  if (cpad % 64 != 0) return false;
+  const int ntx = fh_cdiv(W, CH_TW), nty = fh_cdiv(H, CH_TH);
+  const int nco = fh_cdiv(Cout, 64);
+  const long blocks = (long)ntx * nty * nco * N;
+  if (blocks < 320) return false;  // fill-first: generic kernel handles it
+  dim3 grid((unsigned)blocks), block(CH_THREADS);
+  if (act == 1)
+    hipLaunchKernelGGL(conv_halo3_fwd_kernel<1>, grid, block, 0, stream,
+                       (const __bf16*)x, (const __bf16*)x2,
+                       (const __bf16*)wpk, bias, (__bf16*)out,
+                       (const __bf16*)zpage, N, H, W, ld_x, ld_x2, C1, Cin,
+                       Cout, cpad, ldo, ntx, nty, nco);
+  else
+    hipLaunchKernelGGL(conv_halo3_fwd_kernel<0>, grid, block, 0, stream,
+                       (const __bf16*)x, (const __bf16*)x2,
+                       (const __bf16*)wpk, bias, (__bf16*)out,
+                       (const __bf16*)zpage, N, H, W, ld_x, ld_x2, C1, Cin,
+                       Cout, cpad, ldo, ntx, nty, nco);
+  return true;
+}
