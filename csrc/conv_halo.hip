@@ -26,12 +26,10 @@
 
 #include "common.h"
 
-#define CH_TH 4
 #define CH_TW 32
 #define CH_THREADS 256
 #define CH_LW (CH_TW + 2)        // 34
-#define CH_LH (CH_TH + 2)        // 6
-#define CH_PIECES (CH_LH * CH_LW * 8)  // 16-B pieces per slab tile (1632)
+// 16-B pieces per slab tile for tile height TH: (TH+2)*34*8
 
 // conflict-free for the b128 fragment read: 16 consecutive pixels' slot
 // indices (pix%2)*8 + slot^((pix>>1)&7) cover all 16 positions of the
@@ -42,11 +40,13 @@ __device__ __forceinline__ unsigned ch_swz(unsigned pix, unsigned slot) {
 
 // stage one 64-channel slab of the halo tile; OOB pixels/channels read the
 // zero page. Optional second source (virtually-concatenated input).
+template <int THT>
 __device__ __forceinline__ void ch_stage(
     const __bf16* __restrict__ x, const __bf16* __restrict__ x2,
     const __bf16* __restrict__ zpage, char* lds_buf, int n, int y0, int x0,
     int H, int W, int ld_x, int ld_x2, int C1, int Cin, int c0) {
-  for (int piece0 = threadIdx.x; piece0 < CH_PIECES; piece0 += CH_THREADS) {
+  constexpr int PIECES = (THT + 2) * CH_LW * 8;
+  for (int piece0 = threadIdx.x; piece0 < PIECES; piece0 += CH_THREADS) {
     const int pix = piece0 >> 3;        // 0 .. LH*LW-1
     const int slot = piece0 & 7;
     const int sslot = ch_swz(pix, slot);
@@ -68,7 +68,11 @@ __device__ __forceinline__ void ch_stage(
   }
 }
 
-template <int ACT>
+// THT: output rows per tile. THT=4 double-buffers the slab loop
+// (multi-slab Cin); THT=8 single-buffers (Cin <= 64: one slab, nothing to
+// overlap — spend the LDS on a taller tile instead, halving the staging
+// and barrier cost per output pixel).
+template <int ACT, int THT, bool DBUF>
 __global__ __launch_bounds__(CH_THREADS, 4) void conv_halo3_fwd_kernel(
     const __bf16* __restrict__ x, const __bf16* __restrict__ x2,
     const __bf16* __restrict__ wpk,   // (9, Cout, cpad)
@@ -77,29 +81,31 @@ __global__ __launch_bounds__(CH_THREADS, 4) void conv_halo3_fwd_kernel(
     const __bf16* __restrict__ zpage,
     int N, int H, int W, int ld_x, int ld_x2, int C1, int Cin, int Cout,
     int cpad, int ldo, int ntx, int nty, int nco) {
-  __shared__ __attribute__((aligned(16))) char lds[2 * CH_PIECES * 16];
+  constexpr int PIECES = (THT + 2) * CH_LW * 8;
+  __shared__ __attribute__((aligned(16))) char lds[(DBUF ? 2 : 1) * PIECES *
+                                                   16];
 
   int t = blockIdx.x;
   const int tx = t % ntx; t /= ntx;
   const int ty = t % nty; t /= nty;
   const int co_blk = t % nco; t /= nco;
   const int n = t;
-  const int x0 = tx * CH_TW, y0 = ty * CH_TH;
+  const int x0 = tx * CH_TW, y0 = ty * THT;
   const int n0 = co_blk * 64;
 
   const int wave = threadIdx.x >> 6;
   const int lane = threadIdx.x & 63;
-  const int wr = (wave >> 1) * 64;   // pixel offset of the wave (0|64)
-  const int wc = (wave & 1) * 32;    // cout offset (0|32)
+  const int wr = (wave >> 1) * (THT * 16);  // pixel offset of the wave
+  const int wc = (wave & 1) * 32;           // cout offset (0|32)
 
-  f32x4 acc[4][2];
+  f32x4 acc[THT][2];
 #pragma unroll
-  for (int i = 0; i < 4; ++i)
+  for (int i = 0; i < THT; ++i)
 #pragma unroll
     for (int j = 0; j < 2; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
   const int cslabs = cpad / 64;
-  ch_stage(x, x2, zpage, lds, n, y0, x0, H, W, ld_x, ld_x2, C1, Cin, 0);
+  ch_stage<THT>(x, x2, zpage, lds, n, y0, x0, H, W, ld_x, ld_x2, C1, Cin, 0);
   asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
   __syncthreads();
 
@@ -108,11 +114,11 @@ __global__ __launch_bounds__(CH_THREADS, 4) void conv_halo3_fwd_kernel(
 
   int cur = 0;
   for (int cs = 0; cs < cslabs; ++cs) {
-    if (cs + 1 < cslabs)
-      ch_stage(x, x2, zpage, lds + (cur ^ 1) * CH_PIECES * 16, n, y0, x0, H,
-               W, ld_x, ld_x2, C1, Cin, (cs + 1) * 64);
+    if (DBUF && cs + 1 < cslabs)
+      ch_stage<THT>(x, x2, zpage, lds + (cur ^ 1) * PIECES * 16, n, y0, x0,
+                    H, W, ld_x, ld_x2, C1, Cin, (cs + 1) * 64);
 
-    const char* abuf = lds + cur * CH_PIECES * 16;
+    const char* abuf = lds + cur * PIECES * 16;
 #pragma unroll 1
     for (int kyx = 0; kyx < 9; ++kyx) {
       const int ky = kyx / 3, kx = kyx - ky * 3;
@@ -129,9 +135,9 @@ __global__ __launch_bounds__(CH_THREADS, 4) void conv_halo3_fwd_kernel(
                                     fk * 8);
         }
 #pragma unroll
-        for (int i = 0; i < 4; ++i) {
+        for (int i = 0; i < THT; ++i) {
           // A fragment: 16 pixels' tap-shifted 16-B channel chunks
-          const int p = wr + i * 16 + frow;           // tile pixel 0..127
+          const int p = wr + i * 16 + frow;           // tile pixel
           const int py = p >> 5, px = p & 31;
           const unsigned pix = (unsigned)((py + ky) * CH_LW + (px + kx));
           const unsigned slot = (unsigned)(kk * 4 + fk);
@@ -144,9 +150,14 @@ __global__ __launch_bounds__(CH_THREADS, 4) void conv_halo3_fwd_kernel(
         }
       }
     }
-    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-    __syncthreads();
-    cur ^= 1;
+    if (cs + 1 < cslabs) {
+      if (!DBUF)
+        ch_stage<THT>(x, x2, zpage, lds, n, y0, x0, H, W, ld_x, ld_x2, C1,
+                      Cin, (cs + 1) * 64);
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      __syncthreads();
+      if (DBUF) cur ^= 1;
+    }
   }
 
   // epilogue: bias + activation, masked channels-last stores
@@ -158,7 +169,7 @@ __global__ __launch_bounds__(CH_THREADS, 4) void conv_halo3_fwd_kernel(
     if (gco >= Cout) continue;
     const float b = bias ? bias[gco] : 0.f;
 #pragma unroll
-    for (int i = 0; i < 4; ++i) {
+    for (int i = 0; i < THT; ++i) {
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         const int p = wr + i * 16 + frow0 + r;
@@ -180,22 +191,16 @@ bool flowhip_conv_halo3_fwd_launch(const void* x, const void* x2,
                                    int Cin, int Cout, int cpad, int ldo,
                                    int act, hipStream_t stream) {
   if (cpad % 64 != 0) return false;
-  const int ntx = fh_cdiv(W, CH_TW), nty = fh_cdiv(H, CH_TH);
   const int nco = fh_cdiv(Cout, 64);
+  // single-slab inputs (Cin <= 64): taller single-buffered tile
+  const int THT = (cpad == 64) ? 8 : 4;
+  const int ntx = fh_cdiv(W, CH_TW), nty = fh_cdiv(H, THT);
   const long blocks = (long)ntx * nty * nco * N;
   if (blocks < 320) return false;  // fill-first: generic kernel handles it
   dim3 grid((unsigned)blocks), block(CH_THREADS);
-  if (act == 1)
-    hipLaunchKernelGGL(conv_halo3_fwd_kernel<1>, grid, block, 0, stream,
-                       (const __bf16*)x, (const __bf16*)x2,
-                       (const __bf16*)wpk, bias, (__bf16*)out,
-                       (const __bf16*)zpage, N, H, W, ld_x, ld_x2, C1, Cin,
-                       Cout, cpad, ldo, ntx, nty, nco);
-  else
-    hipLaunchKernelGGL(conv_halo3_fwd_kernel<0>, grid, block, 0, stream,
-                       (const __bf16*)x, (const __bf16*)x2,
-                       (const __bf16*)wpk, bias, (__bf16*)out,
-                       (const __bf16*)zpage, N, H, W, ld_x, ld_x2, C1, Cin,
-                       Cout, cpad, ldo, ntx, nty, nco);
+#define CH_LAUNCH(A, T, D)                                                     hipLaunchKernelGGL((conv_halo3_fwd_kernel<A, T, D>), grid, block, 0,                            stream, (const __bf16*)x, (const __bf16*)x2,                                 (const __bf16*)wpk, bias, (__bf16*)out,                                      (const __bf16*)zpage, N, H, W, ld_x, ld_x2, C1, Cin,                         Cout, cpad, ldo, ntx, nty, nco)
+  if (THT == 8) { if (act == 1) CH_LAUNCH(1, 8, false); else CH_LAUNCH(0, 8, false); }
+  else { if (act == 1) CH_LAUNCH(1, 4, true); else CH_LAUNCH(0, 4, true); }
+#undef CH_LAUNCH
   return true;
 }
