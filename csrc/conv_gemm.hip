@@ -485,192 +485,6 @@ __global__ __launch_bounds__(CG_THREADS, 2) void conv_gemm_wrw_kernel(
   }
 }
 
-// ---------------------------------------------------------------------------
-// KW=3 weight-gradient variant: grid.y = ky only; ONE staged dy image + the
-// THREE kx-shifted x images share each m-tile (the generic kernel re-reads
-// dy AND x from HBM once per (ky,kx) tap — 9x at encoder working sets that
-// exceed L2; this is 3x, and the 3 shifted x gathers of one tile hit the
-// same cache lines). Fragments keep the hardware-transpose layout; the
-// per-tap accumulators fit (3 x 2 x 2 f32x4 = 48 VGPRs).
-// partials layout matches the generic kernel: (CHUNKS, KYX, orows, cpad).
-// ---------------------------------------------------------------------------
-__global__ __launch_bounds__(CG_THREADS, 2) void conv_gemm_wrw3_kernel(
-    const __bf16* __restrict__ dy,  // (Mtot, Cout)
-    const __bf16* __restrict__ x,   // (N, srcH, srcW, ld_x)
-    const __bf16* __restrict__ x2,  // second source or nullptr
-    float* __restrict__ partials,
-    const __bf16* __restrict__ zpage,
-    long Mtot, int HH, int WW, int srcH, int srcW, int sH, int sW, int ld_x,
-    int ld_x2, int C1, int Cin, int Cout, int cpad, int KH, int padH,
-    int padW, int tiles_o, int tiles_c, int nchunk) {
-  __shared__ __attribute__((aligned(16))) char lds[2 * 4 * CG_BM * CG_BK * 2];
-
-  const int tid = threadIdx.x;
-  const int wave = tid >> 6;
-  const int lane = tid & 63;
-
-  const int to = blockIdx.x % tiles_o;
-  const int tc = blockIdx.x / tiles_o;
-  const int ky = blockIdx.y;
-  const int chunk = blockIdx.z;
-  const int dyo = ky - padH;
-
-  const int o0 = to * 64;
-  const int c0 = tc * 64;
-
-  const long mtiles = (Mtot + CG_BM - 1) / CG_BM;
-  const long t0 = (mtiles * chunk) / nchunk;
-  const long t1 = (mtiles * (chunk + 1)) / nchunk;
-
-  const int wr = (wave >> 1) * 32;  // o offset
-  const int wc = (wave & 1) * 32;   // c offset
-
-  f32x4 acc[3][2][2];
-#pragma unroll
-  for (int kx = 0; kx < 3; ++kx)
-#pragma unroll
-    for (int i = 0; i < 2; ++i)
-#pragma unroll
-      for (int j = 0; j < 2; ++j) acc[kx][i][j] = {0.f, 0.f, 0.f, 0.f};
-
-  const unsigned TS = CG_BM * CG_BK * 2;
-  // set = [dy | x(kx=0) | x(kx=1) | x(kx=2)], permuted tr images
-  auto stage_set = [&](long mt, char* buf) {
-#pragma unroll
-    for (int j = 0; j < 2; ++j) {
-      const int piece0 = wave * 64 + CG_THREADS * j;
-      const int piece = piece0 + lane;
-      const int sb = piece >> 3;
-      const int pr = (piece >> 1) & 3;
-      const int ch = piece & 1;
-      const int irow = (sb & 15) * 4 + pr;
-      const long m = mt * CG_BM + wrw_row_to_m(irow);
-      const int cb16 = (sb >> 4) * 16 + ch * 8;
-      // dy piece
-      {
-        const __bf16* src = zpage;
-        const int o = o0 + cb16;
-        if (m < Mtot && o < Cout) src = dy + m * Cout + o;
-        __builtin_amdgcn_global_load_lds(
-            (const __attribute__((address_space(1))) void*)src,
-            (__attribute__((address_space(3))) void*)(buf + piece0 * 16),
-            16, 0, 0);
-      }
-      // x pieces for the three kx taps
-      int xx = 0, yy = 0;
-      long n = 0;
-      if (m < Mtot) {
-        xx = (int)(m % WW);
-        yy = (int)((m / WW) % HH);
-        n = m / ((long)WW * HH);
-      }
-      const int sy = yy * sH + dyo;
-      const int c = c0 + cb16;
-#pragma unroll
-      for (int kx = 0; kx < 3; ++kx) {
-        const __bf16* src = zpage;
-        const int sx = xx * sW + kx - padW;
-        if (m < Mtot && sy >= 0 && sy < srcH && sx >= 0 && sx < srcW &&
-            c < Cin) {
-          if (x2 == nullptr || c < C1)
-            src = x + (((long)n * srcH + sy) * srcW + sx) * ld_x + c;
-          else
-            src = x2 + (((long)n * srcH + sy) * srcW + sx) * ld_x2 +
-                  (c - C1);
-        }
-        __builtin_amdgcn_global_load_lds(
-            (const __attribute__((address_space(1))) void*)src,
-            (__attribute__((address_space(3))) void*)(buf + (kx + 1) * TS +
-                                                      piece0 * 16),
-            16, 0, 0);
-      }
-    }
-  };
-
-  if (t0 < t1) stage_set(t0, lds);
-  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-  __syncthreads();
-
-  const unsigned lb = (unsigned)lane * 8;
-  int cur = 0;
-  for (long t = t0; t < t1; ++t) {
-    if (t + 1 < t1) stage_set(t + 1, lds + (cur ^ 1) * 4 * TS);
-
-    const char* dbuf = lds + cur * 4 * TS;
-#pragma unroll
-    for (int kk = 0; kk < 2; ++kk) {
-      bf16x4 a0[2], a1[2];
-#pragma unroll
-      for (int i = 0; i < 2; ++i) {
-        const unsigned cb = (unsigned)(wr >> 4) + i;
-        a0[i] = wrw_tr_read(dbuf, (cb * 16 + kk * 8 + 0) * 128 + lb);
-        a1[i] = wrw_tr_read(dbuf, (cb * 16 + kk * 8 + 4) * 128 + lb);
-      }
-      asm volatile("s_waitcnt lgkmcnt(0)"
-                   : "+v"(a0[0]), "+v"(a0[1]), "+v"(a1[0]), "+v"(a1[1])
-                   :: "memory");
-      __builtin_amdgcn_sched_barrier(0);
-      bf16x8 af[2];
-#pragma unroll
-      for (int i = 0; i < 2; ++i)
-        af[i] = __builtin_shufflevector(a0[i], a1[i], 0, 1, 2, 3, 4, 5, 6,
-                                        7);
-#pragma unroll
-      for (int kx = 0; kx < 3; ++kx) {
-        const char* xbuf = dbuf + (kx + 1) * TS;
-        bf16x4 b0[2], b1[2];
-#pragma unroll
-        for (int j = 0; j < 2; ++j) {
-          const unsigned cb = (unsigned)(wc >> 4) + j;
-          b0[j] = wrw_tr_read(xbuf, (cb * 16 + kk * 8 + 0) * 128 + lb);
-          b1[j] = wrw_tr_read(xbuf, (cb * 16 + kk * 8 + 4) * 128 + lb);
-        }
-        asm volatile("s_waitcnt lgkmcnt(0)"
-                     : "+v"(b0[0]), "+v"(b0[1]), "+v"(b1[0]), "+v"(b1[1])
-                     :: "memory");
-        __builtin_amdgcn_sched_barrier(0);
-#pragma unroll
-        for (int i = 0; i < 2; ++i) {
-#pragma unroll
-          for (int j = 0; j < 2; ++j) {
-            const bf16x8 bf = __builtin_shufflevector(b0[j], b1[j], 0, 1, 2,
-                                                      3, 4, 5, 6, 7);
-            acc[kx][i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                af[i], bf, acc[kx][i][j], 0, 0, 0);
-          }
-        }
-      }
-    }
-
-    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-    __syncthreads();
-    cur ^= 1;
-  }
-
-  // store partial tiles: partials[chunk][ky*3+kx][o0+...][c0+...]
-  const int KYX = KH * 3;
-  const int orows = tiles_o * 64;
-  const int fcol = lane & 15;
-  const int frow0 = (lane >> 4) * 4;
-#pragma unroll
-  for (int kx = 0; kx < 3; ++kx) {
-    float* pbase = partials +
-                   (((long)chunk * KYX + (ky * 3 + kx)) * orows) * cpad;
-#pragma unroll
-    for (int i = 0; i < 2; ++i) {
-#pragma unroll
-      for (int j = 0; j < 2; ++j) {
-#pragma unroll
-        for (int r = 0; r < 4; ++r) {
-          const int o = o0 + wr + i * 16 + frow0 + r;
-          const int c = c0 + wc + j * 16 + fcol;
-          pbase[(long)o * cpad + c] = acc[kx][i][j][r];
-        }
-      }
-    }
-  }
-}
-
 // reduce partials over chunks and scatter into dW (Cout, Cin, KH, KW) fp32
 __global__ __launch_bounds__(CG_THREADS) void conv_gemm_wrw_reduce_kernel(
     const float* __restrict__ partials, float* __restrict__ dw,
@@ -775,24 +589,17 @@ void flowhip_conv_gemm_wrw_launch(const void* dy, const void* x,
   const int tiles_o = fh_cdiv(Cout, 64);
   const int tiles_c = fh_cdiv(cpad, 64);
   dim3 block(CG_THREADS);
-  if (KW == 3 && padW == 1) {
-    // kx taps share one staging set (3x less HBM re-read)
-    dim3 grid3(tiles_o * tiles_c, KH, nchunk);
-    hipLaunchKernelGGL(conv_gemm_wrw3_kernel, grid3, block, 0, stream,
-                       (const __bf16*)dy, (const __bf16*)x,
-                       (const __bf16*)x2, partials, (const __bf16*)zpage,
-                       Mtot, HH, WW, srcH, srcW, sH, sW, ld_x, ld_x2, C1,
-                       Cin, Cout, cpad, KH, padH, padW, tiles_o, tiles_c,
-                       nchunk);
-  } else {
-    dim3 grid(tiles_o * tiles_c, KH * KW, nchunk);
-    hipLaunchKernelGGL(conv_gemm_wrw_kernel, grid, block, 0, stream,
-                       (const __bf16*)dy, (const __bf16*)x,
-                       (const __bf16*)x2, partials, (const __bf16*)zpage,
-                       Mtot, HH, WW, srcH, srcW, sH, sW, ld_x, ld_x2, C1,
-                       Cin, Cout, cpad, KH, KW, padH, padW, tiles_o, tiles_c,
-                       nchunk);
-  }
+  // (a KW=3 tap-sharing wrw variant was measured SLOWER despite 3x less
+  // operand re-read: the 4-image LDS set halves occupancy and the
+  // encoder working sets are largely L3-resident, so the re-reads were
+  // cheaper than modeled — within-box A/B 41.0 vs 42.65 pairs/s)
+  dim3 grid(tiles_o * tiles_c, KH * KW, nchunk);
+  hipLaunchKernelGGL(conv_gemm_wrw_kernel, grid, block, 0, stream,
+                     (const __bf16*)dy, (const __bf16*)x,
+                     (const __bf16*)x2, partials, (const __bf16*)zpage,
+                     Mtot, HH, WW, srcH, srcW, sH, sW, ld_x, ld_x2, C1,
+                     Cin, Cout, cpad, KH, KW, padH, padW, tiles_o, tiles_c,
+                     nchunk);
   const long total = (long)Cout * Cin * KH * KW;
   long rblocks = (total + CG_THREADS - 1) / CG_THREADS;
   if (rblocks > 4096) rblocks = 4096;
