@@ -529,12 +529,13 @@ static void cg_fwd_dispatch(const __bf16* x, const __bf16* x2,
 #undef CG_LAUNCH
 }
 
-bool flowhip_conv_halo3_fwd_launch(const void* x, const void* x2,
-                                   const void* wpk, const float* bias,
-                                   void* out, const void* zpage, int N,
-                                   int H, int W, int ld_x, int ld_x2, int C1,
-                                   int Cin, int Cout, int cpad, int ldo,
-                                   int act, hipStream_t stream);
+bool flowhip_conv_halo_fwd_launch(const void* x, const void* x2,
+                                  const void* wpk, const float* bias,
+                                  void* out, const void* zpage, int N,
+                                  int H, int W, int ld_x, int ld_x2, int C1,
+                                  int Cin, int Cout, int cpad, int ldo,
+                                  int KH, int KW, int act,
+                                  hipStream_t stream);
 
 void flowhip_conv_gemm_fwd_launch(const void* x, const void* x2,
                                   const void* wpk, const float* bias,
@@ -545,14 +546,16 @@ void flowhip_conv_gemm_fwd_launch(const void* x, const void* x2,
                                   int cpad, int KH, int KW, int padH,
                                   int padW, int osplit, int act, int smode,
                                   hipStream_t stream) {
-  // 3x3 stride-1 (the dominant shape incl. its backward-data): the
-  // halo-staged kernel stages the input once for all 9 taps
-  if (smode == 0 && KH == 3 && KW == 3 && out2 == nullptr &&
-      padH == 1 && padW == 1) {
+  // stride-1 3x3 / 1x5 / 5x1 (the dominant shapes incl. their
+  // backward-data): the halo-staged kernel stages the input once for all
+  // KH*KW taps
+  if (smode == 0 && out2 == nullptr && padH == KH / 2 && padW == KW / 2 &&
+      ((KH == 3 && KW == 3) || (KH == 1 && KW == 5) ||
+       (KH == 5 && KW == 1))) {
     const int N = (int)(Mtot / ((long)HH * WW));
-    if (flowhip_conv_halo3_fwd_launch(x, x2, wpk, bias, out, zpage, N, HH,
-                                      WW, ld_x, ld_x2, C1, Cin, Cout, cpad,
-                                      Cout, act, stream))
+    if (flowhip_conv_halo_fwd_launch(x, x2, wpk, bias, out, zpage, N, HH,
+                                     WW, ld_x, ld_x2, C1, Cin, Cout, cpad,
+                                     Cout, KH, KW, act, stream))
       return;
   }
   // tap offsets: direct conv (smode 0/1) reads sy = oy*sH + ky - padH;

@@ -28,8 +28,8 @@
 
 #define CH_TW 32
 #define CH_THREADS 256
-#define CH_LW (CH_TW + 2)        // 34
-// 16-B pieces per slab tile for tile height TH: (TH+2)*34*8
+// LDS tile row stride for kernel width KW: CH_TW + KW - 1
+// 16-B pieces per slab tile: (TH+KH-1) * (TW+KW-1) * 8
 
 // conflict-free for the b128 fragment read: 16 consecutive pixels' slot
 // indices (pix%2)*8 + slot^((pix>>1)&7) cover all 16 positions of the
@@ -40,19 +40,20 @@ __device__ __forceinline__ unsigned ch_swz(unsigned pix, unsigned slot) {
 
 // stage one 64-channel slab of the halo tile; OOB pixels/channels read the
 // zero page. Optional second source (virtually-concatenated input).
-template <int THT>
+template <int THT, int KHH, int KWW>
 __device__ __forceinline__ void ch_stage(
     const __bf16* __restrict__ x, const __bf16* __restrict__ x2,
     const __bf16* __restrict__ zpage, char* lds_buf, int n, int y0, int x0,
     int H, int W, int ld_x, int ld_x2, int C1, int Cin, int c0) {
-  constexpr int PIECES = (THT + 2) * CH_LW * 8;
+  constexpr int LW = CH_TW + KWW - 1;
+  constexpr int PIECES = (THT + KHH - 1) * LW * 8;
   for (int piece0 = threadIdx.x; piece0 < PIECES; piece0 += CH_THREADS) {
     const int pix = piece0 >> 3;        // 0 .. LH*LW-1
     const int slot = piece0 & 7;
     const int sslot = ch_swz(pix, slot);
-    const int py = pix / CH_LW, px = pix - py * CH_LW;
-    const int gy = y0 + py - 1;
-    const int gx = x0 + px - 1;
+    const int py = pix / LW, px = pix - py * LW;
+    const int gy = y0 + py - KHH / 2;
+    const int gx = x0 + px - KWW / 2;
     const int c = c0 + sslot * 8;
     const __bf16* src = zpage;
     if (gy >= 0 && gy < H && gx >= 0 && gx < W && c < Cin) {
@@ -71,17 +72,20 @@ __device__ __forceinline__ void ch_stage(
 // THT: output rows per tile. THT=4 double-buffers the slab loop
 // (multi-slab Cin); THT=8 single-buffers (Cin <= 64: one slab, nothing to
 // overlap — spend the LDS on a taller tile instead, halving the staging
-// and barrier cost per output pixel).
-template <int ACT, int THT, bool DBUF>
+// and barrier cost per output pixel). (KHH, KWW) generalizes the tap
+// geometry: (3,3) encoder/update convs, (1,5)/(5,1) the separable GRU
+// convs (same one-stage-all-taps property).
+template <int ACT, int THT, bool DBUF, int KHH, int KWW>
 __global__ __launch_bounds__(CH_THREADS, 4) void conv_halo3_fwd_kernel(
     const __bf16* __restrict__ x, const __bf16* __restrict__ x2,
-    const __bf16* __restrict__ wpk,   // (9, Cout, cpad)
+    const __bf16* __restrict__ wpk,   // (KH*KW, Cout, cpad)
     const float* __restrict__ bias,
     __bf16* __restrict__ out,         // (N*H*W, ldo) channels-last rows
     const __bf16* __restrict__ zpage,
     int N, int H, int W, int ld_x, int ld_x2, int C1, int Cin, int Cout,
     int cpad, int ldo, int ntx, int nty, int nco) {
-  constexpr int PIECES = (THT + 2) * CH_LW * 8;
+  constexpr int LW = CH_TW + KWW - 1;
+  constexpr int PIECES = (THT + KHH - 1) * LW * 8;
   __shared__ __attribute__((aligned(16))) char lds[(DBUF ? 2 : 1) * PIECES *
                                                    16];
 
@@ -105,7 +109,8 @@ __global__ __launch_bounds__(CH_THREADS, 4) void conv_halo3_fwd_kernel(
     for (int j = 0; j < 2; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
   const int cslabs = cpad / 64;
-  ch_stage<THT>(x, x2, zpage, lds, n, y0, x0, H, W, ld_x, ld_x2, C1, Cin, 0);
+  ch_stage<THT, KHH, KWW>(x, x2, zpage, lds, n, y0, x0, H, W, ld_x, ld_x2,
+                          C1, Cin, 0);
   asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
   __syncthreads();
 
@@ -115,13 +120,14 @@ __global__ __launch_bounds__(CH_THREADS, 4) void conv_halo3_fwd_kernel(
   int cur = 0;
   for (int cs = 0; cs < cslabs; ++cs) {
     if (DBUF && cs + 1 < cslabs)
-      ch_stage<THT>(x, x2, zpage, lds + (cur ^ 1) * PIECES * 16, n, y0, x0,
-                    H, W, ld_x, ld_x2, C1, Cin, (cs + 1) * 64);
+      ch_stage<THT, KHH, KWW>(x, x2, zpage, lds + (cur ^ 1) * PIECES * 16,
+                              n, y0, x0, H, W, ld_x, ld_x2, C1, Cin,
+                              (cs + 1) * 64);
 
     const char* abuf = lds + cur * PIECES * 16;
 #pragma unroll 1
-    for (int kyx = 0; kyx < 9; ++kyx) {
-      const int ky = kyx / 3, kx = kyx - ky * 3;
+    for (int kyx = 0; kyx < KHH * KWW; ++kyx) {
+      const int ky = kyx / KWW, kx = kyx - ky * KWW;
       const __bf16* wsec = wpk + ((long)kyx * Cout + n0) * cpad + cs * 64;
 #pragma unroll
       for (int kk = 0; kk < 2; ++kk) {
@@ -139,7 +145,7 @@ __global__ __launch_bounds__(CH_THREADS, 4) void conv_halo3_fwd_kernel(
           // A fragment: 16 pixels' tap-shifted 16-B channel chunks
           const int p = wr + i * 16 + frow;           // tile pixel
           const int py = p >> 5, px = p & 31;
-          const unsigned pix = (unsigned)((py + ky) * CH_LW + (px + kx));
+          const unsigned pix = (unsigned)((py + ky) * LW + (px + kx));
           const unsigned slot = (unsigned)(kk * 4 + fk);
           const bf16x8 afr = *(const bf16x8*)(
               abuf + pix * 128 + ch_swz(pix, slot) * 16);
@@ -152,8 +158,8 @@ __global__ __launch_bounds__(CH_THREADS, 4) void conv_halo3_fwd_kernel(
     }
     if (cs + 1 < cslabs) {
       if (!DBUF)
-        ch_stage<THT>(x, x2, zpage, lds, n, y0, x0, H, W, ld_x, ld_x2, C1,
-                      Cin, (cs + 1) * 64);
+        ch_stage<THT, KHH, KWW>(x, x2, zpage, lds, n, y0, x0, H, W, ld_x,
+                                ld_x2, C1, Cin, (cs + 1) * 64);
       asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
       __syncthreads();
       if (DBUF) cur ^= 1;

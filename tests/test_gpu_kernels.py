@@ -875,22 +875,27 @@ class TestConvGemm:
         torch.testing.assert_close(w.grad.float(), w2.grad,
                                    atol=2e-1, rtol=5e-2)
 
-    @pytest.mark.parametrize("ci,co", [(96, 64), (64, 96), (128, 126)])
-    def test_halo3_path_matches_torch(self, ci, co):
-        """3x3 stride-1 at grid sizes that take the halo-staged kernel
-        (>=320 workgroups), with ragged Cin/Cout and edge tiles."""
+    @pytest.mark.parametrize("ci,co,kh,kw", [
+        (96, 64, 3, 3), (64, 96, 3, 3), (128, 126, 3, 3),
+        (128, 128, 1, 5), (128, 64, 5, 1),
+    ])
+    def test_halo_path_matches_torch(self, ci, co, kh, kw):
+        """Stride-1 3x3 / 1x5 / 5x1 at grid sizes that take the
+        halo-staged kernel (>=320 workgroups), with ragged Cin/Cout and
+        edge tiles."""
         from flowhip.ops.functional_conv import fused_conv2d
         torch.manual_seed(37)
         B, H, W = 3, 126, 130  # ragged in both tile dims
+        pad = (kh // 2, kw // 2)
         x = (torch.randn(B, ci, H, W, device=_dev()) / 8).to(torch.bfloat16) \
             .contiguous(memory_format=torch.channels_last).requires_grad_(True)
-        w = (torch.randn(co, ci, 3, 3, device=_dev()) /
-             (ci * 9) ** 0.5).requires_grad_(True)
+        w = (torch.randn(co, ci, kh, kw, device=_dev()) /
+             (ci * kh * kw) ** 0.5).requires_grad_(True)
         b = torch.randn(co, device=_dev()).requires_grad_(True)
-        out = fused_conv2d(x, w, b, 1, 1, 1, 1, {})
+        out = fused_conv2d(x, w, b, 1, pad, 1, 1, {})
         ref = torch.nn.functional.conv2d(
             x.detach().float(), w.detach().float(), b.detach().float(),
-            padding=1)
+            padding=pad)
         tol = dict(atol=5e-2, rtol=5e-2)
         torch.testing.assert_close(out.float(), ref, **tol)
         g = torch.randn_like(ref).to(torch.bfloat16) \
@@ -899,7 +904,8 @@ class TestConvGemm:
         x2 = x.detach().float().requires_grad_(True)
         w2 = w.detach().float().requires_grad_(True)
         b2 = b.detach().float().requires_grad_(True)
-        torch.nn.functional.conv2d(x2, w2, b2, padding=1).backward(g.float())
+        torch.nn.functional.conv2d(x2, w2, b2, padding=pad).backward(
+            g.float())
         torch.testing.assert_close(x.grad.float(), x2.grad, **tol)
         torch.testing.assert_close(w.grad.float(), w2.grad,
                                    atol=2e-1, rtol=5e-2)
