@@ -190,12 +190,13 @@ __global__ __launch_bounds__(CH_THREADS, 4) void conv_halo3_fwd_kernel(
   }
 }
 
-bool flowhip_conv_halo3_fwd_launch(const void* x, const void* x2,
-                                   const void* wpk, const float* bias,
-                                   void* out, const void* zpage, int N,
-                                   int H, int W, int ld_x, int ld_x2, int C1,
-                                   int Cin, int Cout, int cpad, int ldo,
-                                   int act, hipStream_t stream) {
+bool flowhip_conv_halo_fwd_launch(const void* x, const void* x2,
+                                  const void* wpk, const float* bias,
+                                  void* out, const void* zpage, int N,
+                                  int H, int W, int ld_x, int ld_x2, int C1,
+                                  int Cin, int Cout, int cpad, int ldo,
+                                  int KH, int KW, int act,
+                                  hipStream_t stream) {
   if (cpad % 64 != 0) return false;
   const int nco = fh_cdiv(Cout, 64);
   // single-slab inputs (Cin <= 64): taller single-buffered tile
@@ -204,9 +205,25 @@ bool flowhip_conv_halo3_fwd_launch(const void* x, const void* x2,
   const long blocks = (long)ntx * nty * nco * N;
   if (blocks < 320) return false;  // fill-first: generic kernel handles it
   dim3 grid((unsigned)blocks), block(CH_THREADS);
-#define CH_LAUNCH(A, T, D)                                                     hipLaunchKernelGGL((conv_halo3_fwd_kernel<A, T, D>), grid, block, 0,                            stream, (const __bf16*)x, (const __bf16*)x2,                                 (const __bf16*)wpk, bias, (__bf16*)out,                                      (const __bf16*)zpage, N, H, W, ld_x, ld_x2, C1, Cin,                         Cout, cpad, ldo, ntx, nty, nco)
-  if (THT == 8) { if (act == 1) CH_LAUNCH(1, 8, false); else CH_LAUNCH(0, 8, false); }
-  else { if (act == 1) CH_LAUNCH(1, 4, true); else CH_LAUNCH(0, 4, true); }
+#define CH_LAUNCH(A, T, D, KHH, KWW)                                         \
+  hipLaunchKernelGGL((conv_halo3_fwd_kernel<A, T, D, KHH, KWW>), grid,       \
+                     block, 0, stream, (const __bf16*)x,                     \
+                     (const __bf16*)x2, (const __bf16*)wpk, bias,            \
+                     (__bf16*)out, (const __bf16*)zpage, N, H, W, ld_x,      \
+                     ld_x2, C1, Cin, Cout, cpad, ldo, ntx, nty, nco)
+#define CH_GEOM(KHH, KWW)                                                    \
+  if (KH == KHH && KW == KWW) {                                              \
+    if (THT == 8) {                                                          \
+      if (act == 1) CH_LAUNCH(1, 8, false, KHH, KWW);                        \
+      else CH_LAUNCH(0, 8, false, KHH, KWW);                                 \
+    } else {                                                                 \
+      if (act == 1) CH_LAUNCH(1, 4, true, KHH, KWW);                         \
+      else CH_LAUNCH(0, 4, true, KHH, KWW);                                  \
+    }                                                                        \
+    return true;                                                             \
+  }
+  CH_GEOM(3, 3) CH_GEOM(1, 5) CH_GEOM(5, 1)
+#undef CH_GEOM
 #undef CH_LAUNCH
-  return true;
+  return false;
 }
