@@ -993,14 +993,10 @@ torch::Tensor col_sum_bf16(torch::Tensor dy) {
   const int C = dy.size(1);
   const long M = dy.numel() / C;
   const int nchunk = (int)((M + 255) / 256);  // must cover ALL rows (kernel bounds come from the chunk index)
-  const int ncb = (C + 63) / 64;
-  auto partials = torch::empty({(long)nchunk * ncb * 64},
-                               dy.options().dtype(torch::kFloat32));
-  auto out = torch::empty({(long)C}, dy.options().dtype(torch::kFloat32));
+  auto out = torch::zeros({(long)C}, dy.options().dtype(torch::kFloat32));
   const c10::cuda::CUDAGuard guard(dy.device());
   hipStream_t stream = at::cuda::getCurrentCUDAStream().stream();
-  bool ok = flowhip_col_sum_launch(dy.data_ptr(),
-                                   partials.data_ptr<float>(),
+  bool ok = flowhip_col_sum_launch(dy.data_ptr(), nullptr,
                                    out.data_ptr<float>(), M, C, nchunk,
                                    stream);
   TORCH_CHECK(ok, "col_sum_bf16: C must be a multiple of 8");
@@ -1058,14 +1054,10 @@ torch::Tensor col_sum2_bf16(torch::Tensor g, torch::Tensor x) {
   const int C = g.size(1);
   const long M = g.numel() / C;
   const int nchunk = (int)((M + 255) / 256);
-  const int ncb = (C + 63) / 64;
-  auto partials = torch::empty({(long)nchunk * 2 * ncb * 64},
-                               g.options().dtype(torch::kFloat32));
-  auto out = torch::empty({2, (long)C}, g.options().dtype(torch::kFloat32));
+  auto out = torch::zeros({2, (long)C}, g.options().dtype(torch::kFloat32));
   const c10::cuda::CUDAGuard guard(g.device());
   hipStream_t stream = at::cuda::getCurrentCUDAStream().stream();
-  bool ok = flowhip_col_sum2_launch(g.data_ptr(), x.data_ptr(),
-                                    partials.data_ptr<float>(),
+  bool ok = flowhip_col_sum2_launch(g.data_ptr(), x.data_ptr(), nullptr,
                                     out.data_ptr<float>(), M, C, nchunk,
                                     stream);
   TORCH_CHECK(ok, "col_sum2_bf16: C must be a multiple of 8");

@@ -48,35 +48,17 @@ __global__ __launch_bounds__(CS_THREADS) void col_sum_partial_kernel(
     float t = 0.f;
     const int smax = S < 32 ? S : 32;
     for (int ss = 0; ss < smax; ++ss) t += red[threadIdx.x * 33 + ss];
-    partials[((long)chunk * ncb + cb) * 64 + threadIdx.x] = t;
+    // one device-scope atomic per (chunk, channel) straight into the
+    // output (out is zero-filled by the caller): drops the separate
+    // finalize launch + the partials round trip (~10 us/call of
+    // latency-bound reduce across ~160 bias grads per step). Accumulation
+    // order across chunks is non-deterministic — like torch's own
+    // multi-block reductions; magnitudes are fp32 partial sums of
+    // comparable scale.
+    atomicAdd(&partials[cb * 64 + threadIdx.x], t);
   }
 }
 
-// One block per 64-channel group; the chunk loop is split across 8 thread
-// slices (tid/64) so the dependent-load chain is 8x shorter and each
-// slice's loads pipeline independently (the old single-wave version was
-// latency-bound at ~43 us for a few hundred chunks).
-__global__ __launch_bounds__(512) void col_sum_finalize_kernel(
-    const float* __restrict__ partials, float* __restrict__ out, int C,
-    int nchunk) {
-  const int ncb = (C + 63) / 64;
-  const int cb = blockIdx.x;
-  const int lane = threadIdx.x & 63;
-  const int slice = threadIdx.x >> 6;  // 0..7
-  float t = 0.f;
-  for (int ch = slice; ch < nchunk; ch += 8)
-    t += partials[((long)ch * ncb + cb) * 64 + lane];
-  __shared__ float red[8][64];
-  red[slice][lane] = t;
-  __syncthreads();
-  const int c = cb * 64 + threadIdx.x;
-  if (threadIdx.x < 64 && c < C) {
-    float s = 0.f;
-#pragma unroll
-    for (int k = 0; k < 8; ++k) s += red[k][threadIdx.x];
-    out[c] = s;
-  }
-}
 
 // Dual column sum for the frozen-BatchNorm backward: sum_m g[m,c] AND
 // sum_m g[m,c]*x[m,c] in one pass over both tensors (torch's
@@ -135,41 +117,13 @@ __global__ __launch_bounds__(CS_THREADS) void col_sum2_partial_kernel(
       t0 += red[0][threadIdx.x * 33 + ss];
       t1 += red[1][threadIdx.x * 33 + ss];
     }
-    float* pc = partials + ((long)chunk * 2) * (ncb * 64);
-    pc[cb * 64 + threadIdx.x] = t0;
-    pc[ncb * 64 + cb * 64 + threadIdx.x] = t1;
+    // atomic chunk reduction straight into the zero-filled (2, C) output
+    // (see col_sum_partial_kernel)
+    atomicAdd(&partials[cb * 64 + threadIdx.x], t0);
+    atomicAdd(&partials[C + cb * 64 + threadIdx.x], t1);
   }
 }
 
-__global__ __launch_bounds__(512) void col_sum2_finalize_kernel(
-    const float* __restrict__ partials, float* __restrict__ out, int C,
-    int nchunk) {
-  const int ncb = (C + 63) / 64;
-  const int cb = blockIdx.x;
-  const int lane = threadIdx.x & 63;
-  const int slice = threadIdx.x >> 6;
-  float t0 = 0.f, t1 = 0.f;
-  for (int ch = slice; ch < nchunk; ch += 8) {
-    const float* pc = partials + ((long)ch * 2) * (ncb * 64);
-    t0 += pc[cb * 64 + lane];
-    t1 += pc[ncb * 64 + cb * 64 + lane];
-  }
-  __shared__ float red[2][8][64];
-  red[0][slice][lane] = t0;
-  red[1][slice][lane] = t1;
-  __syncthreads();
-  const int c = cb * 64 + threadIdx.x;
-  if (threadIdx.x < 64 && c < C) {
-    float s0 = 0.f, s1 = 0.f;
-#pragma unroll
-    for (int k = 0; k < 8; ++k) {
-      s0 += red[0][k][threadIdx.x];
-      s1 += red[1][k][threadIdx.x];
-    }
-    out[c] = s0;
-    out[C + c] = s1;
-  }
-}
 
 // Per-channel plane reduction on NCHW fp32: out[c] = sum_{b,h,w} x (or
 // x*y). Replaces the eager mul+sum pair in the nconv backward's
@@ -266,23 +220,21 @@ bool flowhip_col_sum2_launch(const void* g, const void* x, float* partials,
                              float* out, long M, int C, int nchunk,
                              hipStream_t stream) {
   if (C % 8 != 0 || C < 8) return false;
-  const int ncb = (C + 63) / 64;
-  hipLaunchKernelGGL(col_sum2_partial_kernel, dim3(ncb * nchunk),
+  (void)partials;
+  hipLaunchKernelGGL(col_sum2_partial_kernel, dim3(((C + 63) / 64) * nchunk),
                      dim3(CS_THREADS), 0, stream, (const __bf16*)g,
-                     (const __bf16*)x, partials, M, C, nchunk);
-  hipLaunchKernelGGL(col_sum2_finalize_kernel, dim3(ncb), dim3(512), 0,
-                     stream, partials, out, C, nchunk);
+                     (const __bf16*)x, out, M, C, nchunk);
   return true;
 }
 
 bool flowhip_col_sum_launch(const void* dy, float* partials, float* out,
                             long M, int C, int nchunk, hipStream_t stream) {
   if (C % 8 != 0 || C < 8) return false;
-  const int ncb = (C + 63) / 64;
-  hipLaunchKernelGGL(col_sum_partial_kernel, dim3(ncb * nchunk),
+  (void)partials;
+  // `out` must be zero-filled (the binding allocates torch::zeros); the
+  // partial kernel reduces into it atomically — no finalize pass.
+  hipLaunchKernelGGL(col_sum_partial_kernel, dim3(((C + 63) / 64) * nchunk),
                      dim3(CS_THREADS), 0, stream, (const __bf16*)dy,
-                     partials, M, C, nchunk);
-  hipLaunchKernelGGL(col_sum_finalize_kernel, dim3(ncb), dim3(512), 0, stream,
-                     partials, out, C, nchunk);
+                     out, M, C, nchunk);
   return true;
 }
