@@ -136,15 +136,15 @@ void flowhip_nconv_bwd_prep_launch(const float* gout, const float* gcout,
                                    float* dnomin, float* ddenom, long total,
                                    long plane, int Co, float eps,
                                    hipStream_t stream);
-bool flowhip_col_sum2_launch(const void* g, const void* x, float* partials,
-                             float* out, long M, int C, int nchunk,
+bool flowhip_col_sum2_launch(const void* g, const void* x, float* out,
+                             long M, int C, int nchunk,
                              hipStream_t stream);
 void flowhip_frozen_bn_apply_launch(const void* x, void* y, const float* s,
                                     const float* t, long M, int C,
                                     hipStream_t stream);
 void flowhip_plane_dot_sum_launch(const float* x, const float* y, float* out,
                                   long P, int B, int C, hipStream_t stream);
-bool flowhip_col_sum_launch(const void* dy, float* partials, float* out,
+bool flowhip_col_sum_launch(const void* dy, float* out,
                             long M, int C, int nchunk, hipStream_t stream);
 void flowhip_up2x_cat_fwd_launch(const float* low, const float* skip,
                                  float* out, long total, int C1, int C2,
@@ -996,7 +996,7 @@ torch::Tensor col_sum_bf16(torch::Tensor dy) {
   auto out = torch::zeros({(long)C}, dy.options().dtype(torch::kFloat32));
   const c10::cuda::CUDAGuard guard(dy.device());
   hipStream_t stream = at::cuda::getCurrentCUDAStream().stream();
-  bool ok = flowhip_col_sum_launch(dy.data_ptr(), nullptr,
+  bool ok = flowhip_col_sum_launch(dy.data_ptr(),
                                    out.data_ptr<float>(), M, C, nchunk,
                                    stream);
   TORCH_CHECK(ok, "col_sum_bf16: C must be a multiple of 8");
@@ -1057,7 +1057,7 @@ torch::Tensor col_sum2_bf16(torch::Tensor g, torch::Tensor x) {
   auto out = torch::zeros({2, (long)C}, g.options().dtype(torch::kFloat32));
   const c10::cuda::CUDAGuard guard(g.device());
   hipStream_t stream = at::cuda::getCurrentCUDAStream().stream();
-  bool ok = flowhip_col_sum2_launch(g.data_ptr(), x.data_ptr(), nullptr,
+  bool ok = flowhip_col_sum2_launch(g.data_ptr(), x.data_ptr(),
                                     out.data_ptr<float>(), M, C, nchunk,
                                     stream);
   TORCH_CHECK(ok, "col_sum2_bf16: C must be a multiple of 8");

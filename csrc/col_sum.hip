@@ -1,7 +1,8 @@
 // Column sum for conv bias gradients: (M, C) bf16 channels-last rows ->
 // (C) fp32, dbias[c] = sum_m dy[m, c]. torch's strided reduce over the
 // pixel dims of a channels-last tensor runs ~22 us for 5.5 MB (~100 calls
-// per step); this is a chunked 16-B-load partial reduction + tiny finalize.
+// per step); this is a chunked 16-B-load reduction whose per-chunk rows
+// fold into the zero-filled output with one atomic per (chunk, channel).
 
 #include "common.h"
 
@@ -9,9 +10,8 @@
 #define CS_MCHUNK 256  // small chunks: enough blocks to fill 256 CUs at M~21k
 
 __global__ __launch_bounds__(CS_THREADS) void col_sum_partial_kernel(
-    const __bf16* __restrict__ dy, float* __restrict__ partials, long M,
+    const __bf16* __restrict__ dy, float* __restrict__ out, long M,
     int C, int nchunk) {
-  const int ncb = (C + 63) / 64;
   int b = blockIdx.x;
   const int chunk = b % nchunk; b /= nchunk;
   const int cb = b;
@@ -55,7 +55,7 @@ __global__ __launch_bounds__(CS_THREADS) void col_sum_partial_kernel(
     // order across chunks is non-deterministic — like torch's own
     // multi-block reductions; magnitudes are fp32 partial sums of
     // comparable scale.
-    atomicAdd(&partials[cb * 64 + threadIdx.x], t);
+    atomicAdd(&out[cb * 64 + threadIdx.x], t);
   }
 }
 
@@ -63,11 +63,10 @@ __global__ __launch_bounds__(CS_THREADS) void col_sum_partial_kernel(
 // Dual column sum for the frozen-BatchNorm backward: sum_m g[m,c] AND
 // sum_m g[m,c]*x[m,c] in one pass over both tensors (torch's
 // native_batch_norm_backward spends ~105 us/call on the same reduction
-// pair at encoder shapes). partials layout: (chunk, 2, ncb*64).
+// pair at encoder shapes). out layout: (2, C), zero-filled by the caller.
 __global__ __launch_bounds__(CS_THREADS) void col_sum2_partial_kernel(
     const __bf16* __restrict__ g, const __bf16* __restrict__ x,
-    float* __restrict__ partials, long M, int C, int nchunk) {
-  const int ncb = (C + 63) / 64;
+    float* __restrict__ out, long M, int C, int nchunk) {
   int b = blockIdx.x;
   const int chunk = b % nchunk; b /= nchunk;
   const int cb = b;
@@ -119,8 +118,8 @@ __global__ __launch_bounds__(CS_THREADS) void col_sum2_partial_kernel(
     }
     // atomic chunk reduction straight into the zero-filled (2, C) output
     // (see col_sum_partial_kernel)
-    atomicAdd(&partials[cb * 64 + threadIdx.x], t0);
-    atomicAdd(&partials[C + cb * 64 + threadIdx.x], t1);
+    atomicAdd(&out[cb * 64 + threadIdx.x], t0);
+    atomicAdd(&out[C + cb * 64 + threadIdx.x], t1);
   }
 }
 
@@ -216,23 +215,21 @@ void flowhip_frozen_bn_apply_launch(const void* x, void* y, const float* s,
                      C / 8);
 }
 
-bool flowhip_col_sum2_launch(const void* g, const void* x, float* partials,
-                             float* out, long M, int C, int nchunk,
+bool flowhip_col_sum2_launch(const void* g, const void* x, float* out,
+                             long M, int C, int nchunk,
                              hipStream_t stream) {
   if (C % 8 != 0 || C < 8) return false;
-  (void)partials;
   hipLaunchKernelGGL(col_sum2_partial_kernel, dim3(((C + 63) / 64) * nchunk),
                      dim3(CS_THREADS), 0, stream, (const __bf16*)g,
                      (const __bf16*)x, out, M, C, nchunk);
   return true;
 }
 
-bool flowhip_col_sum_launch(const void* dy, float* partials, float* out,
+bool flowhip_col_sum_launch(const void* dy, float* out,
                             long M, int C, int nchunk, hipStream_t stream) {
   if (C % 8 != 0 || C < 8) return false;
-  (void)partials;
   // `out` must be zero-filled (the binding allocates torch::zeros); the
-  // partial kernel reduces into it atomically — no finalize pass.
+  // chunk kernel reduces into it atomically — no finalize pass.
   hipLaunchKernelGGL(col_sum_partial_kernel, dim3(((C + 63) / 64) * nchunk),
                      dim3(CS_THREADS), 0, stream, (const __bf16*)dy,
                      out, M, C, nchunk);
