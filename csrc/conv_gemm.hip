@@ -366,7 +366,27 @@ __global__ __launch_bounds__(CG_THREADS, 2) void conv_gemm_wrw_kernel(
           16, 0, 0);
     }
   };
-  auto stage_x = [&](long mt, char* buf) {
+  // incremental tile-base coordinates: one 64-bit decomposition of the
+  // first staged m, then +CG_BM carries per k-tile — the per-piece
+  // m%W / m/W divisions were ~30 runtime instructions each on the
+  // staging critical path
+  long base_m = t0 * CG_BM;
+  int bx = 0, by = 0;
+  long bn = 0;
+  if (t0 < t1) {
+    bx = (int)(base_m % WW);
+    by = (int)((base_m / WW) % HH);
+    bn = base_m / ((long)WW * HH);
+  }
+  auto advance_base = [&]() {
+    base_m += CG_BM;
+    bx += CG_BM;
+    while (bx >= WW) {
+      bx -= WW;
+      if (++by == HH) { by = 0; ++bn; }
+    }
+  };
+  auto stage_x = [&](char* buf) {
 #pragma unroll
     for (int j = 0; j < 2; ++j) {
       const int piece0 = wave * 64 + CG_THREADS * j;
@@ -375,12 +395,16 @@ __global__ __launch_bounds__(CG_THREADS, 2) void conv_gemm_wrw_kernel(
       const int pr = (piece >> 1) & 3;
       const int ch = piece & 1;
       const int irow = (sb & 15) * 4 + pr;
-      const long m = mt * CG_BM + wrw_row_to_m(irow);
+      const int dm = wrw_row_to_m(irow);
+      const long m = base_m + dm;
       const __bf16* src = zpage;
       if (m < Mtot) {
-        const int xx = (int)(m % WW);
-        const int yy = (int)((m / WW) % HH);
-        const long n = m / ((long)WW * HH);
+        int xx = bx + dm, yy = by;
+        long n = bn;
+        while (xx >= WW) {
+          xx -= WW;
+          if (++yy == HH) { yy = 0; ++n; }
+        }
         const int sy = yy * sH + dyo, sx = xx * sW + dxo;
         const int c = c0 + (sb >> 4) * 16 + ch * 8;
         if (sy >= 0 && sy < srcH && sx >= 0 && sx < srcW && c < Cin) {
@@ -399,7 +423,8 @@ __global__ __launch_bounds__(CG_THREADS, 2) void conv_gemm_wrw_kernel(
 
   if (t0 < t1) {
     stage_dy(t0, lds);
-    stage_x(t0, lds + TS);
+    stage_x(lds + TS);
+    advance_base();
   }
   asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
   __syncthreads();
@@ -408,7 +433,8 @@ __global__ __launch_bounds__(CG_THREADS, 2) void conv_gemm_wrw_kernel(
   for (long t = t0; t < t1; ++t) {
     if (t + 1 < t1) {
       stage_dy(t + 1, lds + (cur ^ 1) * 2 * TS);
-      stage_x(t + 1, lds + (cur ^ 1) * 2 * TS + TS);
+      stage_x(lds + (cur ^ 1) * 2 * TS + TS);
+      advance_base();
     }
 
     // hardware-transposed fragment reads (ds_read_b64_tr_b16): fragment
