@@ -54,13 +54,32 @@ __device__ __forceinline__ unsigned cg_swz(unsigned row, unsigned slot) {
 // cat (C1 must be a multiple of 64 so a slab never straddles sources).
 // SMODE semantics in the header comment; (dy, dx) carry the per-mode tap
 // offset, (srcH, srcW) the source image dims, (HH, WW) the m-mapping dims.
+// per-thread pixel coordinates of this block's A rows: computed ONCE per
+// block (the m-range is subtile-invariant; recomputing the 64-bit
+// divisions per piece per subtile was a measured ~5% of the step)
+template <int BM, int THREADS>
+__device__ __forceinline__ void cg_coords(
+    int wave, int lane, long m0, long Mtot, int HH, int WW,
+    long* pn, int* pyy, int* pxx) {
+  constexpr int PPT = BM * 8 / THREADS;
+#pragma unroll
+  for (int j = 0; j < PPT; ++j) {
+    const int piece = wave * 64 + THREADS * j + lane;
+    long m = m0 + (piece >> 3);
+    if (m >= Mtot) m = Mtot - 1;
+    pxx[j] = (int)(m % WW);
+    pyy[j] = (int)((m / WW) % HH);
+    pn[j] = m / ((long)WW * HH);
+  }
+}
+
 template <int BM, int THREADS, int SMODE>
 __device__ __forceinline__ void cg_stage_a(
     const __bf16* __restrict__ x, const __bf16* __restrict__ x2,
     const __bf16* __restrict__ zpage,
-    char* lds_buf, int wave, int lane, long m0, long Mtot, int HH, int WW,
-    int srcH, int srcW, int sH, int sW, int ld_x, int ld_x2, int C1, int Cin,
-    int dy, int dx, int c0) {
+    char* lds_buf, int wave, int lane, const long* pn, const int* pyy,
+    const int* pxx, int srcH, int srcW, int sH, int sW, int ld_x, int ld_x2,
+    int C1, int Cin, int dy, int dx, int c0) {
   constexpr int PPT = BM * 8 / THREADS;
 #pragma unroll
   for (int j = 0; j < PPT; ++j) {
@@ -69,11 +88,9 @@ __device__ __forceinline__ void cg_stage_a(
     const int row = piece >> 3;          // 0..BM-1
     const int slot = piece & 7;
     const int sslot = cg_swz(row, slot);
-    long m = m0 + row;
-    if (m >= Mtot) m = Mtot - 1;
-    const int xx = (int)(m % WW);
-    const int yy = (int)((m / WW) % HH);
-    const long n = m / ((long)WW * HH);
+    const int xx = pxx[j];
+    const int yy = pyy[j];
+    const long n = pn[j];
     int sy, sx;
     bool ok;
     if (SMODE == 0) {
@@ -172,9 +189,13 @@ void conv_gemm_fwd_kernel(
   const int cslabs = cpad / CG_BK;
   const int nsub = KH * KW * cslabs;
 
+  long pn[BM * 8 / THREADS];
+  int pyy[BM * 8 / THREADS], pxx[BM * 8 / THREADS];
+  cg_coords<BM, THREADS>(wave, lane, m0, Mtot, HH, WW, pn, pyy, pxx);
+
   // subtile s -> (kyx = s / cslabs, c0 = (s % cslabs) * 64)
-  cg_stage_a<BM, THREADS, SMODE>(x, x2, zpage, lds, wave, lane, m0, Mtot, HH,
-                                 WW, srcH, srcW, sH, sW, ld_x, ld_x2, C1,
+  cg_stage_a<BM, THREADS, SMODE>(x, x2, zpage, lds, wave, lane, pn, pyy,
+                                 pxx, srcH, srcW, sH, sW, ld_x, ld_x2, C1,
                                  Cin, offH, offW, 0);
   cg_stage_b<THREADS>(wpk, lds + TSA, wave, lane, n0, Cout, cpad, 0);
   asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
@@ -187,8 +208,8 @@ void conv_gemm_fwd_kernel(
       const int cs = (s + 1) - kyx * cslabs;
       const int ky = kyx / KW, kx = kyx - ky * KW;
       cg_stage_a<BM, THREADS, SMODE>(
-          x, x2, zpage, lds + (cur ^ 1) * (TSA + TSB), wave, lane, m0, Mtot,
-          HH, WW, srcH, srcW, sH, sW, ld_x, ld_x2, C1, Cin, ky + offH,
+          x, x2, zpage, lds + (cur ^ 1) * (TSA + TSB), wave, lane, pn, pyy,
+          pxx, srcH, srcW, sH, sW, ld_x, ld_x2, C1, Cin, ky + offH,
           kx + offW, cs * CG_BK);
       cg_stage_b<THREADS>(wpk + (long)kyx * Cout * cpad,
                           lds + (cur ^ 1) * (TSA + TSB) + TSA, wave, lane,
