@@ -84,6 +84,7 @@ void flowhip_conv_gemm_fwd_launch(const void* x, const void* x2,
                                   hipStream_t stream);
 void flowhip_conv_gemm_wrw_launch(const void* dy, const void* x,
                                   const void* x2, float* partials, float* dw,
+                                  float* dbias,
                                   const void* zpage, long Mtot, int HH,
                                   int WW, int srcH, int srcW, int sH, int sW,
                                   int ld_x, int ld_x2, int C1,
@@ -936,9 +937,10 @@ torch::Tensor conv_gemm_fwd(torch::Tensor x, torch::Tensor wpk,
                         act, 1, 1, 0, 0, 0)[0];
 }
 
-torch::Tensor conv_gemm_wrw(torch::Tensor dy, torch::Tensor x,
-                            c10::optional<torch::Tensor> x2, int64_t KH,
-                            int64_t KW, int64_t sH, int64_t sW) {
+std::vector<torch::Tensor> conv_gemm_wrw(torch::Tensor dy, torch::Tensor x,
+                                         c10::optional<torch::Tensor> x2,
+                                         int64_t KH, int64_t KW, int64_t sH,
+                                         int64_t sW, bool want_bias) {
   int ld_x, ld_x2 = 0;
   cg_check_x(x, ld_x);
   TORCH_CHECK(dy.is_cuda() && dy.scalar_type() == torch::kBFloat16 &&
@@ -969,21 +971,29 @@ torch::Tensor conv_gemm_wrw(torch::Tensor dy, torch::Tensor x,
   // never more chunks than 64-row m-tiles
   const long mtiles = (Mtot + 63) / 64;
   if (nchunk > mtiles) nchunk = (int)mtiles;
+  // bias partials (nchunk, tiles_o*64) ride at the tail of the same
+  // buffer; dbias is written fully by the reduce kernel — no zero-fill
   auto partials = torch::empty(
-      {(long)nchunk * KH * KW * tiles_o * 64 * cpad},
+      {(long)nchunk * KH * KW * tiles_o * 64 * cpad +
+       (want_bias ? (long)nchunk * tiles_o * 64 : 0)},
       x.options().dtype(torch::kFloat32));
   auto dw = torch::empty({(long)Cout, (long)Cin, KH, KW},
                          x.options().dtype(torch::kFloat32));
+  torch::Tensor dbias;
+  if (want_bias)
+    dbias = torch::empty({(long)Cout}, x.options().dtype(torch::kFloat32));
   const c10::cuda::CUDAGuard guard(x.device());
   hipStream_t stream = at::cuda::getCurrentCUDAStream().stream();
   flowhip_conv_gemm_wrw_launch(dy.data_ptr(), x.data_ptr(), x2p,
                                partials.data_ptr<float>(),
-                               dw.data_ptr<float>(), cg_zero_page(), Mtot,
+                               dw.data_ptr<float>(),
+                               want_bias ? dbias.data_ptr<float>() : nullptr,
+                               cg_zero_page(), Mtot,
                                OH, OW, H, W, (int)sH, (int)sW,
                                ld_x, ld_x2, C1, Cin, Cout, cpad, (int)KH,
                                (int)KW, (int)KH / 2, (int)KW / 2, nchunk,
                                stream);
-  return dw;
+  return {dw, dbias};
 }
 
 torch::Tensor col_sum_bf16(torch::Tensor dy) {
@@ -1297,9 +1307,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("act"), py::arg("sH") = 1, py::arg("sW") = 1,
         py::arg("smode") = 0, py::arg("outH") = 0, py::arg("outW") = 0);
   m.def("conv_gemm_wrw", &conv_gemm_wrw,
-        "implicit-GEMM conv weight gradient (split-M + reduce)",
+        "implicit-GEMM conv weight gradient (split-M + reduce); returns "
+        "[dw, dbias] — dbias undefined unless want_bias",
         py::arg("dy"), py::arg("x"), py::arg("x2"), py::arg("KH"),
-        py::arg("KW"), py::arg("sH") = 1, py::arg("sW") = 1);
+        py::arg("KW"), py::arg("sH") = 1, py::arg("sW") = 1,
+        py::arg("want_bias") = false);
   m.def("instnorm_cl_fwd", &instnorm_cl_fwd,
         "channels-last InstanceNorm2d forward (y, mean, rstd)");
   m.def("instnorm_cl_bwd", &instnorm_cl_bwd,

@@ -328,6 +328,7 @@ __global__ __launch_bounds__(CG_THREADS, 2) void conv_gemm_wrw_kernel(
     const __bf16* __restrict__ x,   // (N, srcH, srcW, ld_x)
     const __bf16* __restrict__ x2,  // second source or nullptr
     float* __restrict__ partials,
+    float* __restrict__ biasp,  // (nchunk, tiles_o*64) or nullptr
     const __bf16* __restrict__ zpage,
     long Mtot, int HH, int WW, int srcH, int srcW, int sH, int sW, int ld_x,
     int ld_x2, int C1, int Cin, int Cout, int cpad, int KH, int KW, int padH,
@@ -360,6 +361,14 @@ __global__ __launch_bounds__(CG_THREADS, 2) void conv_gemm_wrw_kernel(
   for (int i = 0; i < 2; ++i)
 #pragma unroll
     for (int j = 0; j < 2; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  // bias gradient for free: dbias[o] = sum_m dy[m, o], and the (tc==0,
+  // kyx==0) blocks' even waves already hold every dy fragment exactly
+  // once — summing the a-operand elements as they pass gives the
+  // per-chunk column sums with no extra launch and no extra global reads
+  // (replaces a separate col_sum + torch.zeros pair per conv backward).
+  const bool do_bias = biasp != nullptr && tc == 0 && kyx == 0 && wc == 0;
+  float bacc[2] = {0.f, 0.f};
 
   const unsigned TS = CG_BM * CG_BK * 2;
   // piece -> permuted-image coordinates: subblock sb = cb*16 + mq holds
@@ -495,6 +504,10 @@ __global__ __launch_bounds__(CG_THREADS, 2) void conv_gemm_wrw_kernel(
       for (int i = 0; i < 2; ++i) {
         const bf16x8 af = __builtin_shufflevector(a0[i], a1[i], 0, 1, 2, 3,
                                                   4, 5, 6, 7);
+        if (do_bias) {
+#pragma unroll
+          for (int e = 0; e < 8; ++e) bacc[i] += (float)af[e];
+        }
 #pragma unroll
         for (int j = 0; j < 2; ++j) {
           const bf16x8 bf = __builtin_shufflevector(b0[j], b1[j], 0, 1, 2,
@@ -530,12 +543,33 @@ __global__ __launch_bounds__(CG_THREADS, 2) void conv_gemm_wrw_kernel(
       }
     }
   }
+
+  if (do_bias) {
+    // a-fragment lane l holds o = l&15; lane groups l>>4 hold disjoint
+    // m runs — xor-fold them, lane<16 owns the per-chunk column sum
+#pragma unroll
+    for (int i = 0; i < 2; ++i) {
+      float s = bacc[i];
+      s += __shfl_xor(s, 16, 64);
+      s += __shfl_xor(s, 32, 64);
+      if (lane < 16)
+        biasp[(long)chunk * (tiles_o * 64) + o0 + wr + i * 16 + lane] = s;
+    }
+  }
 }
 
 // reduce partials over chunks and scatter into dW (Cout, Cin, KH, KW) fp32
 __global__ __launch_bounds__(CG_THREADS) void conv_gemm_wrw_reduce_kernel(
     const float* __restrict__ partials, float* __restrict__ dw,
+    const float* __restrict__ biasp, float* __restrict__ dbias,
     int nchunk, int KYX, int orows, int cpad, int Cout, int Cin) {
+  if (dbias)
+    for (int o = blockIdx.x * CG_THREADS + threadIdx.x; o < Cout;
+         o += gridDim.x * CG_THREADS) {
+      float s = 0.f;
+      for (int ch = 0; ch < nchunk; ++ch) s += biasp[(long)ch * orows + o];
+      dbias[o] = s;
+    }
   const long total = (long)Cout * Cin * KYX;
   for (long idx = (long)blockIdx.x * CG_THREADS + threadIdx.x; idx < total;
        idx += (long)gridDim.x * CG_THREADS) {
@@ -630,7 +664,8 @@ void flowhip_conv_gemm_fwd_launch(const void* x, const void* x2,
 
 void flowhip_conv_gemm_wrw_launch(const void* dy, const void* x,
                                   const void* x2, float* partials, float* dw,
-                                  const void* zpage, long Mtot, int HH,
+                                  float* dbias, const void* zpage, long Mtot,
+                                  int HH,
                                   int WW, int srcH, int srcW, int sH, int sW,
                                   int ld_x, int ld_x2, int C1, int Cin,
                                   int Cout, int cpad, int KH, int KW,
@@ -638,6 +673,10 @@ void flowhip_conv_gemm_wrw_launch(const void* dy, const void* x,
                                   hipStream_t stream) {
   const int tiles_o = fh_cdiv(Cout, 64);
   const int tiles_c = fh_cdiv(cpad, 64);
+  // bias partials live past the weight partials (caller sized the buffer)
+  float* biasp = dbias
+      ? partials + (long)nchunk * KH * KW * tiles_o * 64 * cpad
+      : nullptr;
   dim3 block(CG_THREADS);
   // (a KW=3 tap-sharing wrw variant was measured SLOWER despite 3x less
   // operand re-read: the 4-image LDS set halves occupancy and the
@@ -646,7 +685,8 @@ void flowhip_conv_gemm_wrw_launch(const void* dy, const void* x,
   dim3 grid(tiles_o * tiles_c, KH * KW, nchunk);
   hipLaunchKernelGGL(conv_gemm_wrw_kernel, grid, block, 0, stream,
                      (const __bf16*)dy, (const __bf16*)x,
-                     (const __bf16*)x2, partials, (const __bf16*)zpage,
+                     (const __bf16*)x2, partials, biasp,
+                     (const __bf16*)zpage,
                      Mtot, HH, WW, srcH, srcW, sH, sW, ld_x, ld_x2, C1,
                      Cin, Cout, cpad, KH, KW, padH, padW, tiles_o, tiles_c,
                      nchunk);
@@ -654,8 +694,8 @@ void flowhip_conv_gemm_wrw_launch(const void* dy, const void* x,
   long rblocks = (total + CG_THREADS - 1) / CG_THREADS;
   if (rblocks > 4096) rblocks = 4096;
   hipLaunchKernelGGL(conv_gemm_wrw_reduce_kernel, dim3((int)rblocks), block,
-                     0, stream, partials, dw, nchunk, KH * KW, tiles_o * 64,
-                     cpad, Cout, Cin);
+                     0, stream, partials, dw, biasp, dbias, nchunk, KH * KW,
+                     tiles_o * 64, cpad, Cout, Cin);
 }
 
 // ---------------------------------------------------------------------------

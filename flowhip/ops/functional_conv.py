@@ -51,14 +51,6 @@ def _pad_dy8(dy):
     return torch.cat([dy, z], dim=1), O
 
 
-def _col_sum(dy):
-    """bias grad: one chunked column-sum kernel (torch's strided reduce on a
-    channels-last tensor is ~4x slower)."""
-    if dy.shape[1] % 8 == 0:
-        return _ext.ext().col_sum_bf16(dy)
-    return dy.sum(dim=(0, 2, 3), dtype=torch.float32)
-
-
 def _pack_fwd(weight):
     """(O, I, KH, KW) fp32 -> (KYX, O, pad64(I)) bf16 contiguous."""
     w = weight.detach()
@@ -124,10 +116,11 @@ class ConvGemmFn(torch.autograd.Function):
                 dx = _ext.ext().conv_gemm_fwd2(dyp, None, wpk_bwd, None, I,
                                                KH, KW, 0, 0, sH, sW, 2,
                                                x.shape[2], x.shape[3])[0]
-        dw = _ext.ext().conv_gemm_wrw(dyp, x, None, KH, KW, sH, sW)
+        dw, db = _ext.ext().conv_gemm_wrw(dyp, x, None, KH, KW, sH, sW,
+                                          has_bias)
         if dw.shape[0] != O_real:
             dw = dw[:O_real].contiguous()
-        dbias = _col_sum(dy) if has_bias else None
+        dbias = db[:O_real] if has_bias else None
         return dx, dw, dbias, None, None, None, None
 
 
@@ -220,10 +213,11 @@ class ConvGemmCat2Fn(torch.autograd.Function):
         dyp, O_real = _pad_dy8(dy)
         dx1, dx2 = _ext.ext().conv_gemm_fwd2(dyp, None, wpk_bwd, None, I, KH,
                                              KW, C1, 0)
-        dw = _ext.ext().conv_gemm_wrw(dyp, x1, x2, KH, KW)
+        dw, db = _ext.ext().conv_gemm_wrw(dyp, x1, x2, KH, KW, 1, 1,
+                                          has_bias)
         if dw.shape[0] != O_real:
             dw = dw[:O_real].contiguous()
-        dbias = _col_sum(dy) if has_bias else None
+        dbias = db[:O_real] if has_bias else None
         return dx1, dx2, dw, dbias, None, None
 
 
@@ -255,8 +249,7 @@ class ConvGemmCat2ZrFn(torch.autograd.Function):
             dy = dy.to(torch.bfloat16)
         dx1, dx2 = _ext.ext().conv_gemm_fwd2(dy, None, wpk_bwd, None, I, KH,
                                              KW, C1, 0)
-        dw = _ext.ext().conv_gemm_wrw(dy, x1, x2, KH, KW, 1, 1)
-        db = _col_sum(dy)
+        dw, db = _ext.ext().conv_gemm_wrw(dy, x1, x2, KH, KW, 1, 1, True)
         return (dx1, dx2, dw[:Oz], dw[Oz:].contiguous(), db[:Oz],
                 db[Oz:].contiguous(), None, None, None)
 
