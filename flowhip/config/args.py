@@ -212,6 +212,10 @@ def build_train_parser(argv=None):
     parser.add_argument("--resume_full", default=None,
                         help="resume full train state (model/opt/sched/step)")
     parser.add_argument("--seed", type=int, default=1234)
+    parser.add_argument("--profile_dir", default=None,
+                        help="rank-0 torch.profiler capture: wait 1 / warmup "
+                        "2 / active 3 steps, chrome trace + op table written "
+                        "here (SURVEY.md §5.1)")
 
     add_ncup_module_flags(parser, argv=argv)
     return parser

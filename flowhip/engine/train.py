@@ -69,6 +69,29 @@ def fetch_optimizer(args, model):
     return optimizer, scheduler
 
 
+def _make_profiler(out_dir):
+    """Rank-0 in-loop torch.profiler: wait 1 / warmup 2 /
+    active 3 steps, then a chrome trace + a self-time op table land in
+    `out_dir`. One cycle only — profiling stays out of steady-state steps."""
+    os.makedirs(out_dir, exist_ok=True)
+
+    def _on_ready(prof):
+        prof.export_chrome_trace(os.path.join(out_dir, "train_trace.json"))
+        table = prof.key_averages().table(
+            sort_by="self_cuda_time_total" if torch.cuda.is_available()
+            else "self_cpu_time_total", row_limit=50)
+        with open(os.path.join(out_dir, "train_ops.txt"), "w") as f:
+            f.write(table)
+
+    activities = [torch.profiler.ProfilerActivity.CPU]
+    if torch.cuda.is_available():
+        activities.append(torch.profiler.ProfilerActivity.CUDA)
+    return torch.profiler.profile(
+        activities=activities,
+        schedule=torch.profiler.schedule(wait=1, warmup=2, active=3, repeat=1),
+        on_trace_ready=_on_ready)
+
+
 def train(args):
     rank, world_size, device = distributed.init_distributed()
 
@@ -109,6 +132,11 @@ def train(args):
         print("Parameter Count: %d" % num_params)
         logger.txt_file.write("Parameter Count: %d\n" % num_params)
 
+    profiler = None
+    if getattr(args, "profile_dir", None) and distributed.is_main():
+        profiler = _make_profiler(args.profile_dir)
+        profiler.__enter__()
+
     should_keep_training = True
     epoch = 0
     while should_keep_training:
@@ -140,6 +168,9 @@ def train(args):
             scheduler.step()
             scaler.update()
 
+            if profiler is not None:
+                profiler.step()
+
             if logger is not None:
                 logger.push(metrics, n_imgs=args.batch_size)
 
@@ -154,6 +185,9 @@ def train(args):
             if total_steps > args.num_steps:
                 should_keep_training = False
                 break
+
+    if profiler is not None:
+        profiler.__exit__(None, None, None)
 
     if distributed.is_main():
         logger.close()

@@ -373,3 +373,25 @@ def test_logger_accepts_numpy_scalars(tmp_path):
     logger.close()
     rec = json.loads(open(tmp_path / "run" / "metrics.jsonl").read())
     assert rec["validation"]["kitti-epe"] == 1.25
+
+
+def test_train_profile_dir_writes_trace(tmp_path, monkeypatch):
+    """--profile_dir captures one torch.profiler cycle inside the train loop
+    (SURVEY §5.1): wait 1 / warmup 2 / active 3 over 7 steps, chrome trace +
+    op table written, training otherwise unaffected."""
+    monkeypatch.chdir(tmp_path)
+
+    argv = ["--name", "prof", "--model", "raft_nc_dbl", "--stage", "synthetic",
+            "--small", "--num_steps", "7", "--batch_size", "1",
+            "--image_size", "64", "64", "--iters", "2", "--lr", "1e-4",
+            "--num_workers", "0", "--profile_dir", str(tmp_path / "prof_out")]
+    parser = build_train_parser(argv=argv)
+    args = finalize_args(parser.parse_args(argv))
+
+    from flowhip.engine.train import train
+    path = train(args)
+    assert os.path.exists(path)
+    trace = tmp_path / "prof_out" / "train_trace.json"
+    table = tmp_path / "prof_out" / "train_ops.txt"
+    assert trace.exists() and trace.stat().st_size > 0
+    assert "Self CPU" in table.read_text()
