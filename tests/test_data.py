@@ -498,3 +498,37 @@ def test_read_disp_kitti(tmp_path):
     np.testing.assert_allclose(flow[..., 0], -disp.astype(np.float32) / 256.0)
     np.testing.assert_array_equal(flow[..., 1], 0)
     np.testing.assert_array_equal(valid, disp > 0)
+
+
+def test_read_pfm_big_endian_and_gray(tmp_path):
+    """readPFM's scale-sign endianness branch (scale > 0 => big-endian) and
+    the single-channel 'Pf' header, both used by FlyingThings disparity/GT
+    variants (reference frame_utils.py readPFM)."""
+    from flowhip.data import frame_utils
+
+    rng = np.random.default_rng(3)
+    arr = rng.standard_normal((5, 7)).astype(np.float32)
+    p = tmp_path / "be.pfm"
+    with open(p, "wb") as f:
+        f.write(b"Pf\n")
+        f.write(b"7 5\n")
+        f.write(b"1.0\n")  # positive scale = big-endian
+        arr[::-1].astype(">f4").tofile(f)
+    out = frame_utils.readPFM(str(p))
+    assert out.shape == (5, 7)
+    np.testing.assert_allclose(out, arr, rtol=0, atol=0)
+
+
+def test_flo_roundtrip_extreme_values(tmp_path):
+    """.flo round-trip preserves exact fp32 bits for extreme magnitudes
+    (large displacements near the MAX_FLOW=400 exclusion boundary and
+    subnormal-small values)."""
+    from flowhip.data import frame_utils
+
+    flow = np.array(
+        [[[399.9, -399.9], [1e-30, -1e-30]],
+         [[65504.0, -65504.0], [0.0, -0.0]]], dtype=np.float32)
+    p = tmp_path / "x.flo"
+    frame_utils.writeFlow(str(p), flow)
+    back = frame_utils.readFlow(str(p))
+    np.testing.assert_array_equal(back, flow)

@@ -395,3 +395,20 @@ def test_train_profile_dir_writes_trace(tmp_path, monkeypatch):
     table = tmp_path / "prof_out" / "train_ops.txt"
     assert trace.exists() and trace.stat().st_size > 0
     assert "Self CPU" in table.read_text()
+
+
+def test_train_add_noise_branch(tmp_path, monkeypatch):
+    """--add_noise applies per-step gaussian noise (stdv ~ U[0,5], clamped to
+    [0,255] — reference train.py:206-209) without breaking the step."""
+    monkeypatch.chdir(tmp_path)
+
+    argv = ["--name", "noise", "--model", "raft_nc_dbl", "--stage",
+            "synthetic", "--small", "--num_steps", "2", "--batch_size", "1",
+            "--image_size", "64", "64", "--iters", "2", "--lr", "1e-4",
+            "--num_workers", "0", "--add_noise"]
+    parser = build_train_parser(argv=argv)
+    args = finalize_args(parser.parse_args(argv))
+
+    from flowhip.engine.train import train
+    path = train(args)
+    assert os.path.exists(path)

@@ -214,3 +214,23 @@ def test_raft_nc_dbl_non_square_shape():
     preds[-1].abs().mean().backward()
     g = model.fnet.conv1.weight.grad
     assert g is not None and torch.isfinite(g).all()
+
+
+def test_raft_nc_dbl_small_defined_behavior():
+    """--small raft_nc_dbl is broken in the reference (guidance channel
+    mismatch); our defined behavior runs it with
+    guidance = hidden_dim. Forward + backward on the small config."""
+    from flowhip.config.args import default_ncup_args
+    from flowhip.models import build_model
+
+    args = default_ncup_args(model="raft_nc_dbl")
+    args.small = True
+    args.mixed_precision = False
+    model = build_model(args)
+    im1 = torch.randn(1, 3, 128, 128) * 40 + 127
+    im2 = torch.randn(1, 3, 128, 128) * 40 + 127
+    preds = model(im1, im2, iters=2)
+    assert preds[-1].shape == (1, 2, 128, 128)
+    preds[-1].float().sum().backward()
+    grads = [p.grad for p in model.parameters() if p.grad is not None]
+    assert grads and all(torch.isfinite(g).all() for g in grads)
