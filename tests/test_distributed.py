@@ -148,3 +148,29 @@ def test_bench_default_invocation(tmp_path):
     assert rec["metric"] == "train_image_pairs_per_sec"
     assert rec["value"] > 0
     assert rec["config"]["global_batch"] == 1
+
+
+@pytest.mark.timeout(900)
+def test_train_entry_world_size_2(tmp_path):
+    """The full train.py CLI under torchrun with 2 CPU ranks (gloo):
+    global --batch_size 2 shards to 1/rank, rank 0 writes the checkpoint
+    and log, both ranks exit 0 — the same launch shape the driver uses for
+    the round-end SCALE pass (with nccl/RCCL on GPUs)."""
+    import subprocess
+    import sys
+
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    cmd = [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+           "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+           "--master-port", "29719", os.path.join(repo, "train.py"),
+           "--name", "ddp2", "--model", "raft_nc_dbl", "--stage", "synthetic",
+           "--small", "--num_steps", "2", "--batch_size", "2",
+           "--image_size", "64", "64", "--iters", "2", "--lr", "1e-4",
+           "--num_workers", "0"]
+    res = subprocess.run(cmd, capture_output=True, text=True, timeout=840,
+                         cwd=str(tmp_path))
+    assert res.returncode == 0, res.stderr[-2000:]
+    ckpt = tmp_path / "checkpoints" / "ddp2" / "final_model.pth"
+    assert ckpt.exists()
+    log = (tmp_path / "checkpoints" / "ddp2" / "log.txt").read_text()
+    assert "Parameter Count" in log
