@@ -412,3 +412,17 @@ def test_train_add_noise_branch(tmp_path, monkeypatch):
     from flowhip.engine.train import train
     path = train(args)
     assert os.path.exists(path)
+
+
+def test_tools_compile_and_guard_gpu():
+    """Every tools/*.py compiles and its module-level code does not require
+    a GPU (each tool guards its CUDA work behind main()/run() so CPU boxes
+    can parse-check them)."""
+    import py_compile
+
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    tools = sorted(f for f in os.listdir(os.path.join(repo, "tools"))
+                   if f.endswith(".py"))
+    assert len(tools) >= 7
+    for f in tools:
+        py_compile.compile(os.path.join(repo, "tools", f), doraise=True)
