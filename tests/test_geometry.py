@@ -93,3 +93,38 @@ def test_input_padder_noop_when_divisible():
         (y,) = p.pad(x.clone())
         assert y.shape == x.shape
         assert torch.equal(p.unpad(y), x)
+
+
+from hypothesis import given, settings, strategies as st
+
+
+@settings(max_examples=40, deadline=None)
+@given(st.integers(3, 100), st.integers(3, 100),
+       st.sampled_from(["sintel", "kitti"]))
+def test_input_padder_property(h, w, mode):
+    """For ANY input size: padded dims divide by 8, per-side pads < 8,
+    and unpad(pad(x)) is exactly x (reference utils.py:7-26 contract)."""
+    from flowhip.utils.geometry import InputPadder
+
+    x = torch.arange(h * w, dtype=torch.float32).reshape(1, 1, h, w)
+    padder = InputPadder(x.shape, mode=mode)
+    (y,) = padder.pad(x)
+    assert y.shape[-2] % 8 == 0 and y.shape[-1] % 8 == 0
+    assert y.shape[-2] - h < 8 and y.shape[-1] - w < 8
+    back = padder.unpad(y)
+    assert back.shape == x.shape
+    torch.testing.assert_close(back, x, rtol=0, atol=0)
+
+
+@settings(max_examples=25, deadline=None)
+@given(st.integers(4, 40), st.integers(4, 40), st.integers(0, 2 ** 31 - 1))
+def test_bilinear_sampler_identity_property(h, w, seed):
+    """Sampling at the identity grid returns the input exactly for any
+    size (align_corners=True pixel-coordinate convention)."""
+    from flowhip.utils.geometry import bilinear_sampler, coords_grid
+
+    g = torch.Generator().manual_seed(seed)
+    img = torch.randn(1, 3, h, w, generator=g)
+    coords = coords_grid(1, h, w).permute(0, 2, 3, 1)
+    out = bilinear_sampler(img, coords)
+    torch.testing.assert_close(out, img, rtol=0, atol=1e-5)
