@@ -532,3 +532,44 @@ def test_flo_roundtrip_extreme_values(tmp_path):
     frame_utils.writeFlow(str(p), flow)
     back = frame_utils.readFlow(str(p))
     np.testing.assert_array_equal(back, flow)
+
+
+def test_flow_augmentor_arbitrary_input_sizes():
+    """The augmentor crops to crop_size for a RANGE of input geometries
+    (taller/wider/near-crop-size inputs), with outputs always uint8 images
+    + fp32 flow — the invariant train-time batching depends on."""
+    from flowhip.data.augmentor import FlowAugmentor
+
+    for seed, (H, W) in enumerate([(65, 65), (64, 200), (200, 64),
+                                   (97, 131), (300, 70)]):
+        np.random.seed(seed)
+        aug = FlowAugmentor(crop_size=(64, 64), min_scale=-0.1, max_scale=0.4)
+        img1 = np.random.randint(0, 255, (H, W, 3), dtype=np.uint8)
+        img2 = np.random.randint(0, 255, (H, W, 3), dtype=np.uint8)
+        flow = np.random.randn(H, W, 2).astype(np.float32)
+        a, b, f = aug(img1, img2, flow)
+        assert a.shape == (64, 64, 3) and b.shape == (64, 64, 3)
+        assert f.shape == (64, 64, 2)
+        assert a.dtype == np.uint8 and f.dtype == np.float32
+        assert np.isfinite(f).all()
+
+
+def test_flow_viz_invariants():
+    """flow_to_color: uint8 HxWx3; zero flow renders (near-)white center
+    colors; scaling the whole field by the max preserves hue structure
+    (Middlebury wheel convention)."""
+    from flowhip.data.flow_viz import flow_to_color
+
+    z = np.zeros((8, 10, 2), dtype=np.float32)
+    img = flow_to_color(z)
+    assert img.shape == (8, 10, 3) and img.dtype == np.uint8
+    assert (img > 220).all()  # zero displacement = unsaturated (white-ish)
+
+    rng = np.random.default_rng(5)
+    f = rng.standard_normal((16, 16, 2)).astype(np.float32) * 3
+    img1 = flow_to_color(f)
+    assert img1.shape == (16, 16, 3) and img1.dtype == np.uint8
+    # the rendering normalizes by the max magnitude: scaling the whole
+    # field leaves the image identical
+    img2 = flow_to_color(f * 4.0)
+    np.testing.assert_array_equal(img1, img2)
