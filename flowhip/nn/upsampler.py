@@ -37,8 +37,15 @@ def get_upsampler(in_ch, guidance_ch, args, respect_choice=False):
         upsampler_name = "nconvupsampler"  # hardcoded in the reference (:12)
 
     if upsampler_name == "nconvupsampler":
+        # channels_to_batch folds the data channels into the batch, so the
+        # NConv U-Net sees 1 channel (the reference hardcodes in_ch=1 and
+        # leaves --final_upsampling_channels_to_batch=False broken, with a
+        # TODO at upsampler.py:16; defined behavior here: honor the flag by
+        # building the net at the real channel count).
+        interp_in_ch = 1 if getattr(
+            args, "final_upsampling_channels_to_batch", True) else in_ch
         interpolation_net = NConvUNet(
-            in_ch=1,  # channels folded to batch; NConvUNet supports in_ch=1
+            in_ch=interp_in_ch,
             channels_multiplier=args.interp_net_channels_multiplier,
             num_downsampling=args.interp_net_num_downsampling,
             encoder_filter_sz=args.interp_net_encoder_filter_sz,
@@ -174,6 +181,13 @@ class NConvUpsampler(nn.Module):
         x_highres = self._inject(x_lowres)
 
         if self.est_on_high_res:
+            # Reference contract (upsampler.py:147-149): the guidance must
+            # already sit at the output resolution. In the flow models it
+            # arrives at the data's low resolution instead, which crashes
+            # the reference's cat — defined behavior here: resize it up
+            # (area), the exact mirror of the low-res branch below.
+            if x_guidance.shape[2:] != x_highres.shape[2:]:
+                x_guidance = ops.area_resize(x_guidance, x_highres.size()[2:])
             x_data_for_guidance = x_highres
         else:
             x_guidance = ops.area_resize(x_guidance, x_lowres.size()[2:])

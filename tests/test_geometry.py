@@ -127,4 +127,5 @@ def test_bilinear_sampler_identity_property(h, w, seed):
     img = torch.randn(1, 3, h, w, generator=g)
     coords = coords_grid(1, h, w).permute(0, 2, 3, 1)
     out = bilinear_sampler(img, coords)
-    torch.testing.assert_close(out, img, rtol=0, atol=1e-5)
+    # pixel -> [-1,1] -> pixel round-trip carries ~1e-5 fp32 rounding
+    torch.testing.assert_close(out, img, rtol=0, atol=1e-4)

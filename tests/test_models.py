@@ -234,3 +234,37 @@ def test_raft_nc_dbl_small_defined_behavior():
     preds[-1].float().sum().backward()
     grads = [p.grad for p in model.parameters() if p.grad is not None]
     assert grads and all(torch.isfinite(g).all() for g in grads)
+
+
+@pytest.mark.parametrize("overrides", [
+    {"interp_net_num_downsampling": 2},
+    {"interp_net_use_bias": True, "interp_net_out_filter_sz": 3},
+    {"interp_net_shared_encoder": False, "interp_net_use_double_conv": False},
+    {"interp_net_data_pooling": "max_pooling"},
+    {"final_upsampling_use_residuals": True},
+    {"final_upsampling_use_data_for_guidance": False},
+    {"final_upsampling_est_on_high_res": True},
+    {"final_upsampling_channels_to_batch": False},
+])
+def test_ncup_variant_configs_run(overrides):
+    """Every reflective-CLI NCUP variant a reference user can request
+    (--interp_net_* / --final_upsampling_* grid) builds and completes a
+    forward+backward — non-default configs must work, not just parse
+    (SURVEY.md §2.5/§5.6)."""
+    from flowhip.config.args import default_ncup_args
+    from flowhip.models import build_model
+
+    args = default_ncup_args(model="raft_nc_dbl", small=True)
+    args.mixed_precision = False
+    for k, v in overrides.items():
+        assert hasattr(args, k), k
+        setattr(args, k, v)
+    torch.manual_seed(7)
+    model = build_model(args)
+    im1 = torch.randn(1, 3, 128, 128) * 40 + 127
+    im2 = torch.randn(1, 3, 128, 128) * 40 + 127
+    preds = model(im1, im2, iters=2)
+    assert preds[-1].shape == (1, 2, 128, 128)
+    preds[-1].float().abs().mean().backward()
+    assert all(torch.isfinite(p.grad).all() for p in model.parameters()
+               if p.grad is not None)
