@@ -245,6 +245,11 @@ def test_raft_nc_dbl_small_defined_behavior():
     {"final_upsampling_use_data_for_guidance": False},
     {"final_upsampling_est_on_high_res": True},
     {"final_upsampling_channels_to_batch": False},
+    {"weights_est_net": "UNet", "final_upsampling_use_data_for_guidance": False},
+    {"weights_est_net_num_ch": [32], "weights_est_net_filter_sz": [5, 3],
+     "weights_est_net_dilation": [2, 1]},
+    {"interp_net_channels_multiplier": 4, "interp_net_encoder_filter_sz": 3,
+     "interp_net_decoder_filter_sz": 5},
 ])
 def test_ncup_variant_configs_run(overrides):
     """Every reflective-CLI NCUP variant a reference user can request
