@@ -426,3 +426,22 @@ def test_tools_compile_and_guard_gpu():
     assert len(tools) >= 7
     for f in tools:
         py_compile.compile(os.path.join(repo, "tools", f), doraise=True)
+
+
+def test_train_raft_baseline_model(tmp_path, monkeypatch):
+    """--model raft (convex-upsample mask head, reference train.py flow)
+    trains end-to-end — the baseline family, not just the NCUP one."""
+    monkeypatch.chdir(tmp_path)
+
+    argv = ["--name", "raftb", "--model", "raft", "--stage", "synthetic",
+            "--small", "--num_steps", "2", "--batch_size", "1",
+            "--image_size", "128", "128", "--iters", "2", "--lr", "1e-4",
+            "--num_workers", "0"]
+    parser = build_train_parser(argv=argv)
+    args = finalize_args(parser.parse_args(argv))
+
+    from flowhip.engine.train import train
+    path = train(args)
+    assert os.path.exists(path)
+    sd = torch.load(path, weights_only=True)
+    assert all(k.startswith("module.") for k in sd)
