@@ -273,3 +273,28 @@ def test_ncup_variant_configs_run(overrides):
     preds[-1].float().abs().mean().backward()
     assert all(torch.isfinite(p.grad).all() for p in model.parameters()
                if p.grad is not None)
+
+
+@pytest.mark.parametrize("pos_fn", ["SoftPlus", "Exp", "Sigmoid", "SoftMax"])
+def test_nconv_pos_fn_variants(pos_fn):
+    """Every EnforcePos non-negativity map the reference supports
+    (nconv_modules.py:254-269, case-insensitive) yields strictly
+    non-negative effective weights and a working forward."""
+    from flowhip.nn.nconv import NConv2d, pos_transform
+
+    m = NConv2d(1, 2, (3, 3), stride=(1, 1), pos_fn=pos_fn, groups=1,
+                bias=False)
+    w = pos_transform(m.weight_p, pos_fn)
+    assert (w >= 0).all()
+    d = torch.rand(1, 1, 12, 12)
+    c = (torch.rand(1, 1, 12, 12) > 0.5).float()
+    out, cout = m((d, c))
+    assert out.shape == (1, 2, 12, 12) and torch.isfinite(out).all()
+    assert (cout >= 0).all()
+
+
+def test_nconv_pos_fn_unknown_raises():
+    from flowhip.nn.nconv import pos_transform
+
+    with pytest.raises(ValueError):
+        pos_transform(torch.zeros(2), "relu")
