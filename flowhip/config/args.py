@@ -281,3 +281,15 @@ def default_ncup_args(**overrides):
     for k, v in overrides.items():
         setattr(ns, k, v)
     return ns
+
+
+def filter_list_of_strings(lst, include=("*",), exclude=()):
+    """fnmatch include/exclude filter (reference args.py:152-156 API)."""
+    import fnmatch
+    import itertools
+
+    matches = set(itertools.chain.from_iterable(
+        fnmatch.filter(lst, pat) for pat in include))
+    nomatch = set(itertools.chain.from_iterable(
+        fnmatch.filter(lst, pat) for pat in exclude))
+    return list(matches - nomatch)

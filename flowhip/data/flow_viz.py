@@ -119,3 +119,16 @@ def flow_to_image(flow):
     img = flow_compute_color(u, v)
     img[idxUnknown] = 0
     return np.uint8(img)
+
+
+def make_color_wheel():
+    """Name-compat with the reference's VCN-port naming (flow_viz.py:158):
+    both reference ports build the identical 55-entry wheel."""
+    return make_colorwheel()
+
+
+def compute_color(u, v):
+    """Name-compat with the reference's VCN port (flow_viz.py:199-237): the
+    faithful 0-based wheel indexing — our default variant. The reference
+    returns a float array of integral values; matched here."""
+    return flow_compute_color(u, v).astype(np.float64)

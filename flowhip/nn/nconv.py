@@ -199,3 +199,12 @@ class NConvUNet(nn.Module):
             x[i + nds + 1], c[i + nds + 1] = self.decoder[i]((xin, cin))
 
         return self.nconv_out((x[-1], c[-1]))
+
+
+def retrieve_elements_from_indices(tensor, indices):
+    """Gather per-pixel elements at pooling argmax indices (reference
+    nconv_modules.py:19-22 public helper; the conf-based pooling itself
+    runs through ops.conf_pool / the HIP kernel)."""
+    flat = tensor.flatten(start_dim=2)
+    return flat.gather(dim=2, index=indices.flatten(start_dim=2)) \
+        .view_as(indices)

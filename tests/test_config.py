@@ -131,3 +131,13 @@ def test_intlist_and_bool_flag_parsing():
     assert args.weights_est_net_num_ch == [8, 4]
     assert args.interp_net_use_bias is False
     assert args.interp_net_shared_encoder is True
+
+
+def test_filter_list_of_strings():
+    """Reference args.py:152-156 helper API."""
+    from flowhip.config.args import filter_list_of_strings
+
+    lst = ["alpha", "beta", "alpine", "gamma"]
+    out = filter_list_of_strings(lst, include=("al*",), exclude=("alpine",))
+    assert sorted(out) == ["alpha"]
+    assert sorted(filter_list_of_strings(lst)) == sorted(lst)

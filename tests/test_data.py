@@ -573,3 +573,18 @@ def test_flow_viz_invariants():
     # field leaves the image identical
     img2 = flow_to_color(f * 4.0)
     np.testing.assert_array_equal(img1, img2)
+
+
+def test_flow_viz_name_compat_helpers():
+    """VCN-naming aliases (reference flow_viz.py:158/199) agree with the
+    canonical variants."""
+    from flowhip.data.flow_viz import (compute_color, flow_compute_color,
+                                       make_color_wheel, make_colorwheel)
+
+    np.testing.assert_array_equal(make_color_wheel(), make_colorwheel())
+    rng = np.random.default_rng(2)
+    u = rng.standard_normal((5, 6)) * 0.4
+    v = rng.standard_normal((5, 6)) * 0.4
+    c = compute_color(u, v)
+    assert c.dtype == np.float64
+    np.testing.assert_array_equal(np.uint8(c), flow_compute_color(u, v))
