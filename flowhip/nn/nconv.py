@@ -208,3 +208,56 @@ def retrieve_elements_from_indices(tensor, indices):
     flat = tensor.flatten(start_dim=2)
     return flat.gather(dim=2, index=indices.flatten(start_dim=2)) \
         .view_as(indices)
+
+
+class EnforcePos:
+    """Hook-based non-negativity reparameterization for ARBITRARY modules
+    (reference nconv_modules.py:218-251 public API). ``apply(module,
+    'weight', pos_fn)`` replaces ``weight`` with a ``weight_p`` parameter
+    and recomputes ``weight = pos(weight_p)`` in a forward pre-hook.
+
+    NConv2d applies the same reparameterization functionally inside its
+    forward (identical ``weight_p`` state-dict surface); this class exists
+    for user code that used the reference hook on other modules. The
+    reference's observable initialization is preserved: ``weight_p`` starts
+    as ``pos(weight)`` — NOT the inverse — so the first effective weight is
+    ``pos(pos(weight))`` (a reference quirk checkpoints depend on).
+    """
+
+    def __init__(self, name, pos_fn):
+        self.name = name
+        self.pos_fn = pos_fn
+
+    def compute_weight(self, module):
+        return pos_transform(getattr(module, self.name + "_p"), self.pos_fn)
+
+    @staticmethod
+    def apply(module, name, pos_fn):
+        fn = EnforcePos(name, pos_fn)
+        weight = getattr(module, name)
+        del module._parameters[name]
+        module.register_parameter(
+            name + "_p", nn.Parameter(pos_transform(weight, pos_fn).detach()))
+        setattr(module, name, fn.compute_weight(module))
+        module.register_forward_pre_hook(fn)
+        return fn
+
+    def remove(self, module):
+        weight = self.compute_weight(module)
+        delattr(module, self.name)
+        del module._parameters[self.name + "_p"]
+        module.register_parameter(self.name, nn.Parameter(weight.detach()))
+
+    def __call__(self, module, inputs):
+        setattr(module, self.name, self.compute_weight(module))
+
+
+def remove_weight_pos(module, name="weight"):
+    """Remove an EnforcePos reparameterization, freezing the effective
+    weight back into a plain parameter (reference nconv_modules.py:272)."""
+    for k, hook in list(module._forward_pre_hooks.items()):
+        if isinstance(hook, EnforcePos) and hook.name == name:
+            hook.remove(module)
+            del module._forward_pre_hooks[k]
+            return module
+    raise ValueError(f"no EnforcePos reparameterization of {name!r} found")

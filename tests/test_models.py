@@ -298,3 +298,29 @@ def test_nconv_pos_fn_unknown_raises():
 
     with pytest.raises(ValueError):
         pos_transform(torch.zeros(2), "relu")
+
+
+def test_enforce_pos_hook_api():
+    """The generic hook-based EnforcePos API (reference
+    nconv_modules.py:218-283): apply -> weight_p surface + positive
+    effective weight each forward; remove -> plain weight restored."""
+    import torch.nn as nn
+    from flowhip.nn.nconv import EnforcePos, remove_weight_pos
+
+    m = nn.Conv2d(2, 3, 3, bias=False)
+    with torch.no_grad():
+        m.weight.uniform_(-1, 1)
+    EnforcePos.apply(m, "weight", "SoftPlus")
+    assert "weight_p" in dict(m.named_parameters())
+    assert "weight" not in dict(m.named_parameters())
+    out = m(torch.randn(1, 2, 8, 8))
+    assert torch.isfinite(out).all()
+    assert (m.weight > 0).all()  # recomputed by the pre-hook
+
+    sd_keys = set(m.state_dict().keys())
+    assert "weight_p" in sd_keys
+
+    remove_weight_pos(m)
+    assert "weight" in dict(m.named_parameters())
+    assert (m.weight > 0).all()
+    m(torch.randn(1, 2, 8, 8))
