@@ -22,6 +22,7 @@ from flowhip.data import flow_viz
 from flowhip.engine import checkpoints
 from flowhip.models import build_model
 from flowhip.utils.geometry import InputPadder
+from flowhip.utils.geometry import InputPadder
 
 DEVICE = "cuda" if torch.cuda.is_available() else "cpu"
 
@@ -30,6 +31,13 @@ def load_image(imfile):
     img = np.array(Image.open(imfile)).astype(np.uint8)
     img = torch.from_numpy(img).permute(2, 0, 1).float()
     return img[None].to(DEVICE)
+
+
+def load_image_list(image_files):
+    """Batch-load + pad a sorted frame list (reference demo.py:26-35)."""
+    images = torch.cat([load_image(f) for f in sorted(image_files)], dim=0)
+    padder = InputPadder(images.shape)
+    return padder.pad(images)[0]
 
 
 def viz(img, flo, out_path):

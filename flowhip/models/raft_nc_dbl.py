@@ -84,3 +84,7 @@ class RAFT_NC_DBL(_RAFTBase):
         if test_mode:
             return coords1 - coords0, flow_up
         return flow_predictions
+
+# The reference names this class `RAFT` inside raft_nc_dbl.py (shadowing
+# the baseline's class name across modules); keep that import surface.
+RAFT = RAFT_NC_DBL

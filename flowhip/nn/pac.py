@@ -463,3 +463,15 @@ class PacPool2d(_PacConvNd):
         out = pacpool2d(input_2d, kernel, self.kernel_size, self.stride,
                         self.padding, self.dilation)
         return out if output_mask is None else (out, output_mask)
+
+
+def np_gaussian_2d(width, sigma=-1):
+    """Truncated, normalized 2D Gaussian filter (reference
+    pac_modules.py:38-49 public utility; sigma defaults to width/4)."""
+    assert width % 2 == 1
+    if sigma <= 0:
+        sigma = float(width) / 4
+    r = np.arange(-(width // 2), width // 2 + 1, dtype=np.float32)
+    g = np.exp(-0.5 * r * r / (sigma * sigma))
+    g2 = g.reshape(-1, 1) * g
+    return g2 / g2.sum()

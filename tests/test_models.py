@@ -324,3 +324,11 @@ def test_enforce_pos_hook_api():
     assert "weight" in dict(m.named_parameters())
     assert (m.weight > 0).all()
     m(torch.randn(1, 2, 8, 8))
+
+
+def test_raft_nc_dbl_module_level_raft_alias():
+    """The reference exposes the NCUP model as `RAFT` inside raft_nc_dbl
+    (import-surface parity)."""
+    from flowhip.models import raft_nc_dbl as mod
+
+    assert mod.RAFT is mod.RAFT_NC_DBL

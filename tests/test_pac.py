@@ -97,3 +97,18 @@ def test_pac_state_dict_keys():
     keys = set(m.state_dict().keys())
     assert keys == {"weight", "bias"}
     assert m.weight.shape == (3, 3, 5, 5)  # (in, out, k, k) for transposed
+
+
+def test_np_gaussian_2d():
+    """Reference pac_modules.py:38-49 utility: normalized, symmetric,
+    peak-centered."""
+    from flowhip.nn.pac import np_gaussian_2d
+
+    g = np_gaussian_2d(5)
+    assert g.shape == (5, 5)
+    assert abs(g.sum() - 1.0) < 1e-6
+    assert g[2, 2] == g.max()
+    import numpy as np
+    np.testing.assert_allclose(g, g.T, atol=0)
+    g3 = np_gaussian_2d(3, sigma=1.0)
+    assert g3.shape == (3, 3) and abs(g3.sum() - 1.0) < 1e-6
