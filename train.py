@@ -14,7 +14,9 @@ import numpy as np
 import torch
 
 from flowhip.config import build_train_parser, finalize_args
-from flowhip.engine.train import train
+from flowhip.engine.train import count_parameters, fetch_optimizer, train
+from flowhip.engine.logger import Logger
+from flowhip.ops import sequence_loss
 
 if __name__ == "__main__":
     args = finalize_args(build_train_parser().parse_args())
