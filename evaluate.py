@@ -9,6 +9,8 @@ import torch
 from flowhip.config import build_eval_parser, finalize_args
 from flowhip.engine import checkpoints
 from flowhip.engine.evaluate import (
+    create_kitti_submission,
+    create_sintel_submission,
     validate_chairs,
     validate_kitti,
     validate_sintel,
