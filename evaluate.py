@@ -38,5 +38,11 @@ if __name__ == "__main__":
             validate_kitti(model, args.iters or 24)
         elif args.dataset == "synthetic":
             validate_synthetic(model, args.iters or 12)
+        elif args.dataset == "sintel_submission":
+            # the reference ships these as commented-out lines the user must
+            # edit in (evaluate.py:267,271); defined behavior: CLI choices
+            create_sintel_submission(model, warm_start=True)
+        elif args.dataset == "kitti_submission":
+            create_kitti_submission(model)
         else:
             raise SystemExit(f"unknown --dataset {args.dataset!r}")
