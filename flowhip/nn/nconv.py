@@ -170,6 +170,14 @@ class NConvUNet(nn.Module):
         self.nconv_out = NConv2d(mid, in_ch, out_filter_sz, stride=(1, 1),
                                  pos_fn=pos_fn, groups=groups, bias=False)
 
+    @staticmethod
+    def downsample_data_conf(data, conf, ds_factor=2,
+                             pooling_type="conf_based"):
+        """Downsample a (data, conf) pair keeping data at the
+        max-confidence positions (reference nconv_modules.py:94-104 public
+        helper; routes through ops.conf_pool and the HIP kernel on GPU)."""
+        return ops.conf_pool(data, conf, ds_factor, pooling_type)
+
     def forward(self, inpt):
         nds = self.num_downsampling
         x = [None] * (nds * 2 + 1)
@@ -186,7 +194,8 @@ class NConvUNet(nn.Module):
             if i == 0:
                 x[i + 1], c[i + 1] = self.encoder[i]((x[i], c[i]))
             else:
-                d_ds, c_ds = ops.conf_pool(x[i], c[i], 2, self.data_pooling)
+                d_ds, c_ds = self.downsample_data_conf(
+                    x[i], c[i], 2, self.data_pooling)
                 x[i + 1], c[i + 1] = self.encoder[i]((d_ds, c_ds))
 
         # Decoder (reference index arithmetic: stage 0 pairs x[nds] with
