@@ -629,3 +629,37 @@ def test_checkpoint_file_interchange_with_reference(ref_modules, tmp_path):
     checkpoints.load_weights(ours2, ref_path, strict=True)
     assert torch.equal(ours2.state_dict()["fnet.conv1.weight"],
                        ours.state_dict()["fnet.conv1.weight"])
+
+
+def test_enforce_pos_hook_bitexact_vs_reference(ref_modules):
+    """The generic EnforcePos hook API vs the reference's
+    (nconv_modules.py:218-283): identical weight_p initialization (the
+    pos(weight) quirk), identical recomputed effective weight after a
+    parameter update, identical remove() freeze."""
+    import nconv_modules as ref_nc
+
+    from flowhip.nn.nconv import EnforcePos, remove_weight_pos
+
+    torch.manual_seed(31)
+    w0 = torch.randn(3, 2, 3, 3)
+
+    m_ref = torch.nn.Conv2d(2, 3, 3, bias=False)
+    m_our = torch.nn.Conv2d(2, 3, 3, bias=False)
+    with torch.no_grad():
+        m_ref.weight.copy_(w0)
+        m_our.weight.copy_(w0)
+
+    ref_nc.EnforcePos.apply(m_ref, "weight", "softplus")
+    EnforcePos.apply(m_our, "weight", "softplus")
+    assert torch.equal(m_ref.weight_p, m_our.weight_p)
+
+    with torch.no_grad():
+        m_ref.weight_p.add_(0.25)
+        m_our.weight_p.add_(0.25)
+    x = torch.randn(1, 2, 8, 8)
+    torch.testing.assert_close(m_ref(x), m_our(x), rtol=0, atol=0)
+    assert torch.equal(m_ref.weight, m_our.weight)
+
+    remove_weight_pos(m_our)
+    assert torch.equal(m_our.weight, m_ref.weight)
+    assert "weight_p" not in dict(m_our.named_parameters())
